@@ -1,0 +1,39 @@
+"""accelerate_amd — an MI355X-native training-loop framework with the
+capabilities of huggingface/accelerate, built from scratch for CDNA4:
+PyTorch-ROCm at the tensor/autograd level; our own RCCL/xGMI gradient
+reducer, CDNA4 HIP kernel pack (fused AdamW, clip, scaler), device-map
+dispatch sized for 288 GB HBM3E; no CUDA shims, no Triton, no plugin
+dispatch to third-party engines.
+"""
+
+__version__ = "0.1.0"
+
+from .accelerator import Accelerator
+from .big_modeling import (
+    cpu_offload,
+    disk_offload,
+    dispatch_model,
+    init_empty_weights,
+    init_on_device,
+    load_checkpoint_and_dispatch,
+)
+from .data_loader import prepare_data_loader, skip_first_batches
+from .launchers import debug_launcher, notebook_launcher
+from .local_sgd import LocalSGD
+from .state import AcceleratorState, GradientState, PartialState
+from .utils.dataclasses import (
+    AutocastKwargs,
+    DistributedDataParallelKwargs,
+    DistributedType,
+    FP8RecipeKwargs,
+    FullyShardedDataParallelPlugin,
+    GradientAccumulationPlugin,
+    GradScalerKwargs,
+    InitProcessGroupKwargs,
+    ProfileKwargs,
+    ProjectConfiguration,
+)
+from .utils.modeling import infer_auto_device_map, load_checkpoint_in_model
+from .utils.memory import find_executable_batch_size
+from .utils.operations import send_to_device
+from .utils.random_utils import set_seed

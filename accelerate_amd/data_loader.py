@@ -1,0 +1,903 @@
+"""Data loading: per-rank sharding, prefetch, device placement
+(reference: data_loader.py — behavior-parity documented per class; the
+implementation is our own).
+
+Sharding modes:
+- default ("deal" mode): batch ``k`` from the inner sampler goes to rank
+  ``k % n``. A one-batch lookahead ensures every rank only yields once the
+  whole cycle has full batches; ``even_batches`` wraps indices around so the
+  tail cycle stays rectangular (reference: data_loader.py:213-272).
+- ``split_batches``: every rank iterates the same global batch and keeps its
+  contiguous slice ``[bs/n*rank : bs/n*(rank+1)]`` (reference: :191-211).
+
+Device placement: batches move H2D with ``non_blocking=True``; when the
+source loader pins memory the copy overlaps the next host-side fetch on the
+HIP copy engine.
+"""
+
+import math
+from contextlib import suppress
+from typing import Callable, List, Optional, Union
+
+import torch
+from torch.utils.data import BatchSampler, DataLoader, IterableDataset, RandomSampler
+
+from .logging import get_logger
+from .state import GradientState, PartialState
+from .utils.dataclasses import DistributedType, RNGType
+from .utils.operations import (
+    broadcast,
+    broadcast_object_list,
+    concatenate,
+    find_batch_size,
+    get_data_structure,
+    initialize_tensors,
+    send_to_device,
+    slice_tensors,
+)
+from .utils.random_utils import synchronize_rng_states
+
+logger = get_logger(__name__)
+
+# kwargs of torch DataLoader that we re-plumb when rebuilding a loader
+_PYTORCH_DATALOADER_KWARGS = {
+    "batch_size": 1,
+    "shuffle": False,
+    "sampler": None,
+    "batch_sampler": None,
+    "num_workers": 0,
+    "collate_fn": None,
+    "pin_memory": False,
+    "drop_last": False,
+    "timeout": 0,
+    "worker_init_fn": None,
+    "multiprocessing_context": None,
+    "generator": None,
+    "prefetch_factor": None,
+    "persistent_workers": False,
+    "pin_memory_device": "",
+}
+
+
+class SeedableRandomSampler(RandomSampler):
+    """RandomSampler reseeded with ``seed + epoch`` each iteration so every
+    process draws identical permutations (reference: data_loader.py:73-108)."""
+
+    def __init__(self, *args, **kwargs):
+        self.epoch = kwargs.pop("epoch", 0)
+        self.initial_seed = kwargs.pop("seed", None)
+        super().__init__(*args, **kwargs)
+        if self.generator is None:
+            self.generator = torch.Generator()
+            if self.initial_seed is None:
+                self.initial_seed = torch.random.default_generator.seed() % 2**31
+        elif self.initial_seed is None:
+            self.initial_seed = self.generator.initial_seed()
+
+    def __iter__(self):
+        self.generator.manual_seed(self.initial_seed + self.epoch)
+        yield from super().__iter__()
+        self.set_epoch(self.epoch + 1)
+
+    def set_epoch(self, epoch: int):
+        self.epoch = epoch
+
+
+class BatchSamplerShard(BatchSampler):
+    """Shard an inner ``BatchSampler`` across ``num_processes`` ranks
+    (reference: data_loader.py:110-272; semantics described in the module
+    docstring)."""
+
+    def __init__(
+        self,
+        batch_sampler: BatchSampler,
+        num_processes: int = 1,
+        process_index: int = 0,
+        split_batches: bool = False,
+        even_batches: bool = True,
+    ):
+        if split_batches and batch_sampler.batch_size % num_processes != 0:
+            raise ValueError(
+                f"To use `BatchSamplerShard` in `split_batches` mode, the batch size ({batch_sampler.batch_size}) "
+                f"needs to be a round multiple of the number of processes ({num_processes})."
+            )
+        self.batch_sampler = batch_sampler
+        self.num_processes = num_processes
+        self.process_index = process_index
+        self.split_batches = split_batches
+        self.even_batches = even_batches
+        self.batch_size = getattr(batch_sampler, "batch_size", None)
+        self.drop_last = getattr(batch_sampler, "drop_last", False)
+        if self.batch_size is None and self.even_batches:
+            raise ValueError("You need to use `even_batches=False` when the batch sampler has no batch size.")
+
+    @property
+    def total_length(self):
+        return len(self.batch_sampler)
+
+    def __len__(self):
+        if self.split_batches:
+            # every rank sees every (sliced) batch
+            return len(self.batch_sampler)
+        if len(self.batch_sampler) % self.num_processes == 0:
+            return len(self.batch_sampler) // self.num_processes
+        length = len(self.batch_sampler) // self.num_processes
+        if self.drop_last:
+            return length
+        elif self.even_batches:
+            return length + 1
+        else:
+            # ranks below the remainder get one extra batch
+            return length + 1 if self.process_index < len(self.batch_sampler) % self.num_processes else length
+
+    def __iter__(self):
+        return self._iter_split() if self.split_batches else self._iter_deal()
+
+    def _iter_split(self):
+        wrap_pool = None
+        per_rank = self.batch_size // self.num_processes
+        lo, hi = per_rank * self.process_index, per_rank * (self.process_index + 1)
+        for global_batch in self.batch_sampler:
+            if wrap_pool is None:
+                wrap_pool = global_batch
+            if len(global_batch) == self.batch_size:
+                yield global_batch[lo:hi]
+            else:
+                # short tail batch
+                if not self.even_batches:
+                    mine = global_batch[lo:hi]
+                    if len(mine) > 0:
+                        yield mine
+                else:
+                    # wrap indices from the first batch until rectangular
+                    padded = list(global_batch)
+                    while len(padded) < self.batch_size:
+                        padded += wrap_pool[: self.batch_size - len(padded)]
+                    yield padded[lo:hi]
+
+    def _iter_deal(self):
+        wrap_pool = []  # indices from the first full cycle, used for tail padding
+        pending = []  # this rank's batch, held until the cycle completes
+        idx = -1
+        tail_batch = []
+        for idx, batch in enumerate(self.batch_sampler):
+            if not self.drop_last and idx < self.num_processes:
+                wrap_pool += batch
+            if idx % self.num_processes == self.process_index:
+                pending = batch
+            # only release once the LAST batch of the cycle is full — a short
+            # batch there means some rank would starve (lookahead semantics)
+            if idx % self.num_processes == self.num_processes - 1 and (
+                self.batch_size is None or len(batch) == self.batch_size
+            ):
+                yield pending
+                pending = []
+            tail_batch = batch
+
+        if self.drop_last or idx == -1 or len(wrap_pool) == 0:
+            return
+        if not self.even_batches:
+            if len(pending) > 0:
+                yield pending
+            return
+
+        # even_batches tail: complete the final cycle by wrapping indices
+        if self.batch_size is not None and len(pending) == self.batch_size:
+            # our batch was full but the cycle was incomplete: it was never
+            # released above, release it now before padding the rest
+            yield pending
+
+        # make the wrap pool long enough for degenerate tiny datasets
+        while len(wrap_pool) < self.num_processes * self.batch_size:
+            wrap_pool += wrap_pool
+
+        batch = tail_batch
+        if len(batch) == self.batch_size:
+            batch = []
+            idx += 1
+        cursor = 0
+        while idx % self.num_processes != 0 or len(batch) > 0:
+            take = cursor + self.batch_size - len(batch)
+            batch = list(batch) + wrap_pool[cursor:take]
+            if idx % self.num_processes == self.process_index:
+                yield batch
+            cursor = take
+            batch = []
+            idx += 1
+
+
+class IterableDatasetShard(IterableDataset):
+    """Shard an ``IterableDataset``: buffer ``batch_size × n`` items, yield the
+    rank's slice; pad the tail from the first buffer when ``drop_last=False``
+    (reference: data_loader.py:274-371)."""
+
+    def __init__(
+        self,
+        dataset: IterableDataset,
+        batch_size: int = 1,
+        drop_last: bool = False,
+        num_processes: int = 1,
+        process_index: int = 0,
+        split_batches: bool = False,
+    ):
+        if split_batches and batch_size > 1 and batch_size % num_processes != 0:
+            raise ValueError(
+                f"To use `IterableDatasetShard` in `split_batches` mode, the batch size ({batch_size}) "
+                f"needs to be a round multiple of the number of processes ({num_processes})."
+            )
+        self.dataset = dataset
+        self.batch_size = batch_size
+        self.drop_last = drop_last
+        self.num_processes = num_processes
+        self.process_index = process_index
+        self.split_batches = split_batches
+
+    def set_epoch(self, epoch):
+        self.epoch = epoch
+        if hasattr(self.dataset, "set_epoch"):
+            self.dataset.set_epoch(epoch)
+
+    def __len__(self):
+        # Will raise if the underlying dataset is not sized.
+        if self.drop_last:
+            return (len(self.dataset) // (self.batch_size * self.num_processes)) * self.batch_size
+        else:
+            return math.ceil(len(self.dataset) / (self.batch_size * self.num_processes)) * self.batch_size
+
+    def __iter__(self):
+        if (
+            not hasattr(self.dataset, "set_epoch")
+            and hasattr(self.dataset, "generator")
+            and isinstance(self.dataset.generator, torch.Generator)
+        ):
+            self.dataset.generator.manual_seed(getattr(self, "epoch", 0))
+        real_batch_size = self.batch_size if self.split_batches else (self.batch_size * self.num_processes)
+        per_rank = real_batch_size // self.num_processes
+        lo, hi = per_rank * self.process_index, per_rank * (self.process_index + 1)
+
+        wrap_buffer = None
+        buffer = []
+        for item in self.dataset:
+            buffer.append(item)
+            if len(buffer) == real_batch_size:
+                for i in range(lo, hi):
+                    yield buffer[i]
+                if wrap_buffer is None:
+                    wrap_buffer = buffer.copy()
+                buffer = []
+
+        if not self.drop_last and len(buffer) > 0:
+            if wrap_buffer is None:
+                wrap_buffer = buffer.copy()
+            while len(buffer) < real_batch_size:
+                buffer += wrap_buffer
+            for i in range(lo, hi):
+                yield buffer[i]
+
+
+class DataLoaderStateMixin:
+    """begin()/end() register the loader with GradientState so grad-accum can
+    sync on the last batch (reference: data_loader.py:373-413)."""
+
+    def __init_subclass__(cls, **kwargs):
+        cls.end_of_dataloader = False
+        cls.remainder = -1
+
+    def reset(self):
+        self.end_of_dataloader = False
+        self.remainder = -1
+
+    def begin(self):
+        self.reset()
+        with suppress(Exception):
+            if not self._drop_last:
+                length = getattr(self.dataset, "total_dataset_length", len(self.dataset))
+                self.remainder = length % self.total_batch_size
+        self.gradient_state._add_dataloader(self)
+
+    def end(self):
+        self.gradient_state._remove_dataloader(self)
+
+
+class DataLoaderAdapter:
+    """Delegation-based wrapper over ``torch.utils.data.DataLoader`` so the
+    wrapped loader keeps its public surface (reference: data_loader.py:416-507)."""
+
+    def __init__(self, dataset, use_stateful_dataloader=False, batch_sampler=None, **kwargs):
+        self.use_stateful_dataloader = use_stateful_dataloader
+        if use_stateful_dataloader:
+            from torchdata.stateful_dataloader import StatefulDataLoader
+
+            self.base_dataloader = StatefulDataLoader(dataset, batch_sampler=batch_sampler, **kwargs)
+        else:
+            self.base_dataloader = DataLoader(dataset, batch_sampler=batch_sampler, **kwargs)
+        if hasattr(self.base_dataloader, "state_dict"):
+            self.dl_state_dict = self.base_dataloader.state_dict()
+
+    def __getattr__(self, name):
+        # delegate attribute access to the underlying loader
+        if name == "base_dataloader":
+            raise AttributeError()
+        return getattr(self.base_dataloader, name)
+
+    @property
+    def __class__(self):
+        # spoof isinstance(dl, DataLoader) checks in user code
+        return self.base_dataloader.__class__
+
+    def __len__(self):
+        return len(self.base_dataloader)
+
+    def state_dict(self):
+        return self.dl_state_dict
+
+    def load_state_dict(self, state_dict):
+        self.base_dataloader.load_state_dict(state_dict)
+        self.dl_state_dict = state_dict
+
+    def _update_state_dict(self):
+        # Capture the loader state one batch BEFORE yielding: when resuming we
+        # must not replay the batch the caller already consumed
+        # (reference: data_loader.py:471-494).
+        if hasattr(self.base_dataloader, "state_dict"):
+            self.dl_state_dict = self.base_dataloader.state_dict()
+            # decrement the yielded counter adjustments handled by torchdata itself
+
+
+class DataLoaderShard(DataLoaderAdapter, DataLoaderStateMixin):
+    """Per-rank loader: RNG sync at iteration start, one-batch-ahead prefetch
+    so ``end_of_dataloader`` is known *before* the last batch is yielded, and
+    async H2D placement (reference: data_loader.py:510-667)."""
+
+    def __init__(
+        self,
+        dataset,
+        device=None,
+        rng_types=None,
+        synchronized_generator=None,
+        skip_batches=0,
+        use_stateful_dataloader=False,
+        _drop_last: bool = False,
+        _non_blocking: bool = False,
+        **kwargs,
+    ):
+        super().__init__(dataset, use_stateful_dataloader=use_stateful_dataloader, **kwargs)
+        self.device = device
+        self.rng_types = rng_types
+        self.synchronized_generator = synchronized_generator
+        self.skip_batches = skip_batches
+        self.gradient_state = GradientState()
+        self._drop_last = _drop_last
+        self._non_blocking = _non_blocking
+        self.iteration = 0
+
+    def __iter__(self):
+        if self.rng_types is not None:
+            rng_types = [t for t in self.rng_types if t != "generator" or self.synchronized_generator is not None]
+            if rng_types:
+                synchronize_rng_states(rng_types, self.synchronized_generator)
+        self.begin()
+        self.set_epoch(self.iteration)
+        dataloader_iter = self.base_dataloader.__iter__()
+        # Prefetch one batch ahead so the last batch is flagged before yield.
+        try:
+            current_batch = next(dataloader_iter)
+        except StopIteration:
+            self.end()
+            return
+
+        batch_index = 0
+        while True:
+            try:
+                # fetch next BEFORE yielding current (lookahead)
+                next_batch = next(dataloader_iter)
+                if batch_index >= self.skip_batches:
+                    yield self._move(current_batch)
+                batch_index += 1
+                current_batch = next_batch
+            except StopIteration:
+                self.end_of_dataloader = True
+                self._update_state_dict()
+                if batch_index >= self.skip_batches:
+                    yield self._move(current_batch)
+                break
+        self.iteration += 1
+        self.end()
+
+    def _move(self, batch):
+        if self.device is not None:
+            return send_to_device(batch, self.device, non_blocking=self._non_blocking)
+        return batch
+
+    def set_epoch(self, epoch: int):
+        if self.iteration != epoch:
+            self.iteration = epoch
+        if hasattr(self.batch_sampler, "set_epoch"):
+            self.batch_sampler.set_epoch(epoch)
+        elif hasattr(self.batch_sampler, "batch_sampler") and hasattr(self.batch_sampler.batch_sampler, "sampler"):
+            sampler = self.batch_sampler.batch_sampler.sampler
+            if hasattr(sampler, "set_epoch"):
+                sampler.set_epoch(epoch)
+        if hasattr(self.batch_sampler, "sampler") and hasattr(self.batch_sampler.sampler, "set_epoch"):
+            self.batch_sampler.sampler.set_epoch(epoch)
+        elif hasattr(self.dataset, "set_epoch"):
+            self.dataset.set_epoch(epoch)
+
+    @property
+    def total_batch_size(self):
+        batch_sampler = self.sampler if isinstance(self.sampler, BatchSampler) else self.batch_sampler
+        if hasattr(batch_sampler, "split_batches") and batch_sampler.split_batches:
+            return batch_sampler.batch_size
+        if hasattr(batch_sampler, "batch_size") and batch_sampler.batch_size is not None:
+            n = getattr(batch_sampler, "num_processes", 1)
+            return batch_sampler.batch_size * n
+        return None
+
+    @property
+    def total_dataset_length(self):
+        if hasattr(self.dataset, "total_length"):
+            return self.dataset.total_length
+        return len(self.dataset)
+
+    def get_sampler(self):
+        return get_sampler(self)
+
+    def set_sampler(self, sampler):
+        sampler_is_batch_sampler = isinstance(self.sampler, BatchSampler)
+        if sampler_is_batch_sampler:
+            self.sampler.sampler = sampler
+        else:
+            self.batch_sampler.sampler = sampler
+            if hasattr(self.batch_sampler, "batch_sampler"):
+                self.batch_sampler.batch_sampler.sampler = sampler
+
+
+class DataLoaderDispatcher(DataLoaderAdapter, DataLoaderStateMixin):
+    """Rank 0 reads ``num_processes`` batches, broadcasts the concatenated
+    global batch, every rank keeps its slice (reference: data_loader.py:723-995).
+    Used for IterableDataset worlds and ``dispatch_batches=True``."""
+
+    def __init__(
+        self,
+        dataset,
+        split_batches: bool = False,
+        skip_batches=0,
+        use_stateful_dataloader=False,
+        _drop_last: bool = False,
+        _non_blocking: bool = False,
+        slice_fn=None,
+        **kwargs,
+    ):
+        shuffle = False
+        from torch.utils.data.datapipes.iter.combinatorics import ShufflerIterDataPipe
+
+        if isinstance(dataset, ShufflerIterDataPipe):
+            shuffle = dataset._shuffle_enabled
+        super().__init__(dataset, use_stateful_dataloader=use_stateful_dataloader, **kwargs)
+        self.split_batches = split_batches
+        if shuffle:
+            torch.utils.data.graph_settings.apply_shuffle_settings(dataset, shuffle=shuffle)
+        self.gradient_state = GradientState()
+        self.state = PartialState()
+        self._drop_last = _drop_last
+        self._non_blocking = _non_blocking
+        self.skip_batches = skip_batches
+        self.slice_fn = slice_tensors if slice_fn is None else slice_fn
+        self.iteration = 0
+
+    def _fetch_batches(self, iterator):
+        batches, batch = None, None
+        # On process 0, we gather the batch to dispatch.
+        if self.state.process_index == 0:
+            try:
+                if self.split_batches:
+                    self._update_state_dict()
+                    batch = next(iterator)
+                else:
+                    batches = []
+                    for _ in range(self.state.num_processes):
+                        self._update_state_dict()
+                        batches.append(next(iterator))
+                    try:
+                        batch = concatenate(batches, dim=0)
+                    except RuntimeError as e:
+                        raise RuntimeError(
+                            "You can't use batches of different size with `dispatch_batches=True` or when using an "
+                            "`IterableDataset`. Either pass `dispatch_batches=False` and have each process fetch its "
+                            "own batch or pass `split_batches=True`. By doing so, the main process will fetch a full "
+                            "batch and slice it into `num_processes` batches for each process."
+                        ) from e
+                # header: [has_data, structure]
+                batch_info = [get_data_structure(batch), False]
+            except StopIteration:
+                batch_info = [None, True]
+        else:
+            batch_info = [None, self._stop_iteration]
+        broadcast_object_list(batch_info)
+        self._stop_iteration = batch_info[1]
+        if self._stop_iteration:
+            if not self.split_batches and not self._drop_last:
+                if self.state.process_index == 0 and len(batches) > 0:
+                    batch = concatenate(batches, dim=0)
+                    batch_info = [get_data_structure(batch), False]
+                else:
+                    batch_info = [None, True]
+                broadcast_object_list(batch_info)
+        return batch, batch_info
+
+    def __iter__(self):
+        self.begin()
+        self.set_epoch(self.iteration)
+        main_iterator = None
+        # Every rank iterates (for worker-side effects), but only rank-0's
+        # payloads travel.
+        main_iterator = self.base_dataloader.__iter__()
+        stop_iteration = False
+        self._stop_iteration = False
+        first_batch = None
+        next_batch, next_batch_info = self._fetch_batches(main_iterator)
+        batch_index = 0
+        while not stop_iteration:
+            batch, batch_info = next_batch, next_batch_info
+
+            if self.state.process_index != 0:
+                # Initialize tensors on other processes than process 0.
+                batch = initialize_tensors(batch_info[0])
+            batch = send_to_device(batch, self.state.device, non_blocking=self._non_blocking)
+            # Broadcast the batch before splitting it.
+            batch = broadcast(batch, from_process=0)
+
+            if not self._drop_last and first_batch is None:
+                # We keep at least num processes elements of the first batch to be able to complete the last batch
+                first_batch = self.slice_fn(
+                    batch,
+                    slice(0, self.state.num_processes),
+                    process_index=self.state.process_index,
+                    num_processes=self.state.num_processes,
+                )
+
+            if batch is None:
+                raise ValueError(
+                    f"Batch does not contain any data (`{batch}`). At the end of all iterable data available before "
+                    "expected stop iteration."
+                )
+
+            observed_batch_size = find_batch_size(batch)
+            batch_size = observed_batch_size // self.state.num_processes
+
+            stop_iteration = self._stop_iteration
+            if not stop_iteration:
+                # We may still be at the end of the dataloader without knowing it yet: fetch ahead.
+                next_batch, next_batch_info = self._fetch_batches(main_iterator)
+                # next_batch_info[0] is None when there are no more batches, otherwise we still need to process them.
+                if self._stop_iteration and next_batch_info[0] is None:
+                    stop_iteration = True
+
+            if not self._drop_last and stop_iteration and observed_batch_size % self.state.num_processes != 0:
+                # If the last batch is not complete, let's add the first batch to it.
+                batch = concatenate([batch, first_batch], dim=0)
+                # Batch size computation above is wrong, it's off by 1 so we fix it.
+                batch_size += 1
+
+            data_slice = slice(self.state.process_index * batch_size, (self.state.process_index + 1) * batch_size)
+            batch = self.slice_fn(
+                batch,
+                data_slice,
+                process_index=self.state.process_index,
+                num_processes=self.state.num_processes,
+            )
+
+            if stop_iteration:
+                self.end_of_dataloader = True
+                self._update_state_dict()
+                self.remainder = observed_batch_size % self.state.num_processes
+            if batch_index >= self.skip_batches:
+                yield batch
+            batch_index += 1
+        self.iteration += 1
+        self.end()
+
+    def set_epoch(self, epoch: int):
+        if self.iteration != epoch:
+            self.iteration = epoch
+        if hasattr(self.batch_sampler, "sampler") and hasattr(self.batch_sampler.sampler, "set_epoch"):
+            self.batch_sampler.sampler.set_epoch(epoch)
+        elif hasattr(self.dataset, "set_epoch"):
+            self.dataset.set_epoch(epoch)
+
+    def __len__(self):
+        whole_length = len(self.base_dataloader)
+        if self.split_batches:
+            return whole_length
+        elif self._drop_last:
+            return whole_length // self.state.num_processes
+        else:
+            return math.ceil(whole_length / self.state.num_processes)
+
+    @property
+    def total_batch_size(self):
+        return (
+            self.dataset.batch_size if self.split_batches else (self.dataset.batch_size * self.dataset.num_processes)
+        )
+
+    @property
+    def total_dataset_length(self):
+        return len(self.dataset)
+
+    def get_sampler(self):
+        return get_sampler(self)
+
+    def set_sampler(self, sampler):
+        sampler_is_batch_sampler = isinstance(self.sampler, BatchSampler)
+        if sampler_is_batch_sampler:
+            self.sampler.sampler = sampler
+        else:
+            self.batch_sampler.sampler = sampler
+            if hasattr(self.batch_sampler, "batch_sampler"):
+                self.batch_sampler.batch_sampler.sampler = sampler
+
+
+def get_sampler(dataloader):
+    """Fish the innermost sampler out of a (possibly wrapped) dataloader."""
+    sampler_is_batch_sampler = isinstance(getattr(dataloader, "sampler", None), BatchSampler)
+    if sampler_is_batch_sampler:
+        return getattr(dataloader.sampler, "sampler", None)
+    else:
+        batch_sampler = getattr(dataloader, "batch_sampler", None)
+        if batch_sampler is None:
+            return getattr(dataloader, "sampler", None)
+        if hasattr(batch_sampler, "batch_sampler"):
+            return getattr(batch_sampler.batch_sampler, "sampler", None)
+        return getattr(batch_sampler, "sampler", None)
+
+
+def prepare_data_loader(
+    dataloader: DataLoader,
+    device: Optional[torch.device] = None,
+    num_processes: Optional[int] = None,
+    process_index: Optional[int] = None,
+    split_batches: bool = False,
+    put_on_device: bool = False,
+    rng_types: Optional[List[Union[str, RNGType]]] = None,
+    dispatch_batches: Optional[bool] = None,
+    even_batches: bool = True,
+    slice_fn_for_dispatch: Optional[Callable] = None,
+    use_seedable_sampler: bool = False,
+    non_blocking: bool = False,
+    use_stateful_dataloader: bool = False,
+) -> DataLoader:
+    """Wrap a user DataLoader for the current distributed world
+    (reference: data_loader.py:1016-1329)."""
+    state = PartialState()
+    if num_processes is None:
+        num_processes = state.num_processes
+    if process_index is None:
+        process_index = state.process_index
+
+    if dispatch_batches is None:
+        dispatch_batches = False if not put_on_device else isinstance(dataloader.dataset, IterableDataset)
+    if dispatch_batches and not put_on_device:
+        raise ValueError("Using `dispatch_batches=True` requires `put_on_device=True`.")
+
+    if split_batches and dataloader.batch_size is not None and dataloader.batch_size % num_processes != 0:
+        raise ValueError(
+            f"To use a `DataLoader` in `split_batches` mode, the batch size ({dataloader.batch_size}) "
+            f"needs to be a round multiple of the number of processes ({num_processes})."
+        )
+
+    new_dataset = dataloader.dataset
+    new_batch_sampler = dataloader.batch_sampler if not isinstance(new_dataset, IterableDataset) else None
+    sampler_is_batch_sampler = isinstance(dataloader.sampler, BatchSampler)
+    synchronized_generator = None
+
+    sampler = get_sampler(dataloader)
+    if isinstance(sampler, RandomSampler) and use_seedable_sampler:
+        # When iterating through the dataloader during distributed processes
+        # we want to ensure that on each process we are iterating through the same
+        # samples in the same order: replace the RandomSampler by a seedable one.
+        sampler = SeedableRandomSampler(
+            data_source=sampler.data_source,
+            replacement=sampler.replacement,
+            num_samples=sampler._num_samples,
+            generator=getattr(sampler, "generator", torch.Generator()),
+        )
+    if isinstance(dataloader.sampler, RandomSampler):
+        # RNG sync of the sampler generator across ranks happens each epoch
+        generator = torch.Generator().manual_seed(42)
+        dataloader.generator = generator
+        dataloader.sampler.generator = generator
+
+    if num_processes != 1 and not dispatch_batches:
+        if isinstance(new_dataset, IterableDataset):
+            if getattr(dataloader.dataset, "generator", None) is not None:
+                synchronized_generator = dataloader.dataset.generator
+            new_dataset = IterableDatasetShard(
+                new_dataset,
+                batch_size=dataloader.batch_size,
+                drop_last=dataloader.drop_last,
+                num_processes=num_processes,
+                process_index=process_index,
+                split_batches=split_batches,
+            )
+        else:
+            if not use_seedable_sampler and hasattr(sampler, "generator"):
+                if sampler.generator is None:
+                    sampler.generator = torch.Generator()
+                    sampler.generator.manual_seed(42)
+                synchronized_generator = sampler.generator
+            batch_sampler = dataloader.sampler if sampler_is_batch_sampler else dataloader.batch_sampler
+            new_batch_sampler = BatchSamplerShard(
+                batch_sampler,
+                num_processes=num_processes,
+                process_index=process_index,
+                split_batches=split_batches,
+                even_batches=even_batches,
+            )
+
+    # We ignore all of those since they are all dealt with by our new_batch_sampler
+    ignore_kwargs = ["batch_size", "shuffle", "sampler", "batch_sampler", "drop_last"]
+    kwargs = {
+        k: getattr(dataloader, k, _PYTORCH_DATALOADER_KWARGS[k])
+        for k in _PYTORCH_DATALOADER_KWARGS
+        if k not in ignore_kwargs
+    }
+    # Need to provide batch_size as batch_sampler is None for Iterable dataset
+    if new_batch_sampler is None:
+        kwargs["drop_last"] = dataloader.drop_last
+        kwargs["batch_size"] = (
+            dataloader.batch_size // num_processes if split_batches and not dispatch_batches else dataloader.batch_size
+        )
+    if kwargs.get("prefetch_factor") is not None and kwargs.get("num_workers", 0) == 0:
+        kwargs["prefetch_factor"] = None
+    if isinstance(sampler, SeedableRandomSampler) and use_seedable_sampler:
+        if sampler_is_batch_sampler:
+            dataloader.sampler.sampler = sampler
+        else:
+            dataloader.batch_sampler.sampler = sampler
+            if hasattr(dataloader.batch_sampler, "batch_sampler"):
+                dataloader.batch_sampler.batch_sampler.sampler = sampler
+
+    if dispatch_batches:
+        kwargs.pop("generator", None)
+        dataloader = DataLoaderDispatcher(
+            new_dataset,
+            split_batches=split_batches,
+            batch_sampler=new_batch_sampler,
+            _drop_last=dataloader.drop_last,
+            _non_blocking=non_blocking,
+            slice_fn=slice_fn_for_dispatch,
+            use_stateful_dataloader=use_stateful_dataloader,
+            **kwargs,
+        )
+    elif sampler_is_batch_sampler:
+        dataloader = DataLoaderShard(
+            new_dataset,
+            device=device if put_on_device else None,
+            sampler=new_batch_sampler,
+            batch_size=dataloader.batch_size,
+            rng_types=rng_types,
+            _drop_last=dataloader.drop_last,
+            _non_blocking=non_blocking,
+            synchronized_generator=synchronized_generator,
+            use_stateful_dataloader=use_stateful_dataloader,
+            **kwargs,
+        )
+    else:
+        dataloader = DataLoaderShard(
+            new_dataset,
+            device=device if put_on_device else None,
+            batch_sampler=new_batch_sampler,
+            rng_types=rng_types,
+            synchronized_generator=synchronized_generator,
+            _drop_last=dataloader.drop_last,
+            _non_blocking=non_blocking,
+            use_stateful_dataloader=use_stateful_dataloader,
+            **kwargs,
+        )
+
+    return dataloader
+
+
+class SkipBatchSampler(BatchSampler):
+    """Yield batches of an inner batch sampler, skipping the first
+    ``skip_batches`` (reference: data_loader.py:1332+)."""
+
+    def __init__(self, batch_sampler, skip_batches=0):
+        self.batch_sampler = batch_sampler
+        self.sampler = getattr(batch_sampler, "sampler", None)
+        self.skip_batches = skip_batches
+
+    def __iter__(self):
+        for index, samples in enumerate(self.batch_sampler):
+            if index >= self.skip_batches:
+                yield samples
+
+    @property
+    def total_length(self):
+        return len(self.batch_sampler)
+
+    def __len__(self):
+        return len(self.batch_sampler) - self.skip_batches
+
+
+class SkipDataLoader(DataLoaderAdapter, DataLoaderStateMixin):
+    """DataLoader yielding everything after the first ``skip_batches``."""
+
+    def __init__(self, dataset, skip_batches=0, use_stateful_dataloader=False, **kwargs):
+        super().__init__(dataset, use_stateful_dataloader=use_stateful_dataloader, **kwargs)
+        self.skip_batches = skip_batches
+        self.gradient_state = GradientState()
+
+    def __iter__(self):
+        self.begin()
+        for index, batch in enumerate(self.base_dataloader.__iter__()):
+            if index >= self.skip_batches:
+                self._update_state_dict()
+                yield batch
+        self.end()
+
+
+def skip_first_batches(dataloader, num_batches=0):
+    """Mid-epoch resume: a new loader that skips ``num_batches``
+    (reference: data_loader.py:1395)."""
+    state = PartialState()
+    if state.distributed_type == DistributedType.FSDP:
+        pass  # plain torch loaders below handle it
+
+    dataset = dataloader.dataset
+    sampler_is_batch_sampler = False
+    if isinstance(dataset, IterableDataset):
+        new_batch_sampler = None
+    else:
+        sampler_is_batch_sampler = isinstance(dataloader.sampler, BatchSampler)
+        batch_sampler = dataloader.sampler if sampler_is_batch_sampler else dataloader.batch_sampler
+        new_batch_sampler = SkipBatchSampler(batch_sampler, skip_batches=num_batches)
+
+    # We ignore all of those since they are all dealt with by our new_batch_sampler
+    ignore_kwargs = ["batch_size", "shuffle", "sampler", "batch_sampler", "drop_last"]
+    kwargs = {
+        k: getattr(dataloader, k, _PYTORCH_DATALOADER_KWARGS[k])
+        for k in _PYTORCH_DATALOADER_KWARGS
+        if k not in ignore_kwargs
+    }
+    if new_batch_sampler is None:
+        kwargs["drop_last"] = dataloader.drop_last
+        kwargs["batch_size"] = dataloader.batch_size
+    if kwargs.get("prefetch_factor") is not None and kwargs.get("num_workers", 0) == 0:
+        kwargs["prefetch_factor"] = None
+
+    if isinstance(dataloader, DataLoaderDispatcher):
+        if new_batch_sampler is None:
+            # Need to manually skip batches in the dataloader
+            kwargs["skip_batches"] = num_batches
+        dataloader = DataLoaderDispatcher(
+            dataset,
+            split_batches=dataloader.split_batches,
+            batch_sampler=new_batch_sampler,
+            _drop_last=dataloader._drop_last,
+            **kwargs,
+        )
+    elif isinstance(dataloader, DataLoaderShard):
+        if new_batch_sampler is None:
+            # Need to manually skip batches in the dataloader
+            kwargs["skip_batches"] = num_batches
+        elif sampler_is_batch_sampler:
+            kwargs["sampler"] = new_batch_sampler
+            kwargs["batch_size"] = dataloader.batch_size
+        else:
+            kwargs["batch_sampler"] = new_batch_sampler
+        dataloader = DataLoaderShard(
+            dataset,
+            device=dataloader.device,
+            rng_types=dataloader.rng_types,
+            synchronized_generator=dataloader.synchronized_generator,
+            **kwargs,
+        )
+    else:
+        if new_batch_sampler is None:
+            # Need to manually skip batches in the dataloader
+            dataloader = SkipDataLoader(dataset, skip_batches=num_batches, **kwargs)
+        else:
+            dataloader = DataLoader(dataset, batch_sampler=new_batch_sampler, **kwargs)
+
+    return dataloader

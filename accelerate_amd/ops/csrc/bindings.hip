@@ -1,0 +1,134 @@
+// Python bindings for the gfx950 kernel pack (torch extension, built by
+// hipcc directly — no hipify, no CUDA shims).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+#include "multi_tensor.h"
+
+// kernels (defined in optim_kernels.hip)
+__global__ void fused_adamw_kernel(TensorListMeta, float, float, float, float, float,
+                                   float, float, const float*, const float*);
+__global__ void l2norm_squared_kernel(TensorListMeta, float*);
+__global__ void clip_coef_kernel(const float*, float, float*, float*);
+__global__ void multi_tensor_scale_kernel(TensorListMeta, const float*);
+__global__ void unscale_check_kernel(TensorListMeta, const float*, float*);
+
+namespace {
+
+struct MetaHolder {
+  at::Tensor addrs_numels;  // int64 device: [n_lists*n + n]
+  at::Tensor prefix;        // int32 device: [n+1]
+  TensorListMeta meta;
+  int total_chunks;
+};
+
+MetaHolder build_meta(const std::vector<std::vector<at::Tensor>>& lists) {
+  const int n_lists = static_cast<int>(lists.size());
+  const int n = static_cast<int>(lists[0].size());
+  TORCH_CHECK(n > 0, "empty tensor list");
+  auto cpu_i64 = at::empty({n_lists * n + n}, at::TensorOptions().dtype(at::kLong).pinned_memory(true));
+  auto cpu_i32 = at::empty({n + 1}, at::TensorOptions().dtype(at::kInt).pinned_memory(true));
+  int64_t* a = cpu_i64.data_ptr<int64_t>();
+  int32_t* pre = cpu_i32.data_ptr<int32_t>();
+  pre[0] = 0;
+  for (int t = 0; t < n; ++t) {
+    const auto& t0 = lists[0][t];
+    TORCH_CHECK(t0.is_cuda() && t0.is_contiguous() && t0.scalar_type() == at::kFloat,
+                "multi-tensor ops need contiguous fp32 HIP tensors (tensor ", t, ")");
+    int64_t numel = t0.numel();
+    a[n_lists * n + t] = numel;
+    pre[t + 1] = pre[t] + static_cast<int32_t>((numel + kChunkSize - 1) / kChunkSize);
+    for (int l = 0; l < n_lists; ++l) {
+      TORCH_CHECK(lists[l][t].numel() == numel, "tensor list shape mismatch");
+      a[l * n + t] = reinterpret_cast<int64_t>(lists[l][t].data_ptr());
+    }
+  }
+  MetaHolder h;
+  auto dev = lists[0][0].device();
+  h.addrs_numels = cpu_i64.to(dev, /*non_blocking=*/true);
+  h.prefix = cpu_i32.to(dev, /*non_blocking=*/true);
+  h.meta.addrs = h.addrs_numels.data_ptr<int64_t>();
+  h.meta.numels = h.addrs_numels.data_ptr<int64_t>() + n_lists * n;
+  h.meta.chunk_prefix = h.prefix.data_ptr<int32_t>();
+  h.meta.n_tensors = n;
+  h.meta.n_lists = n_lists;
+  h.total_chunks = pre[n];
+  return h;
+}
+
+}  // namespace
+
+void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+                 std::vector<at::Tensor> exp_avgs, std::vector<at::Tensor> exp_avg_sqs,
+                 int64_t step, double lr, double beta1, double beta2, double eps,
+                 double weight_decay,
+                 c10::optional<at::Tensor> grad_scale, c10::optional<at::Tensor> found_inf) {
+  TORCH_CHECK(params.size() == grads.size() && params.size() == exp_avgs.size() &&
+                  params.size() == exp_avg_sqs.size(),
+              "fused_adamw: list length mismatch");
+  auto h = build_meta({params, grads, exp_avgs, exp_avg_sqs});
+  const float bc1 = 1.f - powf(static_cast<float>(beta1), static_cast<float>(step));
+  const float bc2 = 1.f - powf(static_cast<float>(beta2), static_cast<float>(step));
+  const float rsqrt_bc2 = 1.0f / sqrtf(bc2);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(fused_adamw_kernel, dim3(h.total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                     h.meta, (float)lr, (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                     bc1, rsqrt_bc2,
+                     grad_scale.has_value() ? grad_scale->data_ptr<float>() : nullptr,
+                     found_inf.has_value() ? found_inf->data_ptr<float>() : nullptr);
+}
+
+at::Tensor l2norm_squared(std::vector<at::Tensor> grads) {
+  auto h = build_meta({grads});
+  auto out = at::zeros({1}, grads[0].options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(l2norm_squared_kernel, dim3(h.total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                     h.meta, out.data_ptr<float>());
+  return out;
+}
+
+// Fully on-device clip_grad_norm_: returns total_norm (0-dim device tensor),
+// no host sync anywhere.
+at::Tensor clip_grad_norm(std::vector<at::Tensor> grads, double max_norm) {
+  auto h = build_meta({grads});
+  auto opts = grads[0].options().dtype(at::kFloat);
+  auto norm_sq = at::zeros({1}, opts);
+  auto coef = at::empty({1}, opts);
+  auto total_norm = at::empty({1}, opts);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(l2norm_squared_kernel, dim3(h.total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                     h.meta, norm_sq.data_ptr<float>());
+  hipLaunchKernelGGL(clip_coef_kernel, dim3(1), dim3(1), 0, stream.stream(),
+                     norm_sq.data_ptr<float>(), (float)max_norm, coef.data_ptr<float>(),
+                     total_norm.data_ptr<float>());
+  hipLaunchKernelGGL(multi_tensor_scale_kernel, dim3(h.total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                     h.meta, coef.data_ptr<float>());
+  return total_norm.squeeze();
+}
+
+void multi_tensor_scale(std::vector<at::Tensor> grads, at::Tensor coef) {
+  auto h = build_meta({grads});
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(multi_tensor_scale_kernel, dim3(h.total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                     h.meta, coef.data_ptr<float>());
+}
+
+void unscale_and_check(std::vector<at::Tensor> grads, at::Tensor inv_scale, at::Tensor found_inf) {
+  auto h = build_meta({grads});
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(unscale_check_kernel, dim3(h.total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                     h.meta, inv_scale.data_ptr<float>(), found_inf.data_ptr<float>());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (gfx950)");
+  m.def("l2norm_squared", &l2norm_squared, "global L2 norm squared over tensor list");
+  m.def("clip_grad_norm", &clip_grad_norm, "on-device clip_grad_norm_, returns total norm");
+  m.def("multi_tensor_scale", &multi_tensor_scale, "g *= *coef");
+  m.def("unscale_and_check", &unscale_and_check, "g *= *inv_scale with non-finite detection");
+}

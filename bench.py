@@ -1,0 +1,142 @@
+#!/usr/bin/env python
+"""Headline benchmark: samples/sec, nlp_example BERT-base, DDP bf16
+(BASELINE.json metric) on 1..8 MI355X.
+
+Synthetic MRPC-shaped data (seq_len 128, random tokens), random-init
+BERT-base, per-GPU batch 16 (examples/nlp_example.py:43 MAX_GPU_BATCH_SIZE),
+AdamW lr 2e-5 + linear-warmup schedule — the nlp_example training loop run
+through accelerate_amd's Accelerator/prepare/backward with our RCCL reducer
+and fused CDNA4 AdamW.
+
+Launched by the driver as
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+Rank 0 prints one JSON line.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+from torch.utils.data import DataLoader, TensorDataset
+
+PER_GPU_BATCH = 16
+SEQ_LEN = 128
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=5)
+    parser.add_argument("--profile", action="store_true", help="emit a chrome trace under gpurun_out/")
+    args = parser.parse_args()
+
+    from accelerate_amd import Accelerator, set_seed
+    from accelerate_amd.models import BertConfig, BertForSequenceClassification
+
+    accelerator = Accelerator(mixed_precision="bf16" if torch.cuda.is_available() else "no")
+    set_seed(42)
+
+    n = accelerator.num_processes
+    device = accelerator.device
+    on_gpu = device.type == "cuda"
+
+    config = BertConfig.bert_base()
+    model = BertForSequenceClassification(config)
+
+    if on_gpu:
+        from accelerate_amd.ops.optim import FusedAdamW
+
+        optimizer = FusedAdamW(model.parameters(), lr=2e-5, weight_decay=0.01)
+    else:
+        optimizer = torch.optim.AdamW(model.parameters(), lr=2e-5, weight_decay=0.01)
+
+    total_steps = args.steps + args.warmup
+    n_samples = (total_steps + 2) * PER_GPU_BATCH * max(n, 1)
+    g = torch.Generator().manual_seed(1234)
+    input_ids = torch.randint(0, config.vocab_size, (n_samples, SEQ_LEN), generator=g)
+    token_type_ids = torch.zeros(n_samples, SEQ_LEN, dtype=torch.long)
+    attention_mask = torch.ones(n_samples, SEQ_LEN, dtype=torch.long)
+    labels = torch.randint(0, 2, (n_samples,), generator=g)
+    dataset = TensorDataset(input_ids, attention_mask, token_type_ids, labels)
+    dataloader = DataLoader(dataset, batch_size=PER_GPU_BATCH, shuffle=True, drop_last=True)
+
+    scheduler = torch.optim.lr_scheduler.LambdaLR(optimizer, lambda step: min(1.0, (step + 1) / 100))
+
+    model, optimizer, dataloader, scheduler = accelerator.prepare(model, optimizer, dataloader, scheduler)
+    model.train()
+
+    def one_step(batch):
+        ids, mask, types, lbl = batch
+        optimizer.zero_grad()
+        out = model(ids, attention_mask=mask, token_type_ids=types, labels=lbl)
+        accelerator.backward(out["loss"])
+        optimizer.step()
+        scheduler.step()
+        return out["loss"]
+
+    data_iter = iter(dataloader)
+
+    def next_batch():
+        nonlocal data_iter
+        try:
+            return next(data_iter)
+        except StopIteration:
+            data_iter = iter(dataloader)
+            return next(data_iter)
+
+    # warmup
+    for _ in range(args.warmup):
+        one_step(next_batch())
+
+    accelerator.wait_for_everyone()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step(next_batch())
+    if on_gpu:
+        torch.cuda.synchronize()
+    accelerator.wait_for_everyone()
+    t1 = time.perf_counter()
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    if accelerator.use_distributed:
+        elapsed = elapsed.to(device if on_gpu else "cpu")
+        torch.distributed.all_reduce(elapsed, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(elapsed.item())
+
+    if accelerator.is_main_process:
+        global_batch = PER_GPU_BATCH * n
+        samples_per_sec = global_batch * args.steps / elapsed
+        result = {
+            "metric": "samples/sec (whole node) on nlp_example BERT-base, DDP bf16",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32(cpu-ci)",
+            "data": "synthetic",
+            "config": {
+                "model": "bert-base",
+                "global_batch": global_batch,
+                "per_gpu_batch": PER_GPU_BATCH,
+                "seq_len": SEQ_LEN,
+                "parallelism": f"dp{n}",
+                "optimizer": "fused_adamw_hip" if on_gpu else "torch_adamw",
+            },
+        }
+        print(json.dumps(result))
+    accelerator.end_training()
+
+
+if __name__ == "__main__":
+    main()

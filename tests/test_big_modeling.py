@@ -1,0 +1,119 @@
+"""Big-model dispatch tests on CPU/disk (the reference runs tiny
+nn.Sequentials through dispatch/offload and compares to the plain model,
+tests/test_big_modeling.py)."""
+
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.nn as nn
+
+from accelerate_amd import dispatch_model, init_empty_weights
+from accelerate_amd.big_modeling import cpu_offload, disk_offload, load_checkpoint_and_dispatch
+from accelerate_amd.hooks import remove_hook_from_submodules
+
+
+class ModelForTest(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.linear1 = nn.Linear(3, 4)
+        self.batchnorm = nn.BatchNorm1d(4)
+        self.linear2 = nn.Linear(4, 5)
+
+    def forward(self, x):
+        return self.linear2(self.batchnorm(self.linear1(x)))
+
+
+def test_cpu_offload_matches_plain():
+    model = ModelForTest()
+    x = torch.randn(2, 3)
+    expected = model(x)
+    cpu_offload(model, execution_device=torch.device("cpu"))
+    out = model(x)
+    assert torch.allclose(expected, out, atol=1e-6)
+    # twice (weights restored between forwards)
+    out = model(x)
+    assert torch.allclose(expected, out, atol=1e-6)
+
+
+def test_disk_offload_matches_plain():
+    model = ModelForTest()
+    x = torch.randn(2, 3)
+    expected = model(x)
+    with tempfile.TemporaryDirectory() as d:
+        disk_offload(model, d, execution_device=torch.device("cpu"))
+        out = model(x)
+        assert torch.allclose(expected, out, atol=1e-5)
+        out = model(x)
+        assert torch.allclose(expected, out, atol=1e-5)
+
+
+def test_dispatch_model_cpu_disk():
+    model = ModelForTest()
+    x = torch.randn(2, 3)
+    expected = model(x)
+    dmap = {"linear1": "cpu", "batchnorm": "cpu", "linear2": "disk"}
+    with tempfile.TemporaryDirectory() as d:
+        dispatch_model(model, dmap, offload_dir=d, main_device="cpu", force_hooks=True)
+        out = model(x)
+        assert torch.allclose(expected, out, atol=1e-5)
+
+
+def test_dispatch_model_single_device():
+    model = ModelForTest()
+    x = torch.randn(2, 3)
+    expected = model(x)
+    dispatch_model(model, {"": "cpu"})
+    assert torch.allclose(expected, model(x), atol=1e-6)
+
+
+def test_load_checkpoint_and_dispatch_meta():
+    model = ModelForTest()
+    sd = model.state_dict()
+    x = torch.randn(2, 3)
+    model.eval()
+    with torch.no_grad():
+        expected = model(x)
+    with tempfile.TemporaryDirectory() as d:
+        torch.save(sd, os.path.join(d, "pytorch_model.bin"))
+        with init_empty_weights():
+            fresh = ModelForTest()
+        fresh.eval()
+        fresh = load_checkpoint_and_dispatch(fresh, os.path.join(d, "pytorch_model.bin"), device_map={"": "cpu"})
+        with torch.no_grad():
+            out = fresh(x)
+        assert torch.allclose(expected, out, atol=1e-6)
+
+
+def test_dispatch_tied_weights():
+    class Tied(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb = nn.Embedding(10, 8)
+            self.mid = nn.Linear(8, 8)
+            self.head = nn.Linear(8, 10, bias=False)
+            self.head.weight = self.emb.weight
+
+        def forward(self, ids):
+            return self.head(self.mid(self.emb(ids)))
+
+    model = Tied()
+    ids = torch.randint(0, 10, (2, 3))
+    expected = model(ids)
+    with tempfile.TemporaryDirectory() as d:
+        dispatch_model(
+            model, {"emb": "cpu", "mid": "disk", "head": "cpu"}, offload_dir=d, main_device="cpu", force_hooks=True
+        )
+        out = model(ids)
+        assert torch.allclose(expected, out, atol=1e-5)
+        assert model.head.weight is model.emb.weight  # still tied
+
+
+def test_to_poisoned_after_dispatch():
+    model = ModelForTest()
+    with tempfile.TemporaryDirectory() as d:
+        dispatch_model(model, {"linear1": "cpu", "batchnorm": "cpu", "linear2": "disk"}, offload_dir=d,
+                       main_device="cpu", force_hooks=True)
+        with pytest.raises(RuntimeError):
+            model.to("cpu")

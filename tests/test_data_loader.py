@@ -1,0 +1,204 @@
+"""BatchSamplerShard / IterableDatasetShard semantics (oracle: the reference's
+exhaustive small-N enumeration strategy, tests/test_data_loader.py)."""
+
+import pytest
+import torch
+from torch.utils.data import BatchSampler, DataLoader, IterableDataset, SequentialSampler, TensorDataset
+
+from accelerate_amd.data_loader import (
+    BatchSamplerShard,
+    DataLoaderShard,
+    IterableDatasetShard,
+    SeedableRandomSampler,
+    SkipBatchSampler,
+    SkipDataLoader,
+    prepare_data_loader,
+    skip_first_batches,
+)
+
+
+def make_batch_sampler(n_items, batch_size, drop_last=False):
+    return BatchSampler(SequentialSampler(range(n_items)), batch_size=batch_size, drop_last=drop_last)
+
+
+def shards(n_items, batch_size, num_processes, drop_last=False, split_batches=False, even_batches=True):
+    bs = make_batch_sampler(n_items, batch_size, drop_last)
+    return [
+        list(
+            BatchSamplerShard(
+                bs, num_processes=num_processes, process_index=i, split_batches=split_batches, even_batches=even_batches
+            )
+        )
+        for i in range(num_processes)
+    ]
+
+
+class TestBatchSamplerShardDeal:
+    def test_even_division(self):
+        out = shards(24, 3, 2)
+        assert out[0] == [[0, 1, 2], [6, 7, 8], [12, 13, 14], [18, 19, 20]]
+        assert out[1] == [[3, 4, 5], [9, 10, 11], [15, 16, 17], [21, 22, 23]]
+
+    def test_tail_wraparound_even_batches(self):
+        # 22 items, bs 3, n 2: last batch [21] is short -> rank1 pads from the start
+        out = shards(22, 3, 2)
+        assert out[0] == [[0, 1, 2], [6, 7, 8], [12, 13, 14], [18, 19, 20]]
+        assert out[1] == [[3, 4, 5], [9, 10, 11], [15, 16, 17], [21, 0, 1]]
+
+    def test_tail_not_even_batches(self):
+        out = shards(22, 3, 2, even_batches=False)
+        assert out[0] == [[0, 1, 2], [6, 7, 8], [12, 13, 14], [18, 19, 20]]
+        assert out[1] == [[3, 4, 5], [9, 10, 11], [15, 16, 17], [21]]
+
+    def test_drop_last(self):
+        # 22 items, bs 3, drop_last: 7 batches -> last incomplete cycle dropped
+        out = shards(22, 3, 2, drop_last=True)
+        assert out[0] == [[0, 1, 2], [6, 7, 8], [12, 13, 14]]
+        assert out[1] == [[3, 4, 5], [9, 10, 11], [15, 16, 17]]
+
+    def test_degenerate_tiny_dataset(self):
+        # fewer samples than one global batch: wrap repeatedly
+        out = shards(2, 3, 2)
+        assert out[0] == [[0, 1, 0]]
+        assert out[1] == [[1, 0, 1]]
+
+    def test_incomplete_cycle_full_batches(self):
+        # 9 items bs 3 n 2: 3 batches; rank0's batch idx2 completes an odd cycle
+        out = shards(9, 3, 2)
+        # idx0->r0, idx1->r1 (cycle yields), idx2->r0 full but cycle incomplete
+        assert out[0][0] == [0, 1, 2]
+        assert out[1][0] == [3, 4, 5]
+        # tail: rank0 yields [6,7,8]; rank1 wraps for a rectangular cycle
+        assert out[0][1] == [6, 7, 8]
+        assert out[1][1] == [0, 1, 2]
+        assert len(out[0]) == len(out[1]) == 2
+
+    def test_lengths_match_iteration(self):
+        for n_items in (2, 9, 22, 23, 24, 30):
+            for bs in (1, 3, 4):
+                for n in (1, 2, 4):
+                    for even in (True, False):
+                        for drop in (True, False):
+                            if drop and n_items < bs * n:
+                                continue
+                            sh = [
+                                BatchSamplerShard(
+                                    make_batch_sampler(n_items, bs, drop),
+                                    num_processes=n,
+                                    process_index=i,
+                                    even_batches=even,
+                                )
+                                for i in range(n)
+                            ]
+                            for s in sh:
+                                produced = list(s)
+                                if even or drop:
+                                    assert len(produced) == len(s), (n_items, bs, n, even, drop, s.process_index)
+
+    def test_even_batches_rectangular(self):
+        # with even_batches, every rank yields the same number of full batches
+        for n_items in (5, 11, 17, 23):
+            out = shards(n_items, 4, 2)
+            assert len(out[0]) == len(out[1])
+            for r in out:
+                for b in r:
+                    assert len(b) == 4
+
+
+class TestBatchSamplerShardSplit:
+    def test_split_even(self):
+        out = shards(16, 4, 2, split_batches=True)
+        assert out[0] == [[0, 1], [4, 5], [8, 9], [12, 13]]
+        assert out[1] == [[2, 3], [6, 7], [10, 11], [14, 15]]
+
+    def test_split_tail_pad(self):
+        out = shards(14, 4, 2, split_batches=True)
+        # last global batch [12,13] -> padded to [12,13,0,1]
+        assert out[0][-1] == [12, 13]
+        assert out[1][-1] == [0, 1]
+
+    def test_split_requires_divisible(self):
+        with pytest.raises(ValueError):
+            BatchSamplerShard(make_batch_sampler(16, 3, False), num_processes=2, process_index=0, split_batches=True)
+
+
+class RandomIterable(IterableDataset):
+    def __init__(self, n):
+        self.n = n
+
+    def __iter__(self):
+        yield from range(self.n)
+
+
+class TestIterableDatasetShard:
+    def test_basic(self):
+        ds = IterableDatasetShard(RandomIterable(16), batch_size=2, num_processes=2, process_index=0, drop_last=False)
+        assert list(ds) == [0, 1, 4, 5, 8, 9, 12, 13]
+        ds1 = IterableDatasetShard(RandomIterable(16), batch_size=2, num_processes=2, process_index=1, drop_last=False)
+        assert list(ds1) == [2, 3, 6, 7, 10, 11, 14, 15]
+
+    def test_tail_pad(self):
+        # 10 items, real batch 4: buffers [0-3],[4-7], tail [8,9] padded from first buffer
+        ds0 = IterableDatasetShard(RandomIterable(10), batch_size=2, num_processes=2, process_index=0, drop_last=False)
+        ds1 = IterableDatasetShard(RandomIterable(10), batch_size=2, num_processes=2, process_index=1, drop_last=False)
+        assert list(ds0) == [0, 1, 4, 5, 8, 9]
+        assert list(ds1) == [2, 3, 6, 7, 0, 1]
+
+    def test_drop_last(self):
+        ds0 = IterableDatasetShard(RandomIterable(10), batch_size=2, num_processes=2, process_index=0, drop_last=True)
+        assert list(ds0) == [0, 1, 4, 5]
+
+
+class TestDataLoaderShard:
+    def test_end_of_dataloader_flag(self):
+        ds = TensorDataset(torch.arange(8).float())
+        dl = DataLoaderShard(ds, batch_size=2)
+        batches = []
+        for i, b in enumerate(dl):
+            batches.append(b)
+            expected = i == 3
+            assert dl.end_of_dataloader == expected
+        assert len(batches) == 4
+
+    def test_rng_sync_same_permutation(self):
+        # single process: seedable sampler reproducibility across epochs with set_epoch
+        ds = TensorDataset(torch.arange(10).float())
+        sampler = SeedableRandomSampler(data_source=ds, generator=torch.Generator().manual_seed(0))
+        order1 = list(sampler)
+        sampler.set_epoch(0)
+        order2 = list(sampler)
+        assert order1 == order2
+
+    def test_prepare_single_process_passthrough(self):
+        from accelerate_amd.state import PartialState
+
+        PartialState()
+        ds = TensorDataset(torch.arange(10).float())
+        dl = DataLoader(ds, batch_size=2)
+        prepared = prepare_data_loader(dl, put_on_device=False)
+        seen = torch.cat([b[0] for b in prepared])
+        assert torch.equal(seen, torch.arange(10).float())
+
+
+class TestSkip:
+    def test_skip_batch_sampler(self):
+        bs = make_batch_sampler(16, 4)
+        skipped = SkipBatchSampler(bs, skip_batches=2)
+        assert list(skipped) == [[8, 9, 10, 11], [12, 13, 14, 15]]
+        assert len(skipped) == 2
+
+    def test_skip_data_loader(self):
+        ds = TensorDataset(torch.arange(16).float())
+        dl = SkipDataLoader(ds, batch_size=4, skip_batches=2)
+        seen = torch.cat([b[0] for b in dl])
+        assert torch.equal(seen, torch.arange(8, 16).float())
+
+    def test_skip_first_batches(self):
+        from accelerate_amd.state import PartialState
+
+        PartialState()
+        ds = TensorDataset(torch.arange(16).float())
+        dl = DataLoader(ds, batch_size=4)
+        new_dl = skip_first_batches(dl, num_batches=2)
+        seen = torch.cat([b[0] for b in new_dl])
+        assert torch.equal(seen, torch.arange(8, 16).float())

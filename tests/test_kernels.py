@@ -1,0 +1,146 @@
+"""Numerics tests for the CDNA4 HIP kernel pack: each kernel compared
+against a plain PyTorch fp32 reference of the same op (gpu-marked)."""
+
+import pytest
+import torch
+
+gpu = pytest.mark.gpu
+
+
+def make_tensors(shapes, device, seed=0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    return [torch.randn(s, generator=g).to(device) for s in shapes]
+
+
+SHAPES = [(768,), (30522, 768), (768, 768), (5,), (3072, 768), (1, 1), (127,), (16385,)]
+
+
+@gpu
+def test_extension_loads():
+    from accelerate_amd.ops import _load_extension
+
+    ext = _load_extension(required=True)
+    assert ext is not None
+
+
+@gpu
+@pytest.mark.parametrize("steps", [1, 3])
+def test_fused_adamw_matches_torch(steps):
+    from accelerate_amd.ops.optim import FusedAdamW
+
+    device = "cuda"
+    params_ref = make_tensors(SHAPES, device, seed=1)
+    params_ours = [p.clone() for p in params_ref]
+    for p in params_ref + params_ours:
+        p.requires_grad_(True)
+
+    opt_ref = torch.optim.AdamW(params_ref, lr=1e-3, weight_decay=0.01, betas=(0.9, 0.999), eps=1e-8)
+    opt_ours = FusedAdamW(params_ours, lr=1e-3, weight_decay=0.01, betas=(0.9, 0.999), eps=1e-8)
+
+    for step in range(steps):
+        grads = make_tensors(SHAPES, device, seed=10 + step)
+        for p, g in zip(params_ref, grads):
+            p.grad = g.clone()
+        for p, g in zip(params_ours, grads):
+            p.grad = g.clone()
+        opt_ref.step()
+        opt_ours.step()
+        torch.cuda.synchronize()
+    for i, (a, b) in enumerate(zip(params_ref, params_ours)):
+        assert torch.allclose(a, b, atol=1e-6, rtol=1e-5), f"param {i} diff {(a-b).abs().max().item()}"
+
+
+@gpu
+def test_clip_grad_norm_matches_torch():
+    from accelerate_amd.ops.clip_grad import clip_grad_norm_
+
+    device = "cuda"
+    params_ref = make_tensors(SHAPES, device, seed=2)
+    params_ours = [p.clone() for p in params_ref]
+    for plist, seed in ((params_ref, 20), (params_ours, 20)):
+        grads = make_tensors(SHAPES, device, seed=seed)
+        for p, g in zip(plist, grads):
+            p.requires_grad_(True)
+            p.grad = g.clone()
+    norm_ref = torch.nn.utils.clip_grad_norm_(params_ref, max_norm=1.0)
+    norm_ours = clip_grad_norm_(params_ours, max_norm=1.0)
+    torch.cuda.synchronize()
+    assert torch.allclose(norm_ref, norm_ours.to(norm_ref.device), rtol=1e-5), (norm_ref, norm_ours)
+    for a, b in zip(params_ref, params_ours):
+        assert torch.allclose(a.grad, b.grad, atol=1e-6, rtol=1e-5)
+
+
+@gpu
+def test_clip_noop_below_max_norm():
+    from accelerate_amd.ops.clip_grad import clip_grad_norm_
+
+    p = torch.randn(100, device="cuda").requires_grad_(True)
+    p.grad = torch.full((100,), 1e-4, device="cuda")
+    before = p.grad.clone()
+    clip_grad_norm_([p], max_norm=1e6)
+    torch.cuda.synchronize()
+    assert torch.equal(before, p.grad)
+
+
+@gpu
+def test_unscale_and_check():
+    from accelerate_amd.ops import _load_extension
+
+    ext = _load_extension(required=True)
+    g1 = torch.full((1000,), 4.0, device="cuda")
+    g2 = torch.full((257,), 8.0, device="cuda")
+    inv_scale = torch.tensor([0.5], device="cuda")
+    found_inf = torch.zeros(1, device="cuda")
+    ext.unscale_and_check([g1, g2], inv_scale, found_inf)
+    torch.cuda.synchronize()
+    assert found_inf.item() == 0.0
+    assert torch.allclose(g1, torch.full_like(g1, 2.0))
+    assert torch.allclose(g2, torch.full_like(g2, 4.0))
+    # now with an inf
+    g1[777] = float("inf")
+    ext.unscale_and_check([g1, g2], inv_scale, found_inf)
+    torch.cuda.synchronize()
+    assert found_inf.item() == 1.0
+
+
+@gpu
+def test_grad_scaler_skips_on_overflow():
+    from accelerate_amd.ops.grad_scaler import GradScaler
+    from accelerate_amd.ops.optim import FusedAdamW
+
+    p = torch.randn(64, device="cuda").requires_grad_(True)
+    opt = FusedAdamW([p], lr=1.0)
+    scaler = GradScaler(init_scale=2.0)
+    loss = (p * 2).sum()
+    scaler.scale(loss).backward()
+    p.grad[0] = float("nan")
+    before = p.detach().clone()
+    scaler.step(opt)
+    scaler.update()
+    torch.cuda.synchronize()
+    assert torch.equal(before, p.detach()), "step must be skipped on overflow"
+    assert scaler.get_scale() == 1.0  # backoff 0.5×
+
+
+@gpu
+def test_fused_adamw_bench_sanity():
+    """Fused AdamW must do the whole param set in ONE launch-bound call:
+    sanity-check wall time is < a few ms for a BERT-base-sized set."""
+    import time
+
+    from accelerate_amd.ops.optim import FusedAdamW
+
+    shapes = [(30522, 768)] + [(768, 768)] * 48 + [(3072, 768)] * 24 + [(768,)] * 200
+    params = [torch.randn(s, device="cuda").requires_grad_(True) for s in shapes]
+    for p in params:
+        p.grad = torch.randn_like(p)
+    opt = FusedAdamW(params, lr=1e-3)
+    opt.step()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(10):
+        opt.step()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / 10
+    # ~110M params fp32: 4 tensors × 4 B × ~2 traffic each ≈ 3.5 GB/step / 8 TB/s ≈ 0.5 ms
+    assert dt < 0.02, f"fused adamw too slow: {dt*1000:.2f} ms/step"

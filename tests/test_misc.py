@@ -1,0 +1,113 @@
+import json
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.nn as nn
+
+from accelerate_amd import find_executable_batch_size
+from accelerate_amd.tracking import JSONLTracker
+from accelerate_amd.utils.environment import patch_environment, str_to_bool
+from accelerate_amd.utils.memory import should_reduce_batch_size
+from accelerate_amd.utils.offload import OffloadedWeightsLoader, offload_state_dict
+
+
+def test_find_executable_batch_size():
+    calls = []
+
+    @find_executable_batch_size(starting_batch_size=128)
+    def train(batch_size):
+        calls.append(batch_size)
+        if batch_size > 16:
+            raise RuntimeError("CUDA out of memory.")
+        return batch_size
+
+    result = train()
+    assert result <= 16
+    assert calls[0] == 128
+    assert all(calls[i + 1] < calls[i] for i in range(len(calls) - 1))
+
+
+def test_non_oom_error_propagates():
+    @find_executable_batch_size(starting_batch_size=8)
+    def train(batch_size):
+        raise ValueError("unrelated")
+
+    with pytest.raises(ValueError):
+        train()
+
+
+def test_should_reduce_batch_size():
+    assert should_reduce_batch_size(RuntimeError("HIP out of memory."))
+    assert should_reduce_batch_size(RuntimeError("CUDA out of memory."))
+    assert not should_reduce_batch_size(RuntimeError("something else"))
+
+
+def test_str_to_bool():
+    assert str_to_bool("yes") == 1
+    assert str_to_bool("FALSE") == 0
+    with pytest.raises(ValueError):
+        str_to_bool("maybe")
+
+
+def test_patch_environment():
+    with patch_environment(test_var_xyz="1"):
+        assert os.environ["TEST_VAR_XYZ"] == "1"
+    assert "TEST_VAR_XYZ" not in os.environ
+
+
+def test_offload_state_dict_roundtrip():
+    sd = {
+        "a": torch.randn(4, 4),
+        "b": torch.randn(3).to(torch.bfloat16),
+        "scalar": torch.tensor(3.5),
+    }
+    with tempfile.TemporaryDirectory() as d:
+        offload_state_dict(d, sd)
+        loader = OffloadedWeightsLoader(save_folder=d)
+        assert set(loader.keys()) == set(sd.keys())
+        for k in sd:
+            assert torch.equal(loader[k], sd[k]), k
+        assert loader["b"].dtype == torch.bfloat16
+
+
+def test_jsonl_tracker():
+    from accelerate_amd.state import PartialState
+
+    PartialState()
+    with tempfile.TemporaryDirectory() as d:
+        tracker = JSONLTracker("run1", logging_dir=d)
+        tracker.store_init_configuration({"lr": 0.1})
+        tracker.log({"loss": 1.5}, step=0)
+        tracker.log({"loss": 1.0}, step=1)
+        tracker.finish()
+        lines = [json.loads(l) for l in open(os.path.join(d, "run1", "metrics.jsonl"))]
+        assert lines[0]["_config"] == {"lr": 0.1}
+        assert lines[1]["loss"] == 1.5
+        assert lines[2]["_step"] == 1
+
+
+def test_scheduler_steps_num_processes_times():
+    # single process: 1 step per step
+    from accelerate_amd import Accelerator
+
+    acc = Accelerator()
+    model = nn.Linear(2, 2)
+    opt = torch.optim.SGD(model.parameters(), lr=1.0)
+    sched = torch.optim.lr_scheduler.LambdaLR(opt, lambda s: 1.0 / (s + 1))
+    model, opt, sched = acc.prepare(model, opt, sched)
+    model(torch.randn(1, 2)).sum().backward()
+    opt.step()
+    sched.step()
+    assert sched.scheduler._step_count == 2
+
+
+def test_local_sgd_single_process_noop():
+    from accelerate_amd import Accelerator, LocalSGD
+
+    acc = Accelerator()
+    model = nn.Linear(2, 2)
+    model = acc.prepare_model(model)
+    with LocalSGD(accelerator=acc, model=model, local_sgd_steps=2) as ls:
+        ls.step()

@@ -1,0 +1,174 @@
+import json
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.nn as nn
+
+from accelerate_amd.big_modeling import init_empty_weights
+from accelerate_amd.utils.modeling import (
+    check_device_map,
+    clean_device_map,
+    compute_module_sizes,
+    dtype_byte_size,
+    find_tied_parameters,
+    get_max_layer_size,
+    infer_auto_device_map,
+    load_checkpoint_in_model,
+    named_module_tensors,
+    retie_parameters,
+    set_module_tensor_to_device,
+    shard_checkpoint,
+)
+
+
+class ModelForTest(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.linear1 = nn.Linear(3, 4)
+        self.batchnorm = nn.BatchNorm1d(4)
+        self.linear2 = nn.Linear(4, 5)
+
+    def forward(self, x):
+        return self.linear2(self.batchnorm(self.linear1(x)))
+
+
+def test_dtype_byte_size():
+    assert dtype_byte_size(torch.float32) == 4
+    assert dtype_byte_size(torch.bfloat16) == 2
+    assert dtype_byte_size(torch.int8) == 1
+    assert dtype_byte_size(torch.bool) == 1 / 8
+
+
+def test_compute_module_sizes():
+    model = ModelForTest()
+    sizes = compute_module_sizes(model)
+    # linear1: 3*4+4 = 16 params * 4B = 64
+    assert sizes["linear1"] == 64
+    # whole model = sum of parts
+    assert sizes[""] == sizes["linear1"] + sizes["batchnorm"] + sizes["linear2"]
+    half = compute_module_sizes(model, dtype=torch.float16)
+    assert half["linear1"] == 32
+
+
+def test_named_module_tensors():
+    model = ModelForTest()
+    names = [n for n, _ in named_module_tensors(model, recurse=True)]
+    assert "linear1.weight" in names
+    assert "batchnorm.running_mean" in names
+    no_buf = [n for n, _ in named_module_tensors(model, include_buffers=False, recurse=True)]
+    assert "batchnorm.running_mean" not in no_buf
+
+
+def test_find_and_retie_tied_parameters():
+    model = nn.Sequential(nn.Embedding(10, 4))
+    head = nn.Linear(4, 10, bias=False)
+    head.weight = model[0].weight
+    full = nn.ModuleDict({"emb": model[0], "head": head})
+    tied = find_tied_parameters(full)
+    assert tied == [["emb.weight", "head.weight"]]
+    # retie after breaking
+    full.head.weight = nn.Parameter(full.head.weight.detach().clone())
+    retie_parameters(full, tied)
+    assert full.head.weight is full.emb.weight
+
+
+def test_infer_auto_device_map_all_fit():
+    model = ModelForTest()
+    # everything fits on device 0
+    dmap = infer_auto_device_map(model, max_memory={0: 10000, "cpu": 10000})
+    assert set(dmap.values()) == {0}
+
+
+def test_infer_auto_device_map_split():
+    model = ModelForTest()
+    sizes = compute_module_sizes(model)
+    # device 0 can only hold linear1 (+ largest-layer headroom)
+    lim0 = sizes["linear1"] + max(sizes["linear1"], sizes["batchnorm"], sizes["linear2"])
+    dmap = infer_auto_device_map(model, max_memory={0: lim0, 1: 10000, "cpu": 10000})
+    assert dmap["linear1"] == 0
+    assert dmap["batchnorm"] == 1
+    assert dmap["linear2"] == 1
+
+
+def test_infer_auto_device_map_offload():
+    class BiggerModel(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.linear1 = nn.Linear(3, 4)  # 64 B
+            self.batchnorm = nn.BatchNorm1d(4)  # 72 B
+            self.linear2 = nn.Linear(4, 5)  # 100 B
+            self.linear3 = nn.Linear(5, 6)  # 144 B (largest layer -> headroom)
+
+    model = BiggerModel()
+    # GPU0 fits linear1 + 144 B headroom; CPU fits batchnorm + headroom;
+    # linear2/linear3 spill to disk
+    dmap = infer_auto_device_map(model, max_memory={0: 250, "cpu": 250})
+    assert dmap["linear1"] == 0
+    assert dmap["batchnorm"] == "cpu"
+    assert dmap["linear2"] == "disk"
+    assert dmap["linear3"] == "disk"
+
+
+def test_check_device_map():
+    model = ModelForTest()
+    check_device_map(model, {"": 0})
+    check_device_map(model, {"linear1": 0, "batchnorm": 1, "linear2": "cpu"})
+    with pytest.raises(ValueError):
+        check_device_map(model, {"linear1": 0, "linear2": 1})
+
+
+def test_clean_device_map():
+    dmap = {"a.0": 0, "a.1": 0, "b": 1}
+    out = clean_device_map(dict(dmap))
+    assert out == {"a": 0, "b": 1}
+
+
+def test_set_module_tensor_to_device_meta_roundtrip():
+    model = ModelForTest()
+    w = model.linear1.weight.detach().clone()
+    set_module_tensor_to_device(model, "linear1.weight", "meta")
+    assert model.linear1.weight.device == torch.device("meta")
+    set_module_tensor_to_device(model, "linear1.weight", "cpu", value=w)
+    assert torch.equal(model.linear1.weight, w)
+
+
+def test_init_empty_weights():
+    with init_empty_weights():
+        model = nn.Linear(10000, 10000)  # 400 MB if real
+    assert model.weight.device == torch.device("meta")
+
+
+def test_shard_checkpoint():
+    sd = {f"w{i}": torch.randn(100, 100) for i in range(4)}  # 40 KB each
+    shards, index = shard_checkpoint(sd, max_shard_size=90000)
+    assert len(shards) == 2
+    assert set(index["weight_map"].keys()) == set(sd.keys())
+
+
+def test_load_checkpoint_in_model_device_map():
+    model = ModelForTest()
+    sd = {k: v.clone() for k, v in model.state_dict().items()}
+    with tempfile.TemporaryDirectory() as d:
+        torch.save(sd, os.path.join(d, "pytorch_model.bin"))
+        fresh = ModelForTest()
+        with torch.no_grad():
+            for p in fresh.parameters():
+                p.zero_()
+        load_checkpoint_in_model(fresh, os.path.join(d, "pytorch_model.bin"), device_map={"": "cpu"})
+        for k, v in fresh.state_dict().items():
+            assert torch.equal(v, sd[k]), k
+
+
+def test_load_checkpoint_in_model_disk_offload():
+    model = ModelForTest()
+    sd = {k: v.clone() for k, v in model.state_dict().items()}
+    with tempfile.TemporaryDirectory() as d, tempfile.TemporaryDirectory() as offdir:
+        torch.save(sd, os.path.join(d, "pytorch_model.bin"))
+        fresh = ModelForTest()
+        dmap = {"linear1": "cpu", "batchnorm": "cpu", "linear2": "disk"}
+        load_checkpoint_in_model(fresh, os.path.join(d, "pytorch_model.bin"), device_map=dmap, offload_folder=offdir)
+        assert fresh.linear2.weight.device == torch.device("meta")
+        assert os.path.isfile(os.path.join(offdir, "linear2.weight.dat"))
+        assert os.path.isfile(os.path.join(offdir, "index.json"))
