@@ -13,6 +13,9 @@
 // kernels (defined in optim_kernels.hip)
 __global__ void fused_adamw_kernel(TensorListMeta, float, float, float, float, float,
                                    float, float, const float*, const float*);
+__global__ void adamw_incr_step_kernel(float*);
+__global__ void fused_adamw_dev_kernel(TensorListMeta, const float*, const float*,
+                                       float, float, float, float, const float*, const float*);
 __global__ void l2norm_squared_kernel(TensorListMeta, float*);
 __global__ void clip_coef_kernel(const float*, float, float*, float*);
 __global__ void multi_tensor_scale_kernel(TensorListMeta, const float*);
@@ -83,6 +86,32 @@ void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                      found_inf.has_value() ? found_inf->data_ptr<float>() : nullptr);
 }
 
+// hipGraph-capturable AdamW over a PRE-BUILT device plan. `addrs_numels` is
+// the int64 device tensor [4*n ptrs | n numels]; `chunk_prefix` int32 [n+1];
+// `step`/`lr` are 1-elem fp32 device tensors. The caller (Python FusedAdamW)
+// caches the plan, so a captured graph replays with stable pointers and the
+// in-graph step increment keeps bias correction advancing across replays.
+void fused_adamw_planned(at::Tensor addrs_numels, at::Tensor chunk_prefix,
+                         int64_t n_tensors, int64_t total_chunks,
+                         at::Tensor step, at::Tensor lr,
+                         double beta1, double beta2, double eps, double weight_decay,
+                         c10::optional<at::Tensor> grad_scale, c10::optional<at::Tensor> found_inf) {
+  TensorListMeta meta;
+  meta.addrs = addrs_numels.data_ptr<int64_t>();
+  meta.numels = addrs_numels.data_ptr<int64_t>() + 4 * n_tensors;
+  meta.chunk_prefix = chunk_prefix.data_ptr<int32_t>();
+  meta.n_tensors = static_cast<int32_t>(n_tensors);
+  meta.n_lists = 4;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(adamw_incr_step_kernel, dim3(1), dim3(1), 0, stream.stream(),
+                     step.data_ptr<float>());
+  hipLaunchKernelGGL(fused_adamw_dev_kernel, dim3(total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                     meta, step.data_ptr<float>(), lr.data_ptr<float>(),
+                     (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                     grad_scale.has_value() ? grad_scale->data_ptr<float>() : nullptr,
+                     found_inf.has_value() ? found_inf->data_ptr<float>() : nullptr);
+}
+
 at::Tensor l2norm_squared(std::vector<at::Tensor> grads) {
   auto h = build_meta({grads});
   auto out = at::zeros({1}, grads[0].options().dtype(at::kFloat));
@@ -127,6 +156,8 @@ void unscale_and_check(std::vector<at::Tensor> grads, at::Tensor inv_scale, at::
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (gfx950)");
+  m.def("fused_adamw_planned", &fused_adamw_planned,
+        "graph-capturable fused AdamW over a cached device plan");
   m.def("l2norm_squared", &l2norm_squared, "global L2 norm squared over tensor list");
   m.def("clip_grad_norm", &clip_grad_norm, "on-device clip_grad_norm_, returns total norm");
   m.def("multi_tensor_scale", &multi_tensor_scale, "g *= *coef");

@@ -78,6 +78,77 @@ __global__ void fused_adamw_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// hipGraph-capturable AdamW: step and lr live in DEVICE memory so a captured
+// graph replays with fresh values (host writes lr before replay; the prep
+// kernel increments step in-graph). Bias corrections are computed per block
+// from *step — two powf per workgroup, noise.
+// ---------------------------------------------------------------------------
+
+__global__ void adamw_incr_step_kernel(float* __restrict__ step) { *step += 1.f; }
+
+__global__ void fused_adamw_dev_kernel(
+    TensorListMeta meta,
+    const float* __restrict__ step_ptr, const float* __restrict__ lr_ptr,
+    float beta1, float beta2, float eps, float weight_decay,
+    const float* __restrict__ grad_scale,
+    const float* __restrict__ found_inf) {
+  if (found_inf != nullptr && *found_inf != 0.f) return;
+
+  const float step_f = *step_ptr;
+  const float lr = *lr_ptr;
+  const float bias_correction1 = 1.f - powf(beta1, step_f);
+  const float rsqrt_bias_correction2 = rsqrtf(1.f - powf(beta2, step_f));
+
+  const int cid = blockIdx.x;
+  const int t = find_tensor(meta.chunk_prefix, meta.n_tensors, cid);
+  const int64_t chunk_in_tensor = cid - meta.chunk_prefix[t];
+  const int64_t numel = meta.numels[t];
+  const int64_t lo = chunk_in_tensor * kChunkSize;
+  const int64_t hi = min(lo + kChunkSize, numel);
+
+  float* __restrict__ p = reinterpret_cast<float*>(meta.addrs[0 * meta.n_tensors + t]);
+  const float* __restrict__ g = reinterpret_cast<const float*>(meta.addrs[1 * meta.n_tensors + t]);
+  float* __restrict__ m = reinterpret_cast<float*>(meta.addrs[2 * meta.n_tensors + t]);
+  float* __restrict__ v = reinterpret_cast<float*>(meta.addrs[3 * meta.n_tensors + t]);
+
+  const float step_size = lr / bias_correction1;
+  const float decay = 1.f - lr * weight_decay;
+  const float gscale = (grad_scale != nullptr) ? *grad_scale : 1.f;
+
+  const int64_t tid = threadIdx.x;
+  int64_t i = lo + tid * 4;
+  const int64_t vec_end = lo + ((hi - lo) & ~int64_t(3));
+  for (; i + 3 < vec_end; i += kBlockThreads * 4) {
+    float4 gp = *reinterpret_cast<const float4*>(g + i);
+    float4 pp = *reinterpret_cast<float4*>(p + i);
+    float4 mp = *reinterpret_cast<float4*>(m + i);
+    float4 vp = *reinterpret_cast<float4*>(v + i);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float gk = (&gp.x)[k] * gscale;
+      float pk = (&pp.x)[k] * decay;
+      float mk = beta1 * (&mp.x)[k] + (1.f - beta1) * gk;
+      float vk = beta2 * (&vp.x)[k] + (1.f - beta2) * gk * gk;
+      float denom = sqrtf(vk) * rsqrt_bias_correction2 + eps;
+      pk -= step_size * mk / denom;
+      (&pp.x)[k] = pk; (&mp.x)[k] = mk; (&vp.x)[k] = vk;
+    }
+    *reinterpret_cast<float4*>(p + i) = pp;
+    *reinterpret_cast<float4*>(m + i) = mp;
+    *reinterpret_cast<float4*>(v + i) = vp;
+  }
+  for (int64_t j = vec_end + tid; j < hi; j += kBlockThreads) {
+    float gk = g[j] * gscale;
+    float pk = p[j] * decay;
+    float mk = beta1 * m[j] + (1.f - beta1) * gk;
+    float vk = beta2 * v[j] + (1.f - beta2) * gk * gk;
+    float denom = sqrtf(vk) * rsqrt_bias_correction2 + eps;
+    pk -= step_size * mk / denom;
+    p[j] = pk; m[j] = mk; v[j] = vk;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // global L2 norm²: per-wave shuffle reduce → per-block LDS reduce → one
 // device-scope atomicAdd per block (Guideline 12).
 // ---------------------------------------------------------------------------

@@ -6,6 +6,13 @@ chunked), float4-vectorized, HBM-bound. Numerics mirror torch.optim.AdamW
 exactly; `tests/test_kernels.py::test_fused_adamw` compares against the
 eager fp32 reference.
 
+hipGraph-capturable by construction: the multi-tensor plan (device pointer
+table + chunk prefix) is built once and cached; `step` and `lr` live in
+device memory, the step increment happens in-graph, and the host writes lr
+into the device scalar only when the scheduler changes it. Capturing
+``opt.step()`` inside a torch.cuda.graph therefore replays correctly with
+advancing bias corrections.
+
 On CPU (unit tests) the step falls back to torch.optim.AdamW math; on a GPU
 the HIP extension is required — no silent eager fallback.
 """
@@ -33,6 +40,43 @@ class FusedAdamW(torch.optim.Optimizer):
         super().__init__(params, defaults)
         self._grad_scale_tensor = grad_scale_tensor
         self._found_inf_tensor = found_inf_tensor
+        # per-group cached execution plans: {group_idx: plan dict}
+        self._plans = {}
+
+    def _build_plan(self, gi, params, grads, exp_avgs, exp_avg_sqs, device):
+        import itertools
+
+        n = len(params)
+        ptrs = [t.data_ptr() for t in itertools.chain(params, grads, exp_avgs, exp_avg_sqs)]
+        numels = [p.numel() for p in params]
+        prefix = [0]
+        for numel in numels:
+            prefix.append(prefix[-1] + (numel + 16383) // 16384)
+        addrs_numels = torch.tensor(ptrs + numels, dtype=torch.int64).to(device, non_blocking=True)
+        chunk_prefix = torch.tensor(prefix, dtype=torch.int32).to(device, non_blocking=True)
+        group = self.param_groups[gi]
+        old = self._plans.get(gi)
+        if old is not None:
+            step_t, lr_t, last_lr = old["step"], old["lr"], old["last_lr"]
+        else:
+            restored = getattr(self, "_restored_steps", {}).pop(gi, None)
+            step_t = restored.to(device) if restored is not None else torch.zeros(1, dtype=torch.float32, device=device)
+            lr_t = torch.full((1,), float(group["lr"]), dtype=torch.float32, device=device)
+            last_lr = float(group["lr"])
+        plan = {
+            "key": tuple(ptrs),
+            "addrs_numels": addrs_numels,
+            "chunk_prefix": chunk_prefix,
+            "n_tensors": n,
+            "total_chunks": prefix[-1],
+            # device-side hyperparams (stable storage across plan rebuilds so
+            # captured graphs keep pointing at live step/lr scalars)
+            "step": step_t,
+            "lr": lr_t,
+            "last_lr": last_lr,
+        }
+        self._plans[gi] = plan
+        return plan
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -41,7 +85,8 @@ class FusedAdamW(torch.optim.Optimizer):
             with torch.enable_grad():
                 loss = closure()
 
-        for group in self.param_groups:
+        ext = None
+        for gi, group in enumerate(self.param_groups):
             params, grads, exp_avgs, exp_avg_sqs = [], [], [], []
             cpu_params = []
             for p in group["params"]:
@@ -49,29 +94,38 @@ class FusedAdamW(torch.optim.Optimizer):
                     continue
                 state = self.state[p]
                 if len(state) == 0:
-                    state["step"] = 0
                     state["exp_avg"] = torch.zeros_like(p, memory_format=torch.preserve_format)
                     state["exp_avg_sq"] = torch.zeros_like(p, memory_format=torch.preserve_format)
-                state["step"] += 1
                 if p.is_cuda and p.dtype == torch.float32:
                     params.append(p)
                     grads.append(p.grad if p.grad.is_contiguous() else p.grad.contiguous())
                     exp_avgs.append(state["exp_avg"])
                     exp_avg_sqs.append(state["exp_avg_sq"])
                 else:
+                    state["step"] = state.get("step", 0) + 1
                     cpu_params.append((p, state))
 
             if params:
-                ext = _load_extension(required=True)
-                step = self.state[params[0]]["step"]
+                if ext is None:
+                    ext = _load_extension(required=True)
+                import itertools
+
+                plan = self._plans.get(gi)
+                key = tuple(t.data_ptr() for t in itertools.chain(params, grads, exp_avgs, exp_avg_sqs))
+                if plan is None or plan["key"] != key:
+                    plan = self._build_plan(gi, params, grads, exp_avgs, exp_avg_sqs, params[0].device)
+                lr = float(group["lr"])
+                if lr != plan["last_lr"]:
+                    plan["lr"].fill_(lr)
+                    plan["last_lr"] = lr
                 beta1, beta2 = group["betas"]
-                ext.fused_adamw(
-                    params,
-                    grads,
-                    exp_avgs,
-                    exp_avg_sqs,
-                    step,
-                    group["lr"],
+                ext.fused_adamw_planned(
+                    plan["addrs_numels"],
+                    plan["chunk_prefix"],
+                    plan["n_tensors"],
+                    plan["total_chunks"],
+                    plan["step"],
+                    plan["lr"],
                     beta1,
                     beta2,
                     group["eps"],
@@ -82,6 +136,42 @@ class FusedAdamW(torch.optim.Optimizer):
             for p, state in cpu_params:
                 self._single_tensor_step(p, state, group)
         return loss
+
+    def refresh_hyperparams(self):
+        """Push host-side group['lr'] changes into the device scalars.
+
+        Call after ``scheduler.step()`` when the optimizer kernel was captured
+        in a hipGraph (the captured kernel reads lr from device memory)."""
+        for gi, group in enumerate(self.param_groups):
+            plan = self._plans.get(gi)
+            if plan is not None and float(group["lr"]) != plan["last_lr"]:
+                plan["lr"].fill_(float(group["lr"]))
+                plan["last_lr"] = float(group["lr"])
+
+    def state_dict(self):
+        self._sync_steps_from_device()
+        return super().state_dict()
+
+    def _sync_steps_from_device(self):
+        # under graph replay the host never sees step increments; read back
+        for gi, group in enumerate(self.param_groups):
+            plan = self._plans.get(gi)
+            if plan is None:
+                continue
+            step = int(plan["step"].item())
+            for p in group["params"]:
+                if p in self.state and p.is_cuda:
+                    self.state[p]["step"] = step
+
+    def load_state_dict(self, state_dict):
+        super().load_state_dict(state_dict)
+        self._plans = {}
+        self._restored_steps = {}
+        # restore device step counters from the loaded per-param steps
+        for gi, group in enumerate(self.param_groups):
+            steps = [self.state[p].get("step", 0) for p in group["params"] if p in self.state]
+            if steps:
+                self._restored_steps[gi] = torch.tensor([float(max(steps))], dtype=torch.float32)
 
     @staticmethod
     def _single_tensor_step(p, state, group):

@@ -69,15 +69,6 @@ def main():
     model, optimizer, dataloader, scheduler = accelerator.prepare(model, optimizer, dataloader, scheduler)
     model.train()
 
-    def one_step(batch):
-        ids, mask, types, lbl = batch
-        optimizer.zero_grad()
-        out = model(ids, attention_mask=mask, token_type_ids=types, labels=lbl)
-        accelerator.backward(out["loss"])
-        optimizer.step()
-        scheduler.step()
-        return out["loss"]
-
     data_iter = iter(dataloader)
 
     def next_batch():
@@ -87,6 +78,61 @@ def main():
         except StopIteration:
             data_iter = iter(dataloader)
             return next(data_iter)
+
+    use_graph = on_gpu and os.environ.get("BENCH_GRAPH", "1") == "1"
+    step_mode = "eager"
+
+    def eager_step(batch):
+        ids, mask, types, lbl = batch
+        optimizer.zero_grad(set_to_none=False)
+        out = model(ids, attention_mask=mask, token_type_ids=types, labels=lbl)
+        accelerator.backward(out["loss"])
+        optimizer.step()
+        scheduler.step()
+        return out["loss"]
+
+    one_step = eager_step
+
+    if use_graph:
+        # Whole-step hipGraph: zero_grad + fwd + bwd (+ RCCL allreduce) +
+        # fused AdamW replay from ONE graph launch. Inputs live in static
+        # device buffers; lr is a device scalar the scheduler refreshes.
+        try:
+            static = [t.to(device, non_blocking=True) for t in next_batch()]
+
+            def graph_body():
+                optimizer.zero_grad(set_to_none=False)
+                out = model(static[0], attention_mask=static[1], token_type_ids=static[2], labels=static[3])
+                accelerator.backward(out["loss"])
+                optimizer.step()
+                return out["loss"]
+
+            # eager warmup on a side stream (materializes grads/opt state/buckets)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    graph_body()
+            torch.cuda.current_stream().wait_stream(side)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                loss_buf = graph_body()
+
+            fused = optimizer.optimizer  # the FusedAdamW under the wrapper
+
+            def graph_step(batch):
+                for dst, src in zip(static, batch):
+                    dst.copy_(src, non_blocking=True)
+                scheduler.step()
+                fused.refresh_hyperparams()
+                graph.replay()
+                return loss_buf
+
+            one_step = graph_step
+            step_mode = "hipgraph"
+        except Exception as e:
+            print(f"[bench] hipGraph capture failed ({e}); falling back to eager", flush=True)
+            one_step = eager_step
 
     # warmup
     for _ in range(args.warmup):
@@ -132,6 +178,7 @@ def main():
                 "seq_len": SEQ_LEN,
                 "parallelism": f"dp{n}",
                 "optimizer": "fused_adamw_hip" if on_gpu else "torch_adamw",
+                "step_mode": step_mode,
             },
         }
         print(json.dumps(result))
