@@ -921,6 +921,13 @@ class Accelerator:
         (reference: dataclasses.py:1033 TorchDynamoPlugin)."""
         if self.device.type != "cuda":
             return step_fn
+        # Free stale autograd graphs: an AccumulateGrad node kept alive from a
+        # pre-capture iteration and pinned to the default stream segfaults
+        # ROCm's hipGraph capture_end (observed on MI355X, torch 2.10+rocm7.0).
+        import gc
+
+        gc.collect()
+        torch.cuda.synchronize()
         side_stream = torch.cuda.Stream()
         side_stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side_stream):

@@ -1,0 +1,48 @@
+"""`accelerate-amd estimate` — per-dtype memory table for a model
+(reference: commands/estimate.py). Meta-loads via init_empty_weights (zero
+RAM); works offline for local checkpoints/configs and for transformers
+models when their config is cached locally.
+"""
+
+import argparse
+
+from ..big_modeling import init_empty_weights
+from ..utils.modeling import compute_module_sizes
+from ..utils.other import convert_bytes
+
+
+def create_empty_model(model_name: str, trust_remote_code: bool = False):
+    import transformers
+
+    config = transformers.AutoConfig.from_pretrained(model_name, trust_remote_code=trust_remote_code)
+    with init_empty_weights():
+        model = transformers.AutoModel.from_config(config, trust_remote_code=trust_remote_code)
+    return model
+
+
+def estimate_command(args):
+    import torch
+
+    model = create_empty_model(args.model_name, trust_remote_code=args.trust_remote_code)
+    sizes = {}
+    for dtype in args.dtypes:
+        torch_dtype = getattr(torch, {"float32": "float32", "float16": "float16", "bfloat16": "bfloat16", "int8": "int8", "fp8": "float8_e4m3fn"}[dtype])
+        total = compute_module_sizes(model, dtype=torch_dtype)[""]
+        # training ≈ params + grads + 2× Adam state (fp32) + activations headroom
+        train_total = total * (4 if dtype in ("float32",) else 6)
+        sizes[dtype] = (total, train_total)
+    largest = max(len(d) for d in sizes)
+    print(f"Memory estimate for {args.model_name} (MI355X: 288 GB HBM3E per GPU):")
+    print(f"{'dtype':<{largest+2}} {'inference':>12} {'training(Adam)':>16}")
+    for dtype, (inf, train) in sizes.items():
+        print(f"{dtype:<{largest+2}} {convert_bytes(inf):>12} {convert_bytes(train):>16}")
+
+
+def add_parser(subparsers):
+    parser = subparsers.add_parser("estimate", help="Estimate model memory usage")
+    parser.add_argument("model_name", help="transformers model name or local path")
+    parser.add_argument("--dtypes", nargs="+", default=["float32", "float16", "int8"],
+                        choices=["float32", "float16", "bfloat16", "int8", "fp8"])
+    parser.add_argument("--trust_remote_code", action="store_true")
+    parser.set_defaults(func=estimate_command)
+    return parser

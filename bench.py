@@ -108,6 +108,10 @@ def main():
                 return out["loss"]
 
             # eager warmup on a side stream (materializes grads/opt state/buckets)
+            import gc
+
+            gc.collect()  # stale autograd graphs break ROCm capture_end
+            torch.cuda.synchronize()
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(side):

@@ -71,17 +71,31 @@ def main():
     print(f"full step:    {t_full:.3f} ms  (optimizer = {t_full - t_fwdbwd:.3f} ms)")
 
     # ---- hipGraph capture of zero+fwd+bwd ----
+    import faulthandler
+    import gc
+
+    faulthandler.enable()
+    # Drop every reference to prior autograd graphs before capture: a stale
+    # AccumulateGrad node pinned to the default stream segfaults ROCm's
+    # capture_end (observed on MI355X, torch 2.10+rocm7.0).
+    loss = None
+    del loss
+    gc.collect()
     torch.cuda.synchronize()
     fwd_bwd()  # materialize grads
+    gc.collect()
+    print("phase: pre-warmup done", flush=True)
     side = torch.cuda.Stream()
     side.wait_stream(torch.cuda.current_stream())
     with torch.cuda.stream(side):
         for _ in range(3):
             fwd_bwd()
     torch.cuda.current_stream().wait_stream(side)
+    print("phase: side-stream warmup done", flush=True)
     g = torch.cuda.CUDAGraph()
     with torch.cuda.graph(g):
         loss_out = fwd_bwd()
+    print("phase: capture done", flush=True)
 
     def replay():
         g.replay()

@@ -1,0 +1,82 @@
+"""CLI tests (reference: tests/test_cli.py pattern): config round-trip,
+env, launch of the bundled test script (2-proc gloo)."""
+
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import yaml
+
+from testing_utils import REPO_ROOT, get_free_port
+
+
+def run_cli(*args, timeout=240):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO_ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    result = subprocess.run(
+        [sys.executable, "-m", "accelerate_amd", *args], capture_output=True, text=True, timeout=timeout, env=env
+    )
+    return result
+
+
+def test_env_command():
+    r = run_cli("env")
+    assert r.returncode == 0
+    assert "accelerate_amd version" in r.stdout
+    assert "PyTorch version" in r.stdout
+
+
+def test_config_default_roundtrip(tmp_path):
+    cfg_file = tmp_path / "cfg.yaml"
+    r = run_cli("config", "--default", "--config_file", str(cfg_file))
+    assert r.returncode == 0, r.stderr
+    data = yaml.safe_load(cfg_file.read_text())
+    assert "distributed_type" in data
+    assert "num_processes" in data
+
+    from accelerate_amd.commands.config import ClusterConfig
+
+    cfg = ClusterConfig.load(cfg_file)
+    assert cfg.num_processes == data["num_processes"]
+
+
+def test_launch_single_process():
+    script = Path(REPO_ROOT) / "accelerate_amd" / "test_utils" / "test_script.py"
+    r = run_cli("launch", "--num_processes", "1", "--cpu", str(script))
+    assert r.returncode == 0, r.stderr
+    assert "All checks passed!" in r.stdout
+
+
+def test_launch_multi_process():
+    script = Path(REPO_ROOT) / "accelerate_amd" / "test_utils" / "test_script.py"
+    r = run_cli(
+        "launch",
+        "--num_processes",
+        "2",
+        "--cpu",
+        "--main_process_port",
+        str(get_free_port()),
+        str(script),
+    )
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "All checks passed!" in r.stdout
+
+
+def test_launch_respects_config_file(tmp_path):
+    cfg_file = tmp_path / "cfg.yaml"
+    cfg_file.write_text(
+        yaml.safe_dump(
+            {
+                "distributed_type": "MULTI_CPU",
+                "num_processes": 2,
+                "use_cpu": True,
+                "mixed_precision": "no",
+                "main_process_port": get_free_port(),
+            }
+        )
+    )
+    script = Path(REPO_ROOT) / "accelerate_amd" / "test_utils" / "test_script.py"
+    r = run_cli("launch", "--config_file", str(cfg_file), str(script))
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "2 process(es)" in r.stdout
