@@ -316,8 +316,10 @@ class Accelerator:
     @contextmanager
     def no_sync(self, model):
         """Disable gradient sync inside the context (reference: accelerator.py:1132)."""
+        from .parallel.fsdp import ShardedModel
+
         context = contextlib.nullcontext
-        if isinstance(model, DistributedDataParallelEngine):
+        if isinstance(model, (DistributedDataParallelEngine, ShardedModel)):
             context = model.no_sync
         with context():
             yield
@@ -537,9 +539,13 @@ class Accelerator:
             loss.backward(**kwargs)
         # deterministic reduction epilogue: we own backward(), so no autograd
         # engine callbacks are needed (reference delegates to DDP's C++ hooks)
+        from .parallel.fsdp import ShardedModel
+
         for model in self._models:
             if isinstance(model, DistributedDataParallelEngine):
                 model.finalize()
+            elif isinstance(model, ShardedModel):
+                model.finalize_backward()
 
     def unscale_gradients(self, optimizer=None):
         """(reference: accelerator.py:2935)"""
@@ -558,6 +564,19 @@ class Accelerator:
         L2-norm kernel; under DDP grads are already reduced and identical
         across ranks so no extra collective is required."""
         self.unscale_gradients()
+        from .parallel.fsdp import ShardedModel
+
+        parameters = list(parameters) if not isinstance(parameters, torch.Tensor) else [parameters]
+        for model in self._models:
+            if isinstance(model, ShardedModel):
+                # sharded world: the clip must happen on the gradient SHARDS
+                # with a cross-rank norm reduction (reference: FSDP
+                # model.clip_grad_norm_, accelerator.py:2977-3007)
+                param_ids = {id(p) for p in parameters}
+                if any(id(p) in param_ids for u in model.units for p in u.params) or any(
+                    id(u.shard) in param_ids for u in model.units
+                ):
+                    return model.clip_grad_norm_(max_norm, norm_type)
         from .ops.clip_grad import clip_grad_norm_ as _clip
 
         return _clip(parameters, max_norm, norm_type=norm_type)

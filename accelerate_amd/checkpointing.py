@@ -80,9 +80,17 @@ def save_accelerator_state(
         logger.info(f"Model weights saved in {output_dir / weights_name}")
     # Optimizer states
     for i, opt in enumerate(optimizers):
-        optimizer_name = f"{OPTIMIZER_NAME}.bin" if i == 0 else f"{OPTIMIZER_NAME}_{i}.bin"
-        if PartialStateIsMain(save_on_each_node):
+        if getattr(opt, "_sharded", False):
+            # sharded world: the optimizer state IS per-rank (master shards);
+            # every rank writes its own file, no cross-rank communication
+            # (reference rationale: fsdp_utils.py:107-118)
+            suffix = f"_{i}" if i > 0 else ""
+            optimizer_name = f"{OPTIMIZER_NAME}{suffix}_rank{process_index}.bin"
             torch.save(opt.state_dict(), output_dir / optimizer_name)
+        else:
+            optimizer_name = f"{OPTIMIZER_NAME}.bin" if i == 0 else f"{OPTIMIZER_NAME}_{i}.bin"
+            if PartialStateIsMain(save_on_each_node):
+                torch.save(opt.state_dict(), output_dir / optimizer_name)
         logger.info(f"Optimizer state saved in {output_dir / optimizer_name}")
     # Scheduler states
     for i, scheduler in enumerate(schedulers):
@@ -172,7 +180,11 @@ def load_accelerator_state(
 
     # Optimizers
     for i, opt in enumerate(optimizers):
-        optimizer_name = f"{OPTIMIZER_NAME}.bin" if i == 0 else f"{OPTIMIZER_NAME}_{i}.bin"
+        if getattr(opt, "_sharded", False):
+            suffix = f"_{i}" if i > 0 else ""
+            optimizer_name = f"{OPTIMIZER_NAME}{suffix}_rank{process_index}.bin"
+        else:
+            optimizer_name = f"{OPTIMIZER_NAME}.bin" if i == 0 else f"{OPTIMIZER_NAME}_{i}.bin"
         optimizer_state = torch.load(input_dir / optimizer_name, map_location=map_location, weights_only=True)
         optimizers[i].load_state_dict(optimizer_state)
     logger.info("All optimizer states loaded successfully")

@@ -1,37 +1,589 @@
-"""MI355X sharded-parameter engine (FSDP2-equivalent) — see SURVEY.md §2.9 N4.
+"""MI355X sharded-parameter engine — the FSDP2 equivalent
+(reference: SURVEY.md §2.9 N4; torch `fully_shard` at fsdp_utils.py:741).
 
-Design (build plan §7 step 9): per-parameter shards resident in HBM3E,
-forward all-gather / backward reduce-scatter over RCCL on xGMI with
-prefetch streams, bf16 compute + fp32 master policy.
+Design (MI355X-first, not a torch-FSDP translation):
 
-This module currently provides the scaffolding used by the rest of the
-framework; the full engine lands later in the round.
+- The model is partitioned into **units** (transformer blocks by default).
+  Each unit's parameters are flattened into ONE flat fp32 master buffer,
+  padded to ``world_size`` and sharded 1/n per rank — big contiguous
+  shards, few large RCCL calls, sized for 288 GB HBM3E per GPU.
+- **Forward**: the unit's full flat buffer is re-materialized by a single
+  ``all_gather_into_tensor`` in the *compute dtype* (bf16 ⇒ half the xGMI
+  traffic of fp32), issued on a dedicated comm stream one unit AHEAD of
+  compute (prefetch). Parameters are views into the full buffer; its
+  storage is resized to 0 on reshard so activation-saved references keep
+  pointing at the same storage object and re-validate on the next gather.
+- **Backward**: a pre-backward hook re-gathers (prefetching in reverse
+  order); per-param post-accumulate hooks count grads, and when the unit
+  is complete its grads are flattened and **reduce_scatter**'d (in
+  ``reduce_dtype``) into the fp32 gradient shard that the optimizer sees.
+- **Optimizer** operates on the fp32 master shards (the optimizer-param
+  swap, reference: accelerator.py:1714-1726).
+- CPU/gloo worlds (unit tests) run the same code path with collective
+  fallbacks (gloo has no reduce_scatter_tensor → all_reduce + slice).
 """
 
-from typing import Iterable
+import contextlib
+from typing import Dict, List, Optional
 
 import torch
+import torch.distributed as dist
 import torch.nn as nn
+
+from ..logging import get_logger
+
+logger = get_logger(__name__)
+
+
+def _is_nccl(group) -> bool:
+    return dist.is_initialized() and dist.get_backend(group) == "nccl"
+
+
+def _all_gather_flat(out_full: torch.Tensor, shard: torch.Tensor, group):
+    if _is_nccl(group):
+        dist.all_gather_into_tensor(out_full, shard, group=group)
+    else:
+        world = dist.get_world_size(group)
+        chunks = list(out_full.chunk(world))
+        dist.all_gather(chunks, shard, group=group)
+        for i, c in enumerate(chunks):  # all_gather may copy; ensure placement
+            out = out_full.narrow(0, i * shard.numel(), shard.numel())
+            if out.data_ptr() != c.data_ptr():
+                out.copy_(c)
+
+
+def _reduce_scatter_flat(out_shard: torch.Tensor, full: torch.Tensor, group, op):
+    if _is_nccl(group):
+        dist.reduce_scatter_tensor(out_shard, full, op=op, group=group)
+    else:
+        dist.all_reduce(full, op=dist.ReduceOp.SUM, group=group)
+        rank = dist.get_rank(group)
+        out_shard.copy_(full.narrow(0, rank * out_shard.numel(), out_shard.numel()))
+        if op == dist.ReduceOp.AVG:
+            out_shard.div_(dist.get_world_size(group))
+
+
+class _FlatUnit:
+    """One sharded unit: a module subtree whose params live in one flat buffer."""
+
+    def __init__(self, name, module, params_with_names, group, device, compute_dtype, reduce_dtype):
+        self.name = name
+        self.module = module
+        self.group = group
+        self.world = dist.get_world_size(group) if dist.is_initialized() else 1
+        self.rank = dist.get_rank(group) if dist.is_initialized() else 0
+        self.device = device
+        self.compute_dtype = compute_dtype
+        self.reduce_dtype = reduce_dtype
+
+        self.param_names = [n for n, _ in params_with_names]
+        self.params: List[nn.Parameter] = [p for _, p in params_with_names]
+        self.numels = [p.numel() for p in self.params]
+        self.shapes = [p.shape for p in self.params]
+        total = sum(self.numels)
+        self.padded = -(-total // self.world) * self.world  # ceil to world multiple
+        self.total = total
+        self.shard_len = self.padded // self.world
+
+        # build the fp32 master shard from current values
+        flat = torch.zeros(self.padded, dtype=torch.float32, device=device)
+        off = 0
+        for p in self.params:
+            flat[off : off + p.numel()].copy_(p.detach().reshape(-1).to(device, torch.float32))
+            off += p.numel()
+        lo = self.rank * self.shard_len
+        self.shard = nn.Parameter(flat[lo : lo + self.shard_len].clone())
+        del flat
+
+        # persistent full buffer (storage resized 0<->padded across reshard)
+        self.full = torch.empty(self.padded, dtype=self.compute_dtype, device=device)
+        # wire params as views into the full buffer ONCE; storage identity is
+        # stable so activation-saved tensors survive reshard/unshard cycles
+        off = 0
+        for p, shape in zip(self.params, self.shapes):
+            p.data = self.full[off : off + shape.numel()].view(shape)
+            off += shape.numel()
+        self._resident = True
+        self.reshard()
+
+        self.grads_ready = 0
+        self.grad_hooks = []
+        self._prefetched = False
+
+    # -- collective ops ---------------------------------------------------
+
+    def unshard(self):
+        """all_gather the compute-dtype full buffer (idempotent)."""
+        if self._resident:
+            return
+        self.full.untyped_storage().resize_(self.padded * self.full.element_size())
+        shard_c = self.shard.detach().to(self.compute_dtype)
+        if dist.is_initialized() and self.world > 1:
+            _all_gather_flat(self.full, shard_c, self.group)
+        else:
+            self.full.copy_(shard_c)
+        self._resident = True
+
+    def reshard(self):
+        if not self._resident:
+            return
+        self.full.untyped_storage().resize_(0)
+        self._resident = False
+
+    def reduce_grads(self, accumulate: bool = True):
+        """Flatten unit grads, reduce_scatter (mean) into the fp32 shard grad."""
+        flat_grad = torch.zeros(self.padded, dtype=self.reduce_dtype, device=self.device)
+        off = 0
+        for p in self.params:
+            if p.grad is not None:
+                flat_grad[off : off + p.numel()].copy_(p.grad.reshape(-1).to(self.reduce_dtype))
+                p.grad = None
+            off += p.numel()
+        out = torch.empty(self.shard_len, dtype=self.reduce_dtype, device=self.device)
+        if dist.is_initialized() and self.world > 1:
+            # mean over ranks: AVG inside the RCCL kernel; gloo falls back to
+            # all_reduce(SUM)+slice inside _reduce_scatter_flat, divide here
+            op = dist.ReduceOp.AVG if _is_nccl(self.group) else dist.ReduceOp.SUM
+            _reduce_scatter_flat(out, flat_grad, self.group, op)
+            if not _is_nccl(self.group):
+                out.div_(self.world)
+        else:
+            out.copy_(flat_grad[: self.shard_len])
+        out32 = out.to(torch.float32)
+        if self.shard.grad is None or not accumulate:
+            self.shard.grad = out32
+        else:
+            self.shard.grad.add_(out32)
+
+    @torch.no_grad()
+    def gather_full_fp32(self) -> torch.Tensor:
+        """All ranks receive the full fp32 flat buffer (for state dicts)."""
+        if dist.is_initialized() and self.world > 1:
+            full = torch.empty(self.padded, dtype=torch.float32, device=self.device)
+            _all_gather_flat(full, self.shard.detach(), self.group)
+            return full
+        return self.shard.detach().clone()
+
+    @torch.no_grad()
+    def load_from_full_fp32(self, flat: torch.Tensor):
+        lo = self.rank * self.shard_len
+        self.shard.copy_(flat[lo : lo + self.shard_len].to(self.shard.device))
 
 
 class ShardedModel(nn.Module):
-    """Placeholder wrapper type so isinstance checks across the framework are
-    stable while the engine is being built."""
+    """The user-facing wrapper (reference: torch FSDP2 fully_shard semantics)."""
 
-    def __init__(self, module: nn.Module):
+    def __init__(
+        self,
+        module: nn.Module,
+        auto_wrap_policy=None,
+        transformer_cls_names=None,
+        min_num_params: int = 1_000_000,
+        compute_dtype: Optional[torch.dtype] = None,
+        reduce_dtype: Optional[torch.dtype] = None,
+        reshard_after_forward: bool = True,
+        process_group=None,
+        device: Optional[torch.device] = None,
+        activation_checkpointing: bool = False,
+    ):
         super().__init__()
         self.module = module
+        self.group = process_group
+        self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        if device is None:
+            device = torch.device("cuda", torch.cuda.current_device()) if torch.cuda.is_available() else torch.device("cpu")
+        self.device = device
+        if compute_dtype is None:
+            compute_dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+        self.compute_dtype = compute_dtype
+        self.reduce_dtype = reduce_dtype or compute_dtype
+        self.reshard_after_forward = reshard_after_forward
+        self.require_backward_grad_sync = True  # no_sync window flag
+        self._ac = activation_checkpointing
+
+    # unit discovery ------------------------------------------------------
+
+        module.to(device)
+        unit_modules = self._select_units(module, auto_wrap_policy, transformer_cls_names, min_num_params)
+        self.units: List[_FlatUnit] = []
+        claimed = set()
+        for name, m in unit_modules:
+            params = [(f"{name}.{pn}" if name else pn, p) for pn, p in m.named_parameters() if p.requires_grad]
+            params = [(n, p) for n, p in params if id(p) not in claimed]
+            if not params:
+                continue
+            for _, p in params:
+                claimed.add(id(p))
+            self.units.append(
+                _FlatUnit(name, m, params, self.group, device, self.compute_dtype, self.reduce_dtype)
+            )
+        # root unit: all remaining params
+        rest = [(n, p) for n, p in module.named_parameters() if p.requires_grad and id(p) not in claimed]
+        if rest:
+            self.units.append(_FlatUnit("", module, rest, self.group, device, self.compute_dtype, self.reduce_dtype))
+        self._unit_of_module = {id(u.module): u for u in self.units}
+        self._param_to_unit = {}
+        for u in self.units:
+            for p in u.params:
+                self._param_to_unit[id(p)] = u
+
+        self._register_hooks()
+        self._fwd_order: List[_FlatUnit] = []
+        self._order_recorded = False
+        self._comm_stream = torch.cuda.Stream() if device.type == "cuda" else None
+        if self._ac:
+            self._apply_activation_checkpointing()
+
+    @staticmethod
+    def _select_units(module, policy, transformer_cls_names, min_num_params):
+        units = []
+        if callable(policy):
+            for name, m in module.named_modules():
+                if name and policy(m):
+                    units.append((name, m))
+        elif transformer_cls_names:
+            names = set(transformer_cls_names)
+            for name, m in module.named_modules():
+                if name and m.__class__.__name__ in names:
+                    units.append((name, m))
+        else:
+            # default: maximal non-overlapping subtrees with >= min_num_params
+            def walk(prefix, m):
+                n_params = sum(p.numel() for p in m.parameters())
+                children = list(m.named_children())
+                if prefix and n_params >= min_num_params:
+                    big_children = [
+                        (f"{prefix}.{cn}", c)
+                        for cn, c in children
+                        if sum(p.numel() for p in c.parameters()) >= min_num_params
+                    ]
+                    own = n_params - sum(sum(p.numel() for p in c.parameters()) for _, c in big_children)
+                    if len(big_children) <= 1 or own >= min_num_params:
+                        units.append((prefix, m))
+                        return
+                for cn, c in children:
+                    walk(f"{prefix}.{cn}" if prefix else cn, c)
+
+            walk("", module)
+        # drop nested units (keep outermost)
+        kept = []
+        for name, m in units:
+            if not any(name.startswith(other + ".") for other, _ in units if other != name):
+                kept.append((name, m))
+        return kept
+
+    # hooks ---------------------------------------------------------------
+
+    def _register_hooks(self):
+        for u in self.units:
+            u.module.register_forward_pre_hook(self._make_fwd_pre(u))
+            u.module.register_forward_hook(self._make_fwd_post(u))
+            u.module.register_full_backward_pre_hook(self._make_bwd_pre(u))
+            for p in u.params:
+                h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
+                u.grad_hooks.append(h)
+
+    def _stream_unshard(self, unit):
+        """Issue unit's all_gather on the comm stream; compute stream will wait."""
+        if self._comm_stream is None:
+            unit.unshard()
+            return
+        if unit._resident:
+            return
+        self._comm_stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self._comm_stream):
+            unit.unshard()
+        unit._gather_event = torch.cuda.Event()
+        unit._gather_event.record(self._comm_stream)
+        unit._prefetched = True
+
+    def _wait_unshard(self, unit):
+        if self._comm_stream is not None and getattr(unit, "_gather_event", None) is not None:
+            torch.cuda.current_stream().wait_event(unit._gather_event)
+            unit._gather_event = None
+        elif not unit._resident:
+            unit.unshard()
+
+    def _make_fwd_pre(self, unit):
+        def hook(module, args):
+            if not self._order_recorded:
+                self._fwd_order.append(unit)
+            self._stream_unshard(unit)
+            self._wait_unshard(unit)
+            # prefetch the NEXT unit in recorded order
+            if self._order_recorded:
+                idx = self._fwd_index.get(id(unit))
+                if idx is not None and idx + 1 < len(self._fwd_order):
+                    self._stream_unshard(self._fwd_order[idx + 1])
+            return None
+
+        return hook
+
+    def _make_fwd_post(self, unit):
+        def hook(module, args, output):
+            if self.reshard_after_forward and self.world > 1 and self.module.training:
+                unit.reshard()
+            elif not self.module.training and self.reshard_after_forward:
+                unit.reshard()
+            return None
+
+        return hook
+
+    def _make_bwd_pre(self, unit):
+        def hook(module, grad_output):
+            self._wait_for_comm()
+            unit.unshard()
+            # backward prefetch: previous unit in forward order comes next
+            idx = self._fwd_index.get(id(unit)) if self._order_recorded else None
+            if idx is not None and idx - 1 >= 0:
+                self._stream_unshard(self._fwd_order[idx - 1])
+            return None
+
+        return hook
+
+    def _wait_for_comm(self):
+        if self._comm_stream is not None:
+            torch.cuda.current_stream().wait_stream(self._comm_stream)
+
+    def _on_grad_ready(self, param):
+        unit = self._param_to_unit[id(param)]
+        unit.grads_ready += 1
+        if unit.grads_ready >= len(unit.params):
+            unit.grads_ready = 0
+            if self.require_backward_grad_sync:
+                unit.reduce_grads()
+                unit.reshard()
+            # under no_sync keep full grads resident for accumulation
 
     def forward(self, *args, **kwargs):
+        if not self._order_recorded:
+            out = self.module(*args, **kwargs)
+            self._order_recorded = True
+            self._fwd_index = {id(u): i for i, u in enumerate(self._fwd_order)}
+            return out
+        if self._fwd_order:
+            self._stream_unshard(self._fwd_order[0])
         return self.module(*args, **kwargs)
 
+    def finalize_backward(self):
+        """Drain the comm stream and reshard every still-resident unit.
 
-def gather_full_state_dict(model: "ShardedModel"):
-    return model.module.state_dict()
+        Mandatory residency epilogue: a unit left resident across the
+        optimizer step would serve STALE full-buffer values on the next
+        forward (the all_gather is what propagates updated shards). Called
+        from ``Accelerator.backward``; call directly after raw
+        ``loss.backward()``.
+        """
+        self._wait_for_comm()
+        for u in self.units:
+            u.reshard()
+
+    @contextlib.contextmanager
+    def no_sync(self):
+        old = self.require_backward_grad_sync
+        self.require_backward_grad_sync = False
+        try:
+            yield
+        finally:
+            self.require_backward_grad_sync = old
+
+    # gradient clipping (sharded): local norm² + all_reduce ---------------
+
+    def clip_grad_norm_(self, max_norm: float, norm_type: float = 2.0) -> torch.Tensor:
+        assert norm_type == 2.0, "only L2 clipping is supported for sharded models"
+        device = self.device
+        local = torch.zeros(1, device=device)
+        for u in self.units:
+            if u.shard.grad is not None:
+                local += (u.shard.grad.float() ** 2).sum()
+        if dist.is_initialized() and self.world > 1:
+            dist.all_reduce(local, group=self.group)
+        total_norm = local.sqrt().squeeze()
+        clip_coef = (max_norm / (total_norm + 1e-6)).clamp(max=1.0)
+        for u in self.units:
+            if u.shard.grad is not None:
+                u.shard.grad.mul_(clip_coef)
+        return total_norm
+
+    # state dict ----------------------------------------------------------
+
+    def shard_parameters(self):
+        return [u.shard for u in self.units]
+
+    def param_swap_map(self) -> Dict[nn.Parameter, nn.Parameter]:
+        """original param -> master shard, for the optimizer-param swap."""
+        mapping = {}
+        for u in self.units:
+            for p in u.params:
+                mapping[p] = u.shard  # many->one: optimizer dedups below
+        return mapping
+
+    def state_dict(self, *args, **kwargs):
+        return self.full_state_dict()
+
+    @torch.no_grad()
+    def full_state_dict(self):
+        """Full fp32 state dict on every rank (all_gather per unit)."""
+        out = {}
+        for u in self.units:
+            flat = u.gather_full_fp32()
+            off = 0
+            for name, shape in zip(u.param_names, u.shapes):
+                out[name] = flat[off : off + shape.numel()].view(shape).cpu().clone()
+                off += shape.numel()
+            del flat
+        # buffers pass through
+        for name, buf in self.module.named_buffers():
+            out[name] = buf.detach().cpu()
+        return out
+
+    @torch.no_grad()
+    def load_state_dict(self, state_dict, strict: bool = True):
+        for u in self.units:
+            flat = torch.zeros(u.padded, dtype=torch.float32, device=self.device)
+            off = 0
+            for name, shape in zip(u.param_names, u.shapes):
+                if name in state_dict:
+                    flat[off : off + shape.numel()].copy_(state_dict[name].reshape(-1).to(self.device, torch.float32))
+                elif strict:
+                    raise KeyError(f"missing key {name} in sharded load")
+                off += shape.numel()
+            u.load_from_full_fp32(flat)
+            del flat
+        for name, buf in self.module.named_buffers():
+            if name in state_dict:
+                buf.copy_(state_dict[name].to(buf.device, buf.dtype))
+
+    @torch.no_grad()
+    def sharded_state_dict(self):
+        """Per-rank shard state with unflattening metadata — zero cross-rank
+        communication (reference: fsdp_utils.py:107-118 rationale)."""
+        return {
+            "world_size": self.world,
+            "rank": dist.get_rank(self.group) if dist.is_initialized() else 0,
+            "units": {
+                u.name or "<root>": {
+                    "shard": u.shard.detach().cpu().clone(),
+                    "param_names": list(u.param_names),
+                    "shapes": [list(s) for s in u.shapes],
+                    "padded": u.padded,
+                }
+                for u in self.units
+            },
+        }
+
+    @torch.no_grad()
+    def load_sharded_state_dict(self, sd):
+        if sd["world_size"] != self.world:
+            raise ValueError(f"sharded checkpoint world_size {sd['world_size']} != current {self.world}")
+        for u in self.units:
+            u.shard.copy_(sd["units"][u.name or "<root>"]["shard"].to(u.shard.device))
+
+    def _apply_activation_checkpointing(self):
+        from torch.utils.checkpoint import checkpoint
+
+        for u in self.units:
+            if u.module is self.module:
+                continue
+            mod = u.module
+            orig_forward = mod.forward
+
+            def make_ckpt_forward(fwd):
+                def ckpt_forward(*a, **k):
+                    if self.module.training:
+                        return checkpoint(fwd, *a, use_reentrant=False, **k)
+                    return fwd(*a, **k)
+
+                return ckpt_forward
+
+            mod.forward = make_ckpt_forward(orig_forward)
+
+    def train(self, mode: bool = True):
+        super().train(mode)
+        self.module.train(mode)
+        return self
+
+    def extra_repr(self):
+        return f"world={self.world}, units={len(self.units)}, compute={self.compute_dtype}"
+
+
+def gather_full_state_dict(model: ShardedModel):
+    return model.full_state_dict()
 
 
 def fsdp_prepare(accelerator, args, device_placement):
-    raise NotImplementedError(
-        "The MI355X sharded-parameter engine (FSDP equivalent) is under construction this round; "
-        "use the DDP path (default) for now."
-    )
+    """Route Accelerator.prepare for the FSDP world
+    (reference: accelerator.py:1673 _prepare_fsdp2)."""
+    from ..optimizer import AcceleratedOptimizer
+    from ..scheduler import AcceleratedScheduler
+
+    plugin = accelerator.state.fsdp_plugin
+    models = [a for a in args if isinstance(a, nn.Module)]
+    optimizers = [a for a in args if isinstance(a, torch.optim.Optimizer)]
+    if len(models) > 1:
+        raise ValueError("FSDP mode supports preparing exactly one model (with its optimizer) at a time.")
+
+    result = []
+    wrapped_model = None
+    swap_map = None
+    for obj in args:
+        if isinstance(obj, torch.utils.data.DataLoader):
+            result.append(accelerator.prepare_data_loader(obj))
+        elif isinstance(obj, nn.Module):
+            mp = plugin.mixed_precision_policy or {}
+            compute_dtype = mp.get("param_dtype")
+            if compute_dtype is None and accelerator.mixed_precision == "bf16":
+                compute_dtype = torch.bfloat16
+            elif compute_dtype is None and accelerator.mixed_precision == "fp16":
+                compute_dtype = torch.float16
+            wrapped_model = ShardedModel(
+                obj,
+                transformer_cls_names=plugin.transformer_cls_names_to_wrap,
+                min_num_params=plugin.min_num_params or 1_000_000,
+                compute_dtype=compute_dtype,
+                reduce_dtype=mp.get("reduce_dtype"),
+                reshard_after_forward=plugin.reshard_after_forward,
+                device=accelerator.device,
+                activation_checkpointing=plugin.activation_checkpointing,
+            )
+            swap_map = wrapped_model.param_swap_map()
+            accelerator._models.append(wrapped_model)
+            result.append(wrapped_model)
+        else:
+            result.append(obj)
+
+    final = []
+    for obj in result:
+        if isinstance(obj, torch.optim.Optimizer) and not isinstance(obj, AcceleratedOptimizer):
+            if swap_map is not None:
+                _swap_optimizer_params(obj, swap_map)
+            obj = AcceleratedOptimizer(obj, device_placement=False, scaler=accelerator.scaler)
+            obj._sharded = True
+            accelerator._optimizers.append(obj)
+        final.append(obj)
+    result = final
+
+    final = []
+    for obj in result:
+        if (
+            not isinstance(obj, (nn.Module, torch.utils.data.DataLoader, AcceleratedOptimizer))
+            and hasattr(obj, "optimizer")
+            and hasattr(obj, "step")
+        ):
+            obj = accelerator.prepare_scheduler(obj)
+        final.append(obj)
+
+    return tuple(final) if len(final) != 1 else final[0]
+
+
+def _swap_optimizer_params(optimizer, swap_map):
+    """Replace original params with master shards, deduplicated
+    (reference: the FSDP2 optimizer-param swap, accelerator.py:1714-1726)."""
+    for group in optimizer.param_groups:
+        new_params, seen = [], set()
+        for p in group["params"]:
+            target = swap_map.get(p, p)
+            if id(target) not in seen:
+                seen.add(id(target))
+                new_params.append(target)
+        group["params"] = new_params

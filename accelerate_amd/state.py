@@ -318,7 +318,7 @@ class AcceleratorState:
 
         self.use_fsdp = fsdp_plugin is not None or parse_flag_from_env("ACCELERATE_USE_FSDP")
         self.fsdp_plugin = fsdp_plugin
-        if self.use_fsdp and self.distributed_type == DistributedType.MULTI_GPU:
+        if self.use_fsdp and self.distributed_type in (DistributedType.MULTI_GPU, DistributedType.MULTI_CPU):
             self.distributed_type = DistributedType.FSDP
             if self.fsdp_plugin is None:
                 from .utils.dataclasses import FullyShardedDataParallelPlugin
