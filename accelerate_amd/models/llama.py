@@ -1,0 +1,224 @@
+"""Llama-family decoder (Llama-3 8B/70B configs) — the FSDP-bench and
+big-model-inference model (BASELINE configs #3/#4).
+
+MI355X-first: GQA attention as batched rocBLAS GEMMs + softmax (the fused
+CDNA4 flash kernel slots in via ``attn_impl='fused'`` as it lands), SwiGLU
+MLP, RMSNorm, RoPE with a host-precomputed cos/sin table (CDNA4 guide:
+on-device trig turns memory-bound kernels VALU-bound).
+"""
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: int = 8
+    max_position_embeddings: int = 8192
+    rms_norm_eps: float = 1e-5
+    rope_theta: float = 500000.0
+    tie_word_embeddings: bool = False
+
+    @classmethod
+    def llama3_8b(cls, **overrides):
+        return cls(**overrides)
+
+    @classmethod
+    def llama3_70b(cls, **overrides):
+        base = dict(
+            hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
+            num_attention_heads=64, num_key_value_heads=8,
+        )
+        base.update(overrides)
+        return cls(**base)
+
+    @classmethod
+    def tiny(cls, **overrides):
+        base = dict(
+            vocab_size=1024, hidden_size=256, intermediate_size=688, num_hidden_layers=4,
+            num_attention_heads=8, num_key_value_heads=4, max_position_embeddings=512,
+        )
+        base.update(overrides)
+        return cls(**base)
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, hidden_size, eps=1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden_size))
+        self.eps = eps
+
+    def forward(self, x):
+        dtype = x.dtype
+        x = x.float()
+        x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
+        return (self.weight * x.to(dtype)) if self.weight.dtype == dtype else (self.weight.to(dtype) * x.to(dtype))
+
+
+def build_rope_cache(seq_len, head_dim, theta, device, dtype=torch.float32):
+    inv_freq = 1.0 / (theta ** (torch.arange(0, head_dim, 2, device=device).float() / head_dim))
+    t = torch.arange(seq_len, device=device).float()
+    freqs = torch.outer(t, inv_freq)
+    return freqs.cos().to(dtype), freqs.sin().to(dtype)
+
+
+def apply_rope(x, cos, sin):
+    # x: [B, H, S, D]
+    d = x.shape[-1] // 2
+    x1, x2 = x[..., :d], x[..., d:]
+    cos = cos[None, None, : x.shape[2], :]
+    sin = sin[None, None, : x.shape[2], :]
+    return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], dim=-1)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, config: LlamaConfig, attn_impl: str = "math"):
+        super().__init__()
+        self.n_heads = config.num_attention_heads
+        self.n_kv = config.num_key_value_heads
+        self.head_dim = config.hidden_size // config.num_attention_heads
+        self.q_proj = nn.Linear(config.hidden_size, self.n_heads * self.head_dim, bias=False)
+        self.k_proj = nn.Linear(config.hidden_size, self.n_kv * self.head_dim, bias=False)
+        self.v_proj = nn.Linear(config.hidden_size, self.n_kv * self.head_dim, bias=False)
+        self.o_proj = nn.Linear(self.n_heads * self.head_dim, config.hidden_size, bias=False)
+        self.attn_impl = attn_impl
+
+    def forward(self, x, cos, sin, kv_cache=None):
+        B, S, _ = x.shape
+        q = self.q_proj(x).view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
+        k = self.k_proj(x).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
+        v = self.v_proj(x).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
+        q = apply_rope(q, cos, sin)
+        k = apply_rope(k, cos, sin)
+        past_len = 0
+        if kv_cache is not None:
+            past_len = kv_cache["len"]
+            kv_cache["k"][:, :, past_len : past_len + S] = k
+            kv_cache["v"][:, :, past_len : past_len + S] = v
+            kv_cache["len"] = past_len + S
+            k = kv_cache["k"][:, :, : past_len + S]
+            v = kv_cache["v"][:, :, : past_len + S]
+        # GQA: expand kv heads
+        if self.n_kv != self.n_heads:
+            rep = self.n_heads // self.n_kv
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
+
+        if self.attn_impl == "fused" and x.is_cuda and S > 1:
+            from ..ops.attention import flash_attention_forward
+
+            ctx = flash_attention_forward(q, k, v, causal=True)
+        else:
+            scale = 1.0 / math.sqrt(self.head_dim)
+            scores = torch.matmul(q, k.transpose(-1, -2)) * scale
+            total = k.shape[2]
+            if S > 1:
+                causal = torch.ones(S, total, dtype=torch.bool, device=x.device).tril(diagonal=past_len)
+                scores = scores.masked_fill(~causal, torch.finfo(scores.dtype).min)
+            probs = F.softmax(scores.float(), dim=-1).to(q.dtype)
+            ctx = torch.matmul(probs, v)
+        ctx = ctx.transpose(1, 2).reshape(B, S, -1)
+        return self.o_proj(ctx)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, config: LlamaConfig):
+        super().__init__()
+        self.gate_proj = nn.Linear(config.hidden_size, config.intermediate_size, bias=False)
+        self.up_proj = nn.Linear(config.hidden_size, config.intermediate_size, bias=False)
+        self.down_proj = nn.Linear(config.intermediate_size, config.hidden_size, bias=False)
+
+    def forward(self, x):
+        return self.down_proj(F.silu(self.gate_proj(x)) * self.up_proj(x))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, config: LlamaConfig, attn_impl: str = "math"):
+        super().__init__()
+        self.input_layernorm = RMSNorm(config.hidden_size, config.rms_norm_eps)
+        self.self_attn = LlamaAttention(config, attn_impl)
+        self.post_attention_layernorm = RMSNorm(config.hidden_size, config.rms_norm_eps)
+        self.mlp = LlamaMLP(config)
+
+    def forward(self, x, cos, sin, kv_cache=None):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin, kv_cache)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, config: LlamaConfig = None, attn_impl: str = "math"):
+        super().__init__()
+        self.config = config or LlamaConfig()
+        c = self.config
+        self.embed_tokens = nn.Embedding(c.vocab_size, c.hidden_size)
+        self.layers = nn.ModuleList([LlamaDecoderLayer(c, attn_impl) for _ in range(c.num_hidden_layers)])
+        self.norm = RMSNorm(c.hidden_size, c.rms_norm_eps)
+        self.lm_head = nn.Linear(c.hidden_size, c.vocab_size, bias=False)
+        if c.tie_word_embeddings:
+            self.lm_head.weight = self.embed_tokens.weight
+        head_dim = c.hidden_size // c.num_attention_heads
+        cos, sin = build_rope_cache(c.max_position_embeddings, head_dim, c.rope_theta, "cpu")
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        if isinstance(module, nn.Linear):
+            module.weight.data.normal_(mean=0.0, std=0.02)
+            if module.bias is not None:
+                module.bias.data.zero_()
+        elif isinstance(module, nn.Embedding):
+            module.weight.data.normal_(mean=0.0, std=0.02)
+
+    def forward(self, input_ids, labels=None, kv_caches=None):
+        x = self.embed_tokens(input_ids)
+        past = kv_caches[0]["len"] if kv_caches is not None else 0
+        cos = self.rope_cos[past:].to(x.device)
+        sin = self.rope_sin[past:].to(x.device)
+        for i, layer in enumerate(self.layers):
+            x = layer(x, cos, sin, kv_caches[i] if kv_caches is not None else None)
+        x = self.norm(x)
+        logits = self.lm_head(x)
+        loss = None
+        if labels is not None:
+            loss = F.cross_entropy(
+                logits[:, :-1].reshape(-1, logits.shape[-1]).float(), labels[:, 1:].reshape(-1)
+            )
+        return {"loss": loss, "logits": logits}
+
+    def make_kv_caches(self, batch_size, max_len, device, dtype):
+        c = self.config
+        head_dim = c.hidden_size // c.num_attention_heads
+        return [
+            {
+                "k": torch.zeros(batch_size, c.num_key_value_heads, max_len, head_dim, device=device, dtype=dtype),
+                "v": torch.zeros(batch_size, c.num_key_value_heads, max_len, head_dim, device=device, dtype=dtype),
+                "len": 0,
+            }
+            for _ in range(c.num_hidden_layers)
+        ]
+
+    @torch.no_grad()
+    def generate(self, input_ids, max_new_tokens: int = 32):
+        device = input_ids.device
+        dtype = self.lm_head.weight.dtype
+        caches = self.make_kv_caches(input_ids.shape[0], input_ids.shape[1] + max_new_tokens, device, dtype)
+        out = self(input_ids, kv_caches=caches)
+        tokens = [input_ids]
+        next_tok = out["logits"][:, -1].argmax(-1, keepdim=True)
+        tokens.append(next_tok)
+        for _ in range(max_new_tokens - 1):
+            out = self(next_tok, kv_caches=caches)
+            next_tok = out["logits"][:, -1].argmax(-1, keepdim=True)
+            tokens.append(next_tok)
+        return torch.cat(tokens, dim=1)

@@ -154,7 +154,41 @@ void unscale_and_check(std::vector<at::Tensor> grads, at::Tensor inv_scale, at::
                      h.meta, inv_scale.data_ptr<float>(), found_inf.data_ptr<float>());
 }
 
+// fp8 kernels (fp8_kernels.hip)
+__global__ void fp8_cast_amax_e4m3(const ushort*, unsigned char*, const float*, float*, int64_t);
+__global__ void fp8_cast_amax_e5m2(const ushort*, unsigned char*, const float*, float*, int64_t);
+__global__ void fp8_update_scale(const float*, int, float, float, float*, float*);
+
+void fp8_cast_amax(at::Tensor in, at::Tensor out, at::Tensor scale, at::Tensor amax, bool e5m2) {
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.scalar_type() == at::kBFloat16,
+              "fp8_cast_amax: input must be contiguous bf16 on GPU");
+  TORCH_CHECK(out.numel() == in.numel() && out.element_size() == 1, "fp8_cast_amax: bad output");
+  int64_t n = in.numel();
+  // memory-bound: cap grid and grid-stride (CDNA4 Guideline 11)
+  int grid = (int)std::min<int64_t>((n + kBlockThreads * 8 - 1) / (kBlockThreads * 8), 2048);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto* inp = reinterpret_cast<const ushort*>(in.data_ptr());
+  auto* outp = reinterpret_cast<unsigned char*>(out.data_ptr());
+  if (e5m2) {
+    hipLaunchKernelGGL(fp8_cast_amax_e5m2, dim3(grid), dim3(kBlockThreads), 0, stream.stream(),
+                       inp, outp, scale.data_ptr<float>(), amax.data_ptr<float>(), n);
+  } else {
+    hipLaunchKernelGGL(fp8_cast_amax_e4m3, dim3(grid), dim3(kBlockThreads), 0, stream.stream(),
+                       inp, outp, scale.data_ptr<float>(), amax.data_ptr<float>(), n);
+  }
+}
+
+void fp8_update_scale_fn(at::Tensor history, double fp8_max, double margin_pow2,
+                         at::Tensor scale, at::Tensor scale_inv) {
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(fp8_update_scale, dim3(1), dim3(1), 0, stream.stream(),
+                     history.data_ptr<float>(), (int)history.numel(), (float)fp8_max,
+                     (float)margin_pow2, scale.data_ptr<float>(), scale_inv.data_ptr<float>());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fp8_cast_amax", &fp8_cast_amax, "bf16 -> fp8 cast with fused amax (gfx950)");
+  m.def("fp8_update_scale", &fp8_update_scale_fn, "delayed-scaling scale update");
   m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (gfx950)");
   m.def("fused_adamw_planned", &fused_adamw_planned,
         "graph-capturable fused AdamW over a cached device plan");
