@@ -82,8 +82,8 @@ class BertSelfAttention(nn.Module):
         # batched rocBLAS GEMMs + softmax; S is small for the headline config
         scores = torch.matmul(q, k.transpose(-1, -2)) * (1.0 / math.sqrt(self.head_dim))
         if attention_mask is not None:
-            scores = scores + attention_mask
-        probs = F.softmax(scores, dim=-1)
+            scores = scores + attention_mask.to(scores.dtype)
+        probs = F.softmax(scores.float(), dim=-1).to(v.dtype)
         if self.training and self.dropout_p > 0:
             probs = F.dropout(probs, p=self.dropout_p)
         ctx = torch.matmul(probs, v)
@@ -120,9 +120,12 @@ class BertModel(nn.Module):
     def forward(self, input_ids, attention_mask=None, token_type_ids=None):
         extended_mask = None
         if attention_mask is not None:
-            # [B, S] 1/0 mask -> additive [B, 1, 1, S]
-            extended_mask = (1.0 - attention_mask[:, None, None, :].to(input_ids.device).float()) * torch.finfo(
-                torch.float32
+            # [B, S] 1/0 mask -> additive [B, 1, 1, S] in the compute dtype
+            dtype = self.embeddings.word_embeddings.weight.dtype
+            if dtype not in (torch.float32, torch.bfloat16, torch.float16):
+                dtype = torch.float32
+            extended_mask = (1.0 - attention_mask[:, None, None, :].to(device=input_ids.device, dtype=dtype)) * torch.finfo(
+                dtype
             ).min
         hidden = self.embeddings(input_ids, token_type_ids)
         for layer in self.layers:

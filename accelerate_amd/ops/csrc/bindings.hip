@@ -16,6 +16,8 @@ __global__ void fused_adamw_kernel(TensorListMeta, float, float, float, float, f
 __global__ void adamw_incr_step_kernel(float*);
 __global__ void fused_adamw_dev_kernel(TensorListMeta, const float*, const float*,
                                        float, float, float, float, const float*, const float*);
+__global__ void fused_adamw_bf16_kernel(TensorListMeta, const float*, const float*,
+                                        float, float, float, float, const float*, const float*);
 __global__ void l2norm_squared_kernel(TensorListMeta, float*);
 __global__ void clip_coef_kernel(const float*, float, float*, float*);
 __global__ void multi_tensor_scale_kernel(TensorListMeta, const float*);
@@ -95,21 +97,30 @@ void fused_adamw_planned(at::Tensor addrs_numels, at::Tensor chunk_prefix,
                          int64_t n_tensors, int64_t total_chunks,
                          at::Tensor step, at::Tensor lr,
                          double beta1, double beta2, double eps, double weight_decay,
-                         c10::optional<at::Tensor> grad_scale, c10::optional<at::Tensor> found_inf) {
+                         c10::optional<at::Tensor> grad_scale, c10::optional<at::Tensor> found_inf,
+                         bool bf16_master = false) {
   TensorListMeta meta;
   meta.addrs = addrs_numels.data_ptr<int64_t>();
-  meta.numels = addrs_numels.data_ptr<int64_t>() + 4 * n_tensors;
+  meta.numels = addrs_numels.data_ptr<int64_t>() + (bf16_master ? 5 : 4) * n_tensors;
   meta.chunk_prefix = chunk_prefix.data_ptr<int32_t>();
   meta.n_tensors = static_cast<int32_t>(n_tensors);
-  meta.n_lists = 4;
+  meta.n_lists = bf16_master ? 5 : 4;
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(adamw_incr_step_kernel, dim3(1), dim3(1), 0, stream.stream(),
                      step.data_ptr<float>());
-  hipLaunchKernelGGL(fused_adamw_dev_kernel, dim3(total_chunks), dim3(kBlockThreads), 0, stream.stream(),
-                     meta, step.data_ptr<float>(), lr.data_ptr<float>(),
-                     (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
-                     grad_scale.has_value() ? grad_scale->data_ptr<float>() : nullptr,
-                     found_inf.has_value() ? found_inf->data_ptr<float>() : nullptr);
+  if (bf16_master) {
+    hipLaunchKernelGGL(fused_adamw_bf16_kernel, dim3(total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                       meta, step.data_ptr<float>(), lr.data_ptr<float>(),
+                       (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                       grad_scale.has_value() ? grad_scale->data_ptr<float>() : nullptr,
+                       found_inf.has_value() ? found_inf->data_ptr<float>() : nullptr);
+  } else {
+    hipLaunchKernelGGL(fused_adamw_dev_kernel, dim3(total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                       meta, step.data_ptr<float>(), lr.data_ptr<float>(),
+                       (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                       grad_scale.has_value() ? grad_scale->data_ptr<float>() : nullptr,
+                       found_inf.has_value() ? found_inf->data_ptr<float>() : nullptr);
+  }
 }
 
 at::Tensor l2norm_squared(std::vector<at::Tensor> grads) {
@@ -191,7 +202,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_update_scale", &fp8_update_scale_fn, "delayed-scaling scale update");
   m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (gfx950)");
   m.def("fused_adamw_planned", &fused_adamw_planned,
-        "graph-capturable fused AdamW over a cached device plan");
+        "graph-capturable fused AdamW over a cached device plan",
+        pybind11::arg("addrs_numels"), pybind11::arg("chunk_prefix"), pybind11::arg("n_tensors"),
+        pybind11::arg("total_chunks"), pybind11::arg("step"), pybind11::arg("lr"),
+        pybind11::arg("beta1"), pybind11::arg("beta2"), pybind11::arg("eps"),
+        pybind11::arg("weight_decay"), pybind11::arg("grad_scale") = pybind11::none(),
+        pybind11::arg("found_inf") = pybind11::none(), pybind11::arg("bf16_master") = false);
   m.def("l2norm_squared", &l2norm_squared, "global L2 norm squared over tensor list");
   m.def("clip_grad_norm", &clip_grad_norm, "on-device clip_grad_norm_, returns total norm");
   m.def("multi_tensor_scale", &multi_tensor_scale, "g *= *coef");

@@ -149,6 +149,100 @@ __global__ void fused_adamw_dev_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// bf16-weights AdamW: model params/grads are bf16, the optimizer keeps the
+// fp32 master copy + fp32 m/v (standard production bf16 recipe; kills the
+// per-step autocast weight-cast traffic and halves gradient all-reduce bytes
+// over xGMI). Lists: [0]=param_bf16, [1]=grad_bf16, [2]=m, [3]=v, [4]=master.
+// ---------------------------------------------------------------------------
+
+#include <hip/hip_bf16.h>
+
+__device__ __forceinline__ float bf16_to_f32(ushort u) {
+  union { unsigned int i; float f; } cv;
+  cv.i = ((unsigned int)u) << 16;
+  return cv.f;
+}
+
+__device__ __forceinline__ ushort f32_to_bf16(float f) {
+  // round-to-nearest-even, matching torch's fp32->bf16 cast exactly
+  union { float f; unsigned int i; } cv;
+  cv.f = f;
+  if ((cv.i & 0x7F800000u) == 0x7F800000u) {  // inf/nan: truncate
+    return (ushort)(cv.i >> 16) | (ushort)((cv.i & 0xFFFFu) ? 0x40 : 0);
+  }
+  unsigned int lsb = (cv.i >> 16) & 1u;
+  cv.i += 0x7FFFu + lsb;
+  return (ushort)(cv.i >> 16);
+}
+
+__global__ void fused_adamw_bf16_kernel(
+    TensorListMeta meta,
+    const float* __restrict__ step_ptr, const float* __restrict__ lr_ptr,
+    float beta1, float beta2, float eps, float weight_decay,
+    const float* __restrict__ grad_scale,
+    const float* __restrict__ found_inf) {
+  if (found_inf != nullptr && *found_inf != 0.f) return;
+
+  const float step_f = *step_ptr;
+  const float lr = *lr_ptr;
+  const float bias_correction1 = 1.f - powf(beta1, step_f);
+  const float rsqrt_bias_correction2 = rsqrtf(1.f - powf(beta2, step_f));
+
+  const int cid = blockIdx.x;
+  const int t = find_tensor(meta.chunk_prefix, meta.n_tensors, cid);
+  const int64_t chunk_in_tensor = cid - meta.chunk_prefix[t];
+  const int64_t numel = meta.numels[t];
+  const int64_t lo = chunk_in_tensor * kChunkSize;
+  const int64_t hi = min(lo + kChunkSize, numel);
+
+  ushort* __restrict__ p16 = reinterpret_cast<ushort*>(meta.addrs[0 * meta.n_tensors + t]);
+  const ushort* __restrict__ g16 = reinterpret_cast<const ushort*>(meta.addrs[1 * meta.n_tensors + t]);
+  float* __restrict__ m = reinterpret_cast<float*>(meta.addrs[2 * meta.n_tensors + t]);
+  float* __restrict__ v = reinterpret_cast<float*>(meta.addrs[3 * meta.n_tensors + t]);
+  float* __restrict__ w = reinterpret_cast<float*>(meta.addrs[4 * meta.n_tensors + t]);
+
+  const float step_size = lr / bias_correction1;
+  const float decay = 1.f - lr * weight_decay;
+  const float gscale = (grad_scale != nullptr) ? *grad_scale : 1.f;
+
+  const int64_t tid = threadIdx.x;
+  int64_t i = lo + tid * 4;
+  const int64_t vec_end = lo + ((hi - lo) & ~int64_t(3));
+  for (; i + 3 < vec_end; i += kBlockThreads * 4) {
+    ushort4 gp = *reinterpret_cast<const ushort4*>(g16 + i);
+    float4 wp = *reinterpret_cast<float4*>(w + i);
+    float4 mp = *reinterpret_cast<float4*>(m + i);
+    float4 vp = *reinterpret_cast<float4*>(v + i);
+    ushort4 po;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float gk = bf16_to_f32((&gp.x)[k]) * gscale;
+      float pk = (&wp.x)[k] * decay;
+      float mk = beta1 * (&mp.x)[k] + (1.f - beta1) * gk;
+      float vk = beta2 * (&vp.x)[k] + (1.f - beta2) * gk * gk;
+      float denom = sqrtf(vk) * rsqrt_bias_correction2 + eps;
+      pk -= step_size * mk / denom;
+      (&wp.x)[k] = pk; (&mp.x)[k] = mk; (&vp.x)[k] = vk;
+      (&po.x)[k] = f32_to_bf16(pk);
+    }
+    *reinterpret_cast<float4*>(w + i) = wp;
+    *reinterpret_cast<float4*>(m + i) = mp;
+    *reinterpret_cast<float4*>(v + i) = vp;
+    *reinterpret_cast<ushort4*>(p16 + i) = po;
+  }
+  for (int64_t j = vec_end + tid; j < hi; j += kBlockThreads) {
+    float gk = bf16_to_f32(g16[j]) * gscale;
+    float pk = w[j] * decay;
+    float mk = beta1 * m[j] + (1.f - beta1) * gk;
+    float vk = beta2 * v[j] + (1.f - beta2) * gk * gk;
+    float denom = sqrtf(vk) * rsqrt_bias_correction2 + eps;
+    pk -= step_size * mk / denom;
+    w[j] = pk; m[j] = mk; v[j] = vk;
+    p16[j] = f32_to_bf16(pk);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // global L2 norm²: per-wave shuffle reduce → per-block LDS reduce → one
 // device-scope atomicAdd per block (Guideline 12).
 // ---------------------------------------------------------------------------

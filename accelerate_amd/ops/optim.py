@@ -43,11 +43,13 @@ class FusedAdamW(torch.optim.Optimizer):
         # per-group cached execution plans: {group_idx: plan dict}
         self._plans = {}
 
-    def _build_plan(self, gi, params, grads, exp_avgs, exp_avg_sqs, device):
+    def _build_plan(self, plan_key, params, grads, exp_avgs, exp_avg_sqs, device, masters=None):
         import itertools
 
+        gi = plan_key[0] if isinstance(plan_key, tuple) else plan_key
         n = len(params)
-        ptrs = [t.data_ptr() for t in itertools.chain(params, grads, exp_avgs, exp_avg_sqs)]
+        lists = [params, grads, exp_avgs, exp_avg_sqs] + ([masters] if masters else [])
+        ptrs = [t.data_ptr() for t in itertools.chain(*lists)]
         numels = [p.numel() for p in params]
         prefix = [0]
         for numel in numels:
@@ -55,7 +57,7 @@ class FusedAdamW(torch.optim.Optimizer):
         addrs_numels = torch.tensor(ptrs + numels, dtype=torch.int64).to(device, non_blocking=True)
         chunk_prefix = torch.tensor(prefix, dtype=torch.int32).to(device, non_blocking=True)
         group = self.param_groups[gi]
-        old = self._plans.get(gi)
+        old = self._plans.get(plan_key)
         if old is not None:
             step_t, lr_t, last_lr = old["step"], old["lr"], old["last_lr"]
         else:
@@ -75,7 +77,7 @@ class FusedAdamW(torch.optim.Optimizer):
             "lr": lr_t,
             "last_lr": last_lr,
         }
-        self._plans[gi] = plan
+        self._plans[plan_key] = plan
         return plan
 
     @torch.no_grad()
@@ -87,33 +89,46 @@ class FusedAdamW(torch.optim.Optimizer):
 
         ext = None
         for gi, group in enumerate(self.param_groups):
-            params, grads, exp_avgs, exp_avg_sqs = [], [], [], []
+            # fp32 and bf16(+master) GPU params run as two fused plans
+            buckets = {"fp32": ([], [], [], [], None), "bf16": ([], [], [], [], [])}
             cpu_params = []
             for p in group["params"]:
                 if p.grad is None:
                     continue
                 state = self.state[p]
                 if len(state) == 0:
-                    state["exp_avg"] = torch.zeros_like(p, memory_format=torch.preserve_format)
-                    state["exp_avg_sq"] = torch.zeros_like(p, memory_format=torch.preserve_format)
+                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32, memory_format=torch.preserve_format)
+                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32, memory_format=torch.preserve_format)
+                    if p.is_cuda and p.dtype == torch.bfloat16:
+                        state["master"] = p.detach().to(torch.float32)
                 if p.is_cuda and p.dtype == torch.float32:
-                    params.append(p)
-                    grads.append(p.grad if p.grad.is_contiguous() else p.grad.contiguous())
-                    exp_avgs.append(state["exp_avg"])
-                    exp_avg_sqs.append(state["exp_avg_sq"])
+                    b = buckets["fp32"]
+                elif p.is_cuda and p.dtype == torch.bfloat16 and p.grad.dtype == torch.bfloat16:
+                    b = buckets["bf16"]
                 else:
                     state["step"] = state.get("step", 0) + 1
                     cpu_params.append((p, state))
+                    continue
+                b[0].append(p)
+                b[1].append(p.grad if p.grad.is_contiguous() else p.grad.contiguous())
+                b[2].append(state["exp_avg"])
+                b[3].append(state["exp_avg_sq"])
+                if b[4] is not None:
+                    b[4].append(state["master"])
 
-            if params:
+            for kind, (params, grads, exp_avgs, exp_avg_sqs, masters) in buckets.items():
+                if not params:
+                    continue
                 if ext is None:
                     ext = _load_extension(required=True)
                 import itertools
 
-                plan = self._plans.get(gi)
-                key = tuple(t.data_ptr() for t in itertools.chain(params, grads, exp_avgs, exp_avg_sqs))
+                plan_key = (gi, kind)
+                plan = self._plans.get(plan_key)
+                lists = [params, grads, exp_avgs, exp_avg_sqs] + ([masters] if masters else [])
+                key = tuple(t.data_ptr() for t in itertools.chain(*lists))
                 if plan is None or plan["key"] != key:
-                    plan = self._build_plan(gi, params, grads, exp_avgs, exp_avg_sqs, params[0].device)
+                    plan = self._build_plan(plan_key, params, grads, exp_avgs, exp_avg_sqs, params[0].device, masters)
                 lr = float(group["lr"])
                 if lr != plan["last_lr"]:
                     plan["lr"].fill_(lr)
@@ -132,6 +147,7 @@ class FusedAdamW(torch.optim.Optimizer):
                     group["weight_decay"],
                     self._grad_scale_tensor,
                     self._found_inf_tensor,
+                    kind == "bf16",
                 )
             for p, state in cpu_params:
                 self._single_tensor_step(p, state, group)
@@ -142,9 +158,10 @@ class FusedAdamW(torch.optim.Optimizer):
 
         Call after ``scheduler.step()`` when the optimizer kernel was captured
         in a hipGraph (the captured kernel reads lr from device memory)."""
-        for gi, group in enumerate(self.param_groups):
-            plan = self._plans.get(gi)
-            if plan is not None and float(group["lr"]) != plan["last_lr"]:
+        for plan_key, plan in self._plans.items():
+            gi = plan_key[0] if isinstance(plan_key, tuple) else plan_key
+            group = self.param_groups[gi]
+            if float(group["lr"]) != plan["last_lr"]:
                 plan["lr"].fill_(float(group["lr"]))
                 plan["last_lr"] = float(group["lr"])
 
@@ -154,12 +171,10 @@ class FusedAdamW(torch.optim.Optimizer):
 
     def _sync_steps_from_device(self):
         # under graph replay the host never sees step increments; read back
-        for gi, group in enumerate(self.param_groups):
-            plan = self._plans.get(gi)
-            if plan is None:
-                continue
+        for plan_key, plan in self._plans.items():
+            gi = plan_key[0] if isinstance(plan_key, tuple) else plan_key
             step = int(plan["step"].item())
-            for p in group["params"]:
+            for p in self.param_groups[gi]["params"]:
                 if p in self.state and p.is_cuda:
                     self.state[p]["step"] = step
 

@@ -37,7 +37,12 @@ def main():
     from accelerate_amd import Accelerator, set_seed
     from accelerate_amd.models import BertConfig, BertForSequenceClassification
 
-    accelerator = Accelerator(mixed_precision="bf16" if torch.cuda.is_available() else "no")
+    on_gpu_env = torch.cuda.is_available()
+    # bf16-native weights + fp32 master in the fused optimizer (default):
+    # no autocast cast kernels, bf16 gradient all-reduce over xGMI
+    bf16_weights = on_gpu_env and os.environ.get("BENCH_BF16_WEIGHTS", "1") == "1"
+    mixed = "no" if bf16_weights or not on_gpu_env else "bf16"
+    accelerator = Accelerator(mixed_precision=mixed)
     set_seed(42)
 
     n = accelerator.num_processes
@@ -46,6 +51,8 @@ def main():
 
     config = BertConfig.bert_base()
     model = BertForSequenceClassification(config)
+    if bf16_weights:
+        model = model.to(torch.bfloat16)
 
     if on_gpu:
         from accelerate_amd.ops.optim import FusedAdamW
@@ -183,6 +190,7 @@ def main():
                 "parallelism": f"dp{n}",
                 "optimizer": "fused_adamw_hip" if on_gpu else "torch_adamw",
                 "step_mode": step_mode,
+                "weights": "bf16+fp32_master" if bf16_weights else ("fp32+autocast_bf16" if on_gpu else "fp32"),
             },
         }
         print(json.dumps(result))

@@ -51,6 +51,37 @@ def test_fused_adamw_matches_torch(steps):
 
 
 @gpu
+@pytest.mark.parametrize("steps", [5])
+def test_fused_adamw_bf16_master(steps):
+    """bf16 params + fp32 master must track a pure-fp32 AdamW reference."""
+    from accelerate_amd.ops.optim import FusedAdamW
+
+    device = "cuda"
+    shapes = [(768,), (512, 768), (127,)]
+    ref32 = make_tensors(shapes, device, seed=3)
+    params16 = [p.clone().to(torch.bfloat16).requires_grad_(True) for p in ref32]
+    for p in ref32:
+        p.requires_grad_(True)
+    opt_ref = torch.optim.AdamW(ref32, lr=1e-2, weight_decay=0.01)
+    opt16 = FusedAdamW(params16, lr=1e-2, weight_decay=0.01)
+    for step in range(steps):
+        grads = make_tensors(shapes, device, seed=30 + step)
+        for p, g in zip(ref32, grads):
+            p.grad = g.clone()
+        for p, g in zip(params16, grads):
+            p.grad = g.clone().to(torch.bfloat16)
+        opt_ref.step()
+        opt16.step()
+    torch.cuda.synchronize()
+    for i, (a, b) in enumerate(zip(ref32, params16)):
+        # master tracks fp32 trajectory; bf16 mirror is its rounding
+        master = opt16.state[b]["master"]
+        diff = (a - master).abs().max().item()
+        assert diff < 2e-3, f"master drift {diff} on {i}"  # grads were bf16-rounded
+        assert torch.equal(b.detach(), master.to(torch.bfloat16)), "bf16 mirror != rounded master"
+
+
+@gpu
 def test_clip_grad_norm_matches_torch():
     from accelerate_amd.ops.clip_grad import clip_grad_norm_
 
