@@ -54,8 +54,14 @@ class FusedAdamW(torch.optim.Optimizer):
         prefix = [0]
         for numel in numels:
             prefix.append(prefix[-1] + (numel + 16383) // 16384)
-        addrs_numels = torch.tensor(ptrs + numels, dtype=torch.int64).to(device, non_blocking=True)
-        chunk_prefix = torch.tensor(prefix, dtype=torch.int32).to(device, non_blocking=True)
+        # pinned staging + async H2D: capture-safe (a pageable copy would
+        # implicitly sync and abort hipGraph capture) — sources kept alive in
+        # the plan so replays of a captured copy read valid memory
+        pin = device.type == "cuda" if hasattr(device, "type") else True
+        cpu_an = torch.tensor(ptrs + numels, dtype=torch.int64, pin_memory=pin)
+        cpu_cp = torch.tensor(prefix, dtype=torch.int32, pin_memory=pin)
+        addrs_numels = cpu_an.to(device, non_blocking=True)
+        chunk_prefix = cpu_cp.to(device, non_blocking=True)
         group = self.param_groups[gi]
         old = self._plans.get(plan_key)
         if old is not None:
@@ -76,6 +82,7 @@ class FusedAdamW(torch.optim.Optimizer):
             "step": step_t,
             "lr": lr_t,
             "last_lr": last_lr,
+            "_pinned": (cpu_an, cpu_cp),
         }
         self._plans[plan_key] = plan
         return plan

@@ -432,7 +432,8 @@ def reduce(tensor, reduction="mean", scale=1.0):
         torch.distributed.all_reduce(cloned_tensor, op)
         if reduction == "mean":
             cloned_tensor /= state.num_processes
-        cloned_tensor *= scale
+        if scale != 1.0:
+            cloned_tensor *= scale
         return cloned_tensor
 
     return recursively_apply(

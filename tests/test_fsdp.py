@@ -67,3 +67,9 @@ def test_fsdp_distributed_oracle():
         "FSDP_NOSYNC_PASS",
     ):
         assert marker in out, f"missing {marker}\n{out}"
+
+
+def test_pipeline_inference_oracle():
+    script = Path(__file__).parent / "distributed_scripts" / "pipeline_script.py"
+    out = launch_distributed(script, nproc=2, timeout=180)
+    assert "PIPELINE_PASS" in out
