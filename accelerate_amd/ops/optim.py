@@ -55,11 +55,23 @@ class FusedAdamW(torch.optim.Optimizer):
         for numel in numels:
             prefix.append(prefix[-1] + (numel + 16383) // 16384)
         # pinned staging + async H2D: capture-safe (a pageable copy would
-        # implicitly sync and abort hipGraph capture) — sources kept alive in
-        # the plan so replays of a captured copy read valid memory
-        pin = device.type == "cuda" if hasattr(device, "type") else True
-        cpu_an = torch.tensor(ptrs + numels, dtype=torch.int64, pin_memory=pin)
-        cpu_cp = torch.tensor(prefix, dtype=torch.int32, pin_memory=pin)
+        # implicitly sync and a hipHostMalloc would invalidate hipGraph
+        # capture) — the pinned buffers are allocated ONCE per plan shape at
+        # the first (out-of-capture) build and reused in-place on rebuilds,
+        # so a rebuild triggered INSIDE capture only writes + async-copies.
+        pin = device.type == "cuda" if hasattr(device, "type") else False
+        cache = getattr(self, "_pinned_cache", None)
+        if cache is None:
+            cache = self._pinned_cache = {}
+        cached = cache.get(plan_key)
+        if cached is None or cached[0].numel() != len(ptrs) + len(numels):
+            cpu_an = torch.empty(len(ptrs) + len(numels), dtype=torch.int64, pin_memory=pin)
+            cpu_cp = torch.empty(len(prefix), dtype=torch.int32, pin_memory=pin)
+            cache[plan_key] = (cpu_an, cpu_cp)
+        else:
+            cpu_an, cpu_cp = cached
+        cpu_an.copy_(torch.tensor(ptrs + numels, dtype=torch.int64))
+        cpu_cp.copy_(torch.tensor(prefix, dtype=torch.int32))
         addrs_numels = cpu_an.to(device, non_blocking=True)
         chunk_prefix = cpu_cp.to(device, non_blocking=True)
         group = self.param_groups[gi]

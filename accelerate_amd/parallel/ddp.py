@@ -107,6 +107,7 @@ class DistributedDataParallelEngine(nn.Module):
         self._hooks = []
         self._buckets: List[_Bucket] = []
         self._param_to_bucket = {}
+        self._float_buffers = None
         self._build_buckets()
         self._register_hooks()
         if self._world_size > 1:
@@ -206,9 +207,12 @@ class DistributedDataParallelEngine(nn.Module):
 
     def forward(self, *args, **kwargs):
         if self.broadcast_buffers and self._world_size > 1 and self.module.training:
-            for buf in self.module.buffers():
-                if buf.numel() > 0:
-                    dist.broadcast(buf.data, src=0, group=self.process_group)
+            # per-iteration sync only for mutable float buffers (BN running
+            # stats); constant/int buffers were broadcast once at wrap time
+            if self._float_buffers is None:
+                self._float_buffers = [b for b in self.module.buffers() if b.is_floating_point() and b.numel() > 0]
+            for buf in self._float_buffers:
+                dist.broadcast(buf.data, src=0, group=self.process_group)
         return self.module(*args, **kwargs)
 
     @contextmanager
