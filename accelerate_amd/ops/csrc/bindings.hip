@@ -165,6 +165,86 @@ void unscale_and_check(std::vector<at::Tensor> grads, at::Tensor inv_scale, at::
                      h.meta, inv_scale.data_ptr<float>(), found_inf.data_ptr<float>());
 }
 
+// norm kernels (norm_kernels.hip)
+__global__ void layernorm_fwd_bf16(const ushort*, const ushort*, const ushort*, ushort*,
+                                   float*, float*, int64_t, int, float);
+__global__ void layernorm_bwd_bf16(const ushort*, const ushort*, const ushort*,
+                                   const float*, const float*, ushort*, float*, float*, int64_t, int);
+__global__ void norm_fold_partials(const float*, const float*, ushort*, ushort*, int, int);
+__global__ void rmsnorm_fwd_bf16(const ushort*, const ushort*, ushort*, float*, int64_t, int, float);
+__global__ void rmsnorm_bwd_bf16(const ushort*, const ushort*, const ushort*, const float*,
+                                 ushort*, float*, int64_t, int);
+
+static const ushort* bfp(const at::Tensor& t) { return reinterpret_cast<const ushort*>(t.data_ptr()); }
+static ushort* bfp_mut(at::Tensor& t) { return reinterpret_cast<ushort*>(t.data_ptr()); }
+
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == at::kBFloat16, "layernorm_fwd: bf16 contiguous");
+  const int d = (int)x.size(-1);
+  TORCH_CHECK(d % 8 == 0, "layernorm_fwd: inner dim must be a multiple of 8");
+  const int64_t rows = x.numel() / d;
+  auto y = at::empty_like(x);
+  auto mean = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(layernorm_fwd_bf16, dim3(rows), dim3(64), 0, stream.stream(),
+                     bfp(x), bfp(w), bfp(b), bfp_mut(y), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     rows, d, (float)eps);
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                      at::Tensor mean, at::Tensor rstd) {
+  const int d = (int)x.size(-1);
+  TORCH_CHECK(d <= 2048, "layernorm_bwd: fused path supports inner dim <= 2048");
+  const int64_t rows = x.numel() / d;
+  const int n_blocks = (int)std::min<int64_t>(rows, 512);
+  auto dx = at::empty_like(x);
+  auto dw_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
+  auto db_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({d}, x.options());
+  auto db = at::empty({d}, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  auto dyc = dy.contiguous();
+  hipLaunchKernelGGL(layernorm_bwd_bf16, dim3(n_blocks), dim3(64), 0, stream.stream(),
+                     bfp(dyc), bfp(x), bfp(w), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     bfp_mut(dx), dw_partial.data_ptr<float>(), db_partial.data_ptr<float>(), rows, d);
+  hipLaunchKernelGGL(norm_fold_partials, dim3((d + 255) / 256), dim3(256), 0, stream.stream(),
+                     dw_partial.data_ptr<float>(), db_partial.data_ptr<float>(), bfp_mut(dw), bfp_mut(db), d, n_blocks);
+  return {dx, dw, db};
+}
+
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == at::kBFloat16, "rmsnorm_fwd: bf16 contiguous");
+  const int d = (int)x.size(-1);
+  TORCH_CHECK(d % 8 == 0, "rmsnorm_fwd: inner dim must be a multiple of 8");
+  const int64_t rows = x.numel() / d;
+  auto y = at::empty_like(x);
+  auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(rmsnorm_fwd_bf16, dim3(rows), dim3(64), 0, stream.stream(),
+                     bfp(x), bfp(w), bfp_mut(y), rstd.data_ptr<float>(), rows, d, (float)eps);
+  return {y, rstd};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, at::Tensor rstd) {
+  const int d = (int)x.size(-1);
+  TORCH_CHECK(d <= 8192, "rmsnorm_bwd: fused path supports inner dim <= 8192");
+  const int64_t rows = x.numel() / d;
+  const int n_blocks = (int)std::min<int64_t>(rows, 512);
+  auto dx = at::empty_like(x);
+  auto dw_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({d}, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  auto dyc = dy.contiguous();
+  hipLaunchKernelGGL(rmsnorm_bwd_bf16, dim3(n_blocks), dim3(64), 0, stream.stream(),
+                     bfp(dyc), bfp(x), bfp(w), rstd.data_ptr<float>(), bfp_mut(dx),
+                     dw_partial.data_ptr<float>(), rows, d);
+  hipLaunchKernelGGL(norm_fold_partials, dim3((d + 255) / 256), dim3(256), 0, stream.stream(),
+                     dw_partial.data_ptr<float>(), nullptr, bfp_mut(dw), nullptr, d, n_blocks);
+  return {dx, dw};
+}
+
 // fp8 kernels (fp8_kernels.hip)
 __global__ void fp8_cast_amax_e4m3(const ushort*, unsigned char*, const float*, float*, int64_t);
 __global__ void fp8_cast_amax_e5m2(const ushort*, unsigned char*, const float*, float*, int64_t);
@@ -198,6 +278,10 @@ void fp8_update_scale_fn(at::Tensor history, double fp8_max, double margin_pow2,
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd, "fused bf16 LayerNorm forward");
+  m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused bf16 RMSNorm forward");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused bf16 RMSNorm backward");
   m.def("fp8_cast_amax", &fp8_cast_amax, "bf16 -> fp8 cast with fused amax (gfx950)");
   m.def("fp8_update_scale", &fp8_update_scale_fn, "delayed-scaling scale update");
   m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (gfx950)");

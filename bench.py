@@ -38,6 +38,18 @@ def main():
     from accelerate_amd.models import BertConfig, BertForSequenceClassification
 
     on_gpu_env = torch.cuda.is_available()
+    # hipBLASLt algorithm selection pre-tuned on MI355X (TunableOp CSV,
+    # produced by benchmarks/tune_gemms.py) — read-only at runtime
+    tun_csv = os.path.join(os.path.dirname(os.path.abspath(__file__)), "profiles", "tunableop_mi355x.csv")
+    if on_gpu_env and os.path.exists(tun_csv) and os.environ.get("BENCH_TUNABLEOP", "1") == "1":
+        try:
+            import torch.cuda.tunable as tunable
+
+            tunable.enable(True)
+            tunable.tuning_enable(False)
+            tunable.read_file(tun_csv)
+        except Exception as e:
+            print(f"[bench] TunableOp unavailable: {e}", flush=True)
     # bf16-native weights + fp32 master in the fused optimizer (default):
     # no autocast cast kernels, bf16 gradient all-reduce over xGMI
     bf16_weights = on_gpu_env and os.environ.get("BENCH_BF16_WEIGHTS", "1") == "1"
