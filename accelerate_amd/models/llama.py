@@ -72,11 +72,12 @@ def build_rope_cache(seq_len, head_dim, theta, device, dtype=torch.float32):
 
 
 def apply_rope(x, cos, sin):
-    # x: [B, H, S, D]
+    # x: [B, H, S, D]; rotate in the compute dtype (fp32 tables would
+    # silently promote q/k and push the attention GEMMs off bf16 MFMA)
     d = x.shape[-1] // 2
     x1, x2 = x[..., :d], x[..., d:]
-    cos = cos[None, None, : x.shape[2], :]
-    sin = sin[None, None, : x.shape[2], :]
+    cos = cos[None, None, : x.shape[2], :].to(x.dtype)
+    sin = sin[None, None, : x.shape[2], :].to(x.dtype)
     return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], dim=-1)
 
 

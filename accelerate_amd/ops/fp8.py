@@ -26,17 +26,30 @@ E5M2_MAX = 57344.0
 
 
 class _ScalingState:
-    """Per-tensor delayed-scaling state (scale, inverse, amax history)."""
+    """Per-tensor delayed-scaling state (scale, inverse, amax history).
 
-    def __init__(self, device, hist_len=16, fp8_max=E4M3_MAX, margin=0):
+    The scale refresh is amortized over ``update_interval`` steps (TE's
+    amax-update-interval pattern): between refreshes the cast kernel keeps
+    atomicMax-ing into the current history slot, so the recorded amax is the
+    interval max — conservative, and the hot loop stays at exactly ONE
+    kernel per cast."""
+
+    def __init__(self, device, hist_len=16, fp8_max=E4M3_MAX, margin=0, update_interval=16):
         self.scale = torch.ones(1, device=device)
         self.scale_inv = torch.ones(1, device=device)
         self.amax_history = torch.zeros(hist_len, device=device)
         self.fp8_max = fp8_max
         self.margin_pow2 = float(2**margin)
         self._slot = 0
+        self._calls = 0
+        self.update_interval = update_interval
 
     def roll_and_update(self, ext):
+        self._calls += 1
+        # immediate first refresh (grads may be far from scale=1), then
+        # amortized every `update_interval` calls
+        if self._calls != 1 and self._calls % self.update_interval:
+            return
         # amax_history[slot] was filled by the cast kernel; refresh scale
         ext.fp8_update_scale(self.amax_history, self.fp8_max, self.margin_pow2, self.scale, self.scale_inv)
         self._slot = (self._slot + 1) % self.amax_history.numel()
@@ -60,11 +73,14 @@ class _FP8LinearFn(torch.autograd.Function):
         x2d = x.reshape(-1, in_shape[-1])
         x8 = _cast_fp8(x2d, sx, e5m2=False)
         w8 = _cast_fp8(weight, sw, e5m2=False)
+        # pre-cast W^T for the dgrad GEMM now (bf16 transpose + fused cast),
+        # so backward runs with ZERO layout copies
+        w8t = _cast_fp8(weight.t().contiguous(), sw, e5m2=False)
         # y = (x8 @ w8^T) * (1/sx) * (1/sw)  — hipBLASLt fp8 MFMA GEMM
         y = torch._scaled_mm(
             x8, w8.t(), scale_a=sx.scale_inv, scale_b=sw.scale_inv, bias=bias, out_dtype=torch.bfloat16
         )
-        ctx.save_for_backward(w8, x2d)
+        ctx.save_for_backward(w8t, x2d)
         ctx.scales = (sx, sw, sg)
         ctx.ext = ext
         ctx.has_bias = bias is not None
@@ -74,17 +90,17 @@ class _FP8LinearFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out):
-        w8, x2d = ctx.saved_tensors
+        w8t, x2d = ctx.saved_tensors
         sx, sw, sg = ctx.scales
         ext = ctx.ext
         g2d = grad_out.reshape(-1, grad_out.shape[-1])
-        g8 = _cast_fp8(g2d, sg, e5m2=True)
-        # dgrad: gx = g8 @ w8 (fp8); w8 is [N,K] row-major -> need column-major B
+        g8 = _cast_fp8(g2d.contiguous(), sg, e5m2=True)
+        # dgrad: gx[M,K] = g8[M,N] @ W[N,K]; B column-major = w8t.t()
         gx = torch._scaled_mm(
-            g8, w8.t().contiguous().t(), scale_a=sg.scale_inv, scale_b=sw.scale_inv, out_dtype=torch.bfloat16
+            g8, w8t.t(), scale_a=sg.scale_inv, scale_b=sw.scale_inv, out_dtype=torch.bfloat16
         )
-        # wgrad in bf16 (precision-critical; avoids fp8 transpose round-trip)
-        gw = g2d.t().to(torch.bfloat16) @ x2d.to(torch.bfloat16)
+        # wgrad in bf16 (precision-critical; avoids an fp8 transpose of g)
+        gw = g2d.t() @ x2d
         gb = g2d.sum(0) if ctx.has_bias else None
         sg.roll_and_update(ext)
         return gx.view(grad_out.shape[:-1] + (gx.shape[-1],)), gw, gb, None, None, None, None

@@ -318,7 +318,13 @@ class AcceleratorState:
 
         self.use_fsdp = fsdp_plugin is not None or parse_flag_from_env("ACCELERATE_USE_FSDP")
         self.fsdp_plugin = fsdp_plugin
-        if self.use_fsdp and self.distributed_type in (DistributedType.MULTI_GPU, DistributedType.MULTI_CPU):
+        # sharded engine works at any world size (world 1 = no collectives,
+        # still bf16-compute/fp32-master semantics + sharded checkpoints)
+        if self.use_fsdp and self.distributed_type in (
+            DistributedType.MULTI_GPU,
+            DistributedType.MULTI_CPU,
+            DistributedType.NO,
+        ):
             self.distributed_type = DistributedType.FSDP
             if self.fsdp_plugin is None:
                 from .utils.dataclasses import FullyShardedDataParallelPlugin
