@@ -32,13 +32,32 @@ def build_stack(hidden, ffn, n_blocks, fp8):
 
 
 def run(model, tokens, hidden, iters=30, warmup=10):
+    """hipGraph-captured fwd+bwd so both variants measure GPU time, not
+    python autograd overhead (training runs under graphs too)."""
+    import gc
+
     x = torch.randn(tokens, hidden, device="cuda", dtype=torch.bfloat16, requires_grad=True)
-    for _ in range(warmup):
+
+    def step():
         model(x).sum().backward()
+
+    for _ in range(warmup):
+        step()
+    gc.collect()
+    torch.cuda.synchronize()
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for _ in range(3):
+            step()
+    torch.cuda.current_stream().wait_stream(side)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        step()
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(iters):
-        model(x).sum().backward()
+        g.replay()
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / iters
     return tokens / dt
