@@ -249,14 +249,18 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, a
 __global__ void fp8_cast_amax_e4m3(const ushort*, unsigned char*, const float*, float*, int64_t);
 __global__ void fp8_cast_amax_e5m2(const ushort*, unsigned char*, const float*, float*, int64_t);
 __global__ void fp8_update_scale(const float*, int, float, float, float*, float*);
+__global__ void fp8_cast_transpose_e4m3(const ushort*, unsigned char*, unsigned char*,
+                                        const float*, float*, int64_t, int64_t);
+__global__ void fp8_cast_transpose_e5m2(const ushort*, unsigned char*, unsigned char*,
+                                        const float*, float*, int64_t, int64_t);
 
 void fp8_cast_amax(at::Tensor in, at::Tensor out, at::Tensor scale, at::Tensor amax, bool e5m2) {
   TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.scalar_type() == at::kBFloat16,
               "fp8_cast_amax: input must be contiguous bf16 on GPU");
   TORCH_CHECK(out.numel() == in.numel() && out.element_size() == 1, "fp8_cast_amax: bad output");
   int64_t n = in.numel();
-  // memory-bound: cap grid and grid-stride (CDNA4 Guideline 11)
-  int grid = (int)std::min<int64_t>((n + kBlockThreads * 8 - 1) / (kBlockThreads * 8), 2048);
+  // memory-bound: cap grid and grid-stride (CDNA4 Guideline 11); 16 elems/thread
+  int grid = (int)std::min<int64_t>((n + kBlockThreads * 16 - 1) / (kBlockThreads * 16), 2048);
   auto stream = at::hip::getCurrentHIPStream();
   auto* inp = reinterpret_cast<const ushort*>(in.data_ptr());
   auto* outp = reinterpret_cast<unsigned char*>(out.data_ptr());
@@ -266,6 +270,27 @@ void fp8_cast_amax(at::Tensor in, at::Tensor out, at::Tensor scale, at::Tensor a
   } else {
     hipLaunchKernelGGL(fp8_cast_amax_e4m3, dim3(grid), dim3(kBlockThreads), 0, stream.stream(),
                        inp, outp, scale.data_ptr<float>(), amax.data_ptr<float>(), n);
+  }
+}
+
+void fp8_cast_transpose(at::Tensor in, at::Tensor out, at::Tensor out_t,
+                        at::Tensor scale, at::Tensor amax, bool e5m2) {
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.scalar_type() == at::kBFloat16 && in.dim() == 2,
+              "fp8_cast_transpose: 2-D contiguous bf16 input required");
+  const int64_t R = in.size(0), C = in.size(1);
+  TORCH_CHECK(R % 128 == 0 && C % 128 == 0, "fp8_cast_transpose: dims must be multiples of 128");
+  TORCH_CHECK(out.numel() == R * C && out_t.numel() == R * C, "fp8_cast_transpose: bad outputs");
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid(C / 128, R / 128);
+  auto* inp = reinterpret_cast<const ushort*>(in.data_ptr());
+  auto* o = reinterpret_cast<unsigned char*>(out.data_ptr());
+  auto* ot = reinterpret_cast<unsigned char*>(out_t.data_ptr());
+  if (e5m2) {
+    hipLaunchKernelGGL(fp8_cast_transpose_e5m2, grid, dim3(256), 0, stream.stream(),
+                       inp, o, ot, scale.data_ptr<float>(), amax.data_ptr<float>(), R, C);
+  } else {
+    hipLaunchKernelGGL(fp8_cast_transpose_e4m3, grid, dim3(256), 0, stream.stream(),
+                       inp, o, ot, scale.data_ptr<float>(), amax.data_ptr<float>(), R, C);
   }
 }
 
@@ -283,6 +308,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused bf16 RMSNorm forward");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused bf16 RMSNorm backward");
   m.def("fp8_cast_amax", &fp8_cast_amax, "bf16 -> fp8 cast with fused amax (gfx950)");
+  m.def("fp8_cast_transpose", &fp8_cast_transpose,
+        "bf16 -> fp8 + fp8^T with fused amax (LDS-tiled, gfx950)");
   m.def("fp8_update_scale", &fp8_update_scale_fn, "delayed-scaling scale update");
   m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (gfx950)");
   m.def("fused_adamw_planned", &fused_adamw_planned,

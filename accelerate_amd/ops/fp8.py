@@ -46,9 +46,9 @@ class _ScalingState:
 
     def roll_and_update(self, ext):
         self._calls += 1
-        # immediate first refresh (grads may be far from scale=1), then
-        # amortized every `update_interval` calls
-        if self._calls != 1 and self._calls % self.update_interval:
+        # call 2 = first refresh with real history (grads may be far from
+        # scale=1), then amortized every `update_interval` calls
+        if self._calls != 2 and self._calls % self.update_interval:
             return
         # amax_history[slot] was filled by the cast kernel; refresh scale
         ext.fp8_update_scale(self.amax_history, self.fp8_max, self.margin_pow2, self.scale, self.scale_inv)
@@ -66,43 +66,62 @@ def _cast_fp8(x: torch.Tensor, state: _ScalingState, e5m2: bool):
     return out
 
 
+def _cast_transpose_fp8(x2d: torch.Tensor, state: _ScalingState, e5m2: bool):
+    """One HBM pass -> (fp8 [R,C], fp8 [C,R]) via the LDS-tiled kernel;
+    falls back to two casts when dims aren't 64-aligned."""
+    ext = _load_extension(required=True)
+    dtype = torch.float8_e5m2 if e5m2 else torch.float8_e4m3fn
+    R, C = x2d.shape
+    if R % 128 == 0 and C % 128 == 0:
+        out = torch.empty(R, C, dtype=dtype, device=x2d.device)
+        out_t = torch.empty(C, R, dtype=dtype, device=x2d.device)
+        ext.fp8_cast_transpose(x2d.contiguous(), out, out_t, state.scale, state.amax_slot(), e5m2)
+        return out, out_t
+    out = _cast_fp8(x2d, state, e5m2)
+    out_t = _cast_fp8(x2d.t().contiguous(), state, e5m2)
+    return out, out_t
+
+
 class _FP8LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, sx, sw, sg, ext):
         in_shape = x.shape
         x2d = x.reshape(-1, in_shape[-1])
-        x8 = _cast_fp8(x2d, sx, e5m2=False)
-        w8 = _cast_fp8(weight, sw, e5m2=False)
-        # pre-cast W^T for the dgrad GEMM now (bf16 transpose + fused cast),
-        # so backward runs with ZERO layout copies
-        w8t = _cast_fp8(weight.t().contiguous(), sw, e5m2=False)
+        # refresh scales BEFORE casting: the device scale scalar must stay
+        # untouched between a cast and its GEMM consumers (incl. backward)
+        sx.roll_and_update(ext)
+        sw.roll_and_update(ext)
+        # each operand cast ONCE producing both layouts: x8 (fwd) + x8t (wgrad),
+        # w8 (fwd) + w8t (dgrad)
+        x8, x8t = _cast_transpose_fp8(x2d, sx, e5m2=False)
+        w8, w8t = _cast_transpose_fp8(weight, sw, e5m2=False)
         # y = (x8 @ w8^T) * (1/sx) * (1/sw)  — hipBLASLt fp8 MFMA GEMM
         y = torch._scaled_mm(
             x8, w8.t(), scale_a=sx.scale_inv, scale_b=sw.scale_inv, bias=bias, out_dtype=torch.bfloat16
         )
-        ctx.save_for_backward(w8t, x2d)
+        ctx.save_for_backward(w8t, x8t)
         ctx.scales = (sx, sw, sg)
         ctx.ext = ext
         ctx.has_bias = bias is not None
-        sx.roll_and_update(ext)
-        sw.roll_and_update(ext)
         return y.view(*in_shape[:-1], -1)
 
     @staticmethod
     def backward(ctx, grad_out):
-        w8t, x2d = ctx.saved_tensors
+        w8t, x8t = ctx.saved_tensors
         sx, sw, sg = ctx.scales
         ext = ctx.ext
         g2d = grad_out.reshape(-1, grad_out.shape[-1])
-        g8 = _cast_fp8(g2d.contiguous(), sg, e5m2=True)
+        sg.roll_and_update(ext)
+        g8, g8t = _cast_transpose_fp8(g2d.contiguous(), sg, e5m2=True)
         # dgrad: gx[M,K] = g8[M,N] @ W[N,K]; B column-major = w8t.t()
         gx = torch._scaled_mm(
             g8, w8t.t(), scale_a=sg.scale_inv, scale_b=sw.scale_inv, out_dtype=torch.bfloat16
         )
-        # wgrad in bf16 (precision-critical; avoids an fp8 transpose of g)
-        gw = g2d.t() @ x2d
+        # wgrad (fp8): gw[N,K] = g^T[N,M] @ X[M,K]; B column-major = x8t.t()
+        gw = torch._scaled_mm(
+            g8t, x8t.t(), scale_a=sg.scale_inv, scale_b=sx.scale_inv, out_dtype=torch.bfloat16
+        )
         gb = g2d.sum(0) if ctx.has_bias else None
-        sg.roll_and_update(ext)
         return gx.view(grad_out.shape[:-1] + (gx.shape[-1],)), gw, gb, None, None, None, None
 
 

@@ -114,3 +114,21 @@ def test_fp8_training_tracks_bf16():
     fp8_losses = run(True)
     assert fp8_losses[-1] < bf16_losses[-1] * 1.05, f"fp8 diverges from bf16: {fp8_losses[-1]} vs {bf16_losses[-1]}"
     assert fp8_losses[-1] < fp8_losses[0], "fp8 loss must decrease"
+
+
+@gpu
+def test_fp8_cast_transpose_kernel():
+    from accelerate_amd.ops import _load_extension
+
+    ext = _load_extension(required=True)
+    x = torch.randn(256, 384, device="cuda", dtype=torch.bfloat16) * 2
+    scale = torch.ones(1, device="cuda")
+    amax = torch.zeros(1, device="cuda")
+    out = torch.empty(256, 384, dtype=torch.float8_e4m3fn, device="cuda")
+    out_t = torch.empty(384, 256, dtype=torch.float8_e4m3fn, device="cuda")
+    ext.fp8_cast_transpose(x, out, out_t, scale, amax, False)
+    torch.cuda.synchronize()
+    ref = x.to(torch.float8_e4m3fn)
+    assert torch.equal(out.view(torch.uint8), ref.view(torch.uint8))
+    assert torch.equal(out_t.view(torch.uint8), ref.t().contiguous().view(torch.uint8))
+    assert torch.allclose(amax, x.float().abs().max().reshape(1), rtol=1e-3)
