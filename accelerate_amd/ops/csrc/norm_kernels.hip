@@ -13,6 +13,8 @@
 
 namespace {
 
+typedef ushort ushort8 __attribute__((ext_vector_type(8)));  // 16 B = 8 bf16
+
 __device__ __forceinline__ float bf2f(ushort u) {
   union { unsigned int i; float f; } cv;
   cv.i = ((unsigned int)u) << 16;
@@ -52,11 +54,10 @@ __global__ void layernorm_fwd_bf16(const ushort* __restrict__ x, const ushort* _
 
   float s = 0.f, sq = 0.f;
   for (int base = lane * 8; base < d; base += 64 * 8) {
-    ushort4 a = *reinterpret_cast<const ushort4*>(xr + base);
-    ushort4 c = *reinterpret_cast<const ushort4*>(xr + base + 4);
+    ushort8 av = *reinterpret_cast<const ushort8*>(xr + base);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      float v = bf2f(k < 4 ? (&a.x)[k] : (&c.x)[k - 4]);
+      float v = bf2f(av[k]);
       s += v;
       sq += v * v;
     }
@@ -71,23 +72,19 @@ __global__ void layernorm_fwd_bf16(const ushort* __restrict__ x, const ushort* _
     rstd_out[row] = rstd;
   }
   for (int base = lane * 8; base < d; base += 64 * 8) {
-    ushort4 a = *reinterpret_cast<const ushort4*>(xr + base);
-    ushort4 c = *reinterpret_cast<const ushort4*>(xr + base + 4);
-    ushort4 wa = *reinterpret_cast<const ushort4*>(w + base);
-    ushort4 wc = *reinterpret_cast<const ushort4*>(w + base + 4);
-    ushort4 ba = *reinterpret_cast<const ushort4*>(b + base);
-    ushort4 bc = *reinterpret_cast<const ushort4*>(b + base + 4);
-    ushort4 o0, o1;
+    ushort8 av = *reinterpret_cast<const ushort8*>(xr + base);
+    ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
+    ushort8 bv = *reinterpret_cast<const ushort8*>(b + base);
+    ushort8 ov;
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      float v = bf2f(k < 4 ? (&a.x)[k] : (&c.x)[k - 4]);
-      float wk = bf2f(k < 4 ? (&wa.x)[k] : (&wc.x)[k - 4]);
-      float bk = bf2f(k < 4 ? (&ba.x)[k] : (&bc.x)[k - 4]);
+      float v = bf2f(av[k]);
+      float wk = bf2f(wv[k]);
+      float bk = bf2f(bv[k]);
       ushort r = f2bf((v - mean) * rstd * wk + bk);
-      if (k < 4) (&o0.x)[k] = r; else (&o1.x)[k - 4] = r;
+      ov[k] = r;
     }
-    *reinterpret_cast<ushort4*>(yr + base) = o0;
-    *reinterpret_cast<ushort4*>(yr + base + 4) = o1;
+    *reinterpret_cast<ushort8*>(yr + base) = ov;
   }
 }
 
@@ -126,17 +123,14 @@ __global__ void layernorm_bwd_bf16(const ushort* __restrict__ dy, const ushort* 
 
     float s1 = 0.f, s2 = 0.f;
     for (int base = lane * 8; base < d; base += 64 * 8) {
-      ushort4 g0 = *reinterpret_cast<const ushort4*>(dyr + base);
-      ushort4 g1 = *reinterpret_cast<const ushort4*>(dyr + base + 4);
-      ushort4 x0 = *reinterpret_cast<const ushort4*>(xr + base);
-      ushort4 x1 = *reinterpret_cast<const ushort4*>(xr + base + 4);
-      ushort4 w0 = *reinterpret_cast<const ushort4*>(w + base);
-      ushort4 w1 = *reinterpret_cast<const ushort4*>(w + base + 4);
+      ushort8 gv = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 xv8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = bf2f(k < 4 ? (&g0.x)[k] : (&g1.x)[k - 4]);
-        float xv = bf2f(k < 4 ? (&x0.x)[k] : (&x1.x)[k - 4]);
-        float wk = bf2f(k < 4 ? (&w0.x)[k] : (&w1.x)[k - 4]);
+        float g = bf2f(gv[k]);
+        float xv = bf2f(xv8[k]);
+        float wk = bf2f(wv[k]);
         float xhat = (xv - mu) * rs;
         float gw = g * wk;
         s1 += gw;
@@ -148,27 +142,23 @@ __global__ void layernorm_bwd_bf16(const ushort* __restrict__ dy, const ushort* 
 
     int it = 0;
     for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
-      ushort4 g0 = *reinterpret_cast<const ushort4*>(dyr + base);
-      ushort4 g1 = *reinterpret_cast<const ushort4*>(dyr + base + 4);
-      ushort4 x0 = *reinterpret_cast<const ushort4*>(xr + base);
-      ushort4 x1 = *reinterpret_cast<const ushort4*>(xr + base + 4);
-      ushort4 w0 = *reinterpret_cast<const ushort4*>(w + base);
-      ushort4 w1 = *reinterpret_cast<const ushort4*>(w + base + 4);
-      ushort4 o0, o1;
+      ushort8 gv = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 xv8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
+      ushort8 ov;
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = bf2f(k < 4 ? (&g0.x)[k] : (&g1.x)[k - 4]);
-        float xv = bf2f(k < 4 ? (&x0.x)[k] : (&x1.x)[k - 4]);
-        float wk = bf2f(k < 4 ? (&w0.x)[k] : (&w1.x)[k - 4]);
+        float g = bf2f(gv[k]);
+        float xv = bf2f(xv8[k]);
+        float wk = bf2f(wv[k]);
         float xhat = (xv - mu) * rs;
         float dxv = rs * (g * wk - s1 - xhat * s2);
         ushort r = f2bf(dxv);
-        if (k < 4) (&o0.x)[k] = r; else (&o1.x)[k - 4] = r;
+        ov[k] = r;
         accw[it][k] += g * xhat;
         accb[it][k] += g;
       }
-      *reinterpret_cast<ushort4*>(dxr + base) = o0;
-      *reinterpret_cast<ushort4*>(dxr + base + 4) = o1;
+      *reinterpret_cast<ushort8*>(dxr + base) = ov;
     }
   }
 
@@ -215,11 +205,10 @@ __global__ void rmsnorm_fwd_bf16(const ushort* __restrict__ x, const ushort* __r
   const int lane = threadIdx.x;
   float sq = 0.f;
   for (int base = lane * 8; base < d; base += 64 * 8) {
-    ushort4 a = *reinterpret_cast<const ushort4*>(xr + base);
-    ushort4 c = *reinterpret_cast<const ushort4*>(xr + base + 4);
+    ushort8 av = *reinterpret_cast<const ushort8*>(xr + base);
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      float v = bf2f(k < 4 ? (&a.x)[k] : (&c.x)[k - 4]);
+      float v = bf2f(av[k]);
       sq += v * v;
     }
   }
@@ -227,20 +216,17 @@ __global__ void rmsnorm_fwd_bf16(const ushort* __restrict__ x, const ushort* __r
   const float rstd = rsqrtf(sq / d + eps);
   if (lane == 0) rstd_out[row] = rstd;
   for (int base = lane * 8; base < d; base += 64 * 8) {
-    ushort4 a = *reinterpret_cast<const ushort4*>(xr + base);
-    ushort4 c = *reinterpret_cast<const ushort4*>(xr + base + 4);
-    ushort4 wa = *reinterpret_cast<const ushort4*>(w + base);
-    ushort4 wc = *reinterpret_cast<const ushort4*>(w + base + 4);
-    ushort4 o0, o1;
+    ushort8 av = *reinterpret_cast<const ushort8*>(xr + base);
+    ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
+    ushort8 ov;
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
-      float v = bf2f(k < 4 ? (&a.x)[k] : (&c.x)[k - 4]);
-      float wk = bf2f(k < 4 ? (&wa.x)[k] : (&wc.x)[k - 4]);
+      float v = bf2f(av[k]);
+      float wk = bf2f(wv[k]);
       ushort r = f2bf(v * rstd * wk);
-      if (k < 4) (&o0.x)[k] = r; else (&o1.x)[k - 4] = r;
+      ov[k] = r;
     }
-    *reinterpret_cast<ushort4*>(yr + base) = o0;
-    *reinterpret_cast<ushort4*>(yr + base + 4) = o1;
+    *reinterpret_cast<ushort8*>(yr + base) = ov;
   }
 }
 
@@ -267,17 +253,14 @@ __global__ void rmsnorm_bwd_bf16(const ushort* __restrict__ dy, const ushort* __
 
     float s = 0.f;
     for (int base = lane * 8; base < d; base += 64 * 8) {
-      ushort4 g0 = *reinterpret_cast<const ushort4*>(dyr + base);
-      ushort4 g1 = *reinterpret_cast<const ushort4*>(dyr + base + 4);
-      ushort4 x0 = *reinterpret_cast<const ushort4*>(xr + base);
-      ushort4 x1 = *reinterpret_cast<const ushort4*>(xr + base + 4);
-      ushort4 w0 = *reinterpret_cast<const ushort4*>(w + base);
-      ushort4 w1 = *reinterpret_cast<const ushort4*>(w + base + 4);
+      ushort8 gv = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 xv8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = bf2f(k < 4 ? (&g0.x)[k] : (&g1.x)[k - 4]);
-        float xv = bf2f(k < 4 ? (&x0.x)[k] : (&x1.x)[k - 4]);
-        float wk = bf2f(k < 4 ? (&w0.x)[k] : (&w1.x)[k - 4]);
+        float g = bf2f(gv[k]);
+        float xv = bf2f(xv8[k]);
+        float wk = bf2f(wv[k]);
         s += g * wk * xv;
       }
     }
@@ -286,25 +269,21 @@ __global__ void rmsnorm_bwd_bf16(const ushort* __restrict__ dy, const ushort* __
 
     int it = 0;
     for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
-      ushort4 g0 = *reinterpret_cast<const ushort4*>(dyr + base);
-      ushort4 g1 = *reinterpret_cast<const ushort4*>(dyr + base + 4);
-      ushort4 x0 = *reinterpret_cast<const ushort4*>(xr + base);
-      ushort4 x1 = *reinterpret_cast<const ushort4*>(xr + base + 4);
-      ushort4 w0 = *reinterpret_cast<const ushort4*>(w + base);
-      ushort4 w1 = *reinterpret_cast<const ushort4*>(w + base + 4);
-      ushort4 o0, o1;
+      ushort8 gv = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 xv8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
+      ushort8 ov;
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = bf2f(k < 4 ? (&g0.x)[k] : (&g1.x)[k - 4]);
-        float xv = bf2f(k < 4 ? (&x0.x)[k] : (&x1.x)[k - 4]);
-        float wk = bf2f(k < 4 ? (&w0.x)[k] : (&w1.x)[k - 4]);
+        float g = bf2f(gv[k]);
+        float xv = bf2f(xv8[k]);
+        float wk = bf2f(wv[k]);
         float dxv = rs * g * wk - xv * c;
         ushort r = f2bf(dxv);
-        if (k < 4) (&o0.x)[k] = r; else (&o1.x)[k - 4] = r;
+        ov[k] = r;
         accw[it][k] += g * xv * rs;
       }
-      *reinterpret_cast<ushort4*>(dxr + base) = o0;
-      *reinterpret_cast<ushort4*>(dxr + base + 4) = o1;
+      *reinterpret_cast<ushort8*>(dxr + base) = ov;
     }
   }
 
