@@ -82,7 +82,7 @@ def apply_rope(x, cos, sin):
 
 
 class LlamaAttention(nn.Module):
-    def __init__(self, config: LlamaConfig, attn_impl: str = "math"):
+    def __init__(self, config: LlamaConfig, attn_impl: str = "chunked"):
         super().__init__()
         self.n_heads = config.num_attention_heads
         self.n_kv = config.num_key_value_heads
@@ -114,10 +114,11 @@ class LlamaAttention(nn.Module):
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
 
-        if self.attn_impl == "fused" and x.is_cuda and S > 1:
-            from ..ops.attention import flash_attention_forward
+        if self.attn_impl in ("fused", "chunked") and x.is_cuda and S > 1:
+            # blockwise flash attention: O(S·block) memory, MFMA GEMM blocks
+            from ..ops.attention import flash_attention
 
-            ctx = flash_attention_forward(q, k, v, causal=True)
+            ctx = flash_attention(q, k, v, causal=True)
         else:
             scale = 1.0 / math.sqrt(self.head_dim)
             scores = torch.matmul(q, k.transpose(-1, -2)) * scale
@@ -143,7 +144,7 @@ class LlamaMLP(nn.Module):
 
 
 class LlamaDecoderLayer(nn.Module):
-    def __init__(self, config: LlamaConfig, attn_impl: str = "math"):
+    def __init__(self, config: LlamaConfig, attn_impl: str = "chunked"):
         super().__init__()
         self.input_layernorm = RMSNorm(config.hidden_size, config.rms_norm_eps)
         self.self_attn = LlamaAttention(config, attn_impl)
@@ -157,7 +158,7 @@ class LlamaDecoderLayer(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
-    def __init__(self, config: LlamaConfig = None, attn_impl: str = "math"):
+    def __init__(self, config: LlamaConfig = None, attn_impl: str = "chunked"):
         super().__init__()
         self.config = config or LlamaConfig()
         c = self.config

@@ -1,0 +1,120 @@
+"""Memory-efficient attention for MI355X training (flash-attention algorithm,
+blockwise over Q/K with online softmax + logsumexp-recompute backward).
+
+Replaces the O(S²)-materializing math path for long sequences: peak memory is
+O(S·block) instead of O(S²) and every hot op is an MFMA-backed rocBLAS GEMM
+batch. fp32 softmax statistics, bf16 GEMM operands. The API is stable so the
+fully-fused CDNA4 HIP kernel can replace the torch-ops body without touching
+models (`attn_impl="fused"`).
+
+Forward saves only (q, k, v, out, logsumexp): backward recomputes P per
+block (the flash-attention backward), with
+  D  = rowsum(dout ⊙ out)
+  P  = exp(q kᵀ·scale − L)
+  dv = Pᵀ dout ;  dP = dout vᵀ ;  dS = P ⊙ (dP − D)·scale
+  dq = dS k    ;  dk = dSᵀ q
+"""
+
+import math
+from typing import Optional
+
+import torch
+
+
+def _block(size, pref):
+    return min(size, pref)
+
+
+class _FlashAttentionFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale, q_block, k_block):
+        # q,k,v: [B, H, S, D] (kv may have S_k != S_q for cache decode)
+        B, H, Sq, Dh = q.shape
+        Sk = k.shape[2]
+        out = torch.empty_like(q)
+        lse = torch.empty(B, H, Sq, dtype=torch.float32, device=q.device)
+        past = Sk - Sq  # causal offset when kv includes a prefix
+
+        for q0 in range(0, Sq, q_block):
+            q1 = min(q0 + q_block, Sq)
+            qb = q[:, :, q0:q1]
+            acc = torch.zeros(B, H, q1 - q0, Dh, dtype=torch.float32, device=q.device)
+            m = torch.full((B, H, q1 - q0), -float("inf"), dtype=torch.float32, device=q.device)
+            denom = torch.zeros(B, H, q1 - q0, dtype=torch.float32, device=q.device)
+            k_hi = Sk if not causal else min(Sk, past + q1)
+            for k0 in range(0, k_hi, k_block):
+                k1 = min(k0 + k_block, k_hi)
+                s = torch.matmul(qb, k[:, :, k0:k1].transpose(-1, -2)).float() * scale
+                if causal and k1 > past + q0:
+                    qi = torch.arange(q0, q1, device=q.device)[:, None]
+                    ki = torch.arange(k0, k1, device=q.device)[None, :]
+                    s = s.masked_fill(ki > past + qi, -float("inf"))
+                blk_max = s.amax(-1)
+                new_m = torch.maximum(m, blk_max)
+                # rows still at -inf (fully masked so far) contribute nothing
+                corr = torch.exp(m - new_m).nan_to_num(0.0)
+                p = torch.exp(s - new_m[..., None]).nan_to_num(0.0)
+                acc = acc * corr[..., None] + torch.matmul(p.to(v.dtype), v[:, :, k0:k1]).float()
+                denom = denom * corr + p.sum(-1)
+                m = new_m
+            out[:, :, q0:q1] = (acc / denom[..., None].clamp_min(1e-30)).to(q.dtype)
+            lse[:, :, q0:q1] = m + denom.clamp_min(1e-30).log()
+
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.meta = (causal, scale, q_block, k_block, past)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        causal, scale, q_block, k_block, past = ctx.meta
+        B, H, Sq, Dh = q.shape
+        Sk = k.shape[2]
+        dq = torch.zeros_like(q, dtype=torch.float32)
+        dk = torch.zeros_like(k, dtype=torch.float32)
+        dv = torch.zeros_like(v, dtype=torch.float32)
+        Drow = (dout.float() * out.float()).sum(-1)  # [B,H,Sq]
+
+        for q0 in range(0, Sq, q_block):
+            q1 = min(q0 + q_block, Sq)
+            qb = q[:, :, q0:q1]
+            dob = dout[:, :, q0:q1]
+            Lb = lse[:, :, q0:q1]
+            Db = Drow[:, :, q0:q1]
+            k_hi = Sk if not causal else min(Sk, past + q1)
+            for k0 in range(0, k_hi, k_block):
+                k1 = min(k0 + k_block, k_hi)
+                kb, vb = k[:, :, k0:k1], v[:, :, k0:k1]
+                s = torch.matmul(qb, kb.transpose(-1, -2)).float() * scale
+                if causal and k1 > past + q0:
+                    qi = torch.arange(q0, q1, device=q.device)[:, None]
+                    ki = torch.arange(k0, k1, device=q.device)[None, :]
+                    s = s.masked_fill(ki > past + qi, -float("inf"))
+                p = torch.exp(s - Lb[..., None]).nan_to_num(0.0)
+                pb = p.to(q.dtype)
+                dv[:, :, k0:k1] += torch.matmul(pb.transpose(-1, -2), dob).float()
+                dp = torch.matmul(dob, vb.transpose(-1, -2)).float()
+                ds = (p * (dp - Db[..., None]) * scale).to(q.dtype)
+                dq[:, :, q0:q1] += torch.matmul(ds, kb).float()
+                dk[:, :, k0:k1] += torch.matmul(ds.transpose(-1, -2), qb).float()
+
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None, None, None
+
+
+def flash_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool = True,
+    scale: Optional[float] = None,
+    q_block: int = 1024,
+    k_block: int = 1024,
+) -> torch.Tensor:
+    """Blockwise attention over [B, H, S, D] tensors."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    return _FlashAttentionFn.apply(q, k, v, causal, scale, _block(q.shape[2], q_block), _block(k.shape[2], k_block))
+
+
+# alias used by models (the fused HIP kernel will take over this symbol)
+flash_attention_forward = flash_attention

@@ -1,0 +1,92 @@
+"""Blockwise (flash) attention numerics vs plain fp32 math attention."""
+
+import math
+
+import pytest
+import torch
+
+gpu = pytest.mark.gpu
+
+
+def math_attention(q, k, v, causal):
+    scale = 1.0 / math.sqrt(q.shape[-1])
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        Sq, Sk = q.shape[2], k.shape[2]
+        past = Sk - Sq
+        qi = torch.arange(Sq, device=q.device)[:, None]
+        ki = torch.arange(Sk, device=q.device)[None, :]
+        s = s.masked_fill(ki > past + qi, -float("inf"))
+    p = torch.softmax(s, dim=-1)
+    return torch.matmul(p, v.float())
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_flash_attention_forward_cpu(causal):
+    from accelerate_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    q = torch.randn(2, 3, 65, 16)
+    k = torch.randn(2, 3, 65, 16)
+    v = torch.randn(2, 3, 65, 16)
+    out = flash_attention(q, k, v, causal=causal, q_block=32, k_block=16)
+    ref = math_attention(q, k, v, causal)
+    assert torch.allclose(out, ref.to(out.dtype), atol=1e-5), (out - ref).abs().max()
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_flash_attention_backward_cpu(causal):
+    from accelerate_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    shape = (1, 2, 48, 16)
+    q = torch.randn(*shape, requires_grad=True)
+    k = torch.randn(*shape, requires_grad=True)
+    v = torch.randn(*shape, requires_grad=True)
+    q2 = q.detach().clone().requires_grad_(True)
+    k2 = k.detach().clone().requires_grad_(True)
+    v2 = v.detach().clone().requires_grad_(True)
+    dout = torch.randn(*shape)
+
+    out = flash_attention(q, k, v, causal=causal, q_block=16, k_block=16)
+    out.backward(dout)
+    ref = math_attention(q2, k2, v2, causal)
+    ref.backward(dout)
+    for a, b, name in ((q, q2, "dq"), (k, k2, "dk"), (v, v2, "dv")):
+        assert torch.allclose(a.grad, b.grad, atol=1e-4), (name, (a.grad - b.grad).abs().max())
+
+
+def test_flash_attention_kv_cache_offset_cpu():
+    from accelerate_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    # decode-style: 4 query tokens attending to 20 keys (16 past + 4 new)
+    q = torch.randn(1, 2, 4, 16)
+    k = torch.randn(1, 2, 20, 16)
+    v = torch.randn(1, 2, 20, 16)
+    out = flash_attention(q, k, v, causal=True, q_block=2, k_block=8)
+    ref = math_attention(q, k, v, causal=True)
+    assert torch.allclose(out, ref.to(out.dtype), atol=1e-5)
+
+
+@gpu
+def test_flash_attention_bf16_gpu():
+    from accelerate_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    q = torch.randn(2, 8, 512, 128, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(2, 8, 512, 128, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(2, 8, 512, 128, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    out = flash_attention(q, k, v, causal=True, q_block=128, k_block=128)
+    ref = math_attention(q2, k2, v2, causal=True)
+    assert (out.float() - ref).abs().max() < 0.05
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    ref.backward(dout.float())
+    torch.cuda.synchronize()
+    for a, b in ((q, q2), (k, k2), (v, v2)):
+        rel = (a.grad.float() - b.grad).abs().max() / (b.grad.abs().max() + 1e-6)
+        assert rel < 0.05, rel
