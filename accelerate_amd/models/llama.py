@@ -51,17 +51,9 @@ class LlamaConfig:
         return cls(**base)
 
 
-class RMSNorm(nn.Module):
-    def __init__(self, hidden_size, eps=1e-5):
-        super().__init__()
-        self.weight = nn.Parameter(torch.ones(hidden_size))
-        self.eps = eps
-
-    def forward(self, x):
-        dtype = x.dtype
-        x = x.float()
-        x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
-        return (self.weight * x.to(dtype)) if self.weight.dtype == dtype else (self.weight.to(dtype) * x.to(dtype))
+# RMSNorm: the fused CDNA4 kernel (bf16 one-pass fwd, two-kernel bwd) with
+# a built-in eager fallback for CPU / non-bf16 (accelerate_amd.ops.norms)
+from ..ops.norms import FusedRMSNorm as RMSNorm
 
 
 def build_rope_cache(seq_len, head_dim, theta, device, dtype=torch.float32):
@@ -72,13 +64,19 @@ def build_rope_cache(seq_len, head_dim, theta, device, dtype=torch.float32):
 
 
 def apply_rope(x, cos, sin):
-    # x: [B, H, S, D]; rotate in the compute dtype (fp32 tables would
-    # silently promote q/k and push the attention GEMMs off bf16 MFMA)
+    # x: [B, H, S, D]
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 16 == 0 and cos.dtype == torch.float32:
+        from ..ops.rope import fused_rope
+
+        # one fused kernel instead of the slice/mul/cat chain (~7 kernels)
+        return fused_rope(x, cos, sin)
+    # eager fallback: rotate in the compute dtype (fp32 tables would silently
+    # promote q/k and push the attention GEMMs off bf16 MFMA)
     d = x.shape[-1] // 2
     x1, x2 = x[..., :d], x[..., d:]
-    cos = cos[None, None, : x.shape[2], :].to(x.dtype)
-    sin = sin[None, None, : x.shape[2], :].to(x.dtype)
-    return torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], dim=-1)
+    cosd = cos[None, None, : x.shape[2], :].to(x.dtype)
+    sind = sin[None, None, : x.shape[2], :].to(x.dtype)
+    return torch.cat([x1 * cosd - x2 * sind, x2 * cosd + x1 * sind], dim=-1)
 
 
 class LlamaAttention(nn.Module):

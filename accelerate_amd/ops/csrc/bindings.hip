@@ -245,6 +245,30 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, a
   return {dx, dw};
 }
 
+// rope kernel (rope_kernels.hip)
+__global__ void rope_bf16_kernel(const ushort*, ushort*, const float*, const float*,
+                                 int64_t, int, int, float);
+
+at::Tensor rope_bf16(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, bool inverse) {
+  // x: [..., S, D] bf16 contiguous; cos/sin: [>=S, D/2] fp32 contiguous
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == at::kBFloat16, "rope: bf16 contiguous");
+  const int D = (int)x.size(-1);
+  const int S = (int)x.size(-2);
+  TORCH_CHECK(D % 16 == 0, "rope: head_dim must be a multiple of 16");
+  TORCH_CHECK(cos_t.size(-1) == D / 2 && cos_t.size(0) >= S, "rope: bad cos table");
+  const int64_t n_bh = x.numel() / ((int64_t)S * D);
+  auto y = at::empty_like(x);
+  const int64_t total = n_bh * S * (D / 16);
+  int grid = (int)std::min<int64_t>((total + 255) / 256, 2048);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(rope_bf16_kernel, dim3(grid), dim3(256), 0, stream.stream(),
+                     reinterpret_cast<const ushort*>(x.data_ptr()),
+                     reinterpret_cast<ushort*>(y.data_ptr()),
+                     cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+                     n_bh, S, D, inverse ? -1.f : 1.f);
+  return y;
+}
+
 // fp8 kernels (fp8_kernels.hip)
 __global__ void fp8_cast_amax_e4m3(const ushort*, unsigned char*, const float*, float*, int64_t);
 __global__ void fp8_cast_amax_e5m2(const ushort*, unsigned char*, const float*, float*, int64_t);
@@ -306,6 +330,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "fused bf16 LayerNorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused bf16 RMSNorm forward");
+  m.def("rope_bf16", &rope_bf16, "fused rotary embedding (bf16, half-split layout)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused bf16 RMSNorm backward");
   m.def("fp8_cast_amax", &fp8_cast_amax, "bf16 -> fp8 cast with fused amax (gfx950)");
   m.def("fp8_cast_transpose", &fp8_cast_transpose,

@@ -90,3 +90,30 @@ def test_flash_attention_bf16_gpu():
     for a, b in ((q, q2), (k, k2), (v, v2)):
         rel = (a.grad.float() - b.grad).abs().max() / (b.grad.abs().max() + 1e-6)
         assert rel < 0.05, rel
+
+
+@gpu
+def test_fused_rope_matches_eager():
+    from accelerate_amd.models.llama import build_rope_cache
+    from accelerate_amd.ops.rope import fused_rope
+
+    torch.manual_seed(0)
+    B, H, S, D = 2, 4, 128, 64
+    cos, sin = build_rope_cache(S, D, 10000.0, "cuda")
+    x = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+
+    y = fused_rope(x, cos, sin)
+    # eager reference (fp32)
+    d = D // 2
+    x1f, x2f = x2.float()[..., :d], x2.float()[..., d:]
+    c = cos[None, None, :S, :]
+    s = sin[None, None, :S, :]
+    ref = torch.cat([x1f * c - x2f * s, x2f * c + x1f * s], dim=-1)
+    assert (y.float() - ref).abs().max() < 0.02
+
+    dout = torch.randn_like(y)
+    y.backward(dout)
+    ref.backward(dout.float())
+    torch.cuda.synchronize()
+    assert (x.grad.float() - x2.grad).abs().max() < 0.02
