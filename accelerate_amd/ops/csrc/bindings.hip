@@ -19,6 +19,7 @@ __global__ void fused_adamw_dev_kernel(TensorListMeta, const float*, const float
 __global__ void fused_adamw_bf16_kernel(TensorListMeta, const float*, const float*,
                                         float, float, float, float, const float*, const float*);
 __global__ void l2norm_squared_kernel(TensorListMeta, float*);
+__global__ void multi_tensor_copy_kernel(TensorListMeta, bool);
 __global__ void clip_coef_kernel(const float*, float, float*, float*);
 __global__ void multi_tensor_scale_kernel(TensorListMeta, const float*);
 __global__ void unscale_check_kernel(TensorListMeta, const float*, float*);
@@ -121,6 +122,57 @@ void fused_adamw_planned(at::Tensor addrs_numels, at::Tensor chunk_prefix,
                        grad_scale.has_value() ? grad_scale->data_ptr<float>() : nullptr,
                        found_inf.has_value() ? found_inf->data_ptr<float>() : nullptr);
   }
+}
+
+// gather (to_flat=true) or scatter (false) between tensors and flat[offsets]
+void multi_tensor_copy(std::vector<at::Tensor> tensors, at::Tensor flat,
+                       std::vector<int64_t> offsets, bool to_flat) {
+  const int n = (int)tensors.size();
+  TORCH_CHECK(n > 0 && (int)offsets.size() == n, "multi_tensor_copy: bad lists");
+  const int64_t esize = flat.element_size();
+  auto cpu_i64 = at::empty({3 * n}, at::TensorOptions().dtype(at::kLong).pinned_memory(true));
+  auto cpu_i32 = at::empty({n + 1}, at::TensorOptions().dtype(at::kInt).pinned_memory(true));
+  int64_t* a = cpu_i64.data_ptr<int64_t>();
+  int32_t* pre = cpu_i32.data_ptr<int32_t>();
+  pre[0] = 0;
+  const int64_t flat_base = reinterpret_cast<int64_t>(flat.data_ptr());
+  for (int t = 0; t < n; ++t) {
+    TORCH_CHECK(tensors[t].is_contiguous() && tensors[t].element_size() == esize,
+                "multi_tensor_copy: dtype/layout mismatch at ", t);
+    const int64_t nbytes = tensors[t].numel() * esize;
+    a[t] = reinterpret_cast<int64_t>(tensors[t].data_ptr());
+    a[n + t] = flat_base + offsets[t] * esize;
+    a[2 * n + t] = nbytes;
+    pre[t + 1] = pre[t] + (int32_t)((nbytes + kChunkSize - 1) / kChunkSize);
+  }
+  auto dev = flat.device();
+  auto dev_i64 = cpu_i64.to(dev, true);
+  auto dev_i32 = cpu_i32.to(dev, true);
+  TensorListMeta meta;
+  meta.addrs = dev_i64.data_ptr<int64_t>();
+  meta.numels = dev_i64.data_ptr<int64_t>() + 2 * n;
+  meta.chunk_prefix = dev_i32.data_ptr<int32_t>();
+  meta.n_tensors = n;
+  meta.n_lists = 2;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(multi_tensor_copy_kernel, dim3(pre[n]), dim3(kBlockThreads), 0, stream.stream(),
+                     meta, to_flat);
+}
+
+// capture-safe variant over a pre-built device plan (addrs: [2n ptrs | n byte
+// lengths], prefix over kChunkSize-byte chunks) — the Python caller caches
+// the plan (and its pinned staging) exactly like FusedAdamW's plans.
+void multi_tensor_copy_planned(at::Tensor addrs_dev, at::Tensor prefix_dev,
+                               int64_t n_tensors, int64_t total_chunks, bool to_flat) {
+  TensorListMeta meta;
+  meta.addrs = addrs_dev.data_ptr<int64_t>();
+  meta.numels = addrs_dev.data_ptr<int64_t>() + 2 * n_tensors;
+  meta.chunk_prefix = prefix_dev.data_ptr<int32_t>();
+  meta.n_tensors = (int32_t)n_tensors;
+  meta.n_lists = 2;
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(multi_tensor_copy_kernel, dim3(total_chunks), dim3(kBlockThreads), 0, stream.stream(),
+                     meta, to_flat);
 }
 
 at::Tensor l2norm_squared(std::vector<at::Tensor> grads) {
@@ -347,5 +399,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2norm_squared", &l2norm_squared, "global L2 norm squared over tensor list");
   m.def("clip_grad_norm", &clip_grad_norm, "on-device clip_grad_norm_, returns total norm");
   m.def("multi_tensor_scale", &multi_tensor_scale, "g *= *coef");
+  m.def("multi_tensor_copy", &multi_tensor_copy,
+        "fused gather/scatter between tensor list and a flat bucket");
+  m.def("multi_tensor_copy_planned", &multi_tensor_copy_planned,
+        "capture-safe fused gather/scatter over a cached plan");
   m.def("unscale_and_check", &unscale_and_check, "g *= *inv_scale with non-finite detection");
 }

@@ -243,6 +243,35 @@ __global__ void fused_adamw_bf16_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// multi-tensor gather/scatter between parameter gradients and a flat bucket
+// (the DDP reducer's bucket staging — one launch per bucket instead of one
+// copy kernel per parameter). Dtype-agnostic byte copy, uint4-vectorized;
+// list0 = tensor ptrs, list1 = flat+offset ptrs (both byte addresses).
+// chunk_prefix is over BYTES here (kChunkSize-byte chunks).
+// ---------------------------------------------------------------------------
+
+__global__ void multi_tensor_copy_kernel(TensorListMeta meta, bool to_flat) {
+  const int cid = blockIdx.x;
+  const int t = find_tensor(meta.chunk_prefix, meta.n_tensors, cid);
+  const int64_t chunk_in_tensor = cid - meta.chunk_prefix[t];
+  const int64_t nbytes = meta.numels[t];
+  const int64_t lo = chunk_in_tensor * kChunkSize;
+  const int64_t hi = min(lo + kChunkSize, nbytes);
+  const unsigned char* src = reinterpret_cast<const unsigned char*>(
+      to_flat ? meta.addrs[t] : meta.addrs[meta.n_tensors + t]);
+  unsigned char* dst = reinterpret_cast<unsigned char*>(
+      to_flat ? meta.addrs[meta.n_tensors + t] : meta.addrs[t]);
+
+  const int64_t tid = threadIdx.x;
+  int64_t i = lo + tid * 16;
+  const int64_t vec_end = lo + ((hi - lo) & ~int64_t(15));
+  for (; i + 15 < vec_end; i += kBlockThreads * 16) {
+    *reinterpret_cast<uint4*>(dst + i) = *reinterpret_cast<const uint4*>(src + i);
+  }
+  for (int64_t j = vec_end + tid; j < hi; j += kBlockThreads) dst[j] = src[j];
+}
+
+// ---------------------------------------------------------------------------
 // global L2 norm²: per-wave shuffle reduce → per-block LDS reduce → one
 // device-scope atomicAdd per block (Guideline 12).
 // ---------------------------------------------------------------------------

@@ -49,6 +49,9 @@ class _Bucket:
         "launched",
         "dtype",
         "comm_dtype",
+        "plan",
+        "plan_key",
+        "pinned",
     )
 
     def __init__(self, params: List[torch.nn.Parameter], dtype, comm_dtype, device):
@@ -68,11 +71,51 @@ class _Bucket:
         self.ready = set()
         self.work = None
         self.launched = False
+        self.plan = None
+        self.plan_key = None
+        self.pinned = None
 
     def reset(self):
         self.ready.clear()
         self.work = None
         self.launched = False
+
+    def build_plan(self, grads):
+        """Cache the fused gather/scatter plan (grad ptrs → flat slices).
+
+        Pinned staging is allocated once and reused so a rebuild triggered
+        inside hipGraph capture stays capture-legal (async H2D only)."""
+        import torch as _torch
+
+        n = len(grads)
+        esize = self.flat.element_size()
+        key = tuple(g.data_ptr() for g in grads)
+        if self.plan is not None and self.plan_key == key:
+            return self.plan
+        flat_base = self.flat.data_ptr()
+        ptrs = [g.data_ptr() for g in grads]
+        dsts = [flat_base + off * esize for off in self.offsets]
+        nbytes = [g.numel() * esize for g in grads]
+        chunk = 16384
+        prefix = [0]
+        for b in nbytes:
+            prefix.append(prefix[-1] + (b + chunk - 1) // chunk)
+        if self.pinned is None or self.pinned[0].numel() != 3 * n:
+            self.pinned = (
+                _torch.empty(3 * n, dtype=_torch.int64, pin_memory=True),
+                _torch.empty(n + 1, dtype=_torch.int32, pin_memory=True),
+            )
+        self.pinned[0].copy_(_torch.tensor(ptrs + dsts + nbytes, dtype=_torch.int64))
+        self.pinned[1].copy_(_torch.tensor(prefix, dtype=_torch.int32))
+        dev = self.flat.device
+        self.plan = (
+            self.pinned[0].to(dev, non_blocking=True),
+            self.pinned[1].to(dev, non_blocking=True),
+            n,
+            prefix[-1],
+        )
+        self.plan_key = key
+        return self.plan
 
 
 class DistributedDataParallelEngine(nn.Module):
@@ -150,14 +193,29 @@ class DistributedDataParallelEngine(nn.Module):
 
     # -- steady state ------------------------------------------------------
 
+    def _fused_copy(self, bucket: _Bucket, to_flat: bool) -> bool:
+        """One kernel stages/unstages the whole bucket (vs one copy kernel
+        per parameter — the reducer's native hot path). GPU-only."""
+        if bucket.flat.device.type != "cuda":
+            return False
+        grads = [p.grad for p in bucket.params]
+        if any(g is None or not g.is_contiguous() or g.dtype != bucket.dtype for g in grads):
+            return False
+        from ..ops import _load_extension
+
+        ext = _load_extension(required=False)
+        if ext is None:
+            return False
+        plan = bucket.build_plan(grads)
+        ext.multi_tensor_copy_planned(plan[0], plan[1], plan[2], plan[3], to_flat)
+        return True
+
     def _on_grad_ready(self, param: torch.nn.Parameter):
         if not self.require_backward_grad_sync or self._world_size <= 1:
             return
         bucket, index = self._param_to_bucket[param]
         if index in bucket.ready:
             return
-        lo = bucket.offsets[index]
-        bucket.flat[lo : lo + param.numel()].copy_(param.grad.reshape(-1), non_blocking=True)
         bucket.ready.add(index)
         if len(bucket.ready) == len(bucket.params):
             self._launch(bucket)
@@ -166,6 +224,13 @@ class DistributedDataParallelEngine(nn.Module):
         if bucket.launched:
             return
         bucket.launched = True
+        if not self._fused_copy(bucket, to_flat=True):
+            for i, p in enumerate(bucket.params):
+                lo = bucket.offsets[i]
+                if p.grad is not None:
+                    bucket.flat[lo : lo + p.numel()].copy_(p.grad.reshape(-1), non_blocking=True)
+                else:
+                    bucket.flat[lo : lo + p.numel()].zero_()
         if bucket.comm_flat is not bucket.flat:
             bucket.comm_flat.copy_(bucket.flat)
         op = dist.ReduceOp.AVG if self._use_avg else dist.ReduceOp.SUM
@@ -180,14 +245,10 @@ class DistributedDataParallelEngine(nn.Module):
             return
         for bucket in self._buckets:
             if not bucket.launched:
-                # some params had no grad this step: contribute zeros
-                for i, p in enumerate(bucket.params):
-                    if i not in bucket.ready:
-                        lo = bucket.offsets[i]
-                        if p.grad is not None:
-                            bucket.flat[lo : lo + p.numel()].copy_(p.grad.reshape(-1), non_blocking=True)
-                        else:
-                            bucket.flat[lo : lo + p.numel()].zero_()
+                # stragglers (params that never got a grad contribute zeros)
+                for p in bucket.params:
+                    if p.grad is None:
+                        p.grad = torch.zeros_like(p)
                 self._launch(bucket)
         for bucket in self._buckets:
             if bucket.work is not None:
@@ -196,13 +257,14 @@ class DistributedDataParallelEngine(nn.Module):
                 bucket.flat.copy_(bucket.comm_flat)
             if not self._use_avg:
                 bucket.flat.div_(self._world_size)
-            for i, p in enumerate(bucket.params):
-                lo = bucket.offsets[i]
-                reduced = bucket.flat[lo : lo + p.numel()].view_as(p)
-                if p.grad is None:
-                    p.grad = reduced.clone()
-                else:
-                    p.grad.copy_(reduced, non_blocking=True)
+            if not self._fused_copy(bucket, to_flat=False):
+                for i, p in enumerate(bucket.params):
+                    lo = bucket.offsets[i]
+                    reduced = bucket.flat[lo : lo + p.numel()].view_as(p)
+                    if p.grad is None:
+                        p.grad = reduced.clone()
+                    else:
+                        p.grad.copy_(reduced, non_blocking=True)
             bucket.reset()
 
     def forward(self, *args, **kwargs):

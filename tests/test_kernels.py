@@ -177,3 +177,27 @@ def test_fused_adamw_bench_sanity():
     dt = (time.perf_counter() - t0) / 10
     # ~110M params fp32: 4 tensors × 4 B × ~2 traffic each ≈ 3.5 GB/step / 8 TB/s ≈ 0.5 ms
     assert dt < 0.02, f"fused adamw too slow: {dt*1000:.2f} ms/step"
+
+
+@gpu
+def test_multi_tensor_copy_roundtrip():
+    from accelerate_amd.ops import _load_extension
+
+    ext = _load_extension(required=True)
+    torch.manual_seed(0)
+    tensors = [torch.randn(s, device="cuda", dtype=torch.bfloat16) for s in [(128,), (64, 64), (33,)]]
+    total = sum(t.numel() for t in tensors)
+    flat = torch.zeros(total, device="cuda", dtype=torch.bfloat16)
+    offsets, off = [], 0
+    for t in tensors:
+        offsets.append(off)
+        off += t.numel()
+    ext.multi_tensor_copy(tensors, flat, offsets, True)
+    torch.cuda.synchronize()
+    ref = torch.cat([t.reshape(-1) for t in tensors])
+    assert torch.equal(flat, ref)
+    flat.mul_(2)
+    ext.multi_tensor_copy(tensors, flat, offsets, False)
+    torch.cuda.synchronize()
+    for t, o in zip(tensors, offsets):
+        assert torch.equal(t.reshape(-1), flat[o : o + t.numel()])
