@@ -185,9 +185,32 @@ class ShardedModel(nn.Module):
         process_group=None,
         device: Optional[torch.device] = None,
         activation_checkpointing: bool = False,
+        shard_group_size: Optional[int] = None,
     ):
         super().__init__()
         self.module = module
+        global_world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+
+        # HSDP (hybrid_shard): shard within groups of `shard_group_size`
+        # consecutive ranks (intra-node xGMI reduce_scatter/all_gather),
+        # replicate across groups (inter-group all_reduce of shard grads —
+        # the reference's 2-level mesh, SURVEY.md §2.3 HSDP row)
+        self.replica_group = None
+        if shard_group_size is not None and dist.is_initialized() and 1 < shard_group_size < global_world:
+            if global_world % shard_group_size != 0:
+                raise ValueError(f"world size {global_world} not divisible by shard_group_size {shard_group_size}")
+            rank = dist.get_rank()
+            n_groups = global_world // shard_group_size
+            shard_groups = [
+                dist.new_group(list(range(i * shard_group_size, (i + 1) * shard_group_size)))
+                for i in range(n_groups)
+            ]
+            replica_groups = [
+                dist.new_group(list(range(j, global_world, shard_group_size))) for j in range(shard_group_size)
+            ]
+            process_group = shard_groups[rank // shard_group_size]
+            self.replica_group = replica_groups[rank % shard_group_size]
+
         self.group = process_group
         self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
         if device is None:
@@ -352,6 +375,13 @@ class ShardedModel(nn.Module):
             unit.grads_ready = 0
             if self.require_backward_grad_sync:
                 unit.reduce_grads()
+                if self.replica_group is not None:
+                    # HSDP level 2: average the shard grad across replicas
+                    if _is_nccl(self.replica_group):
+                        dist.all_reduce(unit.shard.grad, op=dist.ReduceOp.AVG, group=self.replica_group)
+                    else:
+                        dist.all_reduce(unit.shard.grad, group=self.replica_group)
+                        unit.shard.grad.div_(dist.get_world_size(self.replica_group))
                 unit.reshard()
             # under no_sync keep full grads resident for accumulation
 
@@ -536,6 +566,11 @@ def fsdp_prepare(accelerator, args, device_placement):
                 compute_dtype = torch.bfloat16
             elif compute_dtype is None and accelerator.mixed_precision == "fp16":
                 compute_dtype = torch.float16
+            shard_group_size = None
+            if plugin.sharding_strategy == "hybrid_shard":
+                import os as _os
+
+                shard_group_size = int(_os.environ.get("FSDP_SHARD_GROUP_SIZE", "0")) or None
             wrapped_model = ShardedModel(
                 obj,
                 transformer_cls_names=plugin.transformer_cls_names_to_wrap,
@@ -545,6 +580,7 @@ def fsdp_prepare(accelerator, args, device_placement):
                 reshard_after_forward=plugin.reshard_after_forward,
                 device=accelerator.device,
                 activation_checkpointing=plugin.activation_checkpointing,
+                shard_group_size=shard_group_size,
             )
             swap_map = wrapped_model.param_swap_map()
             accelerator._models.append(wrapped_model)

@@ -116,6 +116,7 @@ def main():
         # Whole-step hipGraph: zero_grad + fwd + bwd (+ RCCL allreduce) +
         # fused AdamW replay from ONE graph launch. Inputs live in static
         # device buffers; lr is a device scalar the scheduler refreshes.
+        graph = loss_buf = None
         try:
             static = [t.to(device, non_blocking=True) for t in next_batch()]
 
@@ -143,7 +144,33 @@ def main():
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 loss_buf = graph_body()
+            captured = True
+        except Exception as e:
+            print(f"[bench] hipGraph capture failed ({e}); falling back to eager", flush=True)
+            captured = False
 
+        # all-or-none: every rank must run the SAME collective sequence, so a
+        # capture failure anywhere sends the whole job down the eager path
+        if accelerator.use_distributed:
+            oks = [None] * n
+            torch.distributed.all_gather_object(oks, captured)
+            captured = all(oks)
+
+        if captured:
+            # sanity replay (all ranks together): RCCL-in-graph exercised
+            # before committing; finiteness agreed across ranks
+            graph.replay()
+            torch.cuda.synchronize()
+            finite = bool(torch.isfinite(loss_buf).all().item())
+            if accelerator.use_distributed:
+                oks = [None] * n
+                torch.distributed.all_gather_object(oks, finite)
+                finite = all(oks)
+            captured = finite
+            if not finite:
+                print("[bench] captured step produced non-finite loss; eager fallback", flush=True)
+
+        if captured:
             fused = optimizer.optimizer  # the FusedAdamW under the wrapper
 
             def graph_step(batch):
@@ -156,8 +183,7 @@ def main():
 
             one_step = graph_step
             step_mode = "hipgraph"
-        except Exception as e:
-            print(f"[bench] hipGraph capture failed ({e}); falling back to eager", flush=True)
+        else:
             one_step = eager_step
 
     # warmup

@@ -73,3 +73,9 @@ def test_pipeline_inference_oracle():
     script = Path(__file__).parent / "distributed_scripts" / "pipeline_script.py"
     out = launch_distributed(script, nproc=2, timeout=180)
     assert "PIPELINE_PASS" in out
+
+
+def test_hsdp_oracle_4proc():
+    script = Path(__file__).parent / "distributed_scripts" / "hsdp_script.py"
+    out = launch_distributed(script, nproc=4, timeout=240)
+    assert "HSDP_PARITY_PASS" in out
