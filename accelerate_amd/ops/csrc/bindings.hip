@@ -297,6 +297,27 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, a
   return {dx, dw};
 }
 
+// MFMA GEMM (gemm_kernels.hip)
+__global__ void gemm_bt_bf16_kernel(const __bf16*, const __bf16*, const ushort*, ushort*, int, int, int);
+
+at::Tensor mfma_gemm_bt(at::Tensor a, at::Tensor b, c10::optional<at::Tensor> bias) {
+  // C[M,N] = a[M,K] @ b[N,K]^T — the nn.Linear forward layout
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && a.scalar_type() == at::kBFloat16, "mfma_gemm_bt: A bf16 contiguous");
+  TORCH_CHECK(b.is_cuda() && b.is_contiguous() && b.scalar_type() == at::kBFloat16, "mfma_gemm_bt: B bf16 contiguous");
+  const int M = (int)a.size(0), K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK(b.size(1) == K, "mfma_gemm_bt: K mismatch");
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0, "mfma_gemm_bt: need M,N %128==0 and K %64==0");
+  auto c = at::empty({M, N}, a.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  const int grid = (M / 128) * (N / 128);
+  hipLaunchKernelGGL(gemm_bt_bf16_kernel, dim3(grid), dim3(256), 0, stream.stream(),
+                     reinterpret_cast<const __bf16*>(a.data_ptr()),
+                     reinterpret_cast<const __bf16*>(b.data_ptr()),
+                     bias.has_value() ? reinterpret_cast<const ushort*>(bias->data_ptr()) : nullptr,
+                     reinterpret_cast<ushort*>(c.data_ptr()), M, N, K);
+  return c;
+}
+
 // rope kernel (rope_kernels.hip)
 __global__ void rope_bf16_kernel(const ushort*, ushort*, const float*, const float*,
                                  int64_t, int, int, float);
@@ -383,6 +404,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused bf16 RMSNorm forward");
   m.def("rope_bf16", &rope_bf16, "fused rotary embedding (bf16, half-split layout)");
+  m.def("mfma_gemm_bt", &mfma_gemm_bt, "hand-written MFMA bf16 GEMM (A @ B^T + bias)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused bf16 RMSNorm backward");
   m.def("fp8_cast_amax", &fp8_cast_amax, "bf16 -> fp8 cast with fused amax (gfx950)");
   m.def("fp8_cast_transpose", &fp8_cast_transpose,

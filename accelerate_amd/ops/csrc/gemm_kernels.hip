@@ -1,0 +1,130 @@
+// Hand-written MFMA GEMM for gfx950: C[M,N] = A[M,K] · B[N,K]^T (+bias),
+// bf16 inputs, fp32 accumulate, bf16 out — the nn.Linear forward layout.
+//
+// Structure = the CDNA4 guide's verified 128²-tile recipe (§5 ladder step 3):
+// 4 waves per block (2×2), each wave owns a 64×64 output sub-tile of 4×4
+// 16×16 MFMA fragments; BK=64 staged through LDS with
+// __builtin_amdgcn_global_load_lds width 16 (direct HBM→LDS, no VGPR
+// round-trip); single LDS buffer, two barriers per K-step.
+//
+// Fragment mapping (verified on MI355X by benchmarks/mfma_layout_probe.hip):
+// the mfma_f32_16x16x32_bf16 k-pairing is consistent for any shared (lane,j)
+// →k bijection; we use k = (lane>>4)*8 + j so LDS fragment reads are single
+// contiguous 16-byte ds_read_b128s. C/D: col = lane&15, row = (lane>>4)*4+j
+// (guide §3, m89-verified).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+namespace {
+
+typedef __bf16 gbf16;
+typedef gbf16 bf16x8g __attribute__((ext_vector_type(8)));
+typedef float f32x4g __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ ushort gf2bf(float f) {
+  union { float f; unsigned int i; } cv;
+  cv.f = f;
+  if ((cv.i & 0x7F800000u) == 0x7F800000u) return (ushort)(cv.i >> 16) | (ushort)((cv.i & 0xFFFFu) ? 0x40 : 0);
+  cv.i += 0x7FFFu + ((cv.i >> 16) & 1u);
+  return (ushort)(cv.i >> 16);
+}
+
+}  // namespace
+
+// block = 256 threads (4 waves as 2×2), tile 128(M)×128(N), BK=64
+__global__ __launch_bounds__(256) void gemm_bt_bf16_kernel(
+    const gbf16* __restrict__ A, const gbf16* __restrict__ B,
+    const ushort* __restrict__ bias, ushort* __restrict__ C,
+    int M, int N, int K) {
+  constexpr int BM = 128, BN = 128, BK = 64;
+  __shared__ gbf16 As[BM * BK];
+  __shared__ gbf16 Bs[BN * BK];
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;          // 0..3
+  const int lane = tid & 63;
+  const int wm = wave >> 1;           // wave row (0..1) → 64 rows each
+  const int wn = wave & 1;            // wave col (0..1) → 64 cols each
+
+  // XCD-aware block swizzle (guide T1, bijective variant): consecutive
+  // blocks share B-panels; group them per XCD for L2 locality.
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  {
+    const int nx = 8;
+    const int q = nwg / nx, r = nwg % nx;
+    const int xcd = bid % nx, idx = bid / nx;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int tiles_n = N / BN;
+  const int tile_m = (bid / tiles_n) * BM;
+  const int tile_n = (bid % tiles_n) * BN;
+
+  // staging: per round, each wave DMAs 8 rows × 64 cols (8 lanes/row × 16 B)
+  // into a contiguous LDS span (global_load_lds is wave-uniform-base+lane*16)
+  const int s_row_in_wave = lane >> 3;   // 0..7
+  const int s_col = (lane & 7) * 8;      // 0..56, 8 bf16 = 16 B
+
+  f32x4g acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4g{};
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // LDS write address is HARDWARE-fixed at (wave-uniform base + lane*16 B)
+    // (guide §5 caveat): with lane = rr*8+cc that lands exactly row-major
+    // [8 rows][64 cols] from the base — the global address carries the
+    // per-lane row/col.
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = r * 32 + wave * 8 + s_row_in_wave;  // 0..127
+      const gbf16* ga = A + (int64_t)(tile_m + row) * K + k0 + s_col;
+      const gbf16* gb = B + (int64_t)(tile_n + row) * K + k0 + s_col;
+      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)ga,
+                                       (__attribute__((address_space(3))) unsigned int*)(As + (r * 32 + wave * 8) * BK),
+                                       16, 0, 0);
+      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)gb,
+                                       (__attribute__((address_space(3))) unsigned int*)(Bs + (r * 32 + wave * 8) * BK),
+                                       16, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)");
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {  // two K=32 sub-steps
+      // fragment k-slot: this lane reads 8 contiguous bf16 at
+      // k = kk*32 + (lane>>4)*8; fragments hoisted once (8 ds_read_b128
+      // per sub-step, not 20) then 16 MFMAs on registers
+      const int kslot = kk * 32 + (lane >> 4) * 8;
+      bf16x8g af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        af[i] = *reinterpret_cast<const bf16x8g*>(As + (wm * 64 + i * 16 + (lane & 15)) * BK + kslot);
+        bf[i] = *reinterpret_cast<const bf16x8g*>(Bs + (wn * 64 + i * 16 + (lane & 15)) * BK + kslot);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C[row][col], row = (lane>>4)*4 + j within fragment
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int col = tile_n + wn * 64 + ni * 16 + (lane & 15);
+      const float badd = (bias != nullptr) ? (float)(*reinterpret_cast<const __hip_bfloat16*>(bias + col)) : 0.f;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int row = tile_m + wm * 64 + mi * 16 + (lane >> 4) * 4 + j;
+        C[(int64_t)row * N + col] = gf2bf(acc[mi][ni][j] + badd);
+      }
+    }
+  }
+}
