@@ -1,0 +1,102 @@
+"""2-process gloo oracle for auxiliary distributed features:
+- gather_for_metrics drops even_batches tail duplicates (remainder dedup)
+- ACCELERATE_DEBUG_MODE shape verification raises a per-rank table
+- DDP comm_dtype=bf16 wire compression still averages correctly
+- DataLoaderDispatcher (rank-0 fetch + broadcast + slice) covers the data
+"""
+
+import os
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+from torch.utils.data import DataLoader, TensorDataset
+
+from accelerate_amd import Accelerator, set_seed
+from accelerate_amd.utils.dataclasses import DistributedDataParallelKwargs
+from accelerate_amd.utils.operations import DistributedOperationException, gather
+
+
+def test_gather_for_metrics(acc):
+    # 10 samples, batch 2, world 2 -> per-rank loader yields 3 batches with
+    # the tail wrapped; gather_for_metrics must return exactly 10 predictions
+    ds = TensorDataset(torch.arange(10).float())
+    dl = acc.prepare_data_loader(DataLoader(ds, batch_size=2))
+    seen = []
+    for (batch,) in dl:
+        seen.append(acc.gather_for_metrics(batch))
+    total = torch.cat(seen)
+    assert total.numel() == 10, f"expected 10 samples after dedup, got {total.numel()}"
+    assert sorted(total.tolist()) == [float(i) for i in range(10)], total
+    if acc.is_main_process:
+        print("METRICS_DEDUP_PASS")
+
+
+def test_debug_mode(acc):
+    os.environ["ACCELERATE_DEBUG_MODE"] = "1"
+    from accelerate_amd.state import PartialState
+
+    PartialState._shared_state["debug"] = True
+    bad = torch.ones(acc.process_index + 1)  # mismatched shapes across ranks
+    raised = False
+    try:
+        gather(bad)
+    except DistributedOperationException:
+        raised = True
+    assert raised, "debug mode must raise on shape mismatch"
+    PartialState._shared_state["debug"] = False
+    os.environ.pop("ACCELERATE_DEBUG_MODE")
+    if acc.is_main_process:
+        print("DEBUG_MODE_PASS")
+
+
+def test_comm_dtype_bf16(acc):
+    set_seed(0)
+    model = nn.Linear(16, 4)
+    ref = nn.Linear(16, 4)
+    ref.load_state_dict(model.state_dict())
+    model = acc.prepare_model(model)
+    # rebuild engine with bf16 wire dtype
+    from accelerate_amd.parallel.ddp import DistributedDataParallelEngine
+
+    engine = DistributedDataParallelEngine(ref, comm_dtype="bf16")
+    n, r = acc.num_processes, acc.process_index
+    x = torch.randn(8, 16, generator=torch.Generator().manual_seed(1))
+    (engine(x[r::n]) ** 2).mean().backward()
+    engine.finalize()
+    gs = [torch.empty_like(engine.module.weight.grad) for _ in range(n)]
+    dist.all_gather(gs, engine.module.weight.grad)
+    assert torch.allclose(gs[0], gs[1], atol=1e-6), "bf16-wire grads must match across ranks"
+    if acc.is_main_process:
+        print("COMM_DTYPE_PASS")
+
+
+def test_dispatcher(acc):
+    from accelerate_amd.data_loader import DataLoaderDispatcher
+
+    ds = TensorDataset(torch.arange(16).float())
+    base = DataLoader(ds, batch_size=2)
+    dl = acc.prepare_data_loader(base)  # map-style -> shard mode; force dispatch:
+    acc.dispatch_batches = True
+    dl2 = acc.prepare_data_loader(DataLoader(ds, batch_size=2))
+    assert isinstance(dl2, DataLoaderDispatcher)
+    seen = torch.cat([b[0] for b in dl2])
+    everything = gather(seen)
+    assert sorted(everything.tolist()) == [float(i) for i in range(16)]
+    acc.dispatch_batches = None
+    if acc.is_main_process:
+        print("DISPATCHER_PASS")
+
+
+def main():
+    acc = Accelerator(cpu=True)
+    assert acc.num_processes == 2
+    test_gather_for_metrics(acc)
+    test_debug_mode(acc)
+    test_comm_dtype_bf16(acc)
+    test_dispatcher(acc)
+    acc.end_training()
+
+
+if __name__ == "__main__":
+    main()
