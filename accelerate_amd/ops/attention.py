@@ -27,13 +27,16 @@ def _block(size, pref):
 
 class _FlashAttentionFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale, q_block, k_block):
+    def forward(ctx, q, k, v, causal, scale, q_block, k_block, q_start=None):
         # q,k,v: [B, H, S, D] (kv may have S_k != S_q for cache decode)
         B, H, Sq, Dh = q.shape
         Sk = k.shape[2]
         out = torch.empty_like(q)
         lse = torch.empty(B, H, Sq, dtype=torch.float32, device=q.device)
-        past = Sk - Sq  # causal offset when kv includes a prefix
+        # causal offset: query i attends keys <= past + i. Default assumes the
+        # queries are the LAST Sq positions (kv-cache decode); context
+        # parallelism passes the rank's absolute start explicitly.
+        past = (Sk - Sq) if q_start is None else q_start
 
         for q0 in range(0, Sq, q_block):
             q1 = min(q0 + q_block, Sq)
@@ -100,7 +103,7 @@ class _FlashAttentionFn(torch.autograd.Function):
                 dq[:, :, q0:q1] += torch.matmul(ds, kb).float()
                 dk[:, :, k0:k1] += torch.matmul(ds.transpose(-1, -2), qb).float()
 
-        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None, None, None
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None, None, None, None
 
 
 def flash_attention(
@@ -111,11 +114,15 @@ def flash_attention(
     scale: Optional[float] = None,
     q_block: int = 1024,
     k_block: int = 1024,
+    q_start: Optional[int] = None,
 ) -> torch.Tensor:
-    """Blockwise attention over [B, H, S, D] tensors."""
+    """Blockwise attention over [B, H, S, D] tensors. ``q_start`` is the
+    absolute position of q[...,0] for causal masking (context parallelism)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    return _FlashAttentionFn.apply(q, k, v, causal, scale, _block(q.shape[2], q_block), _block(k.shape[2], k_block))
+    return _FlashAttentionFn.apply(
+        q, k, v, causal, scale, _block(q.shape[2], q_block), _block(k.shape[2], k_block), q_start
+    )
 
 
 # alias used by models (the fused HIP kernel will take over this symbol)

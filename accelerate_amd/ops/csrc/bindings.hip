@@ -250,7 +250,7 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   const int d = (int)x.size(-1);
   TORCH_CHECK(d <= 2048, "layernorm_bwd: fused path supports inner dim <= 2048");
   const int64_t rows = x.numel() / d;
-  const int n_blocks = (int)std::min<int64_t>(rows, 2048);
+  const int n_blocks = (int)std::min<int64_t>(rows, 512);
   auto dx = at::empty_like(x);
   auto dw_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
   auto db_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
@@ -283,7 +283,7 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, a
   const int d = (int)x.size(-1);
   TORCH_CHECK(d <= 8192, "rmsnorm_bwd: fused path supports inner dim <= 8192");
   const int64_t rows = x.numel() / d;
-  const int n_blocks = (int)std::min<int64_t>(rows, 2048);
+  const int n_blocks = (int)std::min<int64_t>(rows, 512);
   auto dx = at::empty_like(x);
   auto dw_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
   auto dw = at::empty({d}, x.options());

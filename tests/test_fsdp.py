@@ -79,3 +79,24 @@ def test_hsdp_oracle_4proc():
     script = Path(__file__).parent / "distributed_scripts" / "hsdp_script.py"
     out = launch_distributed(script, nproc=4, timeout=240)
     assert "HSDP_PARITY_PASS" in out
+
+
+def test_tp_oracle_2proc():
+    script = Path(__file__).parent / "distributed_scripts" / "tp_script.py"
+    out = launch_distributed(script, nproc=2, timeout=240)
+    assert "TP_MLP_PASS" in out
+    assert "TP_LLAMA_PASS" in out
+
+
+def test_cp_oracle_2proc():
+    script = Path(__file__).parent / "distributed_scripts" / "cp_script.py"
+    out = launch_distributed(script, nproc=2, timeout=240)
+    assert "CP_ATTN_PASS" in out
+    assert "CP_LLAMA_PASS" in out
+
+
+def test_sp_oracle_2proc():
+    script = Path(__file__).parent / "distributed_scripts" / "sp_script.py"
+    out = launch_distributed(script, nproc=2, timeout=240)
+    assert "SP_ATTN_PASS" in out
+    assert "SP_LLAMA_PASS" in out
