@@ -32,14 +32,17 @@ __device__ __forceinline__ ushort gf2bf(float f) {
 
 }  // namespace
 
-// block = 256 threads (4 waves as 2×2), tile 128(M)×128(N), BK=64
+// block = 256 threads (4 waves as 2×2), tile 128(M)×128(N), BK=64.
+// Double-buffered K-loop with counted prefetch (the CDNA4 guide's minimum
+// 2-phase recipe, §5.5 T3): next tile's global_load_lds issues BEFORE the
+// current tile's ds_read+MFMA, one vmcnt(0)+barrier per tile.
 __global__ __launch_bounds__(256) void gemm_bt_bf16_kernel(
     const gbf16* __restrict__ A, const gbf16* __restrict__ B,
     const ushort* __restrict__ bias, ushort* __restrict__ C,
     int M, int N, int K) {
   constexpr int BM = 128, BN = 128, BK = 64;
-  __shared__ gbf16 As[BM * BK];
-  __shared__ gbf16 Bs[BN * BK];
+  __shared__ gbf16 As[2][BM * BK];
+  __shared__ gbf16 Bs[2][BN * BK];
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;          // 0..3
@@ -72,46 +75,60 @@ __global__ __launch_bounds__(256) void gemm_bt_bf16_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4g{};
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // LDS write address is HARDWARE-fixed at (wave-uniform base + lane*16 B)
-    // (guide §5 caveat): with lane = rr*8+cc that lands exactly row-major
-    // [8 rows][64 cols] from the base — the global address carries the
-    // per-lane row/col.
+  // LDS write address is HARDWARE-fixed at (wave-uniform base + lane*16 B)
+  // (guide §5 caveat): with lane = rr*8+cc that lands exactly row-major
+  // [8 rows][64 cols] from the base — the global address carries the
+  // per-lane row/col.
+  auto stage = [&](int buf, int k0) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = r * 32 + wave * 8 + s_row_in_wave;  // 0..127
       const gbf16* ga = A + (int64_t)(tile_m + row) * K + k0 + s_col;
       const gbf16* gb = B + (int64_t)(tile_n + row) * K + k0 + s_col;
-      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)ga,
-                                       (__attribute__((address_space(3))) unsigned int*)(As + (r * 32 + wave * 8) * BK),
-                                       16, 0, 0);
-      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)gb,
-                                       (__attribute__((address_space(3))) unsigned int*)(Bs + (r * 32 + wave * 8) * BK),
-                                       16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)ga,
+          (__attribute__((address_space(3))) unsigned int*)(&As[buf][(r * 32 + wave * 8) * BK]), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gb,
+          (__attribute__((address_space(3))) unsigned int*)(&Bs[buf][(r * 32 + wave * 8) * BK]), 16, 0, 0);
     }
-    asm volatile("s_waitcnt vmcnt(0)");
-    __syncthreads();
+  };
 
+  auto compute = [&](int buf) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {  // two K=32 sub-steps
-      // fragment k-slot: this lane reads 8 contiguous bf16 at
-      // k = kk*32 + (lane>>4)*8; fragments hoisted once (8 ds_read_b128
-      // per sub-step, not 20) then 16 MFMAs on registers
       const int kslot = kk * 32 + (lane >> 4) * 8;
       bf16x8g af[4], bf[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
-        af[i] = *reinterpret_cast<const bf16x8g*>(As + (wm * 64 + i * 16 + (lane & 15)) * BK + kslot);
-        bf[i] = *reinterpret_cast<const bf16x8g*>(Bs + (wn * 64 + i * 16 + (lane & 15)) * BK + kslot);
+        af[i] = *reinterpret_cast<const bf16x8g*>(&As[buf][(wm * 64 + i * 16 + (lane & 15)) * BK + kslot]);
+        bf[i] = *reinterpret_cast<const bf16x8g*>(&Bs[buf][(wn * 64 + i * 16 + (lane & 15)) * BK + kslot]);
       }
+      asm volatile("s_waitcnt lgkmcnt(0)");
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[mi], bf[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
+  };
+
+  const int nt = K / BK;
+  int cur = 0;
+  // prologue: stage tile 0 and drain it
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __syncthreads();
+  for (int t = 0; t < nt - 1; ++t) {
+    stage(cur ^ 1, (t + 1) * BK);  // issue next tile FIRST (stays in flight)
+    compute(cur);
+    asm volatile("s_waitcnt vmcnt(0)");
     __syncthreads();
+    cur ^= 1;
   }
+  compute(cur);  // epilogue tile (no prefetch)
 
   // epilogue: C[row][col], row = (lane>>4)*4 + j within fragment
 #pragma unroll
