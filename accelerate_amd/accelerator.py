@@ -13,22 +13,18 @@
 
 import contextlib
 import functools
-import json
-import math
 import os
 import shutil
-import sys
 import warnings
 from collections import OrderedDict
 from contextlib import contextmanager
-from types import MethodType
 from typing import Any, Callable, List, Optional, Union
 
 import torch
 import torch.utils.hooks as hooks
 
 from .checkpointing import load_accelerator_state, load_custom_state, save_accelerator_state, save_custom_state
-from .data_loader import DataLoaderDispatcher, DataLoaderShard, prepare_data_loader, skip_first_batches
+from .data_loader import DataLoaderDispatcher, prepare_data_loader
 from .logging import get_logger
 from .optimizer import AcceleratedOptimizer
 from .parallel.ddp import DistributedDataParallelEngine

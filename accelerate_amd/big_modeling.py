@@ -21,17 +21,13 @@ from .hooks import (
     add_hook_to_module,
     attach_align_device_hook,
     attach_align_device_hook_on_blocks,
-    remove_hook_from_submodules,
 )
 from .logging import get_logger
 from .utils.imports import is_hip_available
 from .utils.modeling import (
     check_device_map,
-    clean_device_map,
-    compute_module_sizes,
     find_tied_parameters,
     get_balanced_memory,
-    get_max_memory,
     infer_auto_device_map,
     load_checkpoint_in_model,
     retie_parameters,

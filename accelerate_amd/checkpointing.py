@@ -21,11 +21,9 @@ from .utils.constants import (
     OPTIMIZER_NAME,
     RNG_STATE_NAME,
     SAFE_MODEL_NAME,
-    SAFE_WEIGHTS_NAME,
     SAMPLER_NAME,
     SCALER_NAME,
     SCHEDULER_NAME,
-    WEIGHTS_NAME,
 )
 from .utils.imports import is_safetensors_available
 

@@ -4,8 +4,6 @@ RAM); works offline for local checkpoints/configs and for transformers
 models when their config is cached locally.
 """
 
-import argparse
-
 from ..big_modeling import init_empty_weights
 from ..utils.modeling import compute_module_sizes
 from ..utils.other import convert_bytes

@@ -1,9 +1,6 @@
 """`accelerate-amd merge-weights` — merge sharded FSDP checkpoints into a
 single full state dict (reference: commands/merge.py, fsdp_utils.py:462)."""
 
-import argparse
-import os
-from pathlib import Path
 
 
 def merge_command(args):

@@ -1,7 +1,6 @@
 """`accelerate-amd test` — launch the bundled sanity script through the
 launcher (reference: commands/test.py)."""
 
-import os
 from pathlib import Path
 
 

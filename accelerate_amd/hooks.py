@@ -13,11 +13,7 @@ from typing import Dict, List, Mapping, Optional, Union
 import torch
 import torch.nn as nn
 
-from .utils.modeling import (
-    find_tied_parameters,
-    named_module_tensors,
-    set_module_tensor_to_device,
-)
+from .utils.modeling import named_module_tensors, set_module_tensor_to_device
 from .utils.offload import PrefixedDataset
 from .utils.operations import send_to_device
 

@@ -12,7 +12,6 @@ the last rank concatenates outputs (optionally broadcast back to all ranks,
 reference: inference.py:120-123).
 """
 
-import math
 from typing import List, Optional
 
 import torch
@@ -20,7 +19,6 @@ import torch.distributed as dist
 import torch.nn as nn
 
 from .state import PartialState
-from .utils.operations import broadcast, gather_tensor_shape
 
 
 def _sequential_blocks(model: nn.Module) -> List[nn.Module]:

@@ -12,9 +12,8 @@ import tempfile
 
 import torch
 
-from .state import AcceleratorState, PartialState
+from .state import AcceleratorState
 from .utils.environment import patch_environment
-from .utils.other import get_free_port
 
 
 class PrepareForLaunch:

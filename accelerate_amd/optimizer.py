@@ -8,7 +8,6 @@ works.
 """
 
 import inspect
-import warnings
 
 import torch
 

@@ -14,11 +14,8 @@ The ring (P2P KV rotation) variant maps to RCCL send/recv over the 7 xGMI
 links and is the planned v2 upgrade of `_AllGatherSeq`.
 """
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
-import torch.nn as nn
 
 
 class _AllGatherSeq(torch.autograd.Function):
