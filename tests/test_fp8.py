@@ -132,3 +132,35 @@ def test_fp8_cast_transpose_kernel():
     assert torch.equal(out.view(torch.uint8), ref.view(torch.uint8))
     assert torch.equal(out_t.view(torch.uint8), ref.t().contiguous().view(torch.uint8))
     assert torch.allclose(amax, x.float().abs().max().reshape(1), rtol=1e-3)
+
+
+@gpu
+def test_accelerator_fp8_end_to_end():
+    """mixed_precision='fp8' through Accelerator.prepare: linears converted,
+    loss decreases, grads finite."""
+    import torch.nn as nn
+
+    from accelerate_amd import Accelerator
+    from accelerate_amd.ops.fp8 import FP8Linear
+
+    torch.manual_seed(0)
+    acc = Accelerator(mixed_precision="fp8")
+    model = nn.Sequential(
+        nn.Linear(128, 256), nn.GELU(), nn.Linear(256, 256), nn.GELU(), nn.Linear(256, 128)
+    ).to(torch.bfloat16)
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    model, opt = acc.prepare(model, opt)
+    inner = acc.unwrap_model(model, keep_fp32_wrapper=False)
+    assert any(isinstance(m, FP8Linear) for m in inner.modules()), "fp8 conversion did not run"
+    x = torch.randn(256, 128, device="cuda", dtype=torch.bfloat16)
+    y = torch.randn(256, 128, device="cuda", dtype=torch.bfloat16)
+    losses = []
+    for _ in range(10):
+        opt.zero_grad()
+        loss = ((model(x) - y) ** 2).float().mean()
+        acc.backward(loss)
+        opt.step()
+        losses.append(float(loss))
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0]

@@ -172,3 +172,32 @@ def test_load_checkpoint_in_model_disk_offload():
         assert fresh.linear2.weight.device == torch.device("meta")
         assert os.path.isfile(os.path.join(offdir, "linear2.weight.dat"))
         assert os.path.isfile(os.path.join(offdir, "index.json"))
+
+
+def test_llama70b_device_map_plans_on_8x288gb():
+    """BASELINE config #4: Llama-3-70B must fit across 8×288 GB HBM3E with
+    no CPU/disk spill (meta-init, no weights materialized)."""
+    from accelerate_amd import init_empty_weights
+    from accelerate_amd.models.llama import LlamaConfig, LlamaForCausalLM
+
+    with init_empty_weights():
+        model = LlamaForCausalLM(LlamaConfig.llama3_70b())
+    n_params = sum(p.numel() for p in model.parameters())
+    assert 69e9 < n_params < 72e9, n_params
+
+    budget = int(288e9 * 0.95)
+    max_memory = {i: budget for i in range(8)}
+    max_memory["cpu"] = int(100e9)
+    dmap = infer_auto_device_map(
+        model, max_memory=max_memory, no_split_module_classes=["LlamaDecoderLayer"], dtype=torch.bfloat16
+    )
+    devices = set(dmap.values())
+    assert "disk" not in devices and "cpu" not in devices, f"70B spilled: {devices}"
+    assert devices.issubset(set(range(8)))
+
+    # single-GPU: 70B bf16 (~140 GB) fits on ONE MI355X for inference
+    dmap1 = infer_auto_device_map(
+        model, max_memory={0: budget, "cpu": int(100e9)}, no_split_module_classes=["LlamaDecoderLayer"],
+        dtype=torch.bfloat16,
+    )
+    assert set(dmap1.values()) == {0}, f"70B bf16 should fit one 288GB GPU: {set(dmap1.values())}"
