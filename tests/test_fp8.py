@@ -155,7 +155,7 @@ def test_accelerator_fp8_end_to_end():
     x = torch.randn(256, 128, device="cuda", dtype=torch.bfloat16)
     y = torch.randn(256, 128, device="cuda", dtype=torch.bfloat16)
     losses = []
-    for _ in range(10):
+    for _ in range(30):
         opt.zero_grad()
         loss = ((model(x) - y) ** 2).float().mean()
         acc.backward(loss)
@@ -163,4 +163,6 @@ def test_accelerator_fp8_end_to_end():
         losses.append(float(loss))
     torch.cuda.synchronize()
     assert all(torch.isfinite(torch.tensor(losses)))
-    assert losses[-1] < losses[0]
+    # quality is pinned by test_fp8_training_tracks_bf16; here we assert the
+    # Accelerator wiring trains stably (monotone-ish, no divergence)
+    assert losses[-1] < losses[0], (losses[0], losses[-1])
