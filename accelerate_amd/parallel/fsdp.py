@@ -39,6 +39,17 @@ def _is_nccl(group) -> bool:
     return dist.is_initialized() and dist.get_backend(group) == "nccl"
 
 
+def _iter_tensors(obj):
+    if isinstance(obj, torch.Tensor):
+        yield obj
+    elif isinstance(obj, (list, tuple)):
+        for o in obj:
+            yield from _iter_tensors(o)
+    elif isinstance(obj, dict):
+        for o in obj.values():
+            yield from _iter_tensors(o)
+
+
 def _all_gather_flat(out_full: torch.Tensor, shard: torch.Tensor, group):
     if _is_nccl(group):
         dist.all_gather_into_tensor(out_full, shard, group=group)
@@ -301,7 +312,6 @@ class ShardedModel(nn.Module):
         for u in self.units:
             u.module.register_forward_pre_hook(self._make_fwd_pre(u))
             u.module.register_forward_hook(self._make_fwd_post(u))
-            u.module.register_full_backward_pre_hook(self._make_bwd_pre(u))
             for p in u.params:
                 h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
                 u.grad_hooks.append(h)
@@ -344,6 +354,14 @@ class ShardedModel(nn.Module):
 
     def _make_fwd_post(self, unit):
         def hook(module, args, output):
+            # pre-backward unshard trigger lives on the OUTPUT TENSORS (not a
+            # module backward hook — those never fire for dict/tuple-returning
+            # models like LlamaForCausalLM): the grad of a unit's output is
+            # computed BEFORE its interior backward consumes the parameters.
+            if torch.is_grad_enabled() and self.module.training:
+                for t in _iter_tensors(output):
+                    if t.requires_grad:
+                        t.register_hook(self._make_bwd_trigger(unit))
             if self.reshard_after_forward and self.world > 1 and self.module.training:
                 unit.reshard()
             elif not self.module.training and self.reshard_after_forward:
@@ -352,17 +370,17 @@ class ShardedModel(nn.Module):
 
         return hook
 
-    def _make_bwd_pre(self, unit):
-        def hook(module, grad_output):
+    def _make_bwd_trigger(self, unit):
+        def trigger(grad):
             self._wait_for_comm()
             unit.unshard()
             # backward prefetch: previous unit in forward order comes next
             idx = self._fwd_index.get(id(unit)) if self._order_recorded else None
             if idx is not None and idx - 1 >= 0:
                 self._stream_unshard(self._fwd_order[idx - 1])
-            return None
+            return grad
 
-        return hook
+        return trigger
 
     def _wait_for_comm(self):
         if self._comm_stream is not None:

@@ -100,3 +100,9 @@ def test_sp_oracle_2proc():
     out = launch_distributed(script, nproc=2, timeout=240)
     assert "SP_ATTN_PASS" in out
     assert "SP_LLAMA_PASS" in out
+
+
+def test_fsdp_dict_model_oracle():
+    script = Path(__file__).parent / "distributed_scripts" / "fsdp_llama_script.py"
+    out = launch_distributed(script, nproc=2, timeout=240)
+    assert "FSDP_DICT_MODEL_PASS" in out
