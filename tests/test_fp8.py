@@ -148,12 +148,14 @@ def test_accelerator_fp8_end_to_end():
     model = nn.Sequential(
         nn.Linear(128, 256), nn.GELU(), nn.Linear(256, 256), nn.GELU(), nn.Linear(256, 128)
     ).to(torch.bfloat16)
-    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
     model, opt = acc.prepare(model, opt)
     inner = acc.unwrap_model(model, keep_fp32_wrapper=False)
     assert any(isinstance(m, FP8Linear) for m in inner.modules()), "fp8 conversion did not run"
     x = torch.randn(256, 128, device="cuda", dtype=torch.bfloat16)
-    y = torch.randn(256, 128, device="cuda", dtype=torch.bfloat16)
+    # learnable target (a fixed linear map of x): the loss is clearly reducible
+    w_true = torch.randn(128, 128, device="cuda", dtype=torch.bfloat16) * 0.2
+    y = x @ w_true
     losses = []
     for _ in range(30):
         opt.zero_grad()
@@ -164,5 +166,5 @@ def test_accelerator_fp8_end_to_end():
     torch.cuda.synchronize()
     assert all(torch.isfinite(torch.tensor(losses)))
     # quality is pinned by test_fp8_training_tracks_bf16; here we assert the
-    # Accelerator wiring trains stably (monotone-ish, no divergence)
-    assert losses[-1] < losses[0], (losses[0], losses[-1])
+    # Accelerator wiring trains (loss visibly reduced on a learnable target)
+    assert losses[-1] < losses[0] * 0.9, (losses[0], losses[-1])
