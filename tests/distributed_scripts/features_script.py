@@ -88,6 +88,31 @@ def test_dispatcher(acc):
         print("DISPATCHER_PASS")
 
 
+def test_local_sgd(acc):
+    from accelerate_amd import LocalSGD
+
+    set_seed(0)
+    model = nn.Linear(8, 2)
+    model = acc.prepare_model(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    n, r = acc.num_processes, acc.process_index
+    X = torch.randn(16, 8, generator=torch.Generator().manual_seed(11))
+    with LocalSGD(accelerator=acc, model=model, local_sgd_steps=2) as lsgd:
+        for step in range(4):
+            opt.zero_grad()
+            loss = model(X[r::n]).pow(2).mean()
+            loss.backward()
+            opt.step()
+            lsgd.step()
+    # after the window closes, replicas must hold identical (averaged) params
+    w = acc.unwrap_model(model).weight.detach()
+    ws = [torch.empty_like(w) for _ in range(n)]
+    dist.all_gather(ws, w)
+    assert torch.allclose(ws[0], ws[1], atol=1e-6), "LocalSGD must average params across ranks"
+    if acc.is_main_process:
+        print("LOCALSGD_PASS")
+
+
 def main():
     acc = Accelerator(cpu=True)
     assert acc.num_processes == 2
@@ -95,6 +120,7 @@ def main():
     test_debug_mode(acc)
     test_comm_dtype_bf16(acc)
     test_dispatcher(acc)
+    test_local_sgd(acc)
     acc.end_training()
 
 

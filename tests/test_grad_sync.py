@@ -19,5 +19,5 @@ def test_distributed_grad_sync_oracle():
 def test_distributed_features_oracle():
     script = Path(__file__).parent / "distributed_scripts" / "features_script.py"
     out = launch_distributed(script, nproc=2)
-    for marker in ("METRICS_DEDUP_PASS", "DEBUG_MODE_PASS", "COMM_DTYPE_PASS", "DISPATCHER_PASS"):
+    for marker in ("METRICS_DEDUP_PASS", "DEBUG_MODE_PASS", "COMM_DTYPE_PASS", "DISPATCHER_PASS", "LOCALSGD_PASS"):
         assert marker in out, f"missing {marker}\n{out}"
