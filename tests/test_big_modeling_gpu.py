@@ -75,3 +75,35 @@ def test_fp16_scaler_end_to_end():
     torch.cuda.synchronize()
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]
+
+
+@gpu
+def test_dispatch_memory_discipline():
+    """BASELINE target: peak GPU allocation ≈ the GPU-assigned shard size
+    (no over-allocation) when dispatching with CPU offload."""
+    from accelerate_amd import dispatch_model
+    from accelerate_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from accelerate_amd.utils.modeling import compute_module_sizes
+
+    torch.manual_seed(0)
+    torch.cuda.empty_cache()
+    torch.cuda.reset_peak_memory_stats()
+    model = LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=8)).eval()
+    sizes = compute_module_sizes(model)
+    # half the layers on GPU, the rest offloaded to CPU
+    dmap = {"embed_tokens": 0, "norm": 0, "lm_head": 0}
+    for i in range(8):
+        dmap[f"layers.{i}"] = 0 if i < 4 else "cpu"
+    gpu_bytes = sum(
+        sizes[name] for name, dev in dmap.items() if dev == 0
+    )
+    dispatch_model(model, dmap)
+    ids = torch.randint(0, 1024, (1, 16), device=0)
+    with torch.no_grad():
+        model(ids)
+    peak = torch.cuda.max_memory_allocated()
+    # assigned shard + offloaded-layer onload working set + small activations;
+    # the contract is NO duplicate residency of the whole model
+    assert peak < gpu_bytes + sizes["layers.4"] * 3 + 64 * 2**20, (
+        f"peak {peak/2**20:.1f} MiB vs shard {gpu_bytes/2**20:.1f} MiB"
+    )
