@@ -14,8 +14,13 @@
 
 template <bool kE5M2, bool kWord>
 __device__ __forceinline__ unsigned int cvt_pk2(float lo, float hi, unsigned int old) {
-  // hardware packed fp32x2 -> fp8x2 convert (v_cvt_pk_fp8/bf8_f32), RNE+sat;
-  // the word-select operand must be an immediate
+  // hardware packed fp32x2 -> fp8x2 convert (v_cvt_pk_fp8/bf8_f32). The
+  // convert does NOT saturate and OCP e4m3fn has no inf encoding — overflow
+  // becomes NaN — so clamp to the format max first (delayed scaling can
+  // transiently under-scale between amax refreshes).
+  constexpr float kMax = kE5M2 ? 57344.f : 448.f;
+  lo = fminf(fmaxf(lo, -kMax), kMax);
+  hi = fminf(fmaxf(hi, -kMax), kMax);
   if constexpr (kE5M2) return __builtin_amdgcn_cvt_pk_bf8_f32(lo, hi, old, kWord);
   else return __builtin_amdgcn_cvt_pk_fp8_f32(lo, hi, old, kWord);
 }
