@@ -169,10 +169,6 @@ class Accelerator:
         self._load_model_state_pre_hook = OrderedDict()
         self.step = 0
         self.flag_tensor = None
-        self._dataloader_split_batches = split_batches
-
-        # hipGraph capture support (opt-in via capture_step)
-        self._captured_graphs = {}
 
     # ------------------------------------------------------------------
     # properties

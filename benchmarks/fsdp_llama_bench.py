@@ -26,7 +26,8 @@ def main():
     p.add_argument("--batch", type=int, default=1)
     p.add_argument("--steps", type=int, default=5)
     p.add_argument("--warmup", type=int, default=2)
-    p.add_argument("--ac", action="store_true", default=True, help="activation checkpointing")
+    p.add_argument("--no-ac", dest="ac", action="store_false", help="disable activation checkpointing")
+    p.set_defaults(ac=True)
     args = p.parse_args()
 
     os.environ.setdefault("ACCELERATE_USE_FSDP", "1")
