@@ -399,6 +399,64 @@ void fp8_update_scale_fn(at::Tensor history, double fp8_max, double margin_pow2,
                      (float)margin_pow2, scale.data_ptr<float>(), scale_inv.data_ptr<float>());
 }
 
+// weight-only quantization kernels (quant_kernels.hip)
+__global__ void int8_dequant_kernel(const char*, const float*, ushort*, int64_t, int64_t);
+__global__ void int4_dequant_kernel(const unsigned char*, const float*, ushort*, int64_t, int64_t, int);
+__global__ void w8a16_gemv_kernel(const char*, const float*, const ushort*, const ushort*,
+                                  ushort*, int64_t, int64_t, int);
+
+at::Tensor int8_dequant(at::Tensor q, at::Tensor scale) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.scalar_type() == at::kChar && q.dim() == 2,
+              "int8_dequant: 2-D contiguous int8 on GPU");
+  const int64_t rows = q.size(0), cols = q.size(1);
+  TORCH_CHECK(cols % 16 == 0, "int8_dequant: cols must be a multiple of 16");
+  TORCH_CHECK(scale.numel() == rows && scale.scalar_type() == at::kFloat, "int8_dequant: bad scale");
+  auto out = at::empty({rows, cols}, q.options().dtype(at::kBFloat16));
+  int grid = (int)std::min<int64_t>((rows * cols / 16 + kBlockThreads - 1) / kBlockThreads, 2048);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(int8_dequant_kernel, dim3(grid), dim3(kBlockThreads), 0, stream.stream(),
+                     reinterpret_cast<const char*>(q.data_ptr()), scale.data_ptr<float>(),
+                     reinterpret_cast<ushort*>(out.data_ptr()), rows, cols);
+  return out;
+}
+
+at::Tensor int4_dequant(at::Tensor q, at::Tensor scale, int64_t cols, int64_t group) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.scalar_type() == at::kByte && q.dim() == 2,
+              "int4_dequant: 2-D contiguous uint8 (packed nibbles) on GPU");
+  const int64_t rows = q.size(0);
+  TORCH_CHECK(q.size(1) * 2 == cols, "int4_dequant: packed width mismatch");
+  TORCH_CHECK(cols % 16 == 0 && group % 16 == 0 && cols % group == 0, "int4_dequant: bad group");
+  TORCH_CHECK(scale.numel() == rows * (cols / group) && scale.scalar_type() == at::kFloat,
+              "int4_dequant: bad scale");
+  auto out = at::empty({rows, cols}, q.options().dtype(at::kBFloat16));
+  int grid = (int)std::min<int64_t>((rows * cols / 16 + kBlockThreads - 1) / kBlockThreads, 2048);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(int4_dequant_kernel, dim3(grid), dim3(kBlockThreads), 0, stream.stream(),
+                     reinterpret_cast<const unsigned char*>(q.data_ptr()), scale.data_ptr<float>(),
+                     reinterpret_cast<ushort*>(out.data_ptr()), rows, cols, (int)group);
+  return out;
+}
+
+at::Tensor w8a16_gemv(at::Tensor q, at::Tensor scale, at::Tensor x, c10::optional<at::Tensor> bias) {
+  // y[b, rows] = x[b, cols] @ q^T * scale (decode-shaped: batch <= 8)
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.scalar_type() == at::kChar, "w8a16_gemv: int8 weight");
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == at::kBFloat16, "w8a16_gemv: bf16 x");
+  const int64_t rows = q.size(0), cols = q.size(1);
+  TORCH_CHECK(x.size(-1) == cols && cols % 1024 == 0, "w8a16_gemv: cols must be a multiple of 1024");
+  const int batch = (int)(x.numel() / cols);
+  TORCH_CHECK(batch <= 8, "w8a16_gemv: decode path only (batch <= 8)");
+  auto sizes = x.sizes().vec();
+  sizes.back() = rows;
+  auto y = at::empty(sizes, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(w8a16_gemv_kernel, dim3((rows + 3) / 4), dim3(256), 0, stream.stream(),
+                     reinterpret_cast<const char*>(q.data_ptr()), scale.data_ptr<float>(),
+                     reinterpret_cast<const ushort*>(x.data_ptr()),
+                     bias.has_value() ? reinterpret_cast<const ushort*>(bias->data_ptr()) : nullptr,
+                     reinterpret_cast<ushort*>(y.data_ptr()), rows, cols, batch);
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "fused bf16 LayerNorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
@@ -410,6 +468,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_cast_transpose", &fp8_cast_transpose,
         "bf16 -> fp8 + fp8^T with fused amax (LDS-tiled, gfx950)");
   m.def("fp8_update_scale", &fp8_update_scale_fn, "delayed-scaling scale update");
+  m.def("int8_dequant", &int8_dequant, "int8 weight -> bf16 (per-channel scale)");
+  m.def("int4_dequant", &int4_dequant, "packed int4 weight -> bf16 (group-wise scale)");
+  m.def("w8a16_gemv", &w8a16_gemv, "fused int8-weight x bf16-activation matvec (decode)");
   m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (gfx950)");
   m.def("fused_adamw_planned", &fused_adamw_planned,
         "graph-capturable fused AdamW over a cached device plan",
