@@ -25,18 +25,41 @@ def _block(size, pref):
     return min(size, pref)
 
 
+def _fused_eligible(q, k) -> bool:
+    """The CDNA4 kernel covers bf16, head_dim 64/128, and Sq big enough that
+    a 128-row tile isn't pure padding. Everything else takes the torch path."""
+    return (
+        q.is_cuda
+        and q.dtype == torch.bfloat16
+        and q.shape[-1] in (64, 128)
+        and q.shape[2] >= 32
+        and q.shape[0] * q.shape[1] <= 65535
+    )
+
+
 class _FlashAttentionFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale, q_block, k_block, q_start=None):
         # q,k,v: [B, H, S, D] (kv may have S_k != S_q for cache decode)
         B, H, Sq, Dh = q.shape
         Sk = k.shape[2]
-        out = torch.empty_like(q)
-        lse = torch.empty(B, H, Sq, dtype=torch.float32, device=q.device)
         # causal offset: query i attends keys <= past + i. Default assumes the
         # queries are the LAST Sq positions (kv-cache decode); context
         # parallelism passes the rank's absolute start explicitly.
         past = (Sk - Sq) if q_start is None else q_start
+
+        if _fused_eligible(q, k):
+            from . import _load_extension
+
+            ext = _load_extension(required=True)  # GPU boxes must run native
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+            out, lse = ext.flash_attn_fwd(q, k, v, causal, scale, past)
+            ctx.save_for_backward(q, k, v, out, lse)
+            ctx.meta = (causal, scale, q_block, k_block, past)
+            return out
+
+        out = torch.empty_like(q)
+        lse = torch.empty(B, H, Sq, dtype=torch.float32, device=q.device)
 
         for q0 in range(0, Sq, q_block):
             q1 = min(q0 + q_block, Sq)

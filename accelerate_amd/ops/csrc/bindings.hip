@@ -399,6 +399,33 @@ void fp8_update_scale_fn(at::Tensor history, double fp8_max, double margin_pow2,
                      (float)margin_pow2, scale.data_ptr<float>(), scale_inv.data_ptr<float>());
 }
 
+// fused flash-attention forward (attention_kernels.hip)
+hipError_t launch_fa_fwd(const void*, const void*, const void*, void*, float*,
+                         int64_t, int, int, int, int, int, float, hipStream_t);
+
+std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                       bool causal, double scale, int64_t past) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.scalar_type() == at::kBFloat16 && q.dim() == 4,
+              "flash_attn_fwd: q must be 4-D contiguous bf16 [B,H,Sq,D]");
+  TORCH_CHECK(k.is_contiguous() && v.is_contiguous() && k.scalar_type() == at::kBFloat16 &&
+                  v.scalar_type() == at::kBFloat16,
+              "flash_attn_fwd: k/v must be contiguous bf16");
+  const int64_t B = q.size(0), H = q.size(1), Sq = q.size(2), D = q.size(3);
+  const int64_t Sk = k.size(2);
+  TORCH_CHECK(k.size(0) == B && k.size(1) == H && v.sizes() == k.sizes(),
+              "flash_attn_fwd: expand GQA kv heads before the kernel");
+  TORCH_CHECK(D == 64 || D == 128, "flash_attn_fwd: head_dim must be 64 or 128");
+  TORCH_CHECK(B * H <= 65535, "flash_attn_fwd: grid.y overflow");
+  auto out = at::empty_like(q);
+  auto lse = at::empty({B, H, Sq}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipError_t e = launch_fa_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+                               lse.data_ptr<float>(), B * H, (int)Sq, (int)Sk, (int)D,
+                               (int)past, causal ? 1 : 0, (float)scale, stream.stream());
+  TORCH_CHECK(e == hipSuccess, "flash_attn_fwd launch failed: ", hipGetErrorString(e));
+  return {out, lse};
+}
+
 // weight-only quantization kernels (quant_kernels.hip)
 __global__ void int8_dequant_kernel(const char*, const float*, ushort*, int64_t, int64_t);
 __global__ void int4_dequant_kernel(const unsigned char*, const float*, ushort*, int64_t, int64_t, int);
@@ -468,6 +495,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_cast_transpose", &fp8_cast_transpose,
         "bf16 -> fp8 + fp8^T with fused amax (LDS-tiled, gfx950)");
   m.def("fp8_update_scale", &fp8_update_scale_fn, "delayed-scaling scale update");
+  m.def("flash_attn_fwd", &flash_attn_fwd,
+        "fused flash-attention forward (bf16, head_dim 64/128) -> (out, lse)");
   m.def("int8_dequant", &int8_dequant, "int8 weight -> bf16 (per-channel scale)");
   m.def("int4_dequant", &int4_dequant, "packed int4 weight -> bf16 (group-wise scale)");
   m.def("w8a16_gemv", &w8a16_gemv, "fused int8-weight x bf16-activation matvec (decode)");

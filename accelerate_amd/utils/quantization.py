@@ -109,9 +109,21 @@ def load_and_quantize_model(
     from ..big_modeling import dispatch_model
     from .modeling import infer_auto_device_map, load_checkpoint_in_model
 
+    from ..ops.quant import QuantLinear
+
     if weights_location is not None:
         load_checkpoint_in_model(model, weights_location)
     model = replace_with_quantized_layers(model, quantization_config)
+    # the rest of the model (embeddings, norms, head) computes in
+    # compute_dtype so QuantLinear outputs flow without dtype mismatches
+    model.to(quantization_config.compute_dtype)
+    for mod in model.modules():
+        if isinstance(mod, QuantLinear):
+            mod.scales.data = mod.scales.data.float()  # .to() downcast undone
+    for name in quantization_config.keep_in_fp32_modules:
+        for mod_name, mod in model.named_modules():
+            if mod_name == name or mod_name.endswith("." + name):
+                mod.float()
     model.eval()
     for p in model.parameters():
         p.requires_grad_(False)

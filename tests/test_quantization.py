@@ -125,9 +125,10 @@ def test_int4_dequant_kernel_vs_reference():
     torch.manual_seed(0)
     w = torch.randn(256, 2048, device="cuda")
     q, s = quantize_int4(w, group_size=128)
-    out = dequantize_int4(q, s, group_size=128)  # HIP kernel
+    out = dequantize_int4(q, s, group_size=128)  # HIP kernel (bf16 out)
     ref = dequantize_int4(q.cpu(), s.cpu(), group_size=128, dtype=torch.float32)
-    assert (out.float().cpu() - ref).abs().max() < 1e-2
+    # compare at bf16 granularity — the kernel rounds to bf16 on store
+    assert torch.equal(out.cpu(), ref.to(torch.bfloat16))
 
 
 @gpu
