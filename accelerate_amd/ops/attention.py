@@ -16,6 +16,7 @@ block (the flash-attention backward), with
 """
 
 import math
+import os
 from typing import Optional
 
 import torch
@@ -96,6 +97,13 @@ class _FlashAttentionFn(torch.autograd.Function):
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
         causal, scale, q_block, k_block, past = ctx.meta
+
+        if _fused_eligible(q, k) and os.environ.get("ACCELERATE_AMD_FA_BWD", "1") == "1":
+            from . import _load_extension
+
+            ext = _load_extension(required=True)
+            dq, dk, dv = ext.flash_attn_bwd(dout, q, k, v, out, lse, causal, scale, past)
+            return dq, dk, dv, None, None, None, None, None
         B, H, Sq, Dh = q.shape
         Sk = k.shape[2]
         dq = torch.zeros_like(q, dtype=torch.float32)

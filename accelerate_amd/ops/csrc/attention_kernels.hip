@@ -239,6 +239,387 @@ template __global__ void fa_fwd_kernel<64>(const abf16*, const abf16*, const abf
 template __global__ void fa_fwd_kernel<128>(const abf16*, const abf16*, const abf16*,
                                             ushort*, float*, int, int, int, int, float);
 
+// ===========================================================================
+// Backward: logsumexp-recompute, two passes (no atomics):
+//   pass A (key-outer):  dk, dv — each block owns 128 keys, loops q-blocks
+//   pass B (query-outer): dq    — each block owns 128 queries, loops k-blocks
+// plus a Drow = rowsum(dout . out) precompute. All accumulation fp32 in
+// registers across the whole inner loop; one bf16 store per grad element.
+// Math (ops/attention.py backward):
+//   P  = exp(S*scale − L);  dv = Pᵀ dO;  dP = dO Vᵀ
+//   dS = P ⊙ (dP − Drow)·scale;  dq = dS K;  dk = dSᵀ Q
+// ===========================================================================
+
+__global__ __launch_bounds__(256) void fa_drow_kernel(
+    const ushort* __restrict__ dout, const ushort* __restrict__ out,
+    float* __restrict__ drow, int64_t n_rows, int D) {
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  for (int64_t r = (int64_t)blockIdx.x * 4 + wave; r < n_rows; r += (int64_t)gridDim.x * 4) {
+    float s = 0.f;
+    for (int d = lane; d < D; d += 64) {
+      ushort a = dout[r * D + d], b = out[r * D + d];
+      s += (float)*reinterpret_cast<__hip_bfloat16*>(&a) * (float)*reinterpret_cast<__hip_bfloat16*>(&b);
+    }
+#pragma unroll
+    for (int x = 1; x < 64; x <<= 1) s += __shfl_xor(s, x, 64);
+    if (lane == 0) drow[r] = s;
+  }
+}
+
+// pass A: dk/dv. Block owns keys [kb0, kb0+128) of one (b,h); wave w owns
+// keys [w*32, w*32+32). Inner loop: 64-query tiles. K,V rows live in
+// registers as MFMA A-fragments for the whole block; Q and dO tiles are
+// staged in LDS in BOTH layouts (row-major for B-fragments of S^T/dP^T,
+// transposed for B-fragments of dk/dv).
+template <int D>
+__global__ __launch_bounds__(256) void fa_bwd_dkdv_kernel(
+    const abf16* __restrict__ q, const abf16* __restrict__ k,
+    const abf16* __restrict__ v, const abf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    ushort* __restrict__ dk, ushort* __restrict__ dv,
+    int Sq, int Sk, int past, int causal, float scale) {
+  constexpr int BK = 128, BQ = 64;
+  constexpr int KP = D + 8, QP = BQ + 8;
+  extern __shared__ char smem[];
+  abf16* Qs = reinterpret_cast<abf16*>(smem);   // [BQ][KP]
+  abf16* QTs = Qs + BQ * KP;                    // [D][QP]
+  abf16* dOs = QTs + D * QP;                    // [BQ][KP]
+  abf16* dOTs = dOs + BQ * KP;                  // [D][QP]
+  abf16* Ws = dOTs + D * QP;                    // [BK][QP] P^T then dS^T
+  float* Ls = reinterpret_cast<float*>(Ws + BK * QP);  // [BQ] lse*log2e
+  float* Ds = Ls + BQ;                                 // [BQ] drow
+
+  const int tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+  const int k0 = blockIdx.x * BK;
+  const int64_t bh = blockIdx.y;
+  const abf16* qb = q + bh * Sq * (int64_t)D;
+  const abf16* kb = k + bh * Sk * (int64_t)D;
+  const abf16* vb = v + bh * Sk * (int64_t)D;
+  const abf16* dob = dout + bh * Sq * (int64_t)D;
+  const float* lb = lse + bh * Sq;
+  const float* db = drow + bh * Sq;
+
+  // K,V rows of this wave as A-fragments (row = lane&15 within 16-row tile)
+  bf16x8a kf[2][D / 32], vf[2][D / 32];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+    int kr = k0 + wave * 32 + mi * 16 + (lane & 15);
+    if (kr >= Sk) kr = Sk - 1;
+#pragma unroll
+    for (int kc = 0; kc < D / 32; ++kc) {
+      kf[mi][kc] = *reinterpret_cast<const bf16x8a*>(kb + (int64_t)kr * D + kc * 32 + (lane >> 4) * 8);
+      vf[mi][kc] = *reinterpret_cast<const bf16x8a*>(vb + (int64_t)kr * D + kc * 32 + (lane >> 4) * 8);
+    }
+  }
+  f32x4a dkacc[2][D / 16], dvacc[2][D / 16];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int nd = 0; nd < D / 16; ++nd) { dkacc[mi][nd] = f32x4a{}; dvacc[mi][nd] = f32x4a{}; }
+
+  const float c = scale * kLog2e;
+  const int q_lo = causal ? max(0, ((k0 - past) / BQ) * BQ) : 0;
+
+  for (int q0 = q_lo; q0 < Sq; q0 += BQ) {
+    // stage Q/dO tiles (both layouts) + L*log2e + Drow
+    {
+      constexpr int chunks = BQ * D / 8;
+#pragma unroll
+      for (int it = 0; it < chunks / 256; ++it) {
+        const int cid = tid + it * 256;
+        const int row = cid / (D / 8);
+        const int col8 = (cid % (D / 8)) * 8;
+        int64_t qr = q0 + row;
+        if (qr >= Sq) qr = Sq - 1;  // clamped; masked via p=0 below
+        bf16x8a q8 = *reinterpret_cast<const bf16x8a*>(qb + qr * D + col8);
+        *reinterpret_cast<bf16x8a*>(Qs + row * KP + col8) = q8;
+        bf16x8a d8 = *reinterpret_cast<const bf16x8a*>(dob + qr * D + col8);
+        *reinterpret_cast<bf16x8a*>(dOs + row * KP + col8) = d8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          QTs[(col8 + j) * QP + row] = q8[j];
+          dOTs[(col8 + j) * QP + row] = d8[j];
+        }
+      }
+      if (tid < BQ) {
+        int64_t qr = q0 + tid;
+        if (qr >= Sq) qr = Sq - 1;
+        Ls[tid] = lb[qr] * kLog2e;
+        Ds[tid] = db[qr];
+      }
+    }
+    __syncthreads();
+
+    // S^T = K Q^T ; dP^T = V dO^T  (both A@B^T on MFMA, contraction D)
+    f32x4a st[2][BQ / 16], dpt[2][BQ / 16];
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < BQ / 16; ++ni) { st[mi][ni] = f32x4a{}; dpt[mi][ni] = f32x4a{}; }
+#pragma unroll
+    for (int ni = 0; ni < BQ / 16; ++ni) {
+#pragma unroll
+      for (int kc = 0; kc < D / 32; ++kc) {
+        bf16x8a bq = *reinterpret_cast<const bf16x8a*>(
+            Qs + (ni * 16 + (lane & 15)) * KP + kc * 32 + (lane >> 4) * 8);
+        bf16x8a bd = *reinterpret_cast<const bf16x8a*>(
+            dOs + (ni * 16 + (lane & 15)) * KP + kc * 32 + (lane >> 4) * 8);
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          st[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[mi][kc], bq, st[mi][ni], 0, 0, 0);
+          dpt[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[mi][kc], bd, dpt[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+
+    // P^T = exp2(S^T*c − L[q]) with causal/tail mask; stage P^T (bf16)
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < BQ / 16; ++ni) {
+        const int qcol = q0 + ni * 16 + (lane & 15);
+        const float l2 = Ls[ni * 16 + (lane & 15)];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int kr = k0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
+          const bool dead = qcol >= Sq || kr >= Sk || (causal && kr > past + qcol);
+          const float p = dead ? 0.f : exp2f(st[mi][ni][j] * c - l2);
+          st[mi][ni][j] = p;
+          ushort pb = af2bf(p);
+          Ws[(wave * 32 + mi * 16 + (lane >> 4) * 4 + j) * QP + ni * 16 + (lane & 15)] =
+              *reinterpret_cast<abf16*>(&pb);
+        }
+      }
+    }
+    // dv += P^T dO (A-frags from Ws, B-frags from dO^T; contraction = q)
+#pragma unroll
+    for (int kc = 0; kc < BQ / 32; ++kc) {
+      bf16x8a pf[2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+        pf[mi] = *reinterpret_cast<const bf16x8a*>(
+            Ws + (wave * 32 + mi * 16 + (lane & 15)) * QP + kc * 32 + (lane >> 4) * 8);
+#pragma unroll
+      for (int nd = 0; nd < D / 16; ++nd) {
+        bf16x8a bf = *reinterpret_cast<const bf16x8a*>(
+            dOTs + (nd * 16 + (lane & 15)) * QP + kc * 32 + (lane >> 4) * 8);
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+          dvacc[mi][nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf[mi], bf, dvacc[mi][nd], 0, 0, 0);
+      }
+    }
+
+    // dS^T = P^T ⊙ (dP^T − Drow[q])·scale ; overwrite Ws (wave-private rows)
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < BQ / 16; ++ni) {
+        const float dr = Ds[ni * 16 + (lane & 15)];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const float ds = st[mi][ni][j] * (dpt[mi][ni][j] - dr) * scale;
+          ushort b = af2bf(ds);
+          Ws[(wave * 32 + mi * 16 + (lane >> 4) * 4 + j) * QP + ni * 16 + (lane & 15)] =
+              *reinterpret_cast<abf16*>(&b);
+        }
+      }
+    }
+    // dk += dS^T Q (B-frags from Q^T)
+#pragma unroll
+    for (int kc = 0; kc < BQ / 32; ++kc) {
+      bf16x8a sf[2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+        sf[mi] = *reinterpret_cast<const bf16x8a*>(
+            Ws + (wave * 32 + mi * 16 + (lane & 15)) * QP + kc * 32 + (lane >> 4) * 8);
+#pragma unroll
+      for (int nd = 0; nd < D / 16; ++nd) {
+        bf16x8a bf = *reinterpret_cast<const bf16x8a*>(
+            QTs + (nd * 16 + (lane & 15)) * QP + kc * 32 + (lane >> 4) * 8);
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+          dkacc[mi][nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sf[mi], bf, dkacc[mi][nd], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // next q-tile restages Qs/dOs
+  }
+
+  ushort* dkb = dk + bh * Sk * (int64_t)D;
+  ushort* dvb = dv + bh * Sk * (int64_t)D;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int kr = k0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
+      if (kr >= Sk) continue;
+#pragma unroll
+      for (int nd = 0; nd < D / 16; ++nd) {
+        dkb[(int64_t)kr * D + nd * 16 + (lane & 15)] = af2bf(dkacc[mi][nd][j]);
+        dvb[(int64_t)kr * D + nd * 16 + (lane & 15)] = af2bf(dvacc[mi][nd][j]);
+      }
+    }
+  }
+}
+
+// pass B: dq. Block owns queries [q0, q0+128); wave w owns rows [w*32, +32).
+// Q and dO rows live in registers as A-fragments; K,V tiles of 64 keys are
+// staged per inner step (K in both layouts).
+template <int D>
+__global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
+    const abf16* __restrict__ q, const abf16* __restrict__ k,
+    const abf16* __restrict__ v, const abf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    ushort* __restrict__ dq, int Sq, int Sk, int past, int causal, float scale) {
+  constexpr int BM = 128, BN = 64;
+  constexpr int KP = D + 8, NP = BN + 8;
+  extern __shared__ char smem[];
+  abf16* Ks = reinterpret_cast<abf16*>(smem);  // [BN][KP]
+  abf16* KTs = Ks + BN * KP;                   // [D][NP]
+  abf16* Vs = KTs + D * NP;                    // [BN][KP]
+  abf16* dSs = Vs + BN * KP;                   // [BM][NP]
+
+  const int tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
+  const int q0 = blockIdx.x * BM;
+  const int64_t bh = blockIdx.y;
+  const abf16* qb = q + bh * Sq * (int64_t)D;
+  const abf16* kb = k + bh * Sk * (int64_t)D;
+  const abf16* vb = v + bh * Sk * (int64_t)D;
+  const abf16* dob = dout + bh * Sq * (int64_t)D;
+
+  bf16x8a qf[2][D / 32], dof[2][D / 32];
+  float l2r[2][4], drr[2][4];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+    int qr = q0 + wave * 32 + mi * 16 + (lane & 15);
+    if (qr >= Sq) qr = Sq - 1;
+#pragma unroll
+    for (int kc = 0; kc < D / 32; ++kc) {
+      qf[mi][kc] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr * D + kc * 32 + (lane >> 4) * 8);
+      dof[mi][kc] = *reinterpret_cast<const bf16x8a*>(dob + (int64_t)qr * D + kc * 32 + (lane >> 4) * 8);
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int rr = q0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
+      if (rr >= Sq) rr = Sq - 1;
+      l2r[mi][j] = lse[bh * Sq + rr] * kLog2e;
+      drr[mi][j] = drow[bh * Sq + rr];
+    }
+  }
+  f32x4a dqacc[2][D / 16];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int nd = 0; nd < D / 16; ++nd) dqacc[mi][nd] = f32x4a{};
+
+  const float c = scale * kLog2e;
+  const int k_hi = causal ? min(Sk, past + q0 + BM) : Sk;
+
+  for (int k0 = 0; k0 < k_hi; k0 += BN) {
+    {
+      constexpr int chunks = BN * D / 8;
+#pragma unroll
+      for (int it = 0; it < chunks / 256; ++it) {
+        const int cid = tid + it * 256;
+        const int row = cid / (D / 8);
+        const int col8 = (cid % (D / 8)) * 8;
+        int64_t kr = k0 + row;
+        if (kr >= Sk) kr = Sk - 1;
+        bf16x8a k8 = *reinterpret_cast<const bf16x8a*>(kb + kr * D + col8);
+        *reinterpret_cast<bf16x8a*>(Ks + row * KP + col8) = k8;
+        *reinterpret_cast<bf16x8a*>(Vs + row * KP + col8) =
+            *reinterpret_cast<const bf16x8a*>(vb + kr * D + col8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) KTs[(col8 + j) * NP + row] = k8[j];
+      }
+    }
+    __syncthreads();
+
+    f32x4a st[2][BN / 16], dpt[2][BN / 16];
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < BN / 16; ++ni) { st[mi][ni] = f32x4a{}; dpt[mi][ni] = f32x4a{}; }
+#pragma unroll
+    for (int ni = 0; ni < BN / 16; ++ni) {
+#pragma unroll
+      for (int kc = 0; kc < D / 32; ++kc) {
+        bf16x8a bk = *reinterpret_cast<const bf16x8a*>(
+            Ks + (ni * 16 + (lane & 15)) * KP + kc * 32 + (lane >> 4) * 8);
+        bf16x8a bv = *reinterpret_cast<const bf16x8a*>(
+            Vs + (ni * 16 + (lane & 15)) * KP + kc * 32 + (lane >> 4) * 8);
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          st[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[mi][kc], bk, st[mi][ni], 0, 0, 0);
+          dpt[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[mi][kc], bv, dpt[mi][ni], 0, 0, 0);
+        }
+      }
+    }
+
+    // dS = P ⊙ (dP − Drow)·scale, stage as A-frags (wave-private rows)
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < BN / 16; ++ni) {
+        const int kc = k0 + ni * 16 + (lane & 15);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int qr = q0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
+          const bool dead = kc >= Sk || (causal && kc > past + qr);
+          const float p = dead ? 0.f : exp2f(st[mi][ni][j] * c - l2r[mi][j]);
+          const float ds = p * (dpt[mi][ni][j] - drr[mi][j]) * scale;
+          ushort b = af2bf(ds);
+          dSs[(wave * 32 + mi * 16 + (lane >> 4) * 4 + j) * NP + ni * 16 + (lane & 15)] =
+              *reinterpret_cast<abf16*>(&b);
+        }
+      }
+    }
+    // dq += dS K (B-frags from K^T; contraction = keys)
+#pragma unroll
+    for (int kc = 0; kc < BN / 32; ++kc) {
+      bf16x8a sf[2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+        sf[mi] = *reinterpret_cast<const bf16x8a*>(
+            dSs + (wave * 32 + mi * 16 + (lane & 15)) * NP + kc * 32 + (lane >> 4) * 8);
+#pragma unroll
+      for (int nd = 0; nd < D / 16; ++nd) {
+        bf16x8a bf = *reinterpret_cast<const bf16x8a*>(
+            KTs + (nd * 16 + (lane & 15)) * NP + kc * 32 + (lane >> 4) * 8);
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+          dqacc[mi][nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(sf[mi], bf, dqacc[mi][nd], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  ushort* dqb = dq + bh * Sq * (int64_t)D;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int qr = q0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
+      if (qr >= Sq) continue;
+#pragma unroll
+      for (int nd = 0; nd < D / 16; ++nd)
+        dqb[(int64_t)qr * D + nd * 16 + (lane & 15)] = af2bf(dqacc[mi][nd][j]);
+    }
+  }
+}
+
+template __global__ void fa_bwd_dkdv_kernel<64>(const abf16*, const abf16*, const abf16*,
+                                                const abf16*, const float*, const float*,
+                                                ushort*, ushort*, int, int, int, int, float);
+template __global__ void fa_bwd_dkdv_kernel<128>(const abf16*, const abf16*, const abf16*,
+                                                 const abf16*, const float*, const float*,
+                                                 ushort*, ushort*, int, int, int, int, float);
+template __global__ void fa_bwd_dq_kernel<64>(const abf16*, const abf16*, const abf16*,
+                                              const abf16*, const float*, const float*,
+                                              ushort*, int, int, int, int, float);
+template __global__ void fa_bwd_dq_kernel<128>(const abf16*, const abf16*, const abf16*,
+                                               const abf16*, const float*, const float*,
+                                               ushort*, int, int, int, int, float);
+
 namespace {
 
 template <int D>
@@ -276,5 +657,59 @@ hipError_t launch_fa_fwd(const void* q, const void* k, const void* v, void* out,
     return launch_impl<64>(q, k, v, out, lse, batch_heads, Sq, Sk, past, causal, scale, stream);
   if (head_dim == 128)
     return launch_impl<128>(q, k, v, out, lse, batch_heads, Sq, Sk, past, causal, scale, stream);
+  return hipErrorInvalidValue;
+}
+
+namespace {
+
+template <int D>
+hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const void* dout,
+                           const void* out, const float* lse, float* drow,
+                           void* dq, void* dk, void* dv, int64_t bh, int Sq, int Sk,
+                           int past, int causal, float scale, hipStream_t stream) {
+  const int lds_dkdv = (2 * 64 * (D + 8) + 2 * D * 72 + 128 * 72) * 2 + 2 * 64 * 4;
+  const int lds_dq = (2 * 64 * (D + 8) + D * 72 + 128 * 72) * 2;
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipError_t e = hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dkdv_kernel<D>),
+                                       hipFuncAttributeMaxDynamicSharedMemorySize, lds_dkdv);
+    if (e != hipSuccess) return e;
+    e = hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dq_kernel<D>),
+                            hipFuncAttributeMaxDynamicSharedMemorySize, lds_dq);
+    if (e != hipSuccess) return e;
+    attr_set = true;
+  }
+  const int64_t n_rows = bh * Sq;
+  const int64_t want_drow = (n_rows + 3) / 4;
+  const int grid_drow = (int)(want_drow < 4096 ? want_drow : 4096);
+  hipLaunchKernelGGL(fa_drow_kernel, dim3(grid_drow), dim3(256), 0, stream,
+                     reinterpret_cast<const ushort*>(dout), reinterpret_cast<const ushort*>(out),
+                     drow, n_rows, D);
+  hipLaunchKernelGGL(fa_bwd_dkdv_kernel<D>, dim3((Sk + 127) / 128, (unsigned)bh), dim3(256),
+                     lds_dkdv, stream,
+                     reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                     reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
+                     lse, drow, reinterpret_cast<ushort*>(dk), reinterpret_cast<ushort*>(dv),
+                     Sq, Sk, past, causal, scale);
+  hipLaunchKernelGGL(fa_bwd_dq_kernel<D>, dim3((Sq + 127) / 128, (unsigned)bh), dim3(256),
+                     lds_dq, stream,
+                     reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                     reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
+                     lse, drow, reinterpret_cast<ushort*>(dq), Sq, Sk, past, causal, scale);
+  return hipGetLastError();
+}
+
+}  // namespace
+
+hipError_t launch_fa_bwd(const void* q, const void* k, const void* v, const void* dout,
+                         const void* out, const float* lse, float* drow,
+                         void* dq, void* dk, void* dv, int64_t batch_heads, int Sq, int Sk,
+                         int head_dim, int past, int causal, float scale, hipStream_t stream) {
+  if (head_dim == 64)
+    return launch_bwd_impl<64>(q, k, v, dout, out, lse, drow, dq, dk, dv, batch_heads,
+                               Sq, Sk, past, causal, scale, stream);
+  if (head_dim == 128)
+    return launch_bwd_impl<128>(q, k, v, dout, out, lse, drow, dq, dk, dv, batch_heads,
+                                Sq, Sk, past, causal, scale, stream);
   return hipErrorInvalidValue;
 }

@@ -426,6 +426,33 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {out, lse};
 }
 
+hipError_t launch_fa_bwd(const void*, const void*, const void*, const void*, const void*,
+                         const float*, float*, void*, void*, void*, int64_t, int, int, int,
+                         int, int, float, hipStream_t);
+
+std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
+                                       at::Tensor v, at::Tensor out, at::Tensor lse,
+                                       bool causal, double scale, int64_t past) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.scalar_type() == at::kBFloat16 && q.dim() == 4,
+              "flash_attn_bwd: q must be 4-D contiguous bf16");
+  const int64_t B = q.size(0), H = q.size(1), Sq = q.size(2), D = q.size(3);
+  const int64_t Sk = k.size(2);
+  TORCH_CHECK(D == 64 || D == 128, "flash_attn_bwd: head_dim must be 64 or 128");
+  auto doutc = dout.contiguous();
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto drow = at::empty({B, H, Sq}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipError_t e = launch_fa_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), doutc.data_ptr(),
+                               out.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
+                               dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), B * H,
+                               (int)Sq, (int)Sk, (int)D, (int)past, causal ? 1 : 0,
+                               (float)scale, stream.stream());
+  TORCH_CHECK(e == hipSuccess, "flash_attn_bwd launch failed: ", hipGetErrorString(e));
+  return {dq, dk, dv};
+}
+
 // weight-only quantization kernels (quant_kernels.hip)
 __global__ void int8_dequant_kernel(const char*, const float*, ushort*, int64_t, int64_t);
 __global__ void int4_dequant_kernel(const unsigned char*, const float*, ushort*, int64_t, int64_t, int);
@@ -497,6 +524,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_update_scale", &fp8_update_scale_fn, "delayed-scaling scale update");
   m.def("flash_attn_fwd", &flash_attn_fwd,
         "fused flash-attention forward (bf16, head_dim 64/128) -> (out, lse)");
+  m.def("flash_attn_bwd", &flash_attn_bwd,
+        "fused flash-attention backward (logsumexp recompute) -> (dq, dk, dv)");
   m.def("int8_dequant", &int8_dequant, "int8 weight -> bf16 (per-channel scale)");
   m.def("int4_dequant", &int4_dequant, "packed int4 weight -> bf16 (group-wise scale)");
   m.def("w8a16_gemv", &w8a16_gemv, "fused int8-weight x bf16-activation matvec (decode)");
