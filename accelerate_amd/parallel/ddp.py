@@ -159,7 +159,12 @@ class DistributedDataParallelEngine(nn.Module):
     # -- setup ------------------------------------------------------------
 
     def _build_buckets(self):
-        params = [p for p in self.module.parameters() if p.requires_grad]
+        # params marked _no_ddp_sync are rank-LOCAL (expert-parallel experts:
+        # each rank owns different experts, so averaging them would be wrong)
+        params = [
+            p for p in self.module.parameters()
+            if p.requires_grad and not getattr(p, "_no_ddp_sync", False)
+        ]
         # reverse registration order ≈ order grads become ready in backward
         params = list(reversed(params))
         current, current_bytes, current_key = [], 0, None
@@ -180,7 +185,7 @@ class DistributedDataParallelEngine(nn.Module):
 
     def _register_hooks(self):
         for p in self.module.parameters():
-            if p.requires_grad:
+            if p.requires_grad and not getattr(p, "_no_ddp_sync", False):
                 h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
                 self._hooks.append(h)
 
@@ -188,7 +193,7 @@ class DistributedDataParallelEngine(nn.Module):
         """Broadcast parameters and buffers from rank 0 at wrap time
         (reference: DDP _sync_module_states broadcast)."""
         for t in list(self.module.parameters()) + list(self.module.buffers()):
-            if t.numel() > 0:
+            if t.numel() > 0 and not getattr(t, "_no_ddp_sync", False):
                 dist.broadcast(t.data, src=0, group=self.process_group)
 
     # -- steady state ------------------------------------------------------
