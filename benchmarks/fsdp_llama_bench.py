@@ -27,6 +27,7 @@ def main():
     p.add_argument("--steps", type=int, default=5)
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--no-ac", dest="ac", action="store_false", help="disable activation checkpointing")
+    p.add_argument("--fp8", action="store_true", help="fp8 Linears (OCP e4m3/e5m2, delayed scaling)")
     p.set_defaults(ac=True)
     args = p.parse_args()
 
@@ -44,6 +45,10 @@ def main():
     set_seed(0)
     config = LlamaConfig.llama3_8b(num_hidden_layers=args.layers, max_position_embeddings=args.seq)
     model = LlamaForCausalLM(config)
+    if args.fp8:
+        from accelerate_amd.ops.fp8 import convert_linears_to_fp8
+
+        model = convert_linears_to_fp8(model.to(torch.bfloat16))
     opt = FusedAdamW(model.parameters(), lr=1e-4)
     model, opt = acc.prepare(model, opt)
 
@@ -82,7 +87,7 @@ def main():
                         "seq": args.seq,
                         "batch_per_gpu": args.batch,
                         "activation_checkpointing": args.ac,
-                        "dtype": "bf16 compute + fp32 master shards",
+                        "dtype": ("fp8 linears + " if args.fp8 else "") + "bf16 compute + fp32 master shards",
                     },
                 }
             )
