@@ -7,6 +7,7 @@ main-process-gated.
 """
 
 import functools
+import importlib.util
 import json
 import os
 import time
@@ -266,6 +267,221 @@ class MLflowTracker(GeneralTracker):
         mlflow.end_run()
 
 
+class CometMLTracker(GeneralTracker):
+    """Comet ML (reference: tracking.py:496). Lazy import; the comet_ml
+    package is not bundled — constructing without it raises ImportError."""
+
+    name = "comet_ml"
+    requires_logging_directory = False
+
+    @on_main_process
+    def __init__(self, run_name: str, **kwargs):
+        super().__init__()
+        from comet_ml import ExperimentConfig, start
+
+        self.run_name = run_name
+        self.writer = start(experiment_config=ExperimentConfig(name=run_name, **kwargs))
+
+    @property
+    def tracker(self):
+        return self.writer
+
+    @on_main_process
+    def store_init_configuration(self, values: dict):
+        self.writer.log_parameters(values)
+
+    @on_main_process
+    def log(self, values: dict, step: Optional[int] = None, **kwargs):
+        if step is not None:
+            self.writer.set_step(step)
+        for k, v in listify_values(values).items():
+            if isinstance(v, (int, float)):
+                self.writer.log_metric(k, v, step=step, **kwargs)
+            elif isinstance(v, str):
+                self.writer.log_other(k, v, **kwargs)
+            elif isinstance(v, dict):
+                self.writer.log_metrics(v, step=step, **kwargs)
+
+    @on_main_process
+    def finish(self):
+        self.writer.end()
+
+
+class AimTracker(GeneralTracker):
+    """Aim (reference: tracking.py:590)."""
+
+    name = "aim"
+    requires_logging_directory = True
+
+    @on_main_process
+    def __init__(self, run_name: str, logging_dir=".", **kwargs):
+        super().__init__()
+        from aim import Run
+
+        self.writer = Run(repo=str(logging_dir), **kwargs)
+        self.writer.name = run_name
+
+    @property
+    def tracker(self):
+        return self.writer
+
+    @on_main_process
+    def store_init_configuration(self, values: dict):
+        self.writer["hparams"] = values
+
+    @on_main_process
+    def log(self, values: dict, step: Optional[int] = None, **kwargs):
+        for k, v in listify_values(values).items():
+            self.writer.track(v, name=k, step=step, **kwargs)
+
+    @on_main_process
+    def finish(self):
+        self.writer.close()
+
+
+class ClearMLTracker(GeneralTracker):
+    """ClearML (reference: tracking.py:902)."""
+
+    name = "clearml"
+    requires_logging_directory = False
+
+    @on_main_process
+    def __init__(self, run_name: str = None, **kwargs):
+        super().__init__()
+        from clearml import Task
+
+        current = Task.current_task()
+        self._initialized_externally = current is not None
+        self.task = current or Task.init(
+            project_name=os.environ.get("CLEARML_PROJECT", run_name),
+            task_name=os.environ.get("CLEARML_TASK", run_name),
+            **kwargs,
+        )
+
+    @property
+    def tracker(self):
+        return self.task
+
+    @on_main_process
+    def store_init_configuration(self, values: dict):
+        self.task.connect_configuration(values)
+
+    @on_main_process
+    def log(self, values: dict, step: Optional[int] = None, **kwargs):
+        for k, v in listify_values(values).items():
+            if isinstance(v, (int, float)) and step is None:
+                self.task.get_logger().report_single_value(name=k, value=v, **kwargs)
+            elif isinstance(v, (int, float)):
+                title, _, series = k.rpartition("/") if "/" in k else ("train", "", k)
+                self.task.get_logger().report_scalar(
+                    title=title or "train", series=series, value=v, iteration=step, **kwargs
+                )
+
+    @on_main_process
+    def finish(self):
+        if not self._initialized_externally:
+            self.task.close()
+
+
+class DVCLiveTracker(GeneralTracker):
+    """DVCLive (reference: tracking.py:1060)."""
+
+    name = "dvclive"
+    requires_logging_directory = False
+
+    @on_main_process
+    def __init__(self, run_name: str = None, live=None, **kwargs):
+        super().__init__()
+        from dvclive import Live
+
+        self.live = live if live is not None else Live(**kwargs)
+
+    @property
+    def tracker(self):
+        return self.live
+
+    @on_main_process
+    def store_init_configuration(self, values: dict):
+        self.live.log_params(values)
+
+    @on_main_process
+    def log(self, values: dict, step: Optional[int] = None, **kwargs):
+        if step is not None:
+            self.live.step = step
+        for k, v in listify_values(values).items():
+            if isinstance(v, (int, float)):
+                self.live.log_metric(k, v, **kwargs)
+        self.live.next_step()
+
+    @on_main_process
+    def finish(self):
+        self.live.end()
+
+
+class SwanLabTracker(GeneralTracker):
+    """SwanLab (reference: tracking.py:1148)."""
+
+    name = "swanlab"
+    requires_logging_directory = False
+
+    @on_main_process
+    def __init__(self, run_name: str, **kwargs):
+        super().__init__()
+        import swanlab
+
+        self.run = swanlab.init(project=run_name, **kwargs)
+
+    @property
+    def tracker(self):
+        return self.run
+
+    @on_main_process
+    def store_init_configuration(self, values: dict):
+        import swanlab
+
+        swanlab.config.update(values, allow_val_change=True)
+
+    @on_main_process
+    def log(self, values: dict, step: Optional[int] = None, **kwargs):
+        self.run.log(listify_values(values), step=step)
+
+    @on_main_process
+    def finish(self):
+        self.run.finish()
+
+
+class TrackioTracker(GeneralTracker):
+    """Trackio (reference: tracking.py:419) — wandb-compatible local API."""
+
+    name = "trackio"
+    requires_logging_directory = False
+
+    @on_main_process
+    def __init__(self, run_name: str, **kwargs):
+        super().__init__()
+        import trackio
+
+        self.run = trackio.init(project=run_name, **kwargs)
+
+    @property
+    def tracker(self):
+        return self.run
+
+    @on_main_process
+    def store_init_configuration(self, values: dict):
+        import trackio
+
+        trackio.config.update(values, allow_val_change=True)
+
+    @on_main_process
+    def log(self, values: dict, step: Optional[int] = None, **kwargs):
+        self.run.log(listify_values(values))
+
+    @on_main_process
+    def finish(self):
+        self.run.finish()
+
+
 def listify_values(values: dict) -> dict:
     import torch
 
@@ -282,6 +498,12 @@ LOGGER_TYPE_TO_CLASS = {
     "wandb": WandBTracker,
     "mlflow": MLflowTracker,
     "jsonl": JSONLTracker,
+    "comet_ml": CometMLTracker,
+    "aim": AimTracker,
+    "clearml": ClearMLTracker,
+    "dvclive": DVCLiveTracker,
+    "swanlab": SwanLabTracker,
+    "trackio": TrackioTracker,
 }
 
 
@@ -298,6 +520,10 @@ def filter_trackers(log_with, logging_dir=None):
                 candidates.append("tensorboard")
             if is_wandb_available():
                 candidates.append("wandb")
+            for name, mod in (("comet_ml", "comet_ml"), ("aim", "aim"), ("clearml", "clearml"),
+                              ("dvclive", "dvclive"), ("swanlab", "swanlab"), ("trackio", "trackio")):
+                if importlib.util.find_spec(mod) is not None:
+                    candidates.append(name)
             candidates.append("jsonl")
             log_with = candidates
         for log_type in log_with:
