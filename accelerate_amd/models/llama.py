@@ -106,18 +106,19 @@ class LlamaAttention(nn.Module):
             kv_cache["len"] = past_len + S
             k = kv_cache["k"][:, :, : past_len + S]
             v = kv_cache["v"][:, :, : past_len + S]
-        # GQA: expand kv heads
-        if self.n_kv != self.n_heads:
-            rep = self.n_heads // self.n_kv
-            k = k.repeat_interleave(rep, dim=1)
-            v = v.repeat_interleave(rep, dim=1)
-
         if self.attn_impl in ("fused", "chunked") and x.is_cuda and S > 1:
-            # blockwise flash attention: O(S·block) memory, MFMA GEMM blocks
+            # fused CDNA4 flash attention; kv heads stay UNexpanded — the
+            # kernel maps head h -> kv head h/rep through strides (zero-copy,
+            # 1/rep the K/V HBM traffic and saved-activation bytes)
             from ..ops.attention import flash_attention
 
             ctx = flash_attention(q, k, v, causal=True)
         else:
+            # math path (decode / CPU): expand kv heads
+            if self.n_kv != self.n_heads:
+                rep = self.n_heads // self.n_kv
+                k = k.repeat_interleave(rep, dim=1)
+                v = v.repeat_interleave(rep, dim=1)
             scale = 1.0 / math.sqrt(self.head_dim)
             scores = torch.matmul(q, k.transpose(-1, -2)) * scale
             total = k.shape[2]

@@ -26,15 +26,25 @@ def _block(size, pref):
     return min(size, pref)
 
 
-def _fused_eligible(q, k) -> bool:
-    """The CDNA4 kernel covers bf16, head_dim 64/128, and Sq big enough that
-    a 128-row tile isn't pure padding. Everything else takes the torch path."""
+def _strides_ok(t) -> bool:
+    # kernel reads views directly: innermost dim contiguous, 16 B alignment
+    return t.stride(-1) == 1 and all(s % 8 == 0 for s in t.stride()[:-1])
+
+
+def _fused_eligible(q, k, v) -> bool:
+    """The CDNA4 kernel covers bf16, head_dim 64/128, Sq big enough that a
+    128-row tile isn't pure padding, and GQA kv (Hkv divides Hq) — read
+    zero-copy through strides. Everything else takes the torch path."""
     return (
         q.is_cuda
         and q.dtype == torch.bfloat16
         and q.shape[-1] in (64, 128)
         and q.shape[2] >= 32
         and q.shape[0] * q.shape[1] <= 65535
+        and q.shape[1] % k.shape[1] == 0
+        and _strides_ok(q)
+        and _strides_ok(k)
+        and _strides_ok(v)
     )
 
 
@@ -49,16 +59,22 @@ class _FlashAttentionFn(torch.autograd.Function):
         # parallelism passes the rank's absolute start explicitly.
         past = (Sk - Sq) if q_start is None else q_start
 
-        if _fused_eligible(q, k):
+        if _fused_eligible(q, k, v):
             from . import _load_extension
 
             ext = _load_extension(required=True)  # GPU boxes must run native
-            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
             out, lse = ext.flash_attn_fwd(q, k, v, causal, scale, past)
             ctx.save_for_backward(q, k, v, out, lse)
             ctx.meta = (causal, scale, q_block, k_block, past)
             return out
 
+        # torch path computes with kv expanded to Hq heads; the UNexpanded
+        # tensors are saved (backward re-expands and group-sums dk/dv)
+        k_orig, v_orig = k, v
+        if k.shape[1] != H:
+            rep = H // k.shape[1]
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
         out = torch.empty_like(q)
         lse = torch.empty(B, H, Sq, dtype=torch.float32, device=q.device)
 
@@ -89,7 +105,7 @@ class _FlashAttentionFn(torch.autograd.Function):
             out[:, :, q0:q1] = (acc / denom[..., None].clamp_min(1e-30)).to(q.dtype)
             lse[:, :, q0:q1] = m + denom.clamp_min(1e-30).log()
 
-        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.save_for_backward(q, k_orig, v_orig, out, lse)
         ctx.meta = (causal, scale, q_block, k_block, past)
         return out
 
@@ -97,14 +113,22 @@ class _FlashAttentionFn(torch.autograd.Function):
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
         causal, scale, q_block, k_block, past = ctx.meta
+        B, H, Sq, Dh = q.shape
+        Hkv = k.shape[1]
+        rep = H // Hkv
 
-        if _fused_eligible(q, k) and os.environ.get("ACCELERATE_AMD_FA_BWD", "1") == "1":
+        if _fused_eligible(q, k, v) and os.environ.get("ACCELERATE_AMD_FA_BWD", "1") == "1":
             from . import _load_extension
 
             ext = _load_extension(required=True)
             dq, dk, dv = ext.flash_attn_bwd(dout, q, k, v, out, lse, causal, scale, past)
+            if rep > 1:  # sum the per-q-head partials down to the kv heads
+                dk = dk.view(B, Hkv, rep, *dk.shape[2:]).float().sum(2).to(k.dtype)
+                dv = dv.view(B, Hkv, rep, *dv.shape[2:]).float().sum(2).to(v.dtype)
             return dq, dk, dv, None, None, None, None, None
-        B, H, Sq, Dh = q.shape
+        if rep > 1:
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
         Sk = k.shape[2]
         dq = torch.zeros_like(q, dtype=torch.float32)
         dk = torch.zeros_like(k, dtype=torch.float32)
@@ -134,7 +158,10 @@ class _FlashAttentionFn(torch.autograd.Function):
                 dq[:, :, q0:q1] += torch.matmul(ds, kb).float()
                 dk[:, :, k0:k1] += torch.matmul(ds.transpose(-1, -2), qb).float()
 
-        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None, None, None, None
+        if rep > 1:
+            dk = dk.view(B, Hkv, rep, Sk, Dh).sum(2)
+            dv = dv.view(B, Hkv, rep, Sk, Dh).sum(2)
+        return dq.to(q.dtype), dk.to(q.dtype), dv.to(q.dtype), None, None, None, None, None
 
 
 def flash_attention(
@@ -147,8 +174,11 @@ def flash_attention(
     k_block: int = 1024,
     q_start: Optional[int] = None,
 ) -> torch.Tensor:
-    """Blockwise attention over [B, H, S, D] tensors. ``q_start`` is the
-    absolute position of q[...,0] for causal masking (context parallelism)."""
+    """Blockwise attention over [B, H, S, D] tensors. ``k``/``v`` may carry
+    FEWER heads (GQA): any Hkv dividing H is expanded in-kernel on GPU
+    (zero-copy through strides, no repeat_interleave) or by the torch path
+    on CPU. ``q_start`` is the absolute position of q[...,0] for causal
+    masking (context parallelism)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     return _FlashAttentionFn.apply(

@@ -23,6 +23,8 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
+#include "attention.h"
+
 namespace {
 
 typedef __bf16 abf16;
@@ -42,15 +44,16 @@ constexpr float kLn2 = 0.6931471805599453f;
 
 }  // namespace
 
-// D = head_dim (64 or 128). q:[B,H,Sq,D] k,v:[B,H,Sk,D] bf16 contiguous.
-// out:[B,H,Sq,D] bf16; lse:[B,H,Sq] fp32 natural-log.
+// D = head_dim (64 or 128). q:[B,Hq,Sq,D] k,v:[B,Hkv,Sk,D] bf16 views
+// (strides in elements, dim 3 contiguous). out: written through sO (BSHD
+// storage); lse:[B,Hq,Sq] fp32 natural-log contiguous.
 // past: causal offset — query i attends keys <= past + i.
 template <int D>
 __global__ __launch_bounds__(256) void fa_fwd_kernel(
     const abf16* __restrict__ q, const abf16* __restrict__ k,
     const abf16* __restrict__ v, ushort* __restrict__ out,
     float* __restrict__ lse, int Sq, int Sk, int past, int causal,
-    float scale) {
+    float scale, int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sO) {
   // BN=128 measured faster than BN=64 (10413 vs 10239 tok/s on Llama-8B
   // b4 no-AC): the extra occupancy at 54 KB LDS is VGPR-capped anyway and
   // BN=64 doubles the barrier/softmax-reduction overhead per key.
@@ -66,10 +69,11 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
   const int wave = tid >> 6;
   const int lane = tid & 63;
   const int q0 = blockIdx.x * BM;
-  const int64_t bh = blockIdx.y;  // b*H + h
-  const abf16* qb = q + (bh * Sq) * (int64_t)D;
-  const abf16* kb = k + (bh * Sk) * (int64_t)D;
-  const abf16* vb = v + (bh * Sk) * (int64_t)D;
+  const int64_t bh = blockIdx.y;  // b*Hq + h
+  const int64_t b = bh / Hq, h = bh % Hq, hk = h / (Hq / Hkv);
+  const abf16* qb = q + b * sQ.b + h * sQ.h;
+  const abf16* kb = k + b * sK.b + hk * sK.h;
+  const abf16* vb = v + b * sV.b + hk * sV.h;
 
   // --- Q fragments, loaded once: wave rows [wave*32, wave*32+32) ---
   // A-frag (mi = 16-row tile, kc = 32-wide k chunk): lane holds row
@@ -81,7 +85,7 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
     if (qr >= Sq) qr = Sq - 1;  // clamp (padded rows never stored)
 #pragma unroll
     for (int kc = 0; kc < D / 32; ++kc)
-      qf[mi][kc] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr * D + kc * 32 + (lane >> 4) * 8);
+      qf[mi][kc] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr * sQ.s + kc * 32 + (lane >> 4) * 8);
   }
 
   f32x4a oacc[2][D / 16];
@@ -110,9 +114,9 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
         const int col8 = (cid % (D / 8)) * 8;    // d offset
         int64_t key = kb0 + row;
         if (key >= Sk) key = Sk - 1;             // clamp (masked below)
-        bf16x8a kv8 = *reinterpret_cast<const bf16x8a*>(kb + key * D + col8);
+        bf16x8a kv8 = *reinterpret_cast<const bf16x8a*>(kb + key * sK.s + col8);
         *reinterpret_cast<bf16x8a*>(Ks + row * KP + col8) = kv8;
-        bf16x8a vv8 = *reinterpret_cast<const bf16x8a*>(vb + key * D + col8);
+        bf16x8a vv8 = *reinterpret_cast<const bf16x8a*>(vb + key * sV.s + col8);
 #pragma unroll
         for (int j = 0; j < 8; ++j) VTs[(col8 + j) * NP + row] = vv8[j];
       }
@@ -219,7 +223,7 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
   }
 
   // --- epilogue: out = O/l (bf16), lse = (m + log2 l) * ln2 (natural) ---
-  ushort* ob = out + (bh * Sq) * (int64_t)D;
+  ushort* ob = out + b * sO.b + h * sO.h;
   float* lb = lse + bh * Sq;
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi) {
@@ -230,7 +234,7 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
       const float inv_l = 1.f / fmaxf(l_run[mi][j], 1e-30f);
 #pragma unroll
       for (int nd = 0; nd < D / 16; ++nd)
-        ob[(int64_t)qr * D + nd * 16 + (lane & 15)] = af2bf(oacc[mi][nd][j] * inv_l);
+        ob[(int64_t)qr * sO.s + nd * 16 + (lane & 15)] = af2bf(oacc[mi][nd][j] * inv_l);
       if ((lane & 15) == 0)
         lb[qr] = (m_run[mi][j] + __log2f(fmaxf(l_run[mi][j], 1e-30f))) * kLn2;
     }
@@ -238,9 +242,11 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
 }
 
 template __global__ void fa_fwd_kernel<64>(const abf16*, const abf16*, const abf16*,
-                                           ushort*, float*, int, int, int, int, float);
+                                           ushort*, float*, int, int, int, int, float,
+                                           int, int, Str3, Str3, Str3, Str3);
 template __global__ void fa_fwd_kernel<128>(const abf16*, const abf16*, const abf16*,
-                                            ushort*, float*, int, int, int, int, float);
+                                            ushort*, float*, int, int, int, int, float,
+                                            int, int, Str3, Str3, Str3, Str3);
 
 // ===========================================================================
 // Backward: logsumexp-recompute, two passes (no atomics):
@@ -255,12 +261,16 @@ template __global__ void fa_fwd_kernel<128>(const abf16*, const abf16*, const ab
 
 __global__ __launch_bounds__(256) void fa_drow_kernel(
     const ushort* __restrict__ dout, const ushort* __restrict__ out,
-    float* __restrict__ drow, int64_t n_rows, int D) {
+    float* __restrict__ drow, int64_t n_rows, int D, int Hq, int Sq,
+    Str3 sDo, Str3 sO) {
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
   for (int64_t r = (int64_t)blockIdx.x * 4 + wave; r < n_rows; r += (int64_t)gridDim.x * 4) {
+    const int64_t bb = r / ((int64_t)Hq * Sq), hh = (r / Sq) % Hq, ss = r % Sq;
+    const ushort* dor = dout + bb * sDo.b + hh * sDo.h + ss * sDo.s;
+    const ushort* orr = out + bb * sO.b + hh * sO.h + ss * sO.s;
     float s = 0.f;
     for (int d = lane; d < D; d += 64) {
-      ushort a = dout[r * D + d], b = out[r * D + d];
+      ushort a = dor[d], b = orr[d];
       s += (float)*reinterpret_cast<__hip_bfloat16*>(&a) * (float)*reinterpret_cast<__hip_bfloat16*>(&b);
     }
 #pragma unroll
@@ -280,7 +290,8 @@ __global__ __launch_bounds__(256) void fa_bwd_dkdv_kernel(
     const abf16* __restrict__ v, const abf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
     ushort* __restrict__ dk, ushort* __restrict__ dv,
-    int Sq, int Sk, int past, int causal, float scale) {
+    int Sq, int Sk, int past, int causal, float scale,
+    int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sDo) {
   constexpr int BK = 128, BQ = 64;
   constexpr int KP = D + 8, QP = BQ + 8;
   extern __shared__ char smem[];
@@ -295,10 +306,11 @@ __global__ __launch_bounds__(256) void fa_bwd_dkdv_kernel(
   const int tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
   const int k0 = blockIdx.x * BK;
   const int64_t bh = blockIdx.y;
-  const abf16* qb = q + bh * Sq * (int64_t)D;
-  const abf16* kb = k + bh * Sk * (int64_t)D;
-  const abf16* vb = v + bh * Sk * (int64_t)D;
-  const abf16* dob = dout + bh * Sq * (int64_t)D;
+  const int64_t b = bh / Hq, h = bh % Hq, hk = h / (Hq / Hkv);
+  const abf16* qb = q + b * sQ.b + h * sQ.h;
+  const abf16* kb = k + b * sK.b + hk * sK.h;
+  const abf16* vb = v + b * sV.b + hk * sV.h;
+  const abf16* dob = dout + b * sDo.b + h * sDo.h;
   const float* lb = lse + bh * Sq;
   const float* db = drow + bh * Sq;
 
@@ -310,8 +322,8 @@ __global__ __launch_bounds__(256) void fa_bwd_dkdv_kernel(
     if (kr >= Sk) kr = Sk - 1;
 #pragma unroll
     for (int kc = 0; kc < D / 32; ++kc) {
-      kf[mi][kc] = *reinterpret_cast<const bf16x8a*>(kb + (int64_t)kr * D + kc * 32 + (lane >> 4) * 8);
-      vf[mi][kc] = *reinterpret_cast<const bf16x8a*>(vb + (int64_t)kr * D + kc * 32 + (lane >> 4) * 8);
+      kf[mi][kc] = *reinterpret_cast<const bf16x8a*>(kb + (int64_t)kr * sK.s + kc * 32 + (lane >> 4) * 8);
+      vf[mi][kc] = *reinterpret_cast<const bf16x8a*>(vb + (int64_t)kr * sV.s + kc * 32 + (lane >> 4) * 8);
     }
   }
   f32x4a dkacc[2][D / 16], dvacc[2][D / 16];
@@ -334,9 +346,9 @@ __global__ __launch_bounds__(256) void fa_bwd_dkdv_kernel(
         const int col8 = (cid % (D / 8)) * 8;
         int64_t qr = q0 + row;
         if (qr >= Sq) qr = Sq - 1;  // clamped; masked via p=0 below
-        bf16x8a q8 = *reinterpret_cast<const bf16x8a*>(qb + qr * D + col8);
+        bf16x8a q8 = *reinterpret_cast<const bf16x8a*>(qb + qr * sQ.s + col8);
         *reinterpret_cast<bf16x8a*>(Qs + row * KP + col8) = q8;
-        bf16x8a d8 = *reinterpret_cast<const bf16x8a*>(dob + qr * D + col8);
+        bf16x8a d8 = *reinterpret_cast<const bf16x8a*>(dob + qr * sDo.s + col8);
         *reinterpret_cast<bf16x8a*>(dOs + row * KP + col8) = d8;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -447,6 +459,8 @@ __global__ __launch_bounds__(256) void fa_bwd_dkdv_kernel(
     __syncthreads();  // next q-tile restages Qs/dOs
   }
 
+  // partial dk/dv per Q-HEAD, contiguous [B,Hq,Sk,D]; the host sums the
+  // GQA groups (rep partials -> one kv head) in one cheap fused pass
   ushort* dkb = dk + bh * Sk * (int64_t)D;
   ushort* dvb = dv + bh * Sk * (int64_t)D;
 #pragma unroll
@@ -472,7 +486,8 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
     const abf16* __restrict__ q, const abf16* __restrict__ k,
     const abf16* __restrict__ v, const abf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
-    ushort* __restrict__ dq, int Sq, int Sk, int past, int causal, float scale) {
+    ushort* __restrict__ dq, int Sq, int Sk, int past, int causal, float scale,
+    int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sDo) {
   constexpr int BM = 128, BN = 64;
   constexpr int KP = D + 8, NP = BN + 8;
   extern __shared__ char smem[];
@@ -484,10 +499,11 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
   const int tid = threadIdx.x, wave = tid >> 6, lane = tid & 63;
   const int q0 = blockIdx.x * BM;
   const int64_t bh = blockIdx.y;
-  const abf16* qb = q + bh * Sq * (int64_t)D;
-  const abf16* kb = k + bh * Sk * (int64_t)D;
-  const abf16* vb = v + bh * Sk * (int64_t)D;
-  const abf16* dob = dout + bh * Sq * (int64_t)D;
+  const int64_t b = bh / Hq, h = bh % Hq, hk = h / (Hq / Hkv);
+  const abf16* qb = q + b * sQ.b + h * sQ.h;
+  const abf16* kb = k + b * sK.b + hk * sK.h;
+  const abf16* vb = v + b * sV.b + hk * sV.h;
+  const abf16* dob = dout + b * sDo.b + h * sDo.h;
 
   bf16x8a qf[2][D / 32], dof[2][D / 32];
   float l2r[2][4], drr[2][4];
@@ -497,8 +513,8 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
     if (qr >= Sq) qr = Sq - 1;
 #pragma unroll
     for (int kc = 0; kc < D / 32; ++kc) {
-      qf[mi][kc] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr * D + kc * 32 + (lane >> 4) * 8);
-      dof[mi][kc] = *reinterpret_cast<const bf16x8a*>(dob + (int64_t)qr * D + kc * 32 + (lane >> 4) * 8);
+      qf[mi][kc] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr * sQ.s + kc * 32 + (lane >> 4) * 8);
+      dof[mi][kc] = *reinterpret_cast<const bf16x8a*>(dob + (int64_t)qr * sDo.s + kc * 32 + (lane >> 4) * 8);
     }
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
@@ -527,10 +543,10 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
         const int col8 = (cid % (D / 8)) * 8;
         int64_t kr = k0 + row;
         if (kr >= Sk) kr = Sk - 1;
-        bf16x8a k8 = *reinterpret_cast<const bf16x8a*>(kb + kr * D + col8);
+        bf16x8a k8 = *reinterpret_cast<const bf16x8a*>(kb + kr * sK.s + col8);
         *reinterpret_cast<bf16x8a*>(Ks + row * KP + col8) = k8;
         *reinterpret_cast<bf16x8a*>(Vs + row * KP + col8) =
-            *reinterpret_cast<const bf16x8a*>(vb + kr * D + col8);
+            *reinterpret_cast<const bf16x8a*>(vb + kr * sV.s + col8);
 #pragma unroll
         for (int j = 0; j < 8; ++j) KTs[(col8 + j) * NP + row] = k8[j];
       }
@@ -612,16 +628,20 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
 
 template __global__ void fa_bwd_dkdv_kernel<64>(const abf16*, const abf16*, const abf16*,
                                                 const abf16*, const float*, const float*,
-                                                ushort*, ushort*, int, int, int, int, float);
+                                                ushort*, ushort*, int, int, int, int, float,
+                                                int, int, Str3, Str3, Str3, Str3);
 template __global__ void fa_bwd_dkdv_kernel<128>(const abf16*, const abf16*, const abf16*,
                                                  const abf16*, const float*, const float*,
-                                                 ushort*, ushort*, int, int, int, int, float);
+                                                 ushort*, ushort*, int, int, int, int, float,
+                                                 int, int, Str3, Str3, Str3, Str3);
 template __global__ void fa_bwd_dq_kernel<64>(const abf16*, const abf16*, const abf16*,
                                               const abf16*, const float*, const float*,
-                                              ushort*, int, int, int, int, float);
+                                              ushort*, int, int, int, int, float,
+                                              int, int, Str3, Str3, Str3, Str3);
 template __global__ void fa_bwd_dq_kernel<128>(const abf16*, const abf16*, const abf16*,
                                                const abf16*, const float*, const float*,
-                                               ushort*, int, int, int, int, float);
+                                               ushort*, int, int, int, int, float,
+                                               int, int, Str3, Str3, Str3, Str3);
 
 namespace {
 
@@ -634,7 +654,7 @@ int fa_lds_bytes() {
 template <int D>
 hipError_t launch_impl(const void* q, const void* k, const void* v, void* out, float* lse,
                        int64_t bh, int Sq, int Sk, int past, int causal, float scale,
-                       hipStream_t stream) {
+                       int Hq, int Hkv, const Str3* strides, hipStream_t stream) {
   static bool attr_set = false;
   const int lds = fa_lds_bytes<D>();
   if (!attr_set) {
@@ -647,20 +667,25 @@ hipError_t launch_impl(const void* q, const void* k, const void* v, void* out, f
   hipLaunchKernelGGL(fa_fwd_kernel<D>, grid, dim3(256), lds, stream,
                      reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
                      reinterpret_cast<const abf16*>(v), reinterpret_cast<ushort*>(out),
-                     lse, Sq, Sk, past, causal, scale);
+                     lse, Sq, Sk, past, causal, scale, Hq, Hkv,
+                     strides[0], strides[1], strides[2], strides[3]);
   return hipGetLastError();
 }
 
 }  // namespace
 
-// raw launcher used by bindings.hip (template instantiations stay local)
+// raw launcher used by bindings.hip (template instantiations stay local).
+// strides: q, k, v, out Str3 triples (elements).
 hipError_t launch_fa_fwd(const void* q, const void* k, const void* v, void* out, float* lse,
                          int64_t batch_heads, int Sq, int Sk, int head_dim, int past,
-                         int causal, float scale, hipStream_t stream) {
+                         int causal, float scale, int Hq, int Hkv, const Str3* strides,
+                         hipStream_t stream) {
   if (head_dim == 64)
-    return launch_impl<64>(q, k, v, out, lse, batch_heads, Sq, Sk, past, causal, scale, stream);
+    return launch_impl<64>(q, k, v, out, lse, batch_heads, Sq, Sk, past, causal, scale,
+                           Hq, Hkv, strides, stream);
   if (head_dim == 128)
-    return launch_impl<128>(q, k, v, out, lse, batch_heads, Sq, Sk, past, causal, scale, stream);
+    return launch_impl<128>(q, k, v, out, lse, batch_heads, Sq, Sk, past, causal, scale,
+                            Hq, Hkv, strides, stream);
   return hipErrorInvalidValue;
 }
 
@@ -670,7 +695,8 @@ template <int D>
 hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const void* dout,
                            const void* out, const float* lse, float* drow,
                            void* dq, void* dk, void* dv, int64_t bh, int Sq, int Sk,
-                           int past, int causal, float scale, hipStream_t stream) {
+                           int past, int causal, float scale,
+                           int Hq, int Hkv, const Str3* strides, hipStream_t stream) {
   const int lds_dkdv = (2 * 64 * (D + 8) + 2 * D * 72 + 128 * 72) * 2 + 2 * 64 * 4;
   const int lds_dq = (2 * 64 * (D + 8) + D * 72 + 128 * 72) * 2;
   static bool attr_set = false;
@@ -686,20 +712,23 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
   const int64_t n_rows = bh * Sq;
   const int64_t want_drow = (n_rows + 3) / 4;
   const int grid_drow = (int)(want_drow < 4096 ? want_drow : 4096);
+  // strides[]: q, k, v, dout, out
   hipLaunchKernelGGL(fa_drow_kernel, dim3(grid_drow), dim3(256), 0, stream,
                      reinterpret_cast<const ushort*>(dout), reinterpret_cast<const ushort*>(out),
-                     drow, n_rows, D);
+                     drow, n_rows, D, Hq, Sq, strides[3], strides[4]);
   hipLaunchKernelGGL(fa_bwd_dkdv_kernel<D>, dim3((Sk + 127) / 128, (unsigned)bh), dim3(256),
                      lds_dkdv, stream,
                      reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
                      reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
                      lse, drow, reinterpret_cast<ushort*>(dk), reinterpret_cast<ushort*>(dv),
-                     Sq, Sk, past, causal, scale);
+                     Sq, Sk, past, causal, scale, Hq, Hkv,
+                     strides[0], strides[1], strides[2], strides[3]);
   hipLaunchKernelGGL(fa_bwd_dq_kernel<D>, dim3((Sq + 127) / 128, (unsigned)bh), dim3(256),
                      lds_dq, stream,
                      reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
                      reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
-                     lse, drow, reinterpret_cast<ushort*>(dq), Sq, Sk, past, causal, scale);
+                     lse, drow, reinterpret_cast<ushort*>(dq), Sq, Sk, past, causal, scale,
+                     Hq, Hkv, strides[0], strides[1], strides[2], strides[3]);
   return hipGetLastError();
 }
 
@@ -708,12 +737,13 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
 hipError_t launch_fa_bwd(const void* q, const void* k, const void* v, const void* dout,
                          const void* out, const float* lse, float* drow,
                          void* dq, void* dk, void* dv, int64_t batch_heads, int Sq, int Sk,
-                         int head_dim, int past, int causal, float scale, hipStream_t stream) {
+                         int head_dim, int past, int causal, float scale,
+                         int Hq, int Hkv, const Str3* strides, hipStream_t stream) {
   if (head_dim == 64)
     return launch_bwd_impl<64>(q, k, v, dout, out, lse, drow, dq, dk, dv, batch_heads,
-                               Sq, Sk, past, causal, scale, stream);
+                               Sq, Sk, past, causal, scale, Hq, Hkv, strides, stream);
   if (head_dim == 128)
     return launch_bwd_impl<128>(q, k, v, dout, out, lse, drow, dq, dk, dv, batch_heads,
-                                Sq, Sk, past, causal, scale, stream);
+                                Sq, Sk, past, causal, scale, Hq, Hkv, strides, stream);
   return hipErrorInvalidValue;
 }

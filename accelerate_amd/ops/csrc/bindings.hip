@@ -400,55 +400,78 @@ void fp8_update_scale_fn(at::Tensor history, double fp8_max, double margin_pow2,
 }
 
 // fused flash-attention forward (attention_kernels.hip)
+#include "attention.h"
 hipError_t launch_fa_fwd(const void*, const void*, const void*, void*, float*,
-                         int64_t, int, int, int, int, int, float, hipStream_t);
+                         int64_t, int, int, int, int, int, float, int, int,
+                         const Str3*, hipStream_t);
+
+namespace {
+
+// (batch, head, seq) element strides of a [B,H,S,D] view; dim 3 must be
+// contiguous and every access 16-byte aligned (strides % 8 elements).
+Str3 str3_of(const at::Tensor& t, const char* who) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16 && t.dim() == 4,
+              who, ": 4-D bf16 HIP tensor required");
+  TORCH_CHECK(t.stride(3) == 1, who, ": innermost dim must be contiguous");
+  Str3 s{t.stride(0), t.stride(1), t.stride(2)};
+  TORCH_CHECK(s.b % 8 == 0 && s.h % 8 == 0 && s.s % 8 == 0,
+              who, ": strides must be multiples of 8 elements (16 B)");
+  return s;
+}
+
+}  // namespace
 
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                        bool causal, double scale, int64_t past) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.scalar_type() == at::kBFloat16 && q.dim() == 4,
-              "flash_attn_fwd: q must be 4-D contiguous bf16 [B,H,Sq,D]");
-  TORCH_CHECK(k.is_contiguous() && v.is_contiguous() && k.scalar_type() == at::kBFloat16 &&
-                  v.scalar_type() == at::kBFloat16,
-              "flash_attn_fwd: k/v must be contiguous bf16");
-  const int64_t B = q.size(0), H = q.size(1), Sq = q.size(2), D = q.size(3);
-  const int64_t Sk = k.size(2);
-  TORCH_CHECK(k.size(0) == B && k.size(1) == H && v.sizes() == k.sizes(),
-              "flash_attn_fwd: expand GQA kv heads before the kernel");
+  // q:[B,Hq,Sq,D], k/v:[B,Hkv,Sk,D] — ANY strides with dim 3 contiguous
+  // (transposed BSHD views and un-expanded GQA kv caches read zero-copy).
+  const int64_t B = q.size(0), Hq = q.size(1), Sq = q.size(2), D = q.size(3);
+  const int64_t Hkv = k.size(1), Sk = k.size(2);
+  TORCH_CHECK(k.size(0) == B && k.size(3) == D && v.sizes() == k.sizes(),
+              "flash_attn_fwd: k/v shape mismatch");
+  TORCH_CHECK(Hq % Hkv == 0, "flash_attn_fwd: Hq must be a multiple of Hkv");
   TORCH_CHECK(D == 64 || D == 128, "flash_attn_fwd: head_dim must be 64 or 128");
-  TORCH_CHECK(B * H <= 65535, "flash_attn_fwd: grid.y overflow");
-  auto out = at::empty_like(q);
-  auto lse = at::empty({B, H, Sq}, q.options().dtype(at::kFloat));
+  TORCH_CHECK(B * Hq <= 65535, "flash_attn_fwd: grid.y overflow");
+  // output in BSHD storage: the model's .transpose(1,2).reshape(B,S,H*D)
+  // after attention becomes a free view
+  auto out = at::empty({B, Sq, Hq, D}, q.options()).permute({0, 2, 1, 3});
+  auto lse = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
+  Str3 strides[4] = {str3_of(q, "q"), str3_of(k, "k"), str3_of(v, "v"), str3_of(out, "out")};
   auto stream = at::hip::getCurrentHIPStream();
   hipError_t e = launch_fa_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
-                               lse.data_ptr<float>(), B * H, (int)Sq, (int)Sk, (int)D,
-                               (int)past, causal ? 1 : 0, (float)scale, stream.stream());
+                               lse.data_ptr<float>(), B * Hq, (int)Sq, (int)Sk, (int)D,
+                               (int)past, causal ? 1 : 0, (float)scale, (int)Hq, (int)Hkv,
+                               strides, stream.stream());
   TORCH_CHECK(e == hipSuccess, "flash_attn_fwd launch failed: ", hipGetErrorString(e));
   return {out, lse};
 }
 
 hipError_t launch_fa_bwd(const void*, const void*, const void*, const void*, const void*,
                          const float*, float*, void*, void*, void*, int64_t, int, int, int,
-                         int, int, float, hipStream_t);
+                         int, int, float, int, int, const Str3*, hipStream_t);
 
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                        at::Tensor v, at::Tensor out, at::Tensor lse,
                                        bool causal, double scale, int64_t past) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && q.scalar_type() == at::kBFloat16 && q.dim() == 4,
-              "flash_attn_bwd: q must be 4-D contiguous bf16");
-  const int64_t B = q.size(0), H = q.size(1), Sq = q.size(2), D = q.size(3);
-  const int64_t Sk = k.size(2);
+  // Returns (dq [B,Hq,Sq,D] contiguous, dk/dv PARTIALS [B,Hq,Sk,D]
+  // contiguous — the Python wrapper sums the GQA groups down to Hkv).
+  const int64_t B = q.size(0), Hq = q.size(1), Sq = q.size(2), D = q.size(3);
+  const int64_t Hkv = k.size(1), Sk = k.size(2);
   TORCH_CHECK(D == 64 || D == 128, "flash_attn_bwd: head_dim must be 64 or 128");
-  auto doutc = dout.contiguous();
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
-  auto drow = at::empty({B, H, Sq}, q.options().dtype(at::kFloat));
+  TORCH_CHECK(Hq % Hkv == 0, "flash_attn_bwd: Hq must be a multiple of Hkv");
+  if (dout.stride(3) != 1) dout = dout.contiguous();
+  auto dq = at::empty({B, Hq, Sq, D}, q.options());
+  auto dk = at::empty({B, Hq, Sk, D}, q.options());
+  auto dv = at::empty({B, Hq, Sk, D}, q.options());
+  auto drow = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
+  Str3 strides[5] = {str3_of(q, "q"), str3_of(k, "k"), str3_of(v, "v"),
+                     str3_of(dout, "dout"), str3_of(out, "out")};
   auto stream = at::hip::getCurrentHIPStream();
-  hipError_t e = launch_fa_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), doutc.data_ptr(),
+  hipError_t e = launch_fa_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), dout.data_ptr(),
                                out.data_ptr(), lse.data_ptr<float>(), drow.data_ptr<float>(),
-                               dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), B * H,
+                               dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), B * Hq,
                                (int)Sq, (int)Sk, (int)D, (int)past, causal ? 1 : 0,
-                               (float)scale, stream.stream());
+                               (float)scale, (int)Hq, (int)Hkv, strides, stream.stream());
   TORCH_CHECK(e == hipSuccess, "flash_attn_bwd launch failed: ", hipGetErrorString(e));
   return {dq, dk, dv};
 }
