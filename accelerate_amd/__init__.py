@@ -10,7 +10,9 @@ __version__ = "0.1.0"
 
 from .accelerator import Accelerator
 from .big_modeling import (
+    attach_layerwise_casting_hooks,
     cpu_offload,
+    cpu_offload_with_hook,
     disk_offload,
     dispatch_model,
     init_empty_weights,
