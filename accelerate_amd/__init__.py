@@ -21,10 +21,15 @@ from .big_modeling import (
 )
 from .data_loader import prepare_data_loader, skip_first_batches
 from .launchers import debug_launcher, notebook_launcher
+from .inference import prepare_pipeline, prepare_pippy
 from .local_sgd import LocalSGD
+from .parallelism_config import ParallelismConfig
 from .state import AcceleratorState, GradientState, PartialState
+from .utils.random_utils import synchronize_rng_states  # noqa: F401
 from .utils.dataclasses import (
     AutocastKwargs,
+    DataLoaderConfiguration,
+    DDPCommunicationHookType,
     DistributedDataParallelKwargs,
     DistributedType,
     FP8RecipeKwargs,

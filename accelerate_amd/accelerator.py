@@ -129,6 +129,16 @@ class Accelerator:
         self.even_batches = even_batches
         self.use_seedable_sampler = use_seedable_sampler
         self.non_blocking = non_blocking
+        self.use_stateful_dataloader = False
+        self.data_seed = None
+        if dataloader_config is not None:  # bundled options take precedence
+            self.split_batches = dataloader_config.split_batches
+            self.dispatch_batches = dataloader_config.dispatch_batches
+            self.even_batches = dataloader_config.even_batches
+            self.use_seedable_sampler = dataloader_config.use_seedable_sampler
+            self.non_blocking = dataloader_config.non_blocking
+            self.use_stateful_dataloader = getattr(dataloader_config, "use_stateful_dataloader", False)
+            self.data_seed = getattr(dataloader_config, "data_seed", None)
         self.step_scheduler_with_optimizer = step_scheduler_with_optimizer
 
         # mixed precision

@@ -129,3 +129,7 @@ def prepare_pipeline(model, num_chunks=None, blocks=None, gather_output=False):
 
 # reference-name alias (reference: inference.py:126)
 prepare_pippy = prepare_pipeline
+
+
+# reference-compatible alias (reference inference.py:126 names it prepare_pippy)
+prepare_pippy = prepare_pipeline

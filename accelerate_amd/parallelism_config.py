@@ -1,0 +1,125 @@
+"""Unified multi-dimensional parallelism descriptor (reference:
+parallelism_config.py:34 ``ParallelismConfig``).
+
+Declares the degree of every dimension and builds the RCCL/gloo process
+groups our engines take as ``group=`` arguments:
+
+- dp_replicate × dp_shard — DDP replicas of sharded (FSDP/HSDP) groups
+- tp — tensor parallel (parallel/tp.py Column/RowParallelLinear)
+- cp — context parallel (parallel/cp.py)
+- ep — expert parallel (parallel/ep.py); EP shares ranks with dp_shard in
+  the usual MoE layout, so it is validated against that product rather than
+  multiplied into the world size.
+
+Rank layout is megatron-style innermost-first: tp varies fastest, then cp,
+then dp_shard, then dp_replicate — so TP groups are intra-node neighbors
+(xGMI-adjacent) when launched one-rank-per-GPU.
+"""
+
+from dataclasses import dataclass, field
+from typing import Dict, Optional
+
+import torch.distributed as dist
+
+
+@dataclass
+class ParallelismConfig:
+    dp_replicate_size: int = 1
+    dp_shard_size: int = 1
+    tp_size: int = 1
+    cp_size: int = 1
+    ep_size: int = 1
+
+    _groups: Dict[str, Optional[object]] = field(default_factory=dict, repr=False)
+
+    @property
+    def total_size(self) -> int:
+        return self.dp_replicate_size * self.dp_shard_size * self.tp_size * self.cp_size
+
+    @property
+    def dp_size(self) -> int:
+        return self.dp_replicate_size * self.dp_shard_size
+
+    def validate(self, world_size: int):
+        if self.total_size != world_size:
+            raise ValueError(
+                f"ParallelismConfig total_size ({self.total_size}) != world size ({world_size}); "
+                f"dims: dp_replicate={self.dp_replicate_size} dp_shard={self.dp_shard_size} "
+                f"tp={self.tp_size} cp={self.cp_size}"
+            )
+        if self.ep_size > 1 and self.dp_shard_size % self.ep_size != 0:
+            raise ValueError(
+                f"ep_size ({self.ep_size}) must divide dp_shard_size ({self.dp_shard_size}) — "
+                "experts are sharded across (a subset of) the data-parallel shard group"
+            )
+
+    # -- rank coordinates (innermost-first: tp, cp, dp_shard, dp_replicate) --
+    def coords(self, rank: int):
+        tp = rank % self.tp_size
+        cp = (rank // self.tp_size) % self.cp_size
+        shard = (rank // (self.tp_size * self.cp_size)) % self.dp_shard_size
+        repl = rank // (self.tp_size * self.cp_size * self.dp_shard_size)
+        return {"tp": tp, "cp": cp, "dp_shard": shard, "dp_replicate": repl}
+
+    def build_groups(self) -> Dict[str, Optional[object]]:
+        """Create one process group per dimension containing THIS rank.
+
+        Every rank must call this (new_group is collective). Returns
+        {"tp": pg, "cp": pg, "dp_shard": pg, "dp_replicate": pg, "dp": pg};
+        a size-1 dimension maps to None (engines treat None as 'world' or
+        'off' appropriately — pass the group explicitly).
+        """
+        if not dist.is_initialized():
+            raise RuntimeError("ParallelismConfig.build_groups needs torch.distributed initialized")
+        world = dist.get_world_size()
+        self.validate(world)
+        rank = dist.get_rank()
+        me = self.coords(rank)
+
+        def ranks_varying(dim_sizes, varying):
+            """All rank lists where `varying` sweeps and other dims are fixed."""
+            tp, cp, sh, rp = self.tp_size, self.cp_size, self.dp_shard_size, self.dp_replicate_size
+            out = []
+            for r in range(world):
+                c = self.coords(r)
+                key = tuple(c[d] for d in ("tp", "cp", "dp_shard", "dp_replicate") if d != varying)
+                out.append((key, r))
+            groups = {}
+            for key, r in out:
+                groups.setdefault(key, []).append(r)
+            return list(groups.values())
+
+        result: Dict[str, Optional[object]] = {}
+        for dim, size in (
+            ("tp", self.tp_size),
+            ("cp", self.cp_size),
+            ("dp_shard", self.dp_shard_size),
+            ("dp_replicate", self.dp_replicate_size),
+        ):
+            if size == 1:
+                result[dim] = None
+                continue
+            mine = None
+            for ranks in ranks_varying(None, dim):
+                pg = dist.new_group(ranks)
+                if rank in ranks:
+                    mine = pg
+            result[dim] = mine
+        # combined DP group (replicate x shard): gradient averaging domain
+        if self.dp_size == world:
+            result["dp"] = None  # whole world — use the default group
+        elif self.dp_size == 1:
+            result["dp"] = None
+        else:
+            mine = None
+            tpcp = self.tp_size * self.cp_size
+            lists = {}
+            for r in range(world):
+                lists.setdefault(r % tpcp, []).append(r)
+            for ranks in lists.values():
+                pg = dist.new_group(ranks)
+                if rank in ranks:
+                    mine = pg
+            result["dp"] = mine
+        self._groups = result
+        return result

@@ -98,6 +98,33 @@ class AutocastKwargs(KwargsHandler):
 
 
 @dataclass
+class DDPCommunicationHookType(BaseEnum):
+    """Gradient-compression hook selector (reference: dataclasses.py:202-239).
+    Maps onto the reducer's ``comm_dtype`` wire compression; PowerSGD-style
+    low-rank compression is not implemented (xGMI bandwidth makes bf16
+    compression the better latency/accuracy point on MI355X)."""
+
+    NO = "no"
+    FP16 = "fp16"
+    BF16 = "bf16"
+
+
+@dataclass
+class DataLoaderConfiguration:
+    """Bundled dataloader-preparation options (reference: dataclasses.py:823).
+    Passing this to ``Accelerator(dataloader_config=...)`` overrides the
+    individual constructor flags."""
+
+    split_batches: bool = False
+    dispatch_batches: Optional[bool] = None
+    even_batches: bool = True
+    use_seedable_sampler: bool = False
+    data_seed: Optional[int] = None
+    non_blocking: bool = True
+    use_stateful_dataloader: bool = False
+
+
+@dataclass
 class DistributedDataParallelKwargs(KwargsHandler):
     """Knobs for the MI355X DDP reducer (accelerate_amd/parallel/ddp.py).
 
@@ -114,7 +141,14 @@ class DistributedDataParallelKwargs(KwargsHandler):
     static_graph: bool = False
     broadcast_buffers: bool = True
     comm_dtype: Optional[str] = None  # None | "bf16" | "fp16"
+    comm_hook: "DDPCommunicationHookType" = None  # alias for comm_dtype
     average_in_collective: bool = True
+
+    def __post_init__(self):
+        if self.comm_hook is not None and self.comm_dtype is None:
+            hook = DDPCommunicationHookType(self.comm_hook)
+            if hook != DDPCommunicationHookType.NO:
+                self.comm_dtype = hook.value
 
 
 @dataclass

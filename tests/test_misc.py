@@ -111,3 +111,23 @@ def test_local_sgd_single_process_noop():
     model = acc.prepare_model(model)
     with LocalSGD(accelerator=acc, model=model, local_sgd_steps=2) as ls:
         ls.step()
+
+
+def test_parallelism_config_groups_two_process():
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/pconfig_script.py", nproc=2)
+    assert "PCONFIG_PASS" in out
+
+
+def test_dataloader_configuration_overrides():
+    from accelerate_amd import Accelerator, DataLoaderConfiguration
+    from accelerate_amd.state import PartialState
+
+    PartialState._reset_state()
+    cfg = DataLoaderConfiguration(split_batches=True, even_batches=False, non_blocking=False)
+    acc = Accelerator(cpu=True, dataloader_config=cfg)
+    assert acc.split_batches is True
+    assert acc.even_batches is False
+    assert acc.non_blocking is False
+    PartialState._reset_state()
