@@ -80,3 +80,19 @@ def test_launch_respects_config_file(tmp_path):
     r = run_cli("launch", "--config_file", str(cfg_file), str(script))
     assert r.returncode == 0, r.stdout + r.stderr
     assert "2 process(es)" in r.stdout
+
+
+def test_cv_example_runs_cpu():
+    """examples/cv_example.py (reference cv_example parity): one epoch on
+    synthetic images, single CPU process."""
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, str(Path(REPO_ROOT) / "examples" / "cv_example.py"),
+         "--cpu", "--epochs", "1", "--batch_size", "16"],
+        capture_output=True, text=True, timeout=300,
+        env={**os.environ, "PYTHONPATH": REPO_ROOT},
+    )
+    assert r.returncode == 0, r.stderr
+    assert "eval accuracy" in r.stdout
