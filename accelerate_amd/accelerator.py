@@ -79,6 +79,7 @@ class Accelerator:
         kwargs_handlers: Optional[List[Any]] = None,
         fsdp_plugin: Optional[FullyShardedDataParallelPlugin] = None,
         rng_types: Optional[List[Union[str, RNGType]]] = None,
+        parallelism_config=None,
         dispatch_batches: Optional[bool] = None,
         even_batches: bool = True,
         use_seedable_sampler: bool = False,
@@ -131,6 +132,10 @@ class Accelerator:
         self.non_blocking = non_blocking
         self.use_stateful_dataloader = False
         self.data_seed = None
+        # multi-dimensional parallelism descriptor (reference accelerator.py:301)
+        self.parallelism_config = parallelism_config
+        if parallelism_config is not None and self.state.distributed_type != DistributedType.NO:
+            parallelism_config.validate(self.num_processes)
         if dataloader_config is not None:  # bundled options take precedence
             self.split_batches = dataloader_config.split_batches
             self.dispatch_batches = dataloader_config.dispatch_batches
