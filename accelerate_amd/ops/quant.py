@@ -93,7 +93,8 @@ class QuantLinear(nn.Module):
     @classmethod
     def from_linear(cls, linear: nn.Linear, bits=8, group_size=128, compute_dtype=torch.bfloat16):
         m = cls(linear.in_features, linear.out_features, bias=linear.bias is not None,
-                bits=bits, group_size=group_size, compute_dtype=compute_dtype)
+                bits=bits, group_size=group_size, compute_dtype=compute_dtype,
+                device=linear.weight.device)
         if bits == 8:
             q, s = quantize_int8(linear.weight)
         else:
