@@ -437,6 +437,11 @@ class Accelerator:
 
     def prepare_model(self, model: torch.nn.Module, device_placement: bool = None, evaluation_mode: bool = False):
         """(reference: accelerator.py:1769)"""
+        if getattr(model, "_is_accelerate_prepared", False):
+            # double-wrap protection (reference test_accelerator.py:469)
+            if model not in self._models:
+                self._models.append(model)
+            return model
         if device_placement is None:
             device_placement = self.device_placement
 
@@ -468,6 +473,7 @@ class Accelerator:
             # CPU gloo world (tests): same engine, gloo collectives
             ddp_kwargs = self.ddp_handler.to_dict() if self.ddp_handler is not None else {}
             model = DistributedDataParallelEngine(model, **ddp_kwargs)
+        model._is_accelerate_prepared = True
         self._models.append(model)
         return model
 

@@ -325,6 +325,13 @@ class DataLoaderAdapter:
         # spoof isinstance(dl, DataLoader) checks in user code
         return self.base_dataloader.__class__
 
+    def __reduce__(self):
+        # the __class__ spoof breaks default pickling (copyreg builds the
+        # spoofed class); substitute the REAL type in the reconstructor
+        # (reference: data_loader.py DataLoaderAdapter.__reduce__)
+        args = super().__reduce__()
+        return (args[0], (type(self),) + args[1][1:]) + args[2:]
+
     def __len__(self):
         return len(self.base_dataloader)
 

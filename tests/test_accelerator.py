@@ -195,3 +195,31 @@ def test_unwrap_model_removes_fp32_wrapper():
     unwrapped = accelerator.unwrap_model(model, keep_fp32_wrapper=False)
     out = unwrapped(torch.randn(2, 4))
     assert out.dtype == torch.float32
+
+
+def test_prepare_model_twice_returns_same_wrapper():
+    """Double-wrap protection (reference test_accelerator.py:469): preparing
+    an already-prepared model must not re-wrap it."""
+    import torch.nn as nn
+
+    acc = Accelerator(cpu=True)
+    model = nn.Linear(4, 2)
+    m1 = acc.prepare_model(model)
+    m2 = acc.prepare_model(m1)
+    assert m2 is m1
+    assert len(acc._models) == 1
+
+
+def test_prepared_dataloader_is_picklable():
+    """Prepared loaders must survive pickling (reference: test_accelerator
+    dataloader pickling :673) — needed for spawned dataloader workers."""
+    import pickle
+
+    from torch.utils.data import DataLoader, TensorDataset
+
+    acc = Accelerator(cpu=True)
+    dl = acc.prepare(DataLoader(TensorDataset(torch.arange(10).float()), batch_size=2))
+    dl2 = pickle.loads(pickle.dumps(dl))
+    a = [b[0].tolist() for b in dl]
+    b = [b[0].tolist() for b in dl2]
+    assert a == b
