@@ -147,3 +147,54 @@ def is_amd_kernels_available() -> bool:
         return _load_extension(required=False) is not None
     except Exception:
         return False
+
+
+@lru_cache
+def is_rocm_available() -> bool:
+    import torch
+
+    return bool(getattr(torch.version, "hip", None))
+
+
+@lru_cache
+def is_peft_available() -> bool:
+    return _is_package_available("peft")
+
+
+@lru_cache
+def is_torchvision_available() -> bool:
+    return _is_package_available("torchvision")
+
+
+@lru_cache
+def is_timm_available() -> bool:
+    return _is_package_available("timm")
+
+
+@lru_cache
+def is_pandas_available() -> bool:
+    return _is_package_available("pandas")
+
+
+@lru_cache
+def is_matplotlib_available() -> bool:
+    return _is_package_available("matplotlib")
+
+
+@lru_cache
+def is_pytest_available() -> bool:
+    return _is_package_available("pytest")
+
+
+@lru_cache
+def is_triton_available() -> bool:
+    return _is_package_available("triton")
+
+
+@lru_cache
+def is_torchdata_available() -> bool:
+    return _is_package_available("torchdata")
+
+
+def is_weights_only_available() -> bool:
+    return True  # torch >= 2.4 in this stack
