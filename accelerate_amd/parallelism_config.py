@@ -76,9 +76,8 @@ class ParallelismConfig:
         rank = dist.get_rank()
         me = self.coords(rank)
 
-        def ranks_varying(dim_sizes, varying):
+        def ranks_varying(varying):
             """All rank lists where `varying` sweeps and other dims are fixed."""
-            tp, cp, sh, rp = self.tp_size, self.cp_size, self.dp_shard_size, self.dp_replicate_size
             out = []
             for r in range(world):
                 c = self.coords(r)
@@ -100,7 +99,7 @@ class ParallelismConfig:
                 result[dim] = None
                 continue
             mine = None
-            for ranks in ranks_varying(None, dim):
+            for ranks in ranks_varying(dim):
                 pg = dist.new_group(ranks)
                 if rank in ranks:
                     mine = pg
