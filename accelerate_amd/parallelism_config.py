@@ -74,7 +74,6 @@ class ParallelismConfig:
         world = dist.get_world_size()
         self.validate(world)
         rank = dist.get_rank()
-        me = self.coords(rank)
 
         def ranks_varying(varying):
             """All rank lists where `varying` sweeps and other dims are fixed."""
