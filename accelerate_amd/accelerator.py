@@ -674,6 +674,110 @@ class Accelerator:
     def print(self, *args, **kwargs):
         self.state.print(*args, **kwargs)
 
+    @property
+    def optimizer_step_was_skipped(self) -> bool:
+        """True if any prepared optimizer skipped its last step (fp16 inf/nan
+        or accumulation gating) — reference accelerator.py property."""
+        return any(getattr(opt, "step_was_skipped", False) for opt in self._optimizers)
+
+    @property
+    def multi_device(self) -> bool:
+        return self.use_distributed and self.num_processes > 1
+
+    @property
+    def is_fsdp2(self) -> bool:
+        """Our sharded engine IS the FSDP2-equivalent (single implementation,
+        no FSDP1/FSDP2 split) — true whenever FSDP is active."""
+        return self.distributed_type == DistributedType.FSDP
+
+    @property
+    def fp8_backend(self):
+        """'NATIVE' when our CDNA4 fp8 stack is active (the reference returns
+        TE/AO/MSAMP — all replaced by ops/fp8.py here)."""
+        return "NATIVE" if self.mixed_precision == "fp8" else None
+
+    @property
+    def should_save_model(self) -> bool:
+        """Whether THIS rank must write model weights in save_model
+        (sharded state dicts write per-rank; full dicts write on main)."""
+        plugin = getattr(self.state, "fsdp_plugin", None)
+        if plugin is not None and getattr(plugin, "state_dict_type", "") == "SHARDED_STATE_DICT":
+            return True
+        return self.is_main_process
+
+    # -- parallelism-dimension ranks (reference accelerator.py properties) --
+    def _pc_coord(self, dim: str) -> int:
+        pc = self.parallelism_config
+        if pc is None:
+            return 0
+        return pc.coords(self.process_index)[dim]
+
+    @property
+    def data_parallel_rank(self) -> int:
+        pc = self.parallelism_config
+        if pc is None:
+            return self.process_index
+        c = pc.coords(self.process_index)
+        return c["dp_replicate"] * pc.dp_shard_size + c["dp_shard"]
+
+    @property
+    def data_parallel_shard_rank(self) -> int:
+        return self._pc_coord("dp_shard")
+
+    @property
+    def tensor_parallel_rank(self) -> int:
+        return self._pc_coord("tp")
+
+    @property
+    def context_parallel_rank(self) -> int:
+        return self._pc_coord("cp")
+
+    @property
+    def torch_device_mesh(self):
+        """The per-dimension process groups built from parallelism_config
+        (our mesh equivalent; None until build_groups has run)."""
+        pc = self.parallelism_config
+        return pc._groups if pc is not None and pc._groups else None
+
+    def skip_first_batches(self, dataloader, num_batches: int = 0):
+        """Mid-epoch resume helper (reference accelerator.py:3568)."""
+        from .data_loader import skip_first_batches
+
+        return skip_first_batches(dataloader, num_batches)
+
+    def trigger_sync_in_backward(self, model):
+        """Force gradient sync on the NEXT backward even inside a no_sync /
+        accumulation window (reference accelerator.py:1210-1227) — used to
+        flush accumulated grads early, e.g. before an optimizer swap."""
+        from .parallel.ddp import DistributedDataParallelEngine
+
+        unwrapped = model
+        if isinstance(model, DistributedDataParallelEngine):
+            model.require_backward_grad_sync = True
+            return
+        for m in self._models:
+            if isinstance(m, DistributedDataParallelEngine) and (m is model or m.module is unwrapped):
+                m.require_backward_grad_sync = True
+                return
+
+    @contextmanager
+    def maybe_context_parallel(self, buffers=None, buffer_seq_dims=None):
+        """Shard the given buffers along their sequence dim for this step when
+        a context-parallel group is active (reference accelerator.py:4111
+        maybe_context_parallel); no-op otherwise. Buffers are replaced
+        IN PLACE in the caller's list."""
+        pc = self.parallelism_config
+        group = None
+        if pc is not None and pc.cp_size > 1 and pc._groups:
+            group = pc._groups.get("cp")
+        if group is not None and buffers:
+            from .parallel.cp import shard_sequence
+
+            dims = buffer_seq_dims or [1] * len(buffers)
+            for i, (buf, dim) in enumerate(zip(buffers, dims)):
+                buffers[i] = shard_sequence(buf, group=group, dim=dim)
+        yield
+
     @contextmanager
     def autocast(self, autocast_handler: AutocastKwargs = None):
         """bf16/fp16 autocast context over the ROCm HIP autocast dispatcher

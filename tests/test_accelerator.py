@@ -223,3 +223,33 @@ def test_prepared_dataloader_is_picklable():
     a = [b[0].tolist() for b in dl]
     b = [b[0].tolist() for b in dl2]
     assert a == b
+
+
+def test_parallelism_rank_properties():
+    from accelerate_amd import ParallelismConfig
+
+    acc = Accelerator(cpu=True)
+    assert acc.data_parallel_rank == acc.process_index
+    assert acc.tensor_parallel_rank == 0 and acc.context_parallel_rank == 0
+    acc.parallelism_config = ParallelismConfig(tp_size=1)
+    assert acc.torch_device_mesh is None
+    assert acc.multi_device is False
+    assert acc.optimizer_step_was_skipped is False
+    assert acc.fp8_backend is None
+    assert acc.should_save_model is True
+
+
+def test_skip_first_batches_method():
+    from torch.utils.data import DataLoader, TensorDataset
+
+    acc = Accelerator(cpu=True)
+    dl = acc.prepare(DataLoader(TensorDataset(torch.arange(10).float()), batch_size=2))
+    rest = acc.skip_first_batches(dl, 2)
+    assert [b[0].tolist() for b in rest] == [[4.0, 5.0], [6.0, 7.0], [8.0, 9.0]]
+
+
+def test_maybe_context_parallel_noop():
+    acc = Accelerator(cpu=True)
+    bufs = [torch.randn(2, 8)]
+    with acc.maybe_context_parallel(buffers=bufs):
+        assert bufs[0].shape == (2, 8)  # world 1: untouched
