@@ -28,6 +28,18 @@ def add_parser(subparsers):
     parser.add_argument("--cpu", action="store_true", help="Force CPU-only")
     parser.add_argument("--multi_gpu", action="store_true", help="Force multi-GPU even with config absent")
     parser.add_argument("--use_fsdp", action="store_true", help="Use the sharded-parameter engine")
+    parser.add_argument("--fsdp_sharding_strategy", default=None,
+                        help="full_shard | hybrid_shard (reference --fsdp_sharding_strategy)")
+    parser.add_argument("--fsdp_transformer_layer_cls_to_wrap", default=None,
+                        help="Comma-separated module class names forming shard units")
+    parser.add_argument("--fsdp_shard_group_size", type=int, default=None,
+                        help="HSDP: ranks per shard group (replicas = world/size)")
+    parser.add_argument("--fsdp_state_dict_type", default=None,
+                        help="FULL_STATE_DICT | SHARDED_STATE_DICT")
+    parser.add_argument("--fsdp_activation_checkpointing", default=None,
+                        help="true|false: checkpoint each wrapped unit")
+    parser.add_argument("--num_cpu_threads_per_process", type=int, default=None,
+                        help="Sets OMP_NUM_THREADS per worker")
     parser.add_argument("--gpu_ids", default=None, help="Comma-separated HIP device ids to use")
     parser.add_argument("--gradient_accumulation_steps", type=int, default=None)
     parser.add_argument("--debug", action="store_true")
@@ -97,6 +109,19 @@ def build_env(args, config: ClusterConfig) -> dict:
         env["ACCELERATE_USE_FSDP"] = "1"
         for key, value in (config.fsdp_config or {}).items():
             env[key.replace("fsdp_", "FSDP_").upper()] = str(value)
+        # CLI flags override the config file (reference flag precedence)
+        cli_fsdp = {
+            "FSDP_SHARDING_STRATEGY": args.fsdp_sharding_strategy,
+            "FSDP_TRANSFORMER_CLS_TO_WRAP": args.fsdp_transformer_layer_cls_to_wrap,
+            "FSDP_SHARD_GROUP_SIZE": args.fsdp_shard_group_size,
+            "FSDP_STATE_DICT_TYPE": args.fsdp_state_dict_type,
+            "FSDP_ACTIVATION_CHECKPOINTING": args.fsdp_activation_checkpointing,
+        }
+        for key, value in cli_fsdp.items():
+            if value is not None:
+                env[key] = str(value)
+    if args.num_cpu_threads_per_process is not None:
+        env["OMP_NUM_THREADS"] = str(args.num_cpu_threads_per_process)
     if args.gpu_ids not in (None, "all"):
         env["HIP_VISIBLE_DEVICES"] = str(args.gpu_ids)
     if config.enable_cpu_affinity:
