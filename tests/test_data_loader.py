@@ -2,6 +2,8 @@
 exhaustive small-N enumeration strategy, tests/test_data_loader.py)."""
 
 import pytest
+from hypothesis import given, settings
+from hypothesis import strategies as st
 import torch
 from torch.utils.data import BatchSampler, DataLoader, IterableDataset, SequentialSampler, TensorDataset
 
@@ -202,3 +204,53 @@ class TestSkip:
         new_dl = skip_first_batches(dl, num_batches=2)
         seen = torch.cat([b[0] for b in new_dl])
         assert torch.equal(seen, torch.arange(8, 16).float())
+
+
+class TestBatchSamplerShardProperties:
+    """Property-based invariant sweep (hypothesis): across the whole
+    parameter grid, per-rank shards must partition the sample stream with
+    the documented padding/drop semantics (reference tests enumerate these
+    by hand; the sweep covers the space)."""
+
+    @given(
+        n_samples=st.integers(1, 64),
+        batch_size=st.integers(1, 8),
+        n_ranks=st.integers(1, 4),
+        split_batches=st.booleans(),
+        even_batches=st.booleans(),
+        drop_last=st.booleans(),
+    )
+    @settings(max_examples=200, deadline=None)
+    def test_partition_invariants(self, n_samples, batch_size, n_ranks, split_batches, even_batches, drop_last):
+        from torch.utils.data import BatchSampler, SequentialSampler
+
+        from accelerate_amd.data_loader import BatchSamplerShard
+
+        if split_batches and batch_size % n_ranks != 0:
+            return  # documented constraint (raises)
+        base = BatchSampler(SequentialSampler(range(n_samples)), batch_size, drop_last)
+        shards = [
+            list(BatchSamplerShard(base, num_processes=n_ranks, process_index=r,
+                                   split_batches=split_batches, even_batches=even_batches))
+            for r in range(n_ranks)
+        ]
+        # 1. with even_batches every rank yields the same number of batches
+        #    (lockstep collectives); even_batches=False may be ragged — that
+        #    is exactly the join_uneven_inputs use case
+        if even_batches:
+            lens = {len(s) for s in shards}
+            assert len(lens) == 1, f"ragged shard lengths {lens}"
+        # 2. batches on one rank all have equal size when even_batches
+        if even_batches and not drop_last:
+            sizes = {len(b) for s in shards for b in s}
+            assert len(sizes) <= 1, sizes
+        # 3. union of yielded indices ⊆ dataset, and ⊇ dataset when nothing
+        #    is dropped (drop_last may drop a tail; padding duplicates allowed)
+        seen = {i for s in shards for b in s for i in b}
+        assert seen <= set(range(n_samples))
+        if not drop_last and sum(len(s) for s in shards) > 0:
+            assert seen == set(range(n_samples)), "lost samples without drop_last"
+        # 4. with drop_last, no duplicates at all
+        if drop_last:
+            flat = [i for s in shards for b in s for i in b]
+            assert len(flat) == len(set(flat))
