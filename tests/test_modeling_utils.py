@@ -3,6 +3,8 @@ import os
 import tempfile
 
 import pytest
+from hypothesis import given, settings
+from hypothesis import strategies as st
 import torch
 import torch.nn as nn
 
@@ -201,3 +203,71 @@ def test_llama70b_device_map_plans_on_8x288gb():
         dtype=torch.bfloat16,
     )
     assert set(dmap1.values()) == {0}, f"70B bf16 should fit one 288GB GPU: {set(dmap1.values())}"
+
+
+class TestDeviceMapSolverProperties:
+    """Property sweep over infer_auto_device_map (reference tests enumerate
+    hand cases; this machine-checks the invariants across random models)."""
+
+    @given(
+        n_layers=st.integers(1, 12),
+        hidden=st.sampled_from([8, 16, 32]),
+        n_gpus=st.integers(1, 4),
+        headroom=st.floats(1.1, 4.0),
+    )
+    @settings(max_examples=60, deadline=None)
+    def test_solver_invariants(self, n_layers, hidden, n_gpus, headroom):
+        import torch.nn as nn
+
+        from accelerate_amd.utils import compute_module_sizes, infer_auto_device_map
+
+        class Block(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.fc1 = nn.Linear(hidden, hidden)
+                self.fc2 = nn.Linear(hidden, hidden)
+
+        class Model(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.embed = nn.Embedding(32, hidden)
+                self.blocks = nn.ModuleList(Block() for _ in range(n_layers))
+                self.head = nn.Linear(hidden, 32)
+
+        model = Model()
+        sizes = compute_module_sizes(model)
+        total = sizes[""]
+        per_gpu = int(total / n_gpus * headroom) + 1
+        max_memory = {i: per_gpu for i in range(n_gpus)}
+        max_memory["cpu"] = total * 2  # overflow room
+        dmap = infer_auto_device_map(
+            model, max_memory=max_memory, no_split_module_classes=["Block"]
+        )
+        # 1. every parameter is assigned through exactly one map entry
+        assigned = set()
+        for name, _ in model.named_parameters():
+            owners = [e for e in dmap if name == e or name.startswith(e + ".")] + (
+                [""] if "" in dmap else []
+            )
+            assert owners, f"{name} unassigned"
+            assigned.add(max(owners, key=len))
+        # 2. no-split blocks are never split across devices
+        for name in dmap:
+            for other in dmap:
+                if other != name and other.startswith(name + "."):
+                    raise AssertionError(f"nested map entries {name} / {other}")
+        block_devices = {}
+        for entry, dev in dmap.items():
+            for i in range(n_layers):
+                prefix = f"blocks.{i}"
+                if entry == prefix or entry.startswith(prefix + "."):
+                    block_devices.setdefault(i, set()).add(dev)
+        for i, devs in block_devices.items():
+            assert len(devs) == 1, f"Block {i} split across {devs}"
+        # 3. per-device totals respect max_memory for GPU entries
+        per_dev = {}
+        for entry, dev in dmap.items():
+            per_dev[dev] = per_dev.get(dev, 0) + sizes.get(entry, 0)
+        for dev, used in per_dev.items():
+            if isinstance(dev, int):
+                assert used <= max_memory[dev] * 1.0 + 1, (dev, used, max_memory[dev])
