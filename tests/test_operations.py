@@ -1,5 +1,7 @@
 import numpy as np
 import pytest
+from hypothesis import given, settings
+from hypothesis import strategies as st
 import torch
 
 from accelerate_amd.state import PartialState
@@ -105,3 +107,61 @@ def test_concatenate():
     data = [{"x": torch.ones(2, 3)}, {"x": torch.zeros(1, 3)}]
     out = concatenate(data)
     assert out["x"].shape == (3, 3)
+
+
+class TestStructuredOpsProperties:
+    """Property sweep: structure-preserving ops must round-trip arbitrary
+    nested (dict/list/tuple/tensor) payloads (reference test_utils.py ops)."""
+
+    @staticmethod
+    def _payload(depth_seed, shape):
+        t = torch.arange(float(shape[0] * shape[1])).reshape(shape)
+        if depth_seed % 3 == 0:
+            return {"a": t, "b": [t + 1, (t + 2,)]}
+        if depth_seed % 3 == 1:
+            return [t, {"x": t * 2}]
+        return (t, t + 5)
+
+    @given(depth_seed=st.integers(0, 8), rows=st.integers(1, 6), cols=st.integers(1, 6))
+    @settings(max_examples=50, deadline=None)
+    def test_find_device_and_send(self, depth_seed, rows, cols):
+        from accelerate_amd.utils.operations import find_device, send_to_device
+
+        data = self._payload(depth_seed, (rows, cols))
+        assert find_device(data).type == "cpu"
+        out = send_to_device(data, torch.device("cpu"))
+        flat_in, flat_out = [], []
+
+        def collect(x, acc):
+            if isinstance(x, torch.Tensor):
+                acc.append(x)
+            elif isinstance(x, dict):
+                [collect(v, acc) for v in x.values()]
+            elif isinstance(x, (list, tuple)):
+                [collect(v, acc) for v in x]
+
+        collect(data, flat_in)
+        collect(out, flat_out)
+        assert len(flat_in) == len(flat_out)
+        for a, b in zip(flat_in, flat_out):
+            assert torch.equal(a, b)
+
+    @given(rows=st.integers(1, 7), cols=st.integers(1, 5), pad_index=st.integers(0, 3))
+    @settings(max_examples=50, deadline=None)
+    def test_pad_across_processes_world1(self, rows, cols, pad_index):
+        from accelerate_amd.utils.operations import pad_across_processes
+
+        t = torch.randn(rows, cols)
+        out = pad_across_processes(t, dim=0, pad_index=pad_index)
+        # world 1: max length == own length, content unchanged
+        assert torch.equal(out, t)
+
+    @given(n=st.integers(1, 5))
+    @settings(max_examples=30, deadline=None)
+    def test_concatenate_matches_torch_cat(self, n):
+        from accelerate_amd.utils.operations import concatenate
+
+        parts = [{"x": torch.randn(2, 3), "y": (torch.randn(4),)} for _ in range(n)]
+        out = concatenate(parts)
+        assert out["x"].shape == (2 * n, 3)
+        assert out["y"][0].shape == (4 * n,)
