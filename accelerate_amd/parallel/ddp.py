@@ -141,6 +141,11 @@ class DistributedDataParallelEngine(nn.Module):
         self.comm_dtype = _COMM_DTYPES[comm_dtype] if isinstance(comm_dtype, (str, type(None))) else comm_dtype
         self.bucket_cap_bytes = int(bucket_cap_mb * 1024 * 1024)
         self._world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        # broadcast source must be a MEMBER of the group (global rank 0 is
+        # not in non-trivial subgroups, e.g. the dp group of tp-rank 1)
+        self._src_rank = 0
+        if dist.is_initialized() and process_group is not None:
+            self._src_rank = dist.get_global_rank(process_group, 0)
         # AVG keeps the division inside the RCCL kernel; gloo lacks it.
         self._use_avg = (
             average_in_collective
@@ -194,7 +199,7 @@ class DistributedDataParallelEngine(nn.Module):
         (reference: DDP _sync_module_states broadcast)."""
         for t in list(self.module.parameters()) + list(self.module.buffers()):
             if t.numel() > 0 and not getattr(t, "_no_ddp_sync", False):
-                dist.broadcast(t.data, src=0, group=self.process_group)
+                dist.broadcast(t.data, src=self._src_rank, group=self.process_group)
 
     # -- steady state ------------------------------------------------------
 
@@ -279,7 +284,7 @@ class DistributedDataParallelEngine(nn.Module):
             if self._float_buffers is None:
                 self._float_buffers = [b for b in self.module.buffers() if b.is_floating_point() and b.numel() > 0]
             for buf in self._float_buffers:
-                dist.broadcast(buf.data, src=0, group=self.process_group)
+                dist.broadcast(buf.data, src=self._src_rank, group=self.process_group)
         return self.module(*args, **kwargs)
 
     @contextmanager

@@ -73,3 +73,12 @@ def test_ep_two_process_gloo():
     assert "EP_FWD_PASS" in out
     assert "EP_GRAD_PASS" in out
     assert "EP_DDP_PASS" in out
+
+
+def test_tp_dp_composition_four_process():
+    """ParallelismConfig groups drive TP sharding + DP replication together
+    (4 gloo procs) — parity vs a single-process reference."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/tp_dp_script.py", nproc=4)
+    assert "TP_DP_COMPOSE_PASS" in out
