@@ -1,6 +1,7 @@
 from .bert import BertConfig, BertForSequenceClassification
 from .gpt2 import GPT2Config, GPT2LMHeadModel
 from .llama import LlamaConfig, LlamaForCausalLM
+from .llama_moe import LlamaMoEConfig, LlamaMoEForCausalLM
 
 __all__ = [
     "BertConfig",
@@ -9,4 +10,6 @@ __all__ = [
     "GPT2LMHeadModel",
     "LlamaConfig",
     "LlamaForCausalLM",
+    "LlamaMoEConfig",
+    "LlamaMoEForCausalLM",
 ]

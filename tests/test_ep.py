@@ -82,3 +82,29 @@ def test_tp_dp_composition_four_process():
 
     out = launch_distributed("tests/distributed_scripts/tp_dp_script.py", nproc=4)
     assert "TP_DP_COMPOSE_PASS" in out
+
+
+def test_moe_llama_single_process_trains():
+    from accelerate_amd.models.llama_moe import LlamaMoEConfig, LlamaMoEForCausalLM
+
+    torch.manual_seed(0)
+    model = LlamaMoEForCausalLM(LlamaMoEConfig.tiny_moe())
+    ids = torch.randint(0, 1024, (2, 16))
+    out = model(ids, labels=ids)
+    assert out["logits"].shape == (2, 16, 1024)
+    assert torch.isfinite(out["loss"]) and out["aux_loss"].item() >= 0
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    first = out["loss"].item()
+    for _ in range(6):
+        opt.zero_grad()
+        loss = model(ids, labels=ids)["loss"]
+        loss.backward()
+        opt.step()
+    assert loss.item() < first
+
+
+def test_moe_llama_ep_two_process():
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/moe_model_script.py", nproc=2)
+    assert "MOE_MODEL_PASS" in out
