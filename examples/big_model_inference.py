@@ -7,6 +7,11 @@ load_checkpoint_and_dispatch with CPU/disk offload hooks → generate
 """
 
 import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import tempfile
 
 import torch

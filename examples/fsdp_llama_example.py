@@ -8,6 +8,11 @@
 
 import argparse
 import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import os
 
 import torch
 from torch.utils.data import DataLoader, TensorDataset

@@ -7,6 +7,11 @@ Run:
 """
 
 import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 
 import torch
 from torch.utils.data import DataLoader, TensorDataset
