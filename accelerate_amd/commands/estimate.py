@@ -9,7 +9,27 @@ from ..utils.modeling import compute_module_sizes
 from ..utils.other import convert_bytes
 
 
+# bundled model families resolvable fully offline (models/):
+#   name -> (config factory, model class path)
+_BUILTIN = {
+    "bert-base": ("accelerate_amd.models", "BertConfig", "bert_base", "BertForSequenceClassification"),
+    "gpt2-large": ("accelerate_amd.models", "GPT2Config", "gpt2_large", "GPT2LMHeadModel"),
+    "llama3-8b": ("accelerate_amd.models", "LlamaConfig", "llama3_8b", "LlamaForCausalLM"),
+    "llama3-70b": ("accelerate_amd.models", "LlamaConfig", "llama3_70b", "LlamaForCausalLM"),
+    "mixtral-8x7b": ("accelerate_amd.models", "LlamaMoEConfig", "mixtral_8x7b_shape", "LlamaMoEForCausalLM"),
+    "t5-11b": ("accelerate_amd.models", "T5Config", "t5_11b", "T5ForConditionalGeneration"),
+}
+
+
 def create_empty_model(model_name: str, trust_remote_code: bool = False):
+    if model_name in _BUILTIN:
+        import importlib
+
+        mod_name, cfg_cls, factory, model_cls = _BUILTIN[model_name]
+        mod = importlib.import_module(mod_name)
+        config = getattr(getattr(mod, cfg_cls), factory)()
+        with init_empty_weights():
+            return getattr(mod, model_cls)(config)
     import transformers
 
     config = transformers.AutoConfig.from_pretrained(model_name, trust_remote_code=trust_remote_code)
@@ -38,7 +58,11 @@ def estimate_command(args):
 
 def add_parser(subparsers):
     parser = subparsers.add_parser("estimate", help="Estimate model memory usage")
-    parser.add_argument("model_name", help="transformers model name or local path")
+    parser.add_argument(
+        "model_name",
+        help="bundled family (bert-base, gpt2-large, llama3-8b, llama3-70b, "
+        "mixtral-8x7b, t5-11b), transformers model name, or local path",
+    )
     parser.add_argument("--dtypes", nargs="+", default=["float32", "float16", "int8"],
                         choices=["float32", "float16", "bfloat16", "int8", "fp8"])
     parser.add_argument("--trust_remote_code", action="store_true")

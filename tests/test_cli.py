@@ -96,3 +96,14 @@ def test_cv_example_runs_cpu():
     )
     assert r.returncode == 0, r.stderr
     assert "eval accuracy" in r.stdout
+
+
+def test_estimate_bundled_families():
+    """estimate resolves the bundled model families fully offline; the
+    llama3-70b bf16 figure must match the measured 131.4 GB residency
+    (BENCHMARKS.md 70B demo)."""
+    r = run_cli("estimate", "llama3-70b", "--dtypes", "bfloat16")
+    assert r.returncode == 0, r.stderr
+    assert "131.4" in r.stdout
+    r2 = run_cli("estimate", "mixtral-8x7b", "--dtypes", "bfloat16")
+    assert r2.returncode == 0, r2.stderr
