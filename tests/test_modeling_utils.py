@@ -271,3 +271,35 @@ class TestDeviceMapSolverProperties:
         for dev, used in per_dev.items():
             if isinstance(dev, int):
                 assert used <= max_memory[dev] * 1.0 + 1, (dev, used, max_memory[dev])
+
+
+class TestShardCheckpointProperties:
+    """shard_checkpoint invariants (reference modeling.py:700): every tensor
+    lands in exactly one shard, shard sizes respect the cap (except a single
+    oversize tensor), and the index maps every key to its shard."""
+
+    @given(
+        n_tensors=st.integers(1, 12),
+        rows=st.integers(1, 64),
+        cap_kb=st.integers(1, 64),
+    )
+    @settings(max_examples=50, deadline=None)
+    def test_invariants(self, n_tensors, rows, cap_kb):
+        from accelerate_amd.utils import shard_checkpoint
+
+        sd = {f"w{i}": torch.randn(rows, 16) for i in range(n_tensors)}
+        cap = cap_kb * 1024
+        shards, index = shard_checkpoint(sd, max_shard_size=cap)
+        # every key appears in exactly one shard
+        all_keys = [k for shard in shards.values() for k in shard]
+        assert sorted(all_keys) == sorted(sd)
+        for name, shard in shards.items():
+            size = sum(t.numel() * t.element_size() for t in shard.values())
+            if len(shard) > 1:
+                assert size <= cap, (name, size, cap)
+        if index is not None:
+            assert sorted(index["weight_map"]) == sorted(sd)
+            for k, fname in index["weight_map"].items():
+                assert k in shards[fname]
+        else:
+            assert len(shards) == 1
