@@ -49,12 +49,21 @@ class GPT2Attention(nn.Module):
         self.c_attn = nn.Linear(config.hidden_size, 3 * config.hidden_size)
         self.c_proj = nn.Linear(config.hidden_size, config.hidden_size)
 
-    def forward(self, x):
+    def forward(self, x, kv_cache=None):
         B, S, H = x.shape
         q, k, v = self.c_attn(x).split(H, dim=2)
         q = q.view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
         k = k.view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
         v = v.view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
+        past = 0
+        if kv_cache is not None:
+            past = kv_cache["len"]
+            kv_cache["k"][:, :, past : past + S] = k
+            kv_cache["v"][:, :, past : past + S] = v
+            kv_cache["len"] = past + S
+            k = kv_cache["k"][:, :, : past + S]
+            v = kv_cache["v"][:, :, : past + S]
+        total = k.shape[2]
         if x.is_cuda and S > 1:
             from ..ops.attention import flash_attention
 
@@ -62,8 +71,9 @@ class GPT2Attention(nn.Module):
         else:
             scale = 1.0 / math.sqrt(self.head_dim)
             scores = torch.matmul(q, k.transpose(-1, -2)) * scale
-            mask = torch.ones(S, S, dtype=torch.bool, device=x.device).tril()
-            scores = scores.masked_fill(~mask, torch.finfo(scores.dtype).min)
+            if S > 1:
+                mask = torch.ones(S, total, dtype=torch.bool, device=x.device).tril(past)
+                scores = scores.masked_fill(~mask, torch.finfo(scores.dtype).min)
             probs = F.softmax(scores.float(), dim=-1).to(v.dtype)
             ctx = torch.matmul(probs, v)
         ctx = ctx.transpose(1, 2).reshape(B, S, H)
@@ -88,8 +98,8 @@ class GPT2Block(nn.Module):
         self.ln_2 = FusedLayerNorm(config.hidden_size, eps=config.layer_norm_eps)
         self.mlp = GPT2MLP(config)
 
-    def forward(self, x):
-        x = x + self.attn(self.ln_1(x))
+    def forward(self, x, kv_cache=None):
+        x = x + self.attn(self.ln_1(x), kv_cache)
         x = x + self.mlp(self.ln_2(x))
         return x
 
@@ -112,12 +122,13 @@ class GPT2LMHeadModel(nn.Module):
             if isinstance(m, nn.Linear) and m.bias is not None:
                 nn.init.zeros_(m.bias)
 
-    def forward(self, input_ids, labels=None):
+    def forward(self, input_ids, labels=None, kv_caches=None):
         B, S = input_ids.shape
-        pos = torch.arange(S, device=input_ids.device)
+        past = kv_caches[0]["len"] if kv_caches is not None else 0
+        pos = torch.arange(past, past + S, device=input_ids.device)
         x = self.wte(input_ids) + self.wpe(pos)[None]
-        for block in self.h:
-            x = block(x)
+        for i, block in enumerate(self.h):
+            x = block(x, kv_caches[i] if kv_caches is not None else None)
         x = self.ln_f(x)
         logits = self.lm_head(x)
         out = {"logits": logits}
@@ -129,11 +140,29 @@ class GPT2LMHeadModel(nn.Module):
 
     @torch.no_grad()
     def generate(self, input_ids, max_new_tokens: int = 32):
-        """Greedy decode (full recompute — GPT-2 contexts are short; the
-        KV-cache decode path lives in the Llama family)."""
+        """Greedy decode with per-layer KV caches (prefill once, then one
+        position per step — each decode step reads the weights once)."""
+        c = self.config
+        B = input_ids.shape[0]
+        max_len = min(c.max_position_embeddings, input_ids.shape[1] + max_new_tokens)
+        dtype = self.wte.weight.dtype
+        head_dim = c.hidden_size // c.num_attention_heads
+        caches = [
+            {
+                "k": torch.zeros(B, c.num_attention_heads, max_len, head_dim,
+                                 device=input_ids.device, dtype=dtype),
+                "v": torch.zeros(B, c.num_attention_heads, max_len, head_dim,
+                                 device=input_ids.device, dtype=dtype),
+                "len": 0,
+            }
+            for _ in range(c.num_hidden_layers)
+        ]
         ids = input_ids
+        logits = self.forward(ids[:, -max_len:], kv_caches=caches)["logits"]
         for _ in range(max_new_tokens):
-            ctx = ids[:, -self.config.max_position_embeddings :]
-            logits = self.forward(ctx)["logits"]
-            ids = torch.cat([ids, logits[:, -1].argmax(-1, keepdim=True)], dim=1)
+            nxt = logits[:, -1].argmax(-1, keepdim=True)
+            ids = torch.cat([ids, nxt], dim=1)
+            if caches[0]["len"] >= max_len:
+                break
+            logits = self.forward(nxt, kv_caches=caches)["logits"]
         return ids

@@ -71,3 +71,18 @@ def test_gpt2_gpu_train_bf16():
     torch.cuda.synchronize()
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]
+
+
+def test_generate_cached_matches_full_recompute():
+    """KV-cache decode must produce the same tokens as full recompute."""
+    torch.manual_seed(0)
+    model = GPT2LMHeadModel(GPT2Config.tiny()).eval()
+    ids = torch.randint(0, 1024, (2, 8))
+    cached = model.generate(ids, max_new_tokens=6)
+    # full-recompute reference
+    ref = ids
+    with torch.no_grad():
+        for _ in range(6):
+            logits = model(ref)["logits"]
+            ref = torch.cat([ref, logits[:, -1].argmax(-1, keepdim=True)], dim=1)
+    assert torch.equal(cached, ref)
