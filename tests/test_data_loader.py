@@ -220,7 +220,7 @@ class TestBatchSamplerShardProperties:
         even_batches=st.booleans(),
         drop_last=st.booleans(),
     )
-    @settings(max_examples=200, deadline=None)
+    @settings(max_examples=200, deadline=None, derandomize=True)
     def test_partition_invariants(self, n_samples, batch_size, n_ranks, split_batches, even_batches, drop_last):
         from torch.utils.data import BatchSampler, SequentialSampler
 

@@ -215,7 +215,7 @@ class TestDeviceMapSolverProperties:
         n_gpus=st.integers(1, 4),
         headroom=st.floats(1.1, 4.0),
     )
-    @settings(max_examples=60, deadline=None)
+    @settings(max_examples=60, deadline=None, derandomize=True)
     def test_solver_invariants(self, n_layers, hidden, n_gpus, headroom):
         import torch.nn as nn
 
@@ -283,7 +283,7 @@ class TestShardCheckpointProperties:
         rows=st.integers(1, 64),
         cap_kb=st.integers(1, 64),
     )
-    @settings(max_examples=50, deadline=None)
+    @settings(max_examples=50, deadline=None, derandomize=True)
     def test_invariants(self, n_tensors, rows, cap_kb):
         from accelerate_amd.utils import shard_checkpoint
 

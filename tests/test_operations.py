@@ -123,7 +123,7 @@ class TestStructuredOpsProperties:
         return (t, t + 5)
 
     @given(depth_seed=st.integers(0, 8), rows=st.integers(1, 6), cols=st.integers(1, 6))
-    @settings(max_examples=50, deadline=None)
+    @settings(max_examples=50, deadline=None, derandomize=True)
     def test_find_device_and_send(self, depth_seed, rows, cols):
         from accelerate_amd.utils.operations import find_device, send_to_device
 
@@ -147,7 +147,7 @@ class TestStructuredOpsProperties:
             assert torch.equal(a, b)
 
     @given(rows=st.integers(1, 7), cols=st.integers(1, 5), pad_index=st.integers(0, 3))
-    @settings(max_examples=50, deadline=None)
+    @settings(max_examples=50, deadline=None, derandomize=True)
     def test_pad_across_processes_world1(self, rows, cols, pad_index):
         from accelerate_amd.utils.operations import pad_across_processes
 
@@ -157,7 +157,7 @@ class TestStructuredOpsProperties:
         assert torch.equal(out, t)
 
     @given(n=st.integers(1, 5))
-    @settings(max_examples=30, deadline=None)
+    @settings(max_examples=30, deadline=None, derandomize=True)
     def test_concatenate_matches_torch_cat(self, n):
         from accelerate_amd.utils.operations import concatenate
 
