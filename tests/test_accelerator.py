@@ -253,3 +253,26 @@ def test_maybe_context_parallel_noop():
     bufs = [torch.randn(2, 8)]
     with acc.maybe_context_parallel(buffers=bufs):
         assert bufs[0].shape == (2, 8)  # world 1: untouched
+
+
+def test_automatic_checkpoint_naming_and_total_limit(tmp_path):
+    """save_state rotation (reference: ProjectConfiguration total_limit):
+    automatic checkpoint_N naming, oldest folders deleted past the limit."""
+    import os
+
+    from accelerate_amd.utils import ProjectConfiguration
+
+    acc = Accelerator(
+        cpu=True,
+        project_config=ProjectConfiguration(
+            project_dir=str(tmp_path), automatic_checkpoint_naming=True, total_limit=2
+        ),
+    )
+    model = torch.nn.Linear(4, 2)
+    model = acc.prepare(model)
+    for _ in range(3):
+        acc.save_state()
+    ckpts = sorted(os.listdir(tmp_path / "checkpoints"))
+    assert ckpts == ["checkpoint_1", "checkpoint_2"], ckpts  # checkpoint_0 rotated out
+    # load back the newest
+    acc.load_state(str(tmp_path / "checkpoints" / "checkpoint_2"))
