@@ -30,21 +30,7 @@ from .utils.imports import is_safetensors_available
 logger = get_logger(__name__)
 
 
-def save(obj, f, save_on_each_node: bool = False, safe_serialization: bool = False):
-    """Save ``obj`` on the main process (or every node's main process)."""
-    from .state import PartialState
-
-    state = PartialState()
-    if safe_serialization:
-        import safetensors.torch
-
-        save_func = lambda obj, f: safetensors.torch.save_file(obj, f, metadata={"format": "pt"})
-    else:
-        save_func = torch.save
-    if state.is_main_process and not save_on_each_node:
-        save_func(obj, f)
-    elif state.is_local_main_process and save_on_each_node:
-        save_func(obj, f)
+from .utils.other import save  # noqa: F401  (canonical home: utils.other)
 
 
 def save_accelerator_state(

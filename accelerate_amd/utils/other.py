@@ -199,3 +199,23 @@ def write_basic_config(mixed_precision: str = "no", save_location: str = None):
     with open(save_location, "w") as f:
         yaml.safe_dump(config, f)
     return save_location
+
+
+def save(obj, f, save_on_each_node: bool = False, safe_serialization: bool = False):
+    """Save ``obj`` on the main process — or every node's main process
+    (reference: utils/other.py save; re-exported by checkpointing)."""
+    import torch
+
+    from ..state import PartialState
+
+    state = PartialState()
+    if safe_serialization:
+        import safetensors.torch
+
+        save_func = lambda obj, f: safetensors.torch.save_file(obj, f, metadata={"format": "pt"})
+    else:
+        save_func = torch.save
+    if state.is_main_process and not save_on_each_node:
+        save_func(obj, f)
+    elif state.is_local_main_process and save_on_each_node:
+        save_func(obj, f)
