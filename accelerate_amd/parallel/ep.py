@@ -118,6 +118,10 @@ class ExpertParallelMoE(nn.Module):
         self.aux_loss = torch.zeros(())  # refreshed every forward
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        # NOTE: dropless routing requires the host to know the split sizes
+        # (.tolist() below) — one device sync per MoE layer per step. That is
+        # inherent to variable-split all-to-all; capacity-padded routing
+        # (fixed splits, graph-capturable) is the documented alternative.
         shape = x.shape
         x = x.reshape(-1, self.hidden_size)
         N = x.shape[0]
