@@ -276,3 +276,34 @@ def test_automatic_checkpoint_naming_and_total_limit(tmp_path):
     assert ckpts == ["checkpoint_1", "checkpoint_2"], ckpts  # checkpoint_0 rotated out
     # load back the newest
     acc.load_state(str(tmp_path / "checkpoints" / "checkpoint_2"))
+
+
+def test_fp16_skipped_step_detection_cpu():
+    """step_was_skipped must flip on inf grads (scaler skip) and the
+    AcceleratedScheduler must not advance the LR on skipped steps."""
+    acc = Accelerator(cpu=True, mixed_precision="fp16")
+    model = torch.nn.Linear(4, 2)
+    opt = torch.optim.SGD(model.parameters(), lr=1.0)
+    sched = torch.optim.lr_scheduler.StepLR(opt, step_size=1, gamma=0.5)
+    model, opt, sched = acc.prepare(model, opt, sched)
+
+    # healthy step: advances
+    loss = model(torch.randn(2, 4)).sum()
+    acc.backward(loss)
+    opt.step()
+    assert opt.step_was_skipped is False
+    sched.step()
+    lr_after_good = opt.param_groups[0]["lr"]
+
+    # poisoned grads: scaler must skip and report it
+    loss = model(torch.randn(2, 4)).sum()
+    acc.backward(loss)
+    for p in model.parameters():
+        p.grad.fill_(float("inf"))
+    before = [p.detach().clone() for p in model.parameters()]
+    opt.step()
+    assert opt.step_was_skipped is True
+    for p, b in zip(model.parameters(), before):
+        assert torch.equal(p.detach(), b), "skipped step must not move params"
+    sched.step()
+    assert opt.param_groups[0]["lr"] == lr_after_good, "LR must not advance on a skipped step"
