@@ -1,8 +1,15 @@
-"""LR scheduler wrapper (reference: scheduler.py).
+"""LR-scheduler wrapper keeping the schedule in GLOBAL-batch units.
 
-Steps only when the wrapped optimizer actually stepped, and — when batches
-are not split — steps ``num_processes`` times per call so the LR schedule
-stays in global-batch units (reference: scheduler.py:25-99).
+Behavior parity with the reference's AcceleratedScheduler
+(reference scheduler.py:25-99), restructured for this stack:
+
+- A ``step()`` call is honored only when the linked optimizers actually
+  updated this iteration: accumulation windows tick the internal step
+  counter (so schedulers keyed on `_step_count` stay aligned) without
+  moving the LR, and fp16 skipped steps freeze the schedule entirely.
+- Without ``split_batches`` each process sees 1/N of the global batch, so
+  one training iteration advances the schedule N times — the LR curve a
+  single-process run with the same GLOBAL batch would produce.
 """
 
 from .state import AcceleratorState, GradientState
@@ -11,52 +18,52 @@ from .state import AcceleratorState, GradientState
 class AcceleratedScheduler:
     def __init__(self, scheduler, optimizers, step_with_optimizer: bool = True, split_batches: bool = False):
         self.scheduler = scheduler
-        self.optimizers = optimizers if isinstance(optimizers, (list, tuple)) else [optimizers]
+        self.optimizers = list(optimizers) if isinstance(optimizers, (list, tuple)) else [optimizers]
         self.split_batches = split_batches
         self.step_with_optimizer = step_with_optimizer
         self.gradient_state = GradientState()
 
+    # -- stepping ----------------------------------------------------------
+
+    def _any_optimizer_skipped(self) -> bool:
+        return any(getattr(opt, "step_was_skipped", False) for opt in self.optimizers)
+
+    def _advance(self, *args, **kwargs):
+        # OneCycle-style schedulers with a hard total_steps raise past the
+        # end; clamp when drop_last didn't trim the tail batch.
+        limit = getattr(self.scheduler, "total_steps", None)
+        if limit is not None and self.scheduler._step_count > limit:
+            return
+        self.scheduler.step(*args, **kwargs)
+
     def step(self, *args, **kwargs):
         if not self.step_with_optimizer:
-            # No link between scheduler and optimizer -> just step
-            self.scheduler.step(*args, **kwargs)
+            self.scheduler.step(*args, **kwargs)  # decoupled: always advance
             return
-
-        # Otherwise, first make sure the optimizer was stepped.
         if not self.gradient_state.sync_gradients:
+            # accumulation window: optimizers no-op'd; keep counters aligned
             if self.gradient_state.adjust_scheduler:
                 self.scheduler._step_count += 1
             return
+        if self._any_optimizer_skipped():
+            return  # fp16 inf/nan step: LR stays put
+        ticks = 1 if self.split_batches else AcceleratorState().num_processes
+        for _ in range(ticks):
+            self._advance(*args, **kwargs)
 
-        for opt in self.optimizers:
-            if getattr(opt, "step_was_skipped", False):
-                return
-        if self.split_batches:
-            # Split batches -> the training dataloader batch size is not changed so one step per training step
-            self.scheduler.step(*args, **kwargs)
-        else:
-            # Otherwise the training dataloader batch size was multiplied by `num_processes`, so we need to do
-            # num_processes steps per training step
-            num_processes = AcceleratorState().num_processes
-            for _ in range(num_processes):
-                # Special case when using OneCycle and `drop_last` was not used
-                if hasattr(self.scheduler, "total_steps"):
-                    if self.scheduler._step_count <= self.scheduler.total_steps:
-                        self.scheduler.step(*args, **kwargs)
-                else:
-                    self.scheduler.step(*args, **kwargs)
+    # -- passthrough -------------------------------------------------------
 
     def get_last_lr(self):
         return self.scheduler.get_last_lr()
-
-    def state_dict(self):
-        return self.scheduler.state_dict()
-
-    def load_state_dict(self, state_dict):
-        self.scheduler.load_state_dict(state_dict)
 
     def get_lr(self):
         return self.scheduler.get_lr()
 
     def print_lr(self, *args, **kwargs):
         return self.scheduler.print_lr(*args, **kwargs)
+
+    def state_dict(self):
+        return self.scheduler.state_dict()
+
+    def load_state_dict(self, state_dict):
+        self.scheduler.load_state_dict(state_dict)
