@@ -1,5 +1,13 @@
-"""Disk/CPU offload store: one numpy memmap per tensor + json index
-(reference: utils/offload.py)."""
+"""Disk offload store: one numpy memmap per tensor plus a JSON index.
+
+Same on-disk contract as the reference (utils/offload.py — ``index.json``
+with dtype/shape entries next to ``<name>.dat`` memmaps, safetensors
+entries resolved lazily), implemented around a small dtype-codec:
+numpy has no bf16, so bf16 tensors travel through the store bit-cast to
+int16 and are re-viewed on load. Memmaps keep host RAM flat no matter how
+large the offloaded model is — only the tensors a forward actually touches
+are paged in.
+"""
 
 import json
 import os
@@ -9,128 +17,134 @@ from typing import Dict, List, Optional, Union
 import numpy as np
 import torch
 
+_INDEX_FILE = "index.json"
 
-def offload_weight(weight: torch.Tensor, weight_name: str, offload_folder: Union[str, os.PathLike], index: dict = None):
-    """(reference: offload.py:25)"""
-    dtype = None
-    # Check the string instead of the dtype to be compatible with all torch versions
-    if str(weight.dtype) == "torch.bfloat16":
-        # Need to reinterpret the underlined data as int16 since NumPy does not handle bfloat16s.
-        weight = weight.view(torch.int16)
-        dtype = "bfloat16"
-    array = weight.cpu().numpy()
-    tensor_file = os.path.join(offload_folder, f"{weight_name}.dat")
-    if index is not None:
-        if dtype is None:
-            dtype = str(array.dtype)
-        index[weight_name] = {"dtype": dtype, "shape": list(array.shape)}
+
+def _encode_for_numpy(t: torch.Tensor):
+    """-> (numpy array, logical dtype tag). bf16 is bit-cast to int16."""
+    if t.dtype == torch.bfloat16:
+        return t.view(torch.int16).cpu().numpy(), "bfloat16"
+    arr = t.cpu().numpy()
+    return arr, str(arr.dtype)
+
+
+def _memmap_dtype(tag: str) -> str:
+    return "int16" if tag == "bfloat16" else tag
+
+
+def offload_weight(weight, weight_name, offload_folder, index: dict = None):
+    array, tag = _encode_for_numpy(weight)
     if array.ndim == 0:
-        array = array[None]
-    file_array = np.memmap(tensor_file, dtype=array.dtype, mode="w+", shape=array.shape)
-    file_array[:] = array[:]
-    file_array.flush()
+        array = array[None]  # memmaps cannot be 0-d; shape in the index is
+    if index is not None:
+        index[weight_name] = {"dtype": tag, "shape": list(weight.shape)}
+    mm = np.memmap(
+        os.path.join(offload_folder, f"{weight_name}.dat"),
+        dtype=array.dtype,
+        mode="w+",
+        shape=array.shape,
+    )
+    mm[:] = array[:]
+    mm.flush()
     return index
 
 
-def load_offloaded_weight(weight_file: str, weight_info: dict) -> torch.Tensor:
-    """(reference: offload.py:46)"""
-    shape = tuple(weight_info["shape"])
-    if shape == ():
-        # NumPy memory-mapped arrays can't have 0 dims so it was saved as 1d tensor
-        shape = (1,)
-    dtype = weight_info["dtype"]
-    if dtype == "bfloat16":
-        # NumPy does not support bfloat16 so this was saved as a int16
-        dtype = "int16"
-    weight = np.memmap(weight_file, dtype=dtype, shape=shape, mode="r")
-    if len(weight_info["shape"]) == 0:
-        weight = weight[0]
-    weight = torch.tensor(weight)
+def load_offloaded_weight(weight_file, weight_info: dict) -> torch.Tensor:
+    logical_shape = tuple(weight_info["shape"])
+    mm = np.memmap(
+        weight_file,
+        dtype=_memmap_dtype(weight_info["dtype"]),
+        shape=logical_shape or (1,),
+        mode="r",
+    )
+    t = torch.tensor(mm if logical_shape else mm[0])
     if weight_info["dtype"] == "bfloat16":
-        weight = weight.view(torch.bfloat16)
-    return weight
+        t = t.view(torch.bfloat16)
+    return t
 
 
-def save_offload_index(index: dict, offload_folder: Union[str, os.PathLike]):
-    if index is None or len(index) == 0:
+def save_offload_index(index: dict, offload_folder):
+    if not index:
         return
-    offload_index_file = os.path.join(offload_folder, "index.json")
-    if os.path.isfile(offload_index_file):
-        with open(offload_index_file, encoding="utf-8") as f:
-            current_index = json.load(f)
-    else:
-        current_index = {}
-    current_index.update(index)
-    with open(offload_index_file, "w", encoding="utf-8") as f:
-        json.dump(current_index, f, indent=2)
+    path = os.path.join(offload_folder, _INDEX_FILE)
+    merged = {}
+    if os.path.isfile(path):
+        with open(path, encoding="utf-8") as f:
+            merged = json.load(f)
+    merged.update(index)
+    with open(path, "w", encoding="utf-8") as f:
+        json.dump(merged, f, indent=2)
 
 
-def offload_state_dict(save_dir: Union[str, os.PathLike], state_dict: Dict[str, torch.Tensor]) -> dict:
-    """Offload every tensor of a state dict to memmaps (reference: offload.py:85)."""
+def offload_state_dict(save_dir, state_dict: Dict[str, torch.Tensor]) -> dict:
+    """Offload a whole state dict; returns (and writes) its index."""
     os.makedirs(save_dir, exist_ok=True)
-    index = {}
-    for name, parameter in state_dict.items():
-        index = offload_weight(parameter, name, save_dir, index=index)
+    index: dict = {}
+    for name, tensor in state_dict.items():
+        offload_weight(tensor, name, save_dir, index=index)
     save_offload_index(index, save_dir)
     return index
 
 
 class PrefixedDataset(Mapping):
-    """Restrict a mapping to keys with a given prefix (stripped on access)."""
+    """View of a mapping restricted to keys under ``prefix`` (stripped)."""
 
     def __init__(self, dataset: Mapping, prefix: str):
         self.dataset = dataset
         self.prefix = prefix
 
+    def _matching(self):
+        return [k for k in self.dataset if k.startswith(self.prefix)]
+
     def __getitem__(self, key):
-        return self.dataset[f"{self.prefix}{key}"]
+        return self.dataset[self.prefix + key]
 
     def __iter__(self):
-        return iter([key for key in self.dataset if key.startswith(self.prefix)])
+        return iter(self._matching())
 
     def __len__(self):
-        return len([key for key in self.dataset if key.startswith(self.prefix)])
+        return len(self._matching())
 
 
 class OffloadedWeightsLoader(Mapping):
-    """Lazy Mapping over an in-memory state dict + a memmap/safetensors folder
-    (reference: offload.py:127)."""
+    """Lazy mapping over (in-memory state dict) ∪ (offload folder / index).
 
-    def __init__(
-        self,
-        state_dict: Dict[str, torch.Tensor] = None,
-        save_folder: Optional[Union[str, os.PathLike]] = None,
-        index: Mapping = None,
-        device=None,
-    ):
+    In-memory entries win; index entries resolve to memmap files or to
+    safetensors files when the index says so. Iteration order: state-dict
+    keys first, then index-only keys.
+    """
+
+    def __init__(self, state_dict=None, save_folder=None, index: Mapping = None, device=None):
         if state_dict is None and save_folder is None and index is None:
-            raise ValueError("Need either a `state_dict`, a `save_folder` or an `index` containing offloaded weights.")
-        self.state_dict = {} if state_dict is None else state_dict
+            raise ValueError("OffloadedWeightsLoader needs a state_dict, save_folder, or index")
+        self.state_dict = dict(state_dict) if state_dict else {}
         self.save_folder = save_folder
         if index is None and save_folder is not None:
-            with open(os.path.join(save_folder, "index.json")) as f:
+            with open(os.path.join(save_folder, _INDEX_FILE)) as f:
                 index = json.load(f)
-        self.index = {} if index is None else index
-        self.all_keys = list(self.state_dict.keys())
-        self.all_keys.extend([key for key in self.index if key not in self.all_keys])
+        self.index = dict(index) if index else {}
         self.device = device
+        self.all_keys = list(self.state_dict)
+        self.all_keys += [k for k in self.index if k not in self.state_dict]
+
+    def _from_safetensors(self, key, info):
+        import safetensors.torch
+
+        loaded = safetensors.torch.load_file(
+            info["safetensors_file"], device=str(self.device or "cpu")
+        )
+        tensor = loaded[info.get("weight_name", key)]
+        if info.get("dtype") is not None:
+            tensor = tensor.to(getattr(torch, info["dtype"].replace("torch.", "")))
+        return tensor
 
     def __getitem__(self, key: str):
-        # State dict gets priority
         if key in self.state_dict:
             return self.state_dict[key]
-        weight_info = self.index[key]
-        if weight_info.get("safetensors_file") is not None:
-            import safetensors.torch
-
-            device = "cpu" if self.device is None else self.device
-            tensors = safetensors.torch.load_file(weight_info["safetensors_file"], device=str(device))
-            tensor = tensors[weight_info.get("weight_name", key)]
-            if weight_info.get("dtype") is not None:
-                tensor = tensor.to(getattr(torch, weight_info["dtype"].replace("torch.", "")))
-            return tensor
-        weight_file = os.path.join(self.save_folder, f"{key}.dat")
-        return load_offloaded_weight(weight_file, weight_info)
+        info = self.index[key]
+        if info.get("safetensors_file") is not None:
+            return self._from_safetensors(key, info)
+        return load_offloaded_weight(os.path.join(self.save_folder, f"{key}.dat"), info)
 
     def __iter__(self):
         return iter(self.all_keys)
@@ -140,10 +154,10 @@ class OffloadedWeightsLoader(Mapping):
 
 
 def extract_submodules_state_dict(state_dict: Dict[str, torch.Tensor], submodule_names: List[str]):
-    """(reference: offload.py:194)"""
-    result = {}
-    for module_name in submodule_names:
-        result.update(
-            {key: param for key, param in state_dict.items() if key == module_name or key.startswith(module_name + ".")}
-        )
-    return result
+    """Entries belonging to the named submodules (exact or dotted-prefix)."""
+    picked = {}
+    for name in submodule_names:
+        for key, value in state_dict.items():
+            if key == name or key.startswith(name + "."):
+                picked[key] = value
+    return picked
