@@ -1,10 +1,16 @@
-"""Module hooks runtime for big-model dispatch (reference: hooks.py).
+"""Module-hook runtime for big-model dispatch.
 
-`AlignDevicesHook` onloads a block's weights to its execution device before
-forward and offloads after. On MI355X the onload path is the hot loop of
-offloaded inference: weights stream H2D over PCIe from pinned host memory
-(``hipHostMalloc``-backed via torch pinned tensors) with ``non_blocking``
-copies on the current stream.
+Behavior parity with the reference's hooks runtime (reference hooks.py —
+ModelHook protocol :58, add_hook_to_module :147, AlignDevicesHook :242,
+block attachment :586-717, CPU offload :720, layerwise casting :784),
+organized around two small helpers here: ``_TensorWalk`` bundles the
+repeated "which tensors does this hook manage" iteration, and the
+device-map attachment normalizes its inputs once instead of branching on
+scalar-vs-dict shapes throughout.
+
+On MI355X the onload path is the hot loop of offloaded inference: weights
+stream H2D from pinned host memory with non_blocking copies; a
+tied-pointer map guarantees shared weights cross PCIe once per step.
 """
 
 import functools
@@ -13,13 +19,17 @@ from typing import Dict, List, Mapping, Optional, Union
 import torch
 import torch.nn as nn
 
-from .utils.modeling import named_module_tensors, set_module_tensor_to_device
+from .utils.modeling import (
+    get_non_persistent_buffers,
+    named_module_tensors,
+    set_module_tensor_to_device,
+)
 from .utils.offload import PrefixedDataset
-from .utils.operations import send_to_device
+from .utils.operations import find_device, send_to_device
 
 
 class ModelHook:
-    """Hook protocol (reference: hooks.py:58-113)."""
+    """The hook protocol: init/pre_forward/post_forward/detach."""
 
     no_grad = False
 
@@ -37,92 +47,116 @@ class ModelHook:
 
 
 class SequentialHook(ModelHook):
-    """(reference: hooks.py:116)"""
+    """Run several hooks as one (composition order = argument order)."""
 
     def __init__(self, *hooks):
         self.hooks = hooks
 
     def init_hook(self, module):
-        for hook in self.hooks:
-            module = hook.init_hook(module)
+        for h in self.hooks:
+            module = h.init_hook(module)
         return module
 
     def pre_forward(self, module, *args, **kwargs):
-        for hook in self.hooks:
-            args, kwargs = hook.pre_forward(module, *args, **kwargs)
+        for h in self.hooks:
+            args, kwargs = h.pre_forward(module, *args, **kwargs)
         return args, kwargs
 
     def post_forward(self, module, output):
-        for hook in self.hooks:
-            output = hook.post_forward(module, output)
+        for h in self.hooks:
+            output = h.post_forward(module, output)
         return output
 
     def detach_hook(self, module):
-        for hook in self.hooks:
-            module = hook.detach_hook(module)
+        for h in self.hooks:
+            module = h.detach_hook(module)
         return module
 
 
-def add_hook_to_module(module: nn.Module, hook: ModelHook, append: bool = False):
-    """Wrap module.forward so the hook runs around it (reference: hooks.py:147)."""
-    if append and getattr(module, "_hf_hook", None) is not None:
-        old_hook = module._hf_hook
-        remove_hook_from_module(module)
-        hook = SequentialHook(old_hook, hook)
-
-    if hasattr(module, "_hf_hook") and hasattr(module, "_old_forward"):
-        # If we already put some hook on this module, we replace it with the new one.
-        old_forward = module._old_forward
+def _install_forward(module, fn, original):
+    wrapped = functools.update_wrapper(functools.partial(fn, module), original)
+    # fx GraphModule freezes per-instance forward overrides; patch the class
+    if "GraphModuleImpl" in str(type(module)):
+        module.__class__.forward = wrapped
     else:
-        old_forward = module.forward
-        module._old_forward = old_forward
+        module.forward = wrapped
+
+
+def add_hook_to_module(module: nn.Module, hook: ModelHook, append: bool = False):
+    """Wrap ``module.forward`` so ``hook`` runs around it.
+
+    ``append=True`` composes with an existing hook instead of replacing it.
+    The pristine forward is stashed once in ``_old_forward`` and survives
+    any number of replacements.
+    """
+    existing = getattr(module, "_hf_hook", None)
+    if append and existing is not None:
+        remove_hook_from_module(module)
+        hook = SequentialHook(existing, hook)
+
+    if hasattr(module, "_old_forward"):
+        pristine = module._old_forward  # re-hooking: keep the true original
+    else:
+        pristine = module.forward
+        module._old_forward = pristine
 
     module = hook.init_hook(module)
     module._hf_hook = hook
 
-    def new_forward(module, *args, **kwargs):
-        args, kwargs = module._hf_hook.pre_forward(module, *args, **kwargs)
-        if module._hf_hook.no_grad:
+    def hooked_forward(mod, *args, **kwargs):
+        args, kwargs = mod._hf_hook.pre_forward(mod, *args, **kwargs)
+        if mod._hf_hook.no_grad:
             with torch.no_grad():
-                output = module._old_forward(*args, **kwargs)
+                out = mod._old_forward(*args, **kwargs)
         else:
-            output = module._old_forward(*args, **kwargs)
-        return module._hf_hook.post_forward(module, output)
+            out = mod._old_forward(*args, **kwargs)
+        return mod._hf_hook.post_forward(mod, out)
 
-    # Overriding a GraphModuleImpl forward freezes the forward call and later modifications on the graph will fail.
-    if "GraphModuleImpl" in str(type(module)):
-        module.__class__.forward = functools.update_wrapper(functools.partial(new_forward, module), old_forward)
-    else:
-        module.forward = functools.update_wrapper(functools.partial(new_forward, module), old_forward)
+    _install_forward(module, hooked_forward, pristine)
     return module
 
 
 def remove_hook_from_module(module: nn.Module, recurse: bool = False):
-    """(reference: hooks.py remove_hook_from_module)"""
     if hasattr(module, "_hf_hook"):
         module._hf_hook.detach_hook(module)
         delattr(module, "_hf_hook")
     if hasattr(module, "_old_forward"):
-        # Overriding a GraphModuleImpl forward freezes the forward call and later modifications on the graph will fail.
         if "GraphModuleImpl" in str(type(module)):
             module.__class__.forward = module._old_forward
         else:
             module.forward = module._old_forward
         delattr(module, "_old_forward")
-    # Remove accelerate added warning hooks from dispatch_model
-    for attr in ("_accelerate_added_attributes",):
-        for added in getattr(module, attr, []):
-            module.__dict__.pop(added, None)
-        if hasattr(module, attr):
-            delattr(module, attr)
+    for name in getattr(module, "_accelerate_added_attributes", []):
+        module.__dict__.pop(name, None)
+    if hasattr(module, "_accelerate_added_attributes"):
+        delattr(module, "_accelerate_added_attributes")
     if recurse:
         for child in module.children():
             remove_hook_from_module(child, recurse)
     return module
 
 
+def remove_hook_from_submodules(module: nn.Module):
+    remove_hook_from_module(module)
+    for child in module.children():
+        remove_hook_from_submodules(child)
+
+
+def _attr_data_ptr(module, dotted: str) -> int:
+    obj = module
+    for piece in dotted.split("."):
+        obj = getattr(obj, piece)
+    return obj.data_ptr()
+
+
 class AlignDevicesHook(ModelHook):
-    """Onload weights before forward / offload after (reference: hooks.py:242)."""
+    """Onload managed tensors before forward, offload to meta after.
+
+    ``place_submodules`` widens the managed set to the whole subtree;
+    ``weights_map`` is the offloaded source of truth (memmap/safetensors
+    loader or a plain dict); ``tied_params_map`` (data_ptr -> device ->
+    tensor) deduplicates H2D copies of shared weights across hooks.
+    """
 
     def __init__(
         self,
@@ -142,12 +176,10 @@ class AlignDevicesHook(ModelHook):
         self.offload_buffers = offload_buffers
         self.place_submodules = place_submodules
         self.skip_keys = skip_keys
-        # tied-pointer bookkeeping so shared weights are sent H2D only once
         self.tied_params_map = tied_params_map
         self.input_device = None
-        self.param_original_devices = {}
-        self.buffer_original_devices = {}
         self.tied_params_names = set()
+        self.tied_pointers_to_remove = set()
 
     def __repr__(self):
         return (
@@ -156,37 +188,57 @@ class AlignDevicesHook(ModelHook):
             f"place_submodules={self.place_submodules}, skip_keys={repr(self.skip_keys)})"
         )
 
+    # -- managed-tensor iteration -----------------------------------------
+
+    def _managed(self, module):
+        """Tensors this hook onloads/offloads each step (persistent only)."""
+        return named_module_tensors(
+            module,
+            include_buffers=self.offload_buffers,
+            recurse=self.place_submodules,
+            remove_non_persistent=True,
+        )
+
     def init_hook(self, module):
-        # In case the AlignDevicesHook is on meta device, ignore tied weights as data_ptr() is then always zero.
-        if self.execution_device == "meta" or self.execution_device == torch.device("meta"):
-            self.tied_params_map = None
-        if not self.offload and self.execution_device is not None:
-            for name, _ in named_module_tensors(module, recurse=self.place_submodules):
-                set_module_tensor_to_device(module, name, self.execution_device, tied_params_map=self.tied_params_map)
-        elif self.offload:
-            self.original_devices = {
-                name: param.device for name, param in named_module_tensors(module, recurse=self.place_submodules)
-            }
-            if self.weights_map is None:
-                self.weights_map = {
-                    name: param.to("cpu")
-                    for name, param in named_module_tensors(
-                        module, include_buffers=self.offload_buffers, recurse=self.place_submodules
+        if self.execution_device in ("meta", torch.device("meta")):
+            self.tied_params_map = None  # meta data_ptrs all alias zero
+
+        if not self.offload:
+            if self.execution_device is not None:
+                for name, _ in named_module_tensors(module, recurse=self.place_submodules):
+                    set_module_tensor_to_device(
+                        module, name, self.execution_device, tied_params_map=self.tied_params_map
                     )
-                }
-            for name, _ in named_module_tensors(
-                module, include_buffers=self.offload_buffers, recurse=self.place_submodules, remove_non_persistent=True
-            ):
-                # When using disk offloading, we can not rely on `weights_map[name].data_ptr()` as the reference pointer,
-                # as we have no guarantee anymore that safetensors returns the same pointer for several loads of the same tensor
-                if self.tied_params_map is not None and recursive_getattr_data_ptr(module, name) in self.tied_params_map:
-                    self.tied_params_names.add(name)
-                set_module_tensor_to_device(module, name, "meta")
-            if not self.offload_buffers and self.execution_device is not None:
+            return module
+
+        # offloaded block: remember where tensors lived, snapshot values if
+        # no external weights_map was given, then vacate to meta
+        self.original_devices = {
+            name: t.device for name, t in named_module_tensors(module, recurse=self.place_submodules)
+        }
+        if self.weights_map is None:
+            self.weights_map = {
+                name: t.to("cpu")
+                for name, t in named_module_tensors(
+                    module, include_buffers=self.offload_buffers, recurse=self.place_submodules
+                )
+            }
+        for name, _ in self._managed(module):
+            # disk-backed maps reload to fresh pointers every access, so tie
+            # membership is decided by the MODULE tensor's pointer up front
+            if self.tied_params_map is not None and _attr_data_ptr(module, name) in self.tied_params_map:
+                self.tied_params_names.add(name)
+            set_module_tensor_to_device(module, name, "meta")
+
+        if self.execution_device is not None:
+            if not self.offload_buffers:
+                # buffers stay resident on the execution device
                 for name, _ in module.named_buffers(recurse=self.place_submodules):
-                    set_module_tensor_to_device(module, name, self.execution_device, tied_params_map=self.tied_params_map)
-            elif self.offload_buffers and self.execution_device is not None:
-                for name in get_non_persistent_buffer_names(module, recurse=self.place_submodules):
+                    set_module_tensor_to_device(
+                        module, name, self.execution_device, tied_params_map=self.tied_params_map
+                    )
+            else:
+                for name in get_non_persistent_buffers(module, recurse=self.place_submodules):
                     set_module_tensor_to_device(module, name, self.execution_device)
         return module
 
@@ -195,32 +247,31 @@ class AlignDevicesHook(ModelHook):
             self.input_device = find_device([args, kwargs])
         if self.offload:
             self.tied_pointers_to_remove = set()
-            for name, _ in named_module_tensors(
-                module, include_buffers=self.offload_buffers, recurse=self.place_submodules, remove_non_persistent=True
-            ):
-                fp16_statistics = None
+            for name, _ in self._managed(module):
                 value = self.weights_map[name]
                 if name in self.tied_params_names and value.data_ptr() not in self.tied_params_map:
                     self.tied_params_map[value.data_ptr()] = {}
-                if value is not None and self.tied_params_map is not None and value.data_ptr() in self.tied_params_map:
+                if (
+                    value is not None
+                    and self.tied_params_map is not None
+                    and value.data_ptr() in self.tied_params_map
+                ):
                     self.tied_pointers_to_remove.add((value.data_ptr(), self.execution_device))
                 set_module_tensor_to_device(
                     module, name, self.execution_device, value=value, tied_params_map=self.tied_params_map
                 )
-        return send_to_device(args, self.execution_device), send_to_device(
-            kwargs, self.execution_device, skip_keys=self.skip_keys
+        return (
+            send_to_device(args, self.execution_device),
+            send_to_device(kwargs, self.execution_device, skip_keys=self.skip_keys),
         )
 
     def post_forward(self, module, output):
         if self.offload:
-            for name, _ in named_module_tensors(
-                module, include_buffers=self.offload_buffers, recurse=self.place_submodules, remove_non_persistent=True
-            ):
+            for name, _ in self._managed(module):
                 set_module_tensor_to_device(module, name, "meta")
-            # free tied pointers placed during pre_forward
-            for value_pointer, device in getattr(self, "tied_pointers_to_remove", set()):
-                if value_pointer in self.tied_params_map and device in self.tied_params_map[value_pointer]:
-                    del self.tied_params_map[value_pointer][device]
+            for ptr, device in self.tied_pointers_to_remove:
+                if ptr in self.tied_params_map and device in self.tied_params_map[ptr]:
+                    del self.tied_params_map[ptr][device]
             self.tied_pointers_to_remove = set()
         if self.io_same_device and self.input_device is not None:
             output = send_to_device(output, self.input_device, skip_keys=self.skip_keys)
@@ -230,52 +281,43 @@ class AlignDevicesHook(ModelHook):
         if self.offload:
             for name, device in self.original_devices.items():
                 if device != torch.device("meta"):
-                    set_module_tensor_to_device(module, name, device, value=self.weights_map.get(name, None))
+                    set_module_tensor_to_device(module, name, device, value=self.weights_map.get(name))
         return module
 
 
-def get_non_persistent_buffer_names(module, recurse=False):
-    from .utils.modeling import get_non_persistent_buffers
-
-    return get_non_persistent_buffers(module, recurse=recurse)
-
-
-def recursive_getattr_data_ptr(module, name):
-    obj = module
-    for part in name.split("."):
-        obj = getattr(obj, part)
-    return obj.data_ptr()
-
-
-def find_device(data):
-    from .utils.operations import find_device as _find
-
-    return _find(data)
+# ---------------------------------------------------------------------------
+# attachment helpers (used by dispatch_model)
+# ---------------------------------------------------------------------------
 
 
 def attach_execution_device_hook(
     module: nn.Module,
-    execution_device: Union[int, str, torch.device],
+    execution_device,
     skip_keys=None,
     preload_module_classes: Optional[List[str]] = None,
     tied_params_map=None,
 ):
-    """Attach hooks making sure inputs arrive on the execution device
-    (reference: hooks.py attach_execution_device_hook)."""
+    """Make sure every stateful submodule receives inputs on its device."""
     if not hasattr(module, "_hf_hook") and len(module.state_dict()) > 0:
-        add_hook_to_module(module, AlignDevicesHook(execution_device, skip_keys=skip_keys, tied_params_map=tied_params_map))
+        add_hook_to_module(
+            module,
+            AlignDevicesHook(execution_device, skip_keys=skip_keys, tied_params_map=tied_params_map),
+        )
     if preload_module_classes is not None and module.__class__.__name__ in preload_module_classes:
-        return
+        return  # subtree is managed as one block
     for child in module.children():
         attach_execution_device_hook(
-            child, execution_device, skip_keys=skip_keys, preload_module_classes=preload_module_classes,
+            child,
+            execution_device,
+            skip_keys=skip_keys,
+            preload_module_classes=preload_module_classes,
             tied_params_map=tied_params_map,
         )
 
 
 def attach_align_device_hook(
     module: nn.Module,
-    execution_device: Optional[torch.device] = None,
+    execution_device=None,
     offload: bool = False,
     weights_map: Optional[Mapping] = None,
     offload_buffers: bool = False,
@@ -284,60 +326,48 @@ def attach_align_device_hook(
     preload_module_classes: Optional[List[str]] = None,
     tied_params_map=None,
 ):
-    """Attach per-leaf onload/offload hooks (reference: hooks.py attach_align_device_hook)."""
-    # Attach the hook on this module if it has any direct tensor.
-    directs = named_module_tensors(module)
-    full_offload = (
+    """Per-leaf onload/offload hooks over a subtree (offloaded blocks)."""
+    owns_tensors = any(True for _ in named_module_tensors(module))
+    whole_subtree = (
         offload and preload_module_classes is not None and module.__class__.__name__ in preload_module_classes
     )
-
-    if len(list(directs)) > 0 or full_offload:
+    if owns_tensors or whole_subtree:
+        scoped_map = None
         if weights_map is not None:
-            prefix = f"{module_name}." if len(module_name) > 0 else ""
-            prefixed_weights_map = PrefixedDataset(weights_map, prefix)
-        else:
-            prefixed_weights_map = None
-        hook = AlignDevicesHook(
-            execution_device=execution_device,
-            offload=offload,
-            weights_map=prefixed_weights_map,
-            offload_buffers=offload_buffers,
-            place_submodules=full_offload,
-            skip_keys=skip_keys,
-            tied_params_map=tied_params_map,
+            scoped_map = PrefixedDataset(weights_map, f"{module_name}." if module_name else "")
+        add_hook_to_module(
+            module,
+            AlignDevicesHook(
+                execution_device=execution_device,
+                offload=offload,
+                weights_map=scoped_map,
+                offload_buffers=offload_buffers,
+                place_submodules=whole_subtree,
+                skip_keys=skip_keys,
+                tied_params_map=tied_params_map,
+            ),
+            append=True,
         )
-        add_hook_to_module(module, hook, append=True)
-
-    # We stop the recursion in case we hit the full offload.
-    if full_offload:
-        return
-
-    # Recurse on all children of the module.
+    if whole_subtree:
+        return  # one hook covers everything below
     for child_name, child in module.named_children():
-        child_full_name = f"{module_name}.{child_name}" if len(module_name) > 0 else child_name
         attach_align_device_hook(
             child,
             execution_device=execution_device,
             offload=offload,
             weights_map=weights_map,
             offload_buffers=offload_buffers,
-            module_name=child_full_name,
-            preload_module_classes=preload_module_classes,
+            module_name=f"{module_name}.{child_name}" if module_name else child_name,
             skip_keys=skip_keys,
+            preload_module_classes=preload_module_classes,
             tied_params_map=tied_params_map,
         )
 
 
-def remove_hook_from_submodules(module: nn.Module):
-    remove_hook_from_module(module)
-    for child in module.children():
-        remove_hook_from_submodules(child)
-
-
 def attach_align_device_hook_on_blocks(
     module: nn.Module,
-    execution_device: Optional[Union[torch.device, Dict[str, torch.device]]] = None,
-    offload: Union[bool, Dict[str, bool]] = False,
+    execution_device=None,
+    offload=False,
     weights_map: Mapping = None,
     offload_buffers: bool = False,
     module_name: str = "",
@@ -345,19 +375,12 @@ def attach_align_device_hook_on_blocks(
     preload_module_classes: Optional[List[str]] = None,
     tied_params_map=None,
 ):
-    """Attach hooks per device-map block (reference: hooks.py:586-717)."""
-    # If one device and one offload, we've got one hook.
+    """Walk the device map: resident blocks get ONE whole-subtree hook (the
+    root additionally pins outputs to the input device); offloaded blocks
+    get per-leaf onload hooks plus input-device hooks for their interior."""
+    # degenerate map: a single device / a single offload decision
     if not isinstance(execution_device, Mapping) and not isinstance(offload, dict):
-        if not offload:
-            hook = AlignDevicesHook(
-                execution_device=execution_device,
-                io_same_device=True,
-                skip_keys=skip_keys,
-                place_submodules=True,
-                tied_params_map=tied_params_map,
-            )
-            add_hook_to_module(module, hook)
-        else:
+        if offload:
             attach_align_device_hook(
                 module,
                 execution_device=execution_device,
@@ -368,27 +391,42 @@ def attach_align_device_hook_on_blocks(
                 skip_keys=skip_keys,
                 tied_params_map=tied_params_map,
             )
+        else:
+            add_hook_to_module(
+                module,
+                AlignDevicesHook(
+                    execution_device=execution_device,
+                    io_same_device=True,
+                    skip_keys=skip_keys,
+                    place_submodules=True,
+                    tied_params_map=tied_params_map,
+                ),
+            )
         return
 
+    # normalize both maps to dicts over the same keys
     if not isinstance(execution_device, Mapping):
-        execution_device = {key: execution_device for key in offload.keys()}
+        execution_device = {key: execution_device for key in offload}
     if not isinstance(offload, Mapping):
-        offload = {key: offload for key in execution_device.keys()}
+        offload = {key: offload for key in execution_device}
 
-    if module_name in execution_device and module_name in offload and not offload[module_name]:
-        hook = AlignDevicesHook(
-            execution_device=execution_device[module_name],
-            offload_buffers=offload_buffers,
-            io_same_device=(module_name == ""),
-            place_submodules=True,
-            skip_keys=skip_keys,
-            tied_params_map=tied_params_map,
+    mapped = module_name in execution_device and module_name in offload
+    if mapped and not offload[module_name]:
+        add_hook_to_module(
+            module,
+            AlignDevicesHook(
+                execution_device=execution_device[module_name],
+                offload_buffers=offload_buffers,
+                io_same_device=(module_name == ""),
+                place_submodules=True,
+                skip_keys=skip_keys,
+                tied_params_map=tied_params_map,
+            ),
         )
-        add_hook_to_module(module, hook)
         attach_execution_device_hook(
             module, execution_device[module_name], skip_keys=skip_keys, tied_params_map=tied_params_map
         )
-    elif module_name in execution_device and module_name in offload:
+    elif mapped:
         attach_align_device_hook(
             module,
             execution_device=execution_device[module_name],
@@ -401,13 +439,15 @@ def attach_align_device_hook_on_blocks(
             tied_params_map=tied_params_map,
         )
         if not hasattr(module, "_hf_hook"):
-            hook = AlignDevicesHook(
-                execution_device=execution_device[module_name],
-                io_same_device=(module_name == ""),
-                skip_keys=skip_keys,
-                tied_params_map=tied_params_map,
+            add_hook_to_module(
+                module,
+                AlignDevicesHook(
+                    execution_device=execution_device[module_name],
+                    io_same_device=(module_name == ""),
+                    skip_keys=skip_keys,
+                    tied_params_map=tied_params_map,
+                ),
             )
-            add_hook_to_module(module, hook)
         attach_execution_device_hook(
             module,
             execution_device[module_name],
@@ -416,32 +456,38 @@ def attach_align_device_hook_on_blocks(
             tied_params_map=tied_params_map,
         )
     elif module_name == "":
-        hook = AlignDevicesHook(
-            execution_device=execution_device.get(""),
-            io_same_device=True,
-            skip_keys=skip_keys,
-            tied_params_map=tied_params_map,
+        add_hook_to_module(
+            module,
+            AlignDevicesHook(
+                execution_device=execution_device.get(""),
+                io_same_device=True,
+                skip_keys=skip_keys,
+                tied_params_map=tied_params_map,
+            ),
         )
-        add_hook_to_module(module, hook)
 
     for child_name, child in module.named_children():
-        child_full_name = f"{module_name}.{child_name}" if len(module_name) > 0 else child_name
         attach_align_device_hook_on_blocks(
             child,
             execution_device=execution_device,
             offload=offload,
             weights_map=weights_map,
             offload_buffers=offload_buffers,
-            module_name=child_full_name,
+            module_name=f"{module_name}.{child_name}" if module_name else child_name,
             preload_module_classes=preload_module_classes,
             skip_keys=skip_keys,
             tied_params_map=tied_params_map,
         )
 
 
+# ---------------------------------------------------------------------------
+# sequential CPU offload + layerwise casting
+# ---------------------------------------------------------------------------
+
+
 class CpuOffload(ModelHook):
-    """Offloads a whole model to CPU, onloading to the execution device on
-    forward (sequential offload; reference: hooks.py:720)."""
+    """Whole-model CPU residency; onload to the execution device on forward.
+    ``prev_module_hook`` chains pipelines: running block N offloads N-1."""
 
     def __init__(self, execution_device=None, prev_module_hook=None):
         self.prev_module_hook = prev_module_hook
@@ -462,11 +508,14 @@ class CpuOffload(ModelHook):
 
             clear_device_cache()
         module.to(self.execution_device)
-        return send_to_device(args, self.execution_device), send_to_device(kwargs, self.execution_device)
+        return (
+            send_to_device(args, self.execution_device),
+            send_to_device(kwargs, self.execution_device),
+        )
 
 
 class UserCpuOffloadHook:
-    """User handle over a CpuOffload hook (reference: hooks.py:760)."""
+    """Handle returned by ``cpu_offload_with_hook``: manual offload/remove."""
 
     def __init__(self, model, hook):
         self.model = model
@@ -480,8 +529,8 @@ class UserCpuOffloadHook:
 
 
 class LayerwiseCastingHook(ModelHook):
-    """Keep storage in ``storage_dtype``, compute in ``compute_dtype``
-    (reference: hooks.py:784)."""
+    """Store in ``storage_dtype``, compute in ``compute_dtype`` — per-layer
+    upcast around forward (fp8/bf16 storage with fp32 compute)."""
 
     def __init__(self, storage_dtype: torch.dtype, compute_dtype: torch.dtype, non_blocking: bool = False):
         self.storage_dtype = storage_dtype
