@@ -492,140 +492,133 @@ class DataLoaderDispatcher(DataLoaderAdapter, DataLoaderStateMixin):
         self.slice_fn = slice_tensors if slice_fn is None else slice_fn
         self.iteration = 0
 
+    def _rank0_next(self, iterator, sink):
+        """Rank 0 pulls one GLOBAL batch: num_processes local batches glued
+        together (unless split_batches). Local batches land in ``sink`` as
+        they arrive so a mid-glue StopIteration leaves the partial tail
+        recoverable by the caller."""
+        if self.split_batches:
+            self._update_state_dict()
+            return next(iterator)
+        for _ in range(self.state.num_processes):
+            self._update_state_dict()
+            sink.append(next(iterator))
+        try:
+            return concatenate(sink, dim=0)
+        except RuntimeError as e:
+            raise RuntimeError(
+                "Variable-size batches cannot be dispatched (`dispatch_batches=True` / "
+                "IterableDataset). Use `dispatch_batches=False` so each rank fetches its own "
+                "batch, or `split_batches=True` so rank 0 fetches one full batch and slices it."
+            ) from e
+
     def _fetch_batches(self, iterator):
-        batches, batch = None, None
-        # On process 0, we gather the batch to dispatch.
+        """One lookahead step. Agrees a [structure, exhausted] header across
+        ranks via broadcast_object_list; the payload itself travels later in
+        __iter__ (a tensor broadcast). The collective COUNT here is identical
+        on every rank — including the tail retry — or the job deadlocks."""
+        batch, partial_tail = None, []
         if self.state.process_index == 0:
             try:
-                if self.split_batches:
-                    self._update_state_dict()
-                    batch = next(iterator)
-                else:
-                    batches = []
-                    for _ in range(self.state.num_processes):
-                        self._update_state_dict()
-                        batches.append(next(iterator))
-                    try:
-                        batch = concatenate(batches, dim=0)
-                    except RuntimeError as e:
-                        raise RuntimeError(
-                            "You can't use batches of different size with `dispatch_batches=True` or when using an "
-                            "`IterableDataset`. Either pass `dispatch_batches=False` and have each process fetch its "
-                            "own batch or pass `split_batches=True`. By doing so, the main process will fetch a full "
-                            "batch and slice it into `num_processes` batches for each process."
-                        ) from e
-                # header: [has_data, structure]
-                batch_info = [get_data_structure(batch), False]
+                batch = self._rank0_next(iterator, partial_tail)
+                header = [get_data_structure(batch), False]
             except StopIteration:
-                batch_info = [None, True]
+                header = [None, True]
         else:
-            batch_info = [None, self._stop_iteration]
-        broadcast_object_list(batch_info)
-        self._stop_iteration = batch_info[1]
-        if self._stop_iteration:
-            if not self.split_batches and not self._drop_last:
-                if self.state.process_index == 0 and len(batches) > 0:
-                    batch = concatenate(batches, dim=0)
-                    batch_info = [get_data_structure(batch), False]
-                else:
-                    batch_info = [None, True]
-                broadcast_object_list(batch_info)
-        return batch, batch_info
+            header = [None, self._stop_iteration]
+        broadcast_object_list(header)
+        self._stop_iteration = header[1]
+        if self._stop_iteration and not self.split_batches and not self._drop_last:
+            # rank 0 may hold a PARTIAL tail (StopIteration mid-glue): agree
+            # on whether those leftover rows exist with one more header round
+            if self.state.process_index == 0 and len(partial_tail) > 0:
+                batch = concatenate(partial_tail, dim=0)
+                header = [get_data_structure(batch), False]
+            else:
+                header = [None, True]
+            broadcast_object_list(header)
+        return batch, header
 
     def __iter__(self):
         self.begin()
         self.set_epoch(self.iteration)
-        main_iterator = None
-        # Every rank iterates (for worker-side effects), but only rank-0's
-        # payloads travel.
-        main_iterator = self.base_dataloader.__iter__()
-        stop_iteration = False
+        # every rank drives the underlying loader (worker-side effects run
+        # everywhere) but only rank 0's data is dispatched
+        source = self.base_dataloader.__iter__()
+        exhausted = False
         self._stop_iteration = False
-        first_batch = None
-        next_batch, next_batch_info = self._fetch_batches(main_iterator)
-        batch_index = 0
-        while not stop_iteration:
-            batch, batch_info = next_batch, next_batch_info
-
+        spare_head = None  # first num_processes rows, kept to pad the tail
+        lookahead = self._fetch_batches(source)
+        position = 0
+        while not exhausted:
+            current, header = lookahead
             if self.state.process_index != 0:
-                # Initialize tensors on other processes than process 0.
-                batch = initialize_tensors(batch_info[0])
-            batch = send_to_device(batch, self.state.device, non_blocking=self._non_blocking)
-            # Broadcast the batch before splitting it.
-            batch = broadcast(batch, from_process=0)
+                current = initialize_tensors(header[0])  # right shapes, recv target
+            current = send_to_device(current, self.state.device, non_blocking=self._non_blocking)
+            current = broadcast(current, from_process=0)
 
-            if not self._drop_last and first_batch is None:
-                # We keep at least num processes elements of the first batch to be able to complete the last batch
-                first_batch = self.slice_fn(
-                    batch,
+            if current is None:
+                raise ValueError("dispatched batch is empty before the agreed end of data")
+            if spare_head is None and not self._drop_last:
+                spare_head = self.slice_fn(
+                    current,
                     slice(0, self.state.num_processes),
                     process_index=self.state.process_index,
                     num_processes=self.state.num_processes,
                 )
 
-            if batch is None:
-                raise ValueError(
-                    f"Batch does not contain any data (`{batch}`). At the end of all iterable data available before "
-                    "expected stop iteration."
-                )
+            global_rows = find_batch_size(current)
+            rows_per_rank = global_rows // self.state.num_processes
 
-            observed_batch_size = find_batch_size(batch)
-            batch_size = observed_batch_size // self.state.num_processes
+            exhausted = self._stop_iteration
+            if not exhausted:
+                # fetch ahead: the source may be done without us knowing —
+                # a recovered partial tail still needs dispatching first
+                lookahead = self._fetch_batches(source)
+                if self._stop_iteration and lookahead[1][0] is None:
+                    exhausted = True
 
-            stop_iteration = self._stop_iteration
-            if not stop_iteration:
-                # We may still be at the end of the dataloader without knowing it yet: fetch ahead.
-                next_batch, next_batch_info = self._fetch_batches(main_iterator)
-                # next_batch_info[0] is None when there are no more batches, otherwise we still need to process them.
-                if self._stop_iteration and next_batch_info[0] is None:
-                    stop_iteration = True
+            if exhausted and not self._drop_last and global_rows % self.state.num_processes != 0:
+                # tail not divisible: top it up with the saved head rows so
+                # every rank still gets rows_per_rank+1 (dedup is
+                # gather_for_metrics' job via the remainder below)
+                current = concatenate([current, spare_head], dim=0)
+                rows_per_rank += 1
 
-            if not self._drop_last and stop_iteration and observed_batch_size % self.state.num_processes != 0:
-                # If the last batch is not complete, let's add the first batch to it.
-                batch = concatenate([batch, first_batch], dim=0)
-                # Batch size computation above is wrong, it's off by 1 so we fix it.
-                batch_size += 1
-
-            data_slice = slice(self.state.process_index * batch_size, (self.state.process_index + 1) * batch_size)
-            batch = self.slice_fn(
-                batch,
-                data_slice,
-                process_index=self.state.process_index,
-                num_processes=self.state.num_processes,
+            mine = slice(self.state.process_index * rows_per_rank, (self.state.process_index + 1) * rows_per_rank)
+            current = self.slice_fn(
+                current, mine, process_index=self.state.process_index, num_processes=self.state.num_processes
             )
 
-            if stop_iteration:
+            if exhausted:
                 self.end_of_dataloader = True
                 self._update_state_dict()
-                self.remainder = observed_batch_size % self.state.num_processes
-            if batch_index >= self.skip_batches:
-                yield batch
-            batch_index += 1
+                self.remainder = global_rows % self.state.num_processes
+            if position >= self.skip_batches:
+                yield current
+            position += 1
         self.iteration += 1
         self.end()
 
     def set_epoch(self, epoch: int):
-        if self.iteration != epoch:
-            self.iteration = epoch
-        if hasattr(self.batch_sampler, "sampler") and hasattr(self.batch_sampler.sampler, "set_epoch"):
-            self.batch_sampler.sampler.set_epoch(epoch)
+        self.iteration = epoch
+        inner_sampler = getattr(self.batch_sampler, "sampler", None)
+        if hasattr(inner_sampler, "set_epoch"):
+            inner_sampler.set_epoch(epoch)
         elif hasattr(self.dataset, "set_epoch"):
             self.dataset.set_epoch(epoch)
 
     def __len__(self):
-        whole_length = len(self.base_dataloader)
+        n_global = len(self.base_dataloader)
         if self.split_batches:
-            return whole_length
-        elif self._drop_last:
-            return whole_length // self.state.num_processes
-        else:
-            return math.ceil(whole_length / self.state.num_processes)
+            return n_global
+        div = self.state.num_processes
+        return n_global // div if self._drop_last else math.ceil(n_global / div)
 
     @property
     def total_batch_size(self):
-        return (
-            self.dataset.batch_size if self.split_batches else (self.dataset.batch_size * self.dataset.num_processes)
-        )
+        per_fetch = self.dataset.batch_size
+        return per_fetch if self.split_batches else per_fetch * self.dataset.num_processes
 
     @property
     def total_dataset_length(self):

@@ -83,6 +83,18 @@ def test_dispatcher(acc):
     seen = torch.cat([b[0] for b in dl2])
     everything = gather(seen)
     assert sorted(everything.tolist()) == [float(i) for i in range(16)]
+
+    # non-divisible tail: 18 samples = 9 local batches; rank 0 glues pairs,
+    # the odd 9th local batch is the recovered partial tail — with
+    # gather_for_metrics dedup every sample must appear exactly once
+    ds2 = TensorDataset(torch.arange(18).float())
+    dl3 = acc.prepare_data_loader(DataLoader(ds2, batch_size=2))
+    assert isinstance(dl3, DataLoaderDispatcher)
+    collected = []
+    for (b,) in dl3:
+        collected.append(acc.gather_for_metrics(b))
+    got = sorted(torch.cat(collected).tolist())
+    assert got == [float(i) for i in range(18)], got
     acc.dispatch_batches = None
     if acc.is_main_process:
         print("DISPATCHER_PASS")
