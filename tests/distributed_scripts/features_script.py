@@ -95,6 +95,17 @@ def test_dispatcher(acc):
         collected.append(acc.gather_for_metrics(b))
     got = sorted(torch.cat(collected).tolist())
     assert got == [float(i) for i in range(18)], got
+
+    # split_batches dispatch: rank 0 fetches ONE global batch of 4 and each
+    # rank receives exactly half of every batch
+    acc.split_batches = True
+    dl4 = acc.prepare_data_loader(DataLoader(ds, batch_size=4))
+    assert isinstance(dl4, DataLoaderDispatcher)
+    rows = [b[0] for b in dl4]
+    assert all(r.numel() == 2 for r in rows), [r.numel() for r in rows]
+    covered = gather(torch.cat(rows))
+    assert sorted(covered.tolist()) == [float(i) for i in range(16)]
+    acc.split_batches = False
     acc.dispatch_batches = None
     if acc.is_main_process:
         print("DISPATCHER_PASS")
