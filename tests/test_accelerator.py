@@ -307,3 +307,29 @@ def test_fp16_skipped_step_detection_cpu():
         assert torch.equal(p.detach(), b), "skipped step must not move params"
     sched.step()
     assert opt.param_groups[0]["lr"] == lr_after_good, "LR must not advance on a skipped step"
+
+
+def test_prepare_preserves_order_and_passthrough():
+    """prepare() returns objects in argument order; non-wrappable objects
+    pass through untouched (reference prepare semantics)."""
+    from torch.utils.data import DataLoader, TensorDataset
+
+    acc = Accelerator(cpu=True)
+    model = torch.nn.Linear(4, 2)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    dl = DataLoader(TensorDataset(torch.arange(4).float()), batch_size=2)
+    sentinel = {"not": "preparable"}
+    m, s, o, d = acc.prepare(model, sentinel, opt, dl)
+    assert s is sentinel
+    assert o.optimizer is opt
+    assert hasattr(m, "forward")
+    assert list(d)[0][0].numel() == 2
+
+
+def test_free_memory_clears_references():
+    acc = Accelerator(cpu=True)
+    model = acc.prepare(torch.nn.Linear(4, 2))
+    assert len(acc._models) == 1
+    (model,) = acc.free_memory(model)
+    assert acc._models == []
+    assert model is None  # slots come back None so callers drop references
