@@ -1,0 +1,59 @@
+"""2-process gloo oracle for pipeline-parallel TRAINING: a 2-stage MLP
+trained with GPipe microbatching must match a single-process run of the
+same full model (same grads, same updated weights, same loss).
+The reference raises NotImplementedError for PP training — this is a
+capability beyond it."""
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from accelerate_amd import Accelerator, set_seed
+from accelerate_amd.parallel.pp import PipelineParallelEngine, split_into_stages
+
+
+def main():
+    acc = Accelerator(cpu=True)
+    n, r = acc.num_processes, acc.process_index
+    assert n == 2
+    set_seed(0)
+    full = nn.Sequential(
+        nn.Linear(8, 32), nn.Tanh(), nn.Linear(32, 32), nn.Tanh(), nn.Linear(32, 4)
+    )
+    ref = nn.Sequential(*[nn.Linear(8, 32), nn.Tanh(), nn.Linear(32, 32), nn.Tanh(), nn.Linear(32, 4)])
+    ref.load_state_dict(full.state_dict())
+
+    engine = PipelineParallelEngine(model=full, num_microbatches=4)
+    opt = torch.optim.SGD(engine.parameters(), lr=0.1)
+    ref_opt = torch.optim.SGD(ref.parameters(), lr=0.1)
+    loss_fn = nn.MSELoss()
+
+    g = torch.Generator().manual_seed(4)
+    for step in range(3):
+        X = torch.randn(8, 8, generator=g)
+        T = torch.randn(8, 4, generator=g)
+        opt.zero_grad()
+        loss = engine.train_step(
+            inputs=X if engine.is_first else None,
+            targets=T if engine.is_last else None,
+            loss_fn=loss_fn if engine.is_last else None,
+        )
+        opt.step()
+        ref_opt.zero_grad()
+        ref_loss = loss_fn(ref(X), T)
+        ref_loss.backward()
+        ref_opt.step()
+        if engine.is_last:
+            assert torch.allclose(loss, ref_loss, atol=1e-6), (loss, ref_loss)
+
+    # each rank's stage params must equal the reference's matching slice
+    stages = split_into_stages(ref, 2)
+    for p_eng, p_ref in zip(engine.stage.parameters(), stages[r].parameters()):
+        assert torch.allclose(p_eng, p_ref, atol=1e-6), (p_eng - p_ref).abs().max()
+    if acc.is_main_process:
+        print("PP_TRAIN_PASS")
+    acc.end_training()
+
+
+if __name__ == "__main__":
+    main()
