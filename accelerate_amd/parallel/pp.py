@@ -1,24 +1,30 @@
-"""Pipeline-parallel TRAINING (GPipe schedule over P2P).
+"""Pipeline-parallel TRAINING (GPipe and 1F1B schedules over P2P).
 
 The reference implements pipeline parallelism for inference only and
 explicitly raises for training (reference accelerator.py:795-799
 NotImplementedError; inference lives in inference.py). This engine goes
 beyond it: microbatched forward/backward with activation hand-off over
-``dist.send``/``dist.recv`` — xGMI point-to-point on RCCL, plain TCP on
-gloo (which is how the 2-process CPU oracle runs it).
+point-to-point sends — xGMI-adjacent on RCCL, TCP on gloo (how the
+2-process CPU oracles run).
 
-Schedule (GPipe, all-forward-then-all-backward):
+Schedules (both produce IDENTICAL gradients; they differ in peak memory):
 
-  stage r, microbatch m:   recv a[m] from r-1 → run stage → send to r+1
-  ...all M microbatches...
-  then in reverse:         recv g[m] from r+1 → backward → send input-grad
+- ``gpipe``  — all forwards, then all backwards (reversed order). Peak
+  resident activations = num_microbatches.
+- ``1f1b``   — stage r runs (S-1-r) warmup forwards, then alternates one
+  forward / one backward, then drains backwards. Peak resident
+  activations = S - r: INDEPENDENT of the microbatch count, which is what
+  lets deep pipelines use many microbatches to hide bubble time.
+
+Sends are ``isend`` (buffered, waited at step end) so the blocking-recv
+orderings of adjacent ranks can never form a cycle; per-channel message
+order is monotonic in microbatch index for both schedules.
 
 Gradients accumulate across microbatches in the stage's parameters; the
 caller steps its optimizer once per ``train_step``. The LAST stage owns
-the loss; per-microbatch losses are averaged so the result matches a
-single-process model run on the full batch with a mean-reduced loss.
-Activation shapes are exchanged once (first step) via object send/recv,
-then P2P tensors flow with no per-step metadata.
+the loss; per-microbatch losses are scaled 1/M so the result matches a
+single-process model with a mean-reduced loss. Activation shape/dtype is
+announced downstream exactly once.
 """
 
 from typing import Callable, List, Optional
@@ -48,7 +54,7 @@ def split_into_stages(model: nn.Sequential, n_stages: int) -> List[nn.Sequential
 
 
 class PipelineParallelEngine:
-    """One pipeline stage per rank; ``train_step`` runs a full GPipe cycle.
+    """One pipeline stage per rank; ``train_step`` runs a full cycle.
 
     Every rank constructs the engine with the SAME model (stages are
     selected locally) or with an explicit per-rank ``stage`` module.
@@ -59,14 +65,18 @@ class PipelineParallelEngine:
         model: Optional[nn.Sequential] = None,
         stage: Optional[nn.Module] = None,
         num_microbatches: int = 4,
+        schedule: str = "gpipe",
         group=None,
     ):
         if not dist.is_initialized():
             raise RuntimeError("PipelineParallelEngine needs torch.distributed initialized")
+        if schedule not in ("gpipe", "1f1b"):
+            raise ValueError(f"unknown schedule {schedule!r} (gpipe | 1f1b)")
         self.group = group
         self.world = dist.get_world_size(group)
         self.rank = dist.get_rank(group)
         self.num_microbatches = num_microbatches
+        self.schedule = schedule
         if stage is not None:
             self.stage = stage
         elif model is not None:
@@ -77,29 +87,68 @@ class PipelineParallelEngine:
         self.is_last = self.rank == self.world - 1
         self._recv_shape = None   # learned on the first step (downstream)
         self._announced = False   # shape sent downstream exactly once
+        self._inflight = []       # (work, tensor) pairs kept alive until wait
 
-    # -- P2P helpers (global ranks; group kept for future sub-worlds) -----
+    # -- P2P plumbing ------------------------------------------------------
 
-    def _send(self, tensor, dst):
-        dist.send(tensor.contiguous(), dst=dst)
+    def _send_async(self, tensor, dst):
+        t = tensor.contiguous()
+        self._inflight.append((dist.isend(t, dst=dst), t))
 
-    def _recv(self, shape, dtype):
+    def _drain_sends(self):
+        for work, _ in self._inflight:
+            work.wait()
+        self._inflight = []
+
+    def _recv_from_prev(self):
+        shape, dtype = self._recv_shape
         buf = torch.empty(shape, dtype=dtype)
         dist.recv(buf, src=self.rank - 1)
         return buf
 
-    def _exchange_shape_once(self, example=None):
-        """Downstream ranks learn their input shape/dtype from upstream."""
+    def _learn_shape_once(self):
         if self.is_first or self._recv_shape is not None:
             return
         meta = [None]
         dist.recv_object_list(meta, src=self.rank - 1)
         self._recv_shape = meta[0]
 
-    def _announce_shape(self, out):
+    def _announce_shape_once(self, out):
+        if self.is_last or self._announced:
+            return
         dist.send_object_list([(tuple(out.shape), out.dtype)], dst=self.rank + 1)
+        self._announced = True
 
-    # -- the GPipe cycle ---------------------------------------------------
+    # -- per-microbatch halves --------------------------------------------
+
+    def _forward_one(self, m, ctx):
+        if self.is_first:
+            x = ctx["feeds"][m].detach()
+        else:
+            self._learn_shape_once()
+            x = self._recv_from_prev()
+        x.requires_grad_(not self.is_first)
+        y = self.stage(x)
+        ctx["x"][m], ctx["y"][m] = x, y
+        if not self.is_last:
+            self._announce_shape_once(y)
+            self._send_async(y.detach(), self.rank + 1)
+
+    def _backward_one(self, m, ctx):
+        y = ctx["y"][m]
+        if self.is_last:
+            loss = ctx["loss_fn"](y, ctx["targets"][m]) / self.num_microbatches
+            ctx["losses"].append(loss.detach())
+            loss.backward()
+        else:
+            gout = torch.empty_like(y)
+            dist.recv(gout, src=self.rank + 1)
+            y.backward(gout)
+        if not self.is_first:
+            self._send_async(ctx["x"][m].grad, self.rank - 1)
+        ctx["x"][m] = ctx["y"][m] = None  # release the microbatch's memory
+
+    # -- the training step -------------------------------------------------
 
     def train_step(
         self,
@@ -107,60 +156,47 @@ class PipelineParallelEngine:
         targets: Optional[torch.Tensor] = None,
         loss_fn: Optional[Callable] = None,
     ):
-        """One optimizer-ready step: microbatched fwd + bwd.
+        """One optimizer-ready cycle: M microbatched forwards + backwards.
 
-        - rank 0 passes ``inputs`` (full batch, split on dim 0)
+        - rank 0 passes ``inputs`` (full batch, chunked on dim 0)
         - the LAST rank passes ``targets`` and ``loss_fn(output, target)``
         - returns the mean loss tensor on the last rank, else None
         Parameter ``.grad``s hold the full-batch gradients afterwards.
         """
         M = self.num_microbatches
-        micro_in: List[torch.Tensor] = []
-        micro_out: List[torch.Tensor] = []
-        losses: List[torch.Tensor] = []
-
+        ctx = {"x": [None] * M, "y": [None] * M, "losses": [], "loss_fn": loss_fn}
         if self.is_first:
             if inputs is None:
                 raise ValueError("rank 0 must provide inputs")
-            feeds = list(torch.chunk(inputs, M, dim=0))
-        if self.is_last and (targets is None or loss_fn is None):
-            raise ValueError("last rank must provide targets and loss_fn")
+            ctx["feeds"] = list(torch.chunk(inputs, M, dim=0))
         if self.is_last:
-            target_chunks = list(torch.chunk(targets, M, dim=0))
+            if targets is None or loss_fn is None:
+                raise ValueError("last rank must provide targets and loss_fn")
+            ctx["targets"] = list(torch.chunk(targets, M, dim=0))
 
-        # ---- forward wave ----
-        for m in range(M):
-            if self.is_first:
-                x = feeds[m].detach()
-            else:
-                self._exchange_shape_once()
-                x = self._recv(*self._recv_shape)
-            x.requires_grad_(not self.is_first)
-            y = self.stage(x)
-            micro_in.append(x)
-            micro_out.append(y)
-            if not self.is_last:
-                if not self._announced:
-                    self._announce_shape(y)
-                    self._announced = True
-                self._send(y.detach(), self.rank + 1)
+        if self.schedule == "gpipe":
+            for m in range(M):
+                self._forward_one(m, ctx)
+            for m in reversed(range(M)):
+                self._backward_one(m, ctx)
+        else:  # 1f1b
+            warmup = min(self.world - 1 - self.rank, M)
+            fwd = bwd = 0
+            for _ in range(warmup):
+                self._forward_one(fwd, ctx)
+                fwd += 1
+            while fwd < M:
+                self._forward_one(fwd, ctx)
+                fwd += 1
+                self._backward_one(bwd, ctx)
+                bwd += 1
+            while bwd < M:
+                self._backward_one(bwd, ctx)
+                bwd += 1
 
-        # ---- backward wave (reverse microbatch order) ----
-        for m in reversed(range(M)):
-            y = micro_out[m]
-            if self.is_last:
-                loss = loss_fn(y, target_chunks[m]) / M  # mean over microbatches
-                losses.append(loss.detach())
-                loss.backward()
-            else:
-                gout = torch.empty_like(y)
-                dist.recv(gout, src=self.rank + 1)
-                y.backward(gout)
-            if not self.is_first:
-                self._send(micro_in[m].grad, self.rank - 1)
-
+        self._drain_sends()
         if self.is_last:
-            return torch.stack(losses).sum()
+            return torch.stack(ctx["losses"]).sum()
         return None
 
     def parameters(self):

@@ -12,10 +12,8 @@ from accelerate_amd import Accelerator, set_seed
 from accelerate_amd.parallel.pp import PipelineParallelEngine, split_into_stages
 
 
-def main():
-    acc = Accelerator(cpu=True)
+def run(acc, schedule):
     n, r = acc.num_processes, acc.process_index
-    assert n == 2
     set_seed(0)
     full = nn.Sequential(
         nn.Linear(8, 32), nn.Tanh(), nn.Linear(32, 32), nn.Tanh(), nn.Linear(32, 4)
@@ -23,7 +21,7 @@ def main():
     ref = nn.Sequential(*[nn.Linear(8, 32), nn.Tanh(), nn.Linear(32, 32), nn.Tanh(), nn.Linear(32, 4)])
     ref.load_state_dict(full.state_dict())
 
-    engine = PipelineParallelEngine(model=full, num_microbatches=4)
+    engine = PipelineParallelEngine(model=full, num_microbatches=4, schedule=schedule)
     opt = torch.optim.SGD(engine.parameters(), lr=0.1)
     ref_opt = torch.optim.SGD(ref.parameters(), lr=0.1)
     loss_fn = nn.MSELoss()
@@ -50,6 +48,13 @@ def main():
     stages = split_into_stages(ref, 2)
     for p_eng, p_ref in zip(engine.stage.parameters(), stages[r].parameters()):
         assert torch.allclose(p_eng, p_ref, atol=1e-6), (p_eng - p_ref).abs().max()
+
+
+def main():
+    acc = Accelerator(cpu=True)
+    assert acc.num_processes == 2
+    run(acc, "gpipe")
+    run(acc, "1f1b")
     if acc.is_main_process:
         print("PP_TRAIN_PASS")
     acc.end_training()
