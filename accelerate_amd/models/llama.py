@@ -223,3 +223,60 @@ class LlamaForCausalLM(nn.Module):
             next_tok = out["logits"][:, -1].argmax(-1, keepdim=True)
             tokens.append(next_tok)
         return torch.cat(tokens, dim=1)
+
+
+class LlamaPipelineStage(nn.Module):
+    """One pipeline stage of a LlamaForCausalLM (for parallel/pp.py).
+
+    First stage embeds token ids; middle stages transform hidden states;
+    the last stage applies the final norm + lm_head. Every stage keeps its
+    own RoPE tables (buffers are cheap to replicate; activations are what
+    travel between ranks).
+    """
+
+    def __init__(self, source: "LlamaForCausalLM", layer_lo: int, layer_hi: int,
+                 is_first: bool, is_last: bool):
+        super().__init__()
+        self.is_first, self.is_last = is_first, is_last
+        if is_first:
+            self.embed_tokens = source.embed_tokens
+        self.layers = nn.ModuleList(source.layers[layer_lo:layer_hi])
+        if is_last:
+            self.norm = source.norm
+            self.lm_head = source.lm_head
+        self.register_buffer("rope_cos", source.rope_cos, persistent=False)
+        self.register_buffer("rope_sin", source.rope_sin, persistent=False)
+
+    def forward(self, x):
+        if self.is_first:
+            x = self.embed_tokens(x.long())  # P2P delivers float buffers; ids ride as floats
+        cos = self.rope_cos.to(device=x.device)
+        sin = self.rope_sin.to(device=x.device)
+        for layer in self.layers:
+            x = layer(x, cos, sin)
+        if self.is_last:
+            return self.lm_head(self.norm(x))
+        return x
+
+
+def build_llama_pipeline_stages(model: "LlamaForCausalLM", n_stages: int):
+    """Slice a LlamaForCausalLM into pipeline stages (layer-balanced;
+    embed on the first, norm+head on the last). Pass ``stages[rank]`` as
+    the ``stage=`` of parallel/pp.PipelineParallelEngine."""
+    n_layers = len(model.layers)
+    per = [n_layers // n_stages + (1 if i < n_layers % n_stages else 0) for i in range(n_stages)]
+    stages, lo = [], 0
+    for i, k in enumerate(per):
+        stages.append(
+            LlamaPipelineStage(model, lo, lo + k, is_first=(i == 0), is_last=(i == n_stages - 1))
+        )
+        lo += k
+    return stages
+
+
+def causal_lm_loss(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """The shifted cross-entropy LlamaForCausalLM.forward uses — exposed so
+    a pipeline's last stage can compute the identical loss."""
+    return F.cross_entropy(
+        logits[:, :-1].reshape(-1, logits.shape[-1]).float(), labels.long()[:, 1:].reshape(-1)
+    )

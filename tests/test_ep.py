@@ -130,3 +130,10 @@ def test_pipeline_training_two_process():
 
     out = launch_distributed("tests/distributed_scripts/pp_train_script.py", nproc=2)
     assert "PP_TRAIN_PASS" in out
+
+
+def test_pipeline_llama_two_process():
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/pp_llama_script.py", nproc=2)
+    assert "PP_LLAMA_PASS" in out
