@@ -137,3 +137,17 @@ def test_pipeline_llama_two_process():
 
     out = launch_distributed("tests/distributed_scripts/pp_llama_script.py", nproc=2)
     assert "PP_LLAMA_PASS" in out
+
+
+def test_pp_engine_validation_errors():
+    import pytest as _pytest
+    import torch.nn as nn
+
+    from accelerate_amd.parallel.pp import PipelineParallelEngine, split_into_stages
+
+    # schedule validation happens before any dist requirement
+    with _pytest.raises((ValueError, RuntimeError)):
+        PipelineParallelEngine(model=nn.Sequential(nn.Linear(2, 2)), schedule="wavefront")
+    # degenerate split still returns n stages
+    stages = split_into_stages(nn.Sequential(nn.Linear(2, 2)), 3)
+    assert len(stages) == 3
