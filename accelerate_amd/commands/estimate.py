@@ -16,6 +16,7 @@ _BUILTIN = {
     "gpt2-large": ("accelerate_amd.models", "GPT2Config", "gpt2_large", "GPT2LMHeadModel"),
     "llama3-8b": ("accelerate_amd.models", "LlamaConfig", "llama3_8b", "LlamaForCausalLM"),
     "llama3-70b": ("accelerate_amd.models", "LlamaConfig", "llama3_70b", "LlamaForCausalLM"),
+    "llama3-405b": ("accelerate_amd.models", "LlamaConfig", "llama3_405b", "LlamaForCausalLM"),
     "mixtral-8x7b": ("accelerate_amd.models", "LlamaMoEConfig", "mixtral_8x7b_shape", "LlamaMoEForCausalLM"),
     "t5-11b": ("accelerate_amd.models", "T5Config", "t5_11b", "T5ForConditionalGeneration"),
 }
@@ -60,7 +61,7 @@ def add_parser(subparsers):
     parser = subparsers.add_parser("estimate", help="Estimate model memory usage")
     parser.add_argument(
         "model_name",
-        help="bundled family (bert-base, gpt2-large, llama3-8b, llama3-70b, "
+        help="bundled family (bert-base, gpt2-large, llama3-8b, llama3-70b, llama3-405b, "
         "mixtral-8x7b, t5-11b), transformers model name, or local path",
     )
     parser.add_argument("--dtypes", nargs="+", default=["float32", "float16", "int8"],

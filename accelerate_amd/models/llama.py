@@ -33,6 +33,15 @@ class LlamaConfig:
         return cls(**overrides)
 
     @classmethod
+    def llama3_405b(cls, **overrides):
+        base = dict(
+            hidden_size=16384, intermediate_size=53248, num_hidden_layers=126,
+            num_attention_heads=128, num_key_value_heads=8,
+        )
+        base.update(overrides)
+        return cls(**base)
+
+    @classmethod
     def llama3_70b(cls, **overrides):
         base = dict(
             hidden_size=8192, intermediate_size=28672, num_hidden_layers=80,
