@@ -396,6 +396,11 @@ class DataLoaderShard(DataLoaderAdapter, DataLoaderStateMixin):
         batch_index = 0
         while True:
             try:
+                # Snapshot loader state BEFORE fetching the lookahead batch: the
+                # captured state then equals the number of batches the caller has
+                # consumed, so a mid-epoch save/resume does not replay batches
+                # (reference: data_loader.py:597).
+                self._update_state_dict()
                 # fetch next BEFORE yielding current (lookahead)
                 next_batch = next(dataloader_iter)
                 if batch_index >= self.skip_batches:
@@ -593,7 +598,10 @@ class DataLoaderDispatcher(DataLoaderAdapter, DataLoaderStateMixin):
             if exhausted:
                 self.end_of_dataloader = True
                 self._update_state_dict()
-                self.remainder = global_rows % self.state.num_processes
+                # Real (pre-padding) row count of the final global batch:
+                # gather_for_metrics truncates its gathered tail to this many
+                # rows (reference: data_loader.py:941 observed_batch_size).
+                self.remainder = global_rows
             if position >= self.skip_batches:
                 yield current
             position += 1
@@ -701,12 +709,6 @@ def prepare_data_loader(
             num_samples=sampler._num_samples,
             generator=getattr(sampler, "generator", torch.Generator()),
         )
-    if isinstance(dataloader.sampler, RandomSampler):
-        # RNG sync of the sampler generator across ranks happens each epoch
-        generator = torch.Generator().manual_seed(42)
-        dataloader.generator = generator
-        dataloader.sampler.generator = generator
-
     if num_processes != 1 and not dispatch_batches:
         if isinstance(new_dataset, IterableDataset):
             if getattr(dataloader.dataset, "generator", None) is not None:

@@ -67,6 +67,7 @@ class PipelineParallelEngine:
         num_microbatches: int = 4,
         schedule: str = "gpipe",
         group=None,
+        device: Optional[torch.device] = None,
     ):
         if not dist.is_initialized():
             raise RuntimeError("PipelineParallelEngine needs torch.distributed initialized")
@@ -83,6 +84,20 @@ class PipelineParallelEngine:
             self.stage = split_into_stages(model, self.world)[self.rank]
         else:
             raise ValueError("pass either a Sequential model or this rank's stage")
+        # Under an RCCL world the stage and every P2P buffer must live on this
+        # rank's GPU (recv into a CPU tensor fails on the nccl backend); under
+        # gloo everything stays on CPU. Resolve from the caller, the current
+        # PartialState device, or the stage's own parameters — in that order.
+        if device is None:
+            from ..state import PartialState
+
+            state = PartialState._shared_state
+            if state.get("device") is not None and dist.get_backend() != "gloo":
+                device = state["device"]
+            else:
+                device = next(self.stage.parameters(), torch.empty(0)).device
+        self.device = torch.device(device)
+        self.stage.to(self.device)
         self.is_first = self.rank == 0
         self.is_last = self.rank == self.world - 1
         self._recv_shape = None   # learned on the first step (downstream)
@@ -102,7 +117,7 @@ class PipelineParallelEngine:
 
     def _recv_from_prev(self):
         shape, dtype = self._recv_shape
-        buf = torch.empty(shape, dtype=dtype)
+        buf = torch.empty(shape, dtype=dtype, device=self.device)
         dist.recv(buf, src=self.rank - 1)
         return buf
 
@@ -168,11 +183,11 @@ class PipelineParallelEngine:
         if self.is_first:
             if inputs is None:
                 raise ValueError("rank 0 must provide inputs")
-            ctx["feeds"] = list(torch.chunk(inputs, M, dim=0))
+            ctx["feeds"] = list(torch.chunk(inputs.to(self.device), M, dim=0))
         if self.is_last:
             if targets is None or loss_fn is None:
                 raise ValueError("last rank must provide targets and loss_fn")
-            ctx["targets"] = list(torch.chunk(targets, M, dim=0))
+            ctx["targets"] = list(torch.chunk(targets.to(self.device), M, dim=0))
 
         if self.schedule == "gpipe":
             for m in range(M):

@@ -96,6 +96,18 @@ def test_dispatcher(acc):
     got = sorted(torch.cat(collected).tolist())
     assert got == [float(i) for i in range(18)], got
 
+    # ADVICE-1 regression: tail with MORE real rows than num_processes and
+    # non-divisible — 7 samples, per-rank batch 2, world 2 → final global
+    # batch has 3 real rows; gather_for_metrics must return all 3, not 1
+    ds7 = TensorDataset(torch.arange(7).float())
+    dl7 = acc.prepare_data_loader(DataLoader(ds7, batch_size=2))
+    assert isinstance(dl7, DataLoaderDispatcher)
+    parts = []
+    for (b,) in dl7:
+        parts.append(acc.gather_for_metrics(b))
+    got7 = sorted(torch.cat(parts).tolist())
+    assert got7 == [float(i) for i in range(7)], f"tail dedup lost samples: {got7}"
+
     # split_batches dispatch: rank 0 fetches ONE global batch of 4 and each
     # rank receives exactly half of every batch
     acc.split_batches = True

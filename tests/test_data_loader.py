@@ -254,3 +254,47 @@ class TestBatchSamplerShardProperties:
         if drop_last:
             flat = [i for s in shards for b in s for i in b]
             assert len(flat) == len(set(flat))
+
+
+class CountingDataset(TensorDataset):
+    """TensorDataset counting __getitem__ calls — proxy for items fetched."""
+
+    def __init__(self, n):
+        super().__init__(torch.arange(n).float())
+        self.fetches = 0
+
+    def __getitem__(self, idx):
+        self.fetches += 1
+        return super().__getitem__(idx)
+
+
+class TestAdviceRegressions:
+    """Regression tests for the round-1 advisor findings."""
+
+    def test_user_generator_not_clobbered(self):
+        # prepare_data_loader must not replace a user-supplied shuffle
+        # generator with a fixed seed-42 one (non-XLA framework).
+        ds = TensorDataset(torch.arange(20).float())
+        gen = torch.Generator().manual_seed(7)
+        dl = DataLoader(ds, batch_size=1, shuffle=True, generator=gen)
+        prepared = prepare_data_loader(dl)
+        order = [int(b[0].item()) for b in prepared]
+
+        gen2 = torch.Generator().manual_seed(7)
+        expected = [int(b[0].item()) for b in DataLoader(ds, batch_size=1, shuffle=True, generator=gen2)]
+        assert order == expected, "user generator was clobbered (shuffle order != seed-7 order)"
+
+    def test_shard_state_snapshot_before_yield(self):
+        # With a stateful base loader, the snapshot returned by state_dict()
+        # must reflect batches the CALLER consumed, not the lookahead fetch
+        # and not only end-of-epoch (ADVICE item on _update_state_dict timing).
+        ds = CountingDataset(5)
+        dl = prepare_data_loader(DataLoader(ds, batch_size=1))
+        assert isinstance(dl, DataLoaderShard)
+        dl.base_dataloader.state_dict = lambda: {"fetched": ds.fetches}
+
+        it = iter(dl)
+        next(it)  # consumed 1 (lookahead has fetched 2)
+        assert dl.state_dict()["fetched"] == 1
+        next(it)  # consumed 2
+        assert dl.state_dict()["fetched"] == 2
