@@ -585,10 +585,7 @@ class Accelerator:
                 # sharded world: the clip must happen on the gradient SHARDS
                 # with a cross-rank norm reduction (reference: FSDP
                 # model.clip_grad_norm_, accelerator.py:2977-3007)
-                param_ids = {id(p) for p in parameters}
-                if any(id(p) in param_ids for u in model.units for p in u.params) or any(
-                    id(u.shard) in param_ids for u in model.units
-                ):
+                if any(id(p) in model._unit_param_ids for p in parameters):
                     return model.clip_grad_norm_(max_norm, norm_type)
         from .ops.clip_grad import clip_grad_norm_ as _clip
 

@@ -152,10 +152,18 @@ class DistributedDataParallelEngine(nn.Module):
             and dist.is_initialized()
             and dist.get_backend(process_group) == "nccl"
         )
+        self.find_unused_parameters = find_unused_parameters
         self._hooks = []
         self._buckets: List[_Bucket] = []
         self._param_to_bucket = {}
         self._float_buffers = None
+        self._buffer_flat = None
+        self._param_names = {id(p): n for n, p in module.named_parameters()}
+        # first-iteration observed grad-ready order; buckets are rebuilt from
+        # it once so bucket boundaries align with actual backward completion
+        # order (reference: torch DDP's reducer bucket rebuild)
+        self._ready_order: List[torch.nn.Parameter] = []
+        self._buckets_rebuilt = False
         self._build_buckets()
         self._register_hooks()
         if self._world_size > 1:
@@ -163,15 +171,23 @@ class DistributedDataParallelEngine(nn.Module):
 
     # -- setup ------------------------------------------------------------
 
-    def _build_buckets(self):
+    def _build_buckets(self, ordered_params: Optional[List[torch.nn.Parameter]] = None):
         # params marked _no_ddp_sync are rank-LOCAL (expert-parallel experts:
         # each rank owns different experts, so averaging them would be wrong)
         params = [
             p for p in self.module.parameters()
             if p.requires_grad and not getattr(p, "_no_ddp_sync", False)
         ]
-        # reverse registration order ≈ order grads become ready in backward
-        params = list(reversed(params))
+        if ordered_params is not None:
+            # rebuild from observed ready order; stragglers keep their old
+            # relative (reverse-registration) position at the end
+            seen = {id(p) for p in ordered_params}
+            params = list(ordered_params) + [p for p in reversed(params) if id(p) not in seen]
+        else:
+            # reverse registration order ≈ order grads become ready in backward
+            params = list(reversed(params))
+        self._buckets = []
+        self._param_to_bucket = {}
         current, current_bytes, current_key = [], 0, None
         for p in params:
             key = (p.dtype, p.device)
@@ -223,6 +239,8 @@ class DistributedDataParallelEngine(nn.Module):
     def _on_grad_ready(self, param: torch.nn.Parameter):
         if not self.require_backward_grad_sync or self._world_size <= 1:
             return
+        if not self._buckets_rebuilt:
+            self._ready_order.append(param)
         bucket, index = self._param_to_bucket[param]
         if index in bucket.ready:
             return
@@ -253,9 +271,22 @@ class DistributedDataParallelEngine(nn.Module):
         """
         if not self.require_backward_grad_sync or self._world_size <= 1:
             return
+        if not any(b.ready for b in self._buckets):
+            # this engine saw no grads at all this backward — it did not
+            # participate (e.g. another prepared model's loss), nothing to do
+            return
         for bucket in self._buckets:
             if not bucket.launched:
-                # stragglers (params that never got a grad contribute zeros)
+                # stragglers: params that produced no grad this backward
+                unused = [p for i, p in enumerate(bucket.params) if i not in bucket.ready]
+                if unused and not self.find_unused_parameters:
+                    names = ", ".join(self._param_names.get(id(p), "<unnamed>") for p in unused[:8])
+                    raise RuntimeError(
+                        f"{len(unused)} parameter(s) received no gradient this backward "
+                        f"(first few: {names}). If parts of the model are conditionally "
+                        f"unused, construct the engine with find_unused_parameters=True "
+                        f"(zero contributions are then reduced for them)."
+                    )
                 for p in bucket.params:
                     if p.grad is None:
                         p.grad = torch.zeros_like(p)
@@ -276,15 +307,54 @@ class DistributedDataParallelEngine(nn.Module):
                     else:
                         p.grad.copy_(reduced, non_blocking=True)
             bucket.reset()
+        self._maybe_rebuild_buckets()
+
+    def _maybe_rebuild_buckets(self):
+        """After the first synchronizing backward, re-bucket by the observed
+        grad-ready order so each bucket fills (and its all-reduce launches)
+        as early as possible during backward. Skipped inside hipGraph capture
+        (the rebuild allocates new flat buffers)."""
+        if self._buckets_rebuilt:
+            return
+        if torch.cuda.is_available() and torch.cuda.is_current_stream_capturing():
+            return
+        self._buckets_rebuilt = True
+        if len(self._ready_order) > 1:
+            self._build_buckets(ordered_params=self._ready_order)
+        self._ready_order = []
 
     def forward(self, *args, **kwargs):
         if self.broadcast_buffers and self._world_size > 1 and self.module.training:
             # per-iteration sync only for mutable float buffers (BN running
-            # stats); constant/int buffers were broadcast once at wrap time
+            # stats); constant/int buffers were broadcast once at wrap time.
+            # All float buffers travel in ONE coalesced broadcast — a single
+            # collective launch per step instead of one per buffer (per-step
+            # latency tax at N=8 otherwise).
             if self._float_buffers is None:
                 self._float_buffers = [b for b in self.module.buffers() if b.is_floating_point() and b.numel() > 0]
-            for buf in self._float_buffers:
-                dist.broadcast(buf.data, src=self._src_rank, group=self.process_group)
+                if self._float_buffers:
+                    total = sum(b.numel() for b in self._float_buffers)
+                    # fp32 wire is lossless for bf16/fp16/fp32 buffers; widen
+                    # to fp64 only if any buffer needs it
+                    wire = (
+                        torch.float64
+                        if any(b.dtype == torch.float64 for b in self._float_buffers)
+                        else torch.float32
+                    )
+                    self._buffer_flat = torch.empty(total, dtype=wire, device=self._float_buffers[0].device)
+            if self._float_buffers:
+                flat = self._buffer_flat
+                if dist.get_rank(self.process_group) == 0:
+                    off = 0
+                    for b in self._float_buffers:
+                        flat[off : off + b.numel()].copy_(b.reshape(-1), non_blocking=True)
+                        off += b.numel()
+                dist.broadcast(flat, src=self._src_rank, group=self.process_group)
+                if dist.get_rank(self.process_group) != 0:
+                    off = 0
+                    for b in self._float_buffers:
+                        b.copy_(flat[off : off + b.numel()].view_as(b), non_blocking=True)
+                        off += b.numel()
         return self.module(*args, **kwargs)
 
     @contextmanager

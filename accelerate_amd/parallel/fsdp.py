@@ -141,12 +141,20 @@ class _FlatUnit:
         self.full.untyped_storage().resize_(0)
         self._resident = False
 
-    def reduce_grads(self, accumulate: bool = True):
-        """Flatten unit grads, reduce_scatter (mean) into the fp32 shard grad."""
+    def reduce_grads(self, accumulate: bool = True, use_stream=None):
+        """Flatten unit grads, reduce_scatter (mean) into the fp32 shard grad.
+
+        When ``use_stream`` is given, every op here runs on that stream (the
+        caller made it wait on the compute stream first); grads produced on
+        the compute stream are ``record_stream``-tagged so the caching
+        allocator cannot hand their blocks back while the copy is in flight.
+        """
         flat_grad = torch.zeros(self.padded, dtype=self.reduce_dtype, device=self.device)
         off = 0
         for p in self.params:
             if p.grad is not None:
+                if use_stream is not None:
+                    p.grad.record_stream(use_stream)
                 flat_grad[off : off + p.numel()].copy_(p.grad.reshape(-1).to(self.reduce_dtype))
                 p.grad = None
             off += p.numel()
@@ -264,7 +272,17 @@ class ShardedModel(nn.Module):
         self._register_hooks()
         self._fwd_order: List[_FlatUnit] = []
         self._order_recorded = False
+        # Two side streams: gathers (forward/backward prefetch all_gather)
+        # and reductions (backward flatten + reduce_scatter) — so a unit's
+        # grad reduction overlaps both the remaining backward compute AND the
+        # next unit's unshard instead of serializing with either
+        # (reference semantics: torch FSDP2's separate all-gather /
+        # reduce-scatter streams, fsdp_utils.py:741-903).
         self._comm_stream = torch.cuda.Stream() if device.type == "cuda" else None
+        self._reduce_stream = torch.cuda.Stream() if device.type == "cuda" else None
+        self._unit_param_ids = frozenset(
+            id(p) for u in self.units for p in u.params
+        ) | frozenset(id(u.shard) for u in self.units)
         if self._ac:
             self._apply_activation_checkpointing()
 
@@ -386,22 +404,49 @@ class ShardedModel(nn.Module):
         if self._comm_stream is not None:
             torch.cuda.current_stream().wait_stream(self._comm_stream)
 
+    def _reduce_unit(self, unit):
+        """Reduce-scatter a completed unit's grads off the critical path.
+
+        The whole flatten → reduce_scatter → (HSDP all_reduce) → fp32
+        accumulate chain runs on ``_reduce_stream``; the compute stream only
+        records an event, so backward compute of earlier units and the comm
+        stream's backward-prefetch all_gathers proceed concurrently.
+        ``finalize_backward`` joins the streams before the optimizer runs.
+        """
+        if self._reduce_stream is None:
+            unit.reduce_grads()
+            self._hsdp_allreduce(unit)
+            return
+        self._reduce_stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self._reduce_stream):
+            unit.reduce_grads(use_stream=self._reduce_stream)
+            self._hsdp_allreduce(unit)
+
+    def _hsdp_allreduce(self, unit):
+        if self.replica_group is None:
+            return
+        # HSDP level 2: average the shard grad across replicas
+        if _is_nccl(self.replica_group):
+            dist.all_reduce(unit.shard.grad, op=dist.ReduceOp.AVG, group=self.replica_group)
+        else:
+            dist.all_reduce(unit.shard.grad, group=self.replica_group)
+            unit.shard.grad.div_(dist.get_world_size(self.replica_group))
+
     def _on_grad_ready(self, param):
         unit = self._param_to_unit[id(param)]
         unit.grads_ready += 1
         if unit.grads_ready >= len(unit.params):
             unit.grads_ready = 0
-            if self.require_backward_grad_sync:
-                unit.reduce_grads()
-                if self.replica_group is not None:
-                    # HSDP level 2: average the shard grad across replicas
-                    if _is_nccl(self.replica_group):
-                        dist.all_reduce(unit.shard.grad, op=dist.ReduceOp.AVG, group=self.replica_group)
-                    else:
-                        dist.all_reduce(unit.shard.grad, group=self.replica_group)
-                        unit.shard.grad.div_(dist.get_world_size(self.replica_group))
-                unit.reshard()
-            # under no_sync keep full grads resident for accumulation
+            # Gradient accumulation under sharding reduce-scatters EVERY
+            # microbatch and accumulates into the fp32 SHARD grad — memory
+            # stays O(shard) instead of O(full model) across the accumulate
+            # window. By linearity (mean over ranks of a sum of microbatches
+            # == sum of per-microbatch rank means) the result is bitwise the
+            # semantics of defer-then-reduce; `no_sync` therefore trades one
+            # reduce_scatter per microbatch (1/n traffic of an all_reduce)
+            # for O(full-model) grad memory it would otherwise hold.
+            self._reduce_unit(unit)
+            unit.reshard()
 
     def forward(self, *args, **kwargs):
         if not self._order_recorded:
@@ -423,11 +468,18 @@ class ShardedModel(nn.Module):
         ``loss.backward()``.
         """
         self._wait_for_comm()
+        if self._reduce_stream is not None:
+            torch.cuda.current_stream().wait_stream(self._reduce_stream)
         for u in self.units:
             u.reshard()
 
     @contextlib.contextmanager
     def no_sync(self):
+        """Accumulation window. Under sharding this is NOT a communication
+        blackout: each microbatch still reduce-scatters into the shard grads
+        (see `_on_grad_ready`), so nothing here changes engine behavior — the
+        context exists for API parity with the DDP engine and so
+        `Accelerator.accumulate` composes identically over both."""
         old = self.require_backward_grad_sync
         self.require_backward_grad_sync = False
         try:
@@ -435,18 +487,29 @@ class ShardedModel(nn.Module):
         finally:
             self.require_backward_grad_sync = old
 
-    # gradient clipping (sharded): local norm² + all_reduce ---------------
+    # gradient clipping (sharded): local norm^p + all_reduce --------------
 
     def clip_grad_norm_(self, max_norm: float, norm_type: float = 2.0) -> torch.Tensor:
-        assert norm_type == 2.0, "only L2 clipping is supported for sharded models"
+        """Clip on the gradient SHARDS with a single cross-rank norm
+        reduction (reference: FSDP model.clip_grad_norm_,
+        accelerator.py:2977-3007). Any p-norm: SUM-reduce |g|^p partials
+        (MAX-reduce for inf)."""
+        norm_type = float(norm_type)
         device = self.device
         local = torch.zeros(1, device=device)
         for u in self.units:
             if u.shard.grad is not None:
-                local += (u.shard.grad.float() ** 2).sum()
+                g = u.shard.grad.float()
+                if norm_type == torch.inf:
+                    local = torch.maximum(local, g.abs().max().reshape(1))
+                else:
+                    local += (g.abs() ** norm_type).sum()
         if dist.is_initialized() and self.world > 1:
-            dist.all_reduce(local, group=self.group)
-        total_norm = local.sqrt().squeeze()
+            # the shard group covers every parameter exactly once; HSDP
+            # replicas hold identical (already replica-averaged) shard grads
+            op = dist.ReduceOp.MAX if norm_type == torch.inf else dist.ReduceOp.SUM
+            dist.all_reduce(local, op=op, group=self.group)
+        total_norm = local.squeeze() if norm_type == torch.inf else local.pow(1.0 / norm_type).squeeze()
         clip_coef = (max_norm / (total_norm + 1e-6)).clamp(max=1.0)
         for u in self.units:
             if u.shard.grad is not None:

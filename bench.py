@@ -32,10 +32,17 @@ def main():
     parser.add_argument("--steps", type=int, default=20)
     parser.add_argument("--warmup", type=int, default=5)
     parser.add_argument("--profile", action="store_true", help="emit a chrome trace under gpurun_out/")
+    parser.add_argument(
+        "--bucket-mb",
+        type=int,
+        default=int(os.environ.get("BENCH_BUCKET_MB", "64")),
+        help="DDP reducer bucket size in MiB (sweep 32/64/128 on the first 8-GPU lease)",
+    )
     args = parser.parse_args()
 
     from accelerate_amd import Accelerator, set_seed
     from accelerate_amd.models import BertConfig, BertForSequenceClassification
+    from accelerate_amd.utils.dataclasses import DistributedDataParallelKwargs
 
     on_gpu_env = torch.cuda.is_available()
     # hipBLASLt algorithm selection pre-tuned on MI355X (TunableOp CSV,
@@ -54,7 +61,10 @@ def main():
     # no autocast cast kernels, bf16 gradient all-reduce over xGMI
     bf16_weights = on_gpu_env and os.environ.get("BENCH_BF16_WEIGHTS", "1") == "1"
     mixed = "no" if bf16_weights or not on_gpu_env else "bf16"
-    accelerator = Accelerator(mixed_precision=mixed)
+    accelerator = Accelerator(
+        mixed_precision=mixed,
+        kwargs_handlers=[DistributedDataParallelKwargs(bucket_cap_mb=args.bucket_mb)],
+    )
     set_seed(42)
 
     n = accelerator.num_processes
@@ -231,6 +241,7 @@ def main():
                 "parallelism": f"dp{n}",
                 "optimizer": "fused_adamw_hip" if on_gpu else "torch_adamw",
                 "step_mode": step_mode,
+                "bucket_mb": args.bucket_mb,
                 "weights": "bf16+fp32_master" if bf16_weights else ("fp32+autocast_bf16" if on_gpu else "fp32"),
             },
         }

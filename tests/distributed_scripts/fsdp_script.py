@@ -116,16 +116,27 @@ def main():
                 assert torch.allclose(merged[k], v, atol=1e-6), k
         print("FSDP_MERGE_PASS")
 
-    # no_sync accumulation: grads held locally, reduced at boundary
+    # no_sync accumulation under sharding: every microbatch reduce-scatters
+    # into the SHARD grads (memory stays O(shard), never O(full model)) and
+    # the summed result equals a fresh defer-then-reduce of the same window
     opt.zero_grad()
     with acc.no_sync(model):
         loss = ((model(X[r::n][:4]) - Y[r::n][:4]) ** 2).mean()
         acc.backward(loss)
         for u in model.units:
-            assert u.shard.grad is None, "shard grads must NOT be reduced inside no_sync"
+            assert u.shard.grad is not None, "microbatch grads must land in the shard grad"
+            for p in u.params:
+                assert p.grad is None, "full per-param grads must not stay resident in no_sync"
+        mid = [u.shard.grad.clone() for u in model.units]
     loss = ((model(X[r::n][4:8]) - Y[r::n][4:8]) ** 2).mean()
     acc.backward(loss)
-    assert all(u.shard.grad is not None for u in model.units)
+    accum = [u.shard.grad.clone() for u in model.units]
+    # equivalence oracle: accumulated == sum of the two window microbatches
+    opt.zero_grad()
+    loss = ((model(X[r::n][4:8]) - Y[r::n][4:8]) ** 2).mean()
+    acc.backward(loss)
+    for u, m, a in zip(model.units, mid, accum):
+        assert torch.allclose(a, m + u.shard.grad, atol=1e-6), "no_sync accumulation drifted"
     if acc.is_main_process:
         print("FSDP_NOSYNC_PASS")
 
