@@ -96,15 +96,25 @@ class _FlatUnit:
         self.total = total
         self.shard_len = self.padded // self.world
 
-        # build the fp32 master shard from current values
-        flat = torch.zeros(self.padded, dtype=torch.float32, device=device)
-        off = 0
-        for p in self.params:
-            flat[off : off + p.numel()].copy_(p.detach().reshape(-1).to(device, torch.float32))
-            off += p.numel()
-        lo = self.rank * self.shard_len
-        self.shard = nn.Parameter(flat[lo : lo + self.shard_len].clone())
-        del flat
+        # zero-storage placeholders mark a meta-constructed unit (the
+        # ShardedModel meta path swaps meta params for stride-0 expands)
+        meta_init = any(p.untyped_storage().size() < p.numel() * p.element_size() for p in self.params)
+        if meta_init:
+            # meta-device construction: the full model is NEVER materialized —
+            # the shard starts zeroed and gets its values from a per-rank
+            # checkpoint-slice load or a per-unit init sweep
+            # (ShardedModel.materialize_and_init_ / fsdp_io.load_*)
+            self.shard = nn.Parameter(torch.zeros(self.shard_len, dtype=torch.float32, device=device))
+        else:
+            # build the fp32 master shard from current values
+            flat = torch.zeros(self.padded, dtype=torch.float32, device=device)
+            off = 0
+            for p in self.params:
+                flat[off : off + p.numel()].copy_(p.detach().reshape(-1).to(device, torch.float32))
+                off += p.numel()
+            lo = self.rank * self.shard_len
+            self.shard = nn.Parameter(flat[lo : lo + self.shard_len].clone())
+            del flat
 
         # persistent full buffer (storage resized 0<->padded across reshard)
         self.full = torch.empty(self.padded, dtype=self.compute_dtype, device=device)
@@ -173,6 +183,14 @@ class _FlatUnit:
             self.shard.grad = out32
         else:
             self.shard.grad.add_(out32)
+
+    def param_intervals(self):
+        """[(name, shape, start, end)] — each param's slot in the flat layout."""
+        out, off = [], 0
+        for name, shape, n in zip(self.param_names, self.shapes, self.numels):
+            out.append((name, shape, off, off + n))
+            off += n
+        return out
 
     @torch.no_grad()
     def gather_full_fp32(self) -> torch.Tensor:
@@ -245,7 +263,30 @@ class ShardedModel(nn.Module):
 
     # unit discovery ------------------------------------------------------
 
-        module.to(device)
+        self.meta_init = any(p.is_meta for p in module.parameters())
+        if self.meta_init:
+            # meta-device module (init_empty_weights with include_buffers=False):
+            # the full model is never materialized. A meta Parameter's .data
+            # cannot be re-pointed at a real tensor (incompatible tensor
+            # types), so each meta param is swapped for a ZERO-ALLOCATION
+            # placeholder — a stride-0 expand of a 1-element tensor with the
+            # right shape/dtype — which the unit then re-points into its flat
+            # buffer. Tied params share one placeholder so tying survives.
+            replacement = {}
+            for m in module.modules():
+                for pname, p in list(m.named_parameters(recurse=False)):
+                    if p is not None and p.is_meta:
+                        if id(p) not in replacement:
+                            ph = torch.zeros(1, dtype=p.dtype, device=device).expand(p.shape)
+                            replacement[id(p)] = nn.Parameter(ph, requires_grad=p.requires_grad)
+                        m._parameters[pname] = replacement[id(p)]
+            # buffers are real (include_buffers=False) and small; move them
+            for m in module.modules():
+                for bname, buf in list(m.named_buffers(recurse=False)):
+                    if buf is not None and not buf.is_meta:
+                        m._buffers[bname] = buf.to(device)
+        else:
+            module.to(device)
         unit_modules = self._select_units(module, auto_wrap_policy, transformer_cls_names, min_num_params)
         self.units: List[_FlatUnit] = []
         claimed = set()
@@ -285,6 +326,46 @@ class ShardedModel(nn.Module):
         ) | frozenset(id(u.shard) for u in self.units)
         if self._ac:
             self._apply_activation_checkpointing()
+
+    @classmethod
+    def plan(
+        cls,
+        module: nn.Module,
+        world_size: int,
+        transformer_cls_names=None,
+        min_num_params: int = 1_000_000,
+        compute_dtype: torch.dtype = torch.bfloat16,
+    ) -> dict:
+        """Allocation-free dispatch plan (meta-safe) for a given world size.
+
+        Returns per-rank byte budgets so huge-model feasibility (e.g.
+        llama3-405b on 8×288 GB) can be asserted without touching memory:
+        ``per_rank_master_bytes`` (fp32 shards), ``max_unit_full_bytes``
+        (transient all-gather buffer), ``per_rank_optim_bytes`` (AdamW m+v
+        on the shards).
+        """
+        unit_modules = cls._select_units(module, None, transformer_cls_names, min_num_params)
+        claimed, rows = set(), []
+        for name, m in list(unit_modules) + [("", module)]:
+            ps = [p for p in m.parameters() if p.requires_grad and id(p) not in claimed]
+            for p in ps:
+                claimed.add(id(p))
+            if not ps:
+                continue
+            total = sum(p.numel() for p in ps)
+            padded = -(-total // world_size) * world_size
+            rows.append({"unit": name or "<root>", "numel": total, "padded": padded})
+        esize = torch.finfo(compute_dtype).bits // 8
+        master = sum(r["padded"] // world_size * 4 for r in rows)
+        return {
+            "world_size": world_size,
+            "units": rows,
+            "total_numel": sum(r["numel"] for r in rows),
+            "per_rank_master_bytes": master,
+            "per_rank_optim_bytes": 2 * master,
+            "per_rank_grad_bytes": master,
+            "max_unit_full_bytes": max((r["padded"] * esize for r in rows), default=0),
+        }
 
     @staticmethod
     def _select_units(module, policy, transformer_cls_names, min_num_params):
@@ -564,6 +645,76 @@ class ShardedModel(nn.Module):
         for name, buf in self.module.named_buffers():
             if name in state_dict:
                 buf.copy_(state_dict[name].to(buf.device, buf.dtype))
+
+    @torch.no_grad()
+    def materialize_and_init_(self, init_fn=None, seed: Optional[int] = None):
+        """Initialize a meta-constructed model ONE UNIT AT A TIME.
+
+        Peak extra memory = one unit's fp32 flat buffer, never the model.
+        Every rank runs the identical init sweep (same ``seed``) and keeps
+        only its shard slice — no rank-0 broadcast, no collective at all
+        (the anti-pattern this replaces: reference
+        fsdp_utils.py:563-656 per-param broadcast load).
+
+        ``init_fn(module)`` is called once per *owning submodule* of each
+        unit's params (e.g. your model's ``_init_weights``); default is the
+        submodule's ``reset_parameters``.
+        """
+        if seed is not None:
+            torch.manual_seed(seed)
+        # param id -> owning submodule (the module holding it directly)
+        owner_of = {}
+        for m in self.module.modules():
+            for p in m.parameters(recurse=False):
+                owner_of[id(p)] = m
+        for u in self.units:
+            tmp = torch.zeros(u.padded, dtype=torch.float32, device=self.device)
+            saved = []
+            off = 0
+            for p, shape in zip(u.params, u.shapes):
+                saved.append(p.data)
+                p.data = tmp[off : off + shape.numel()].view(shape)
+                off += shape.numel()
+            owners, seen = [], set()
+            for p in u.params:
+                m = owner_of.get(id(p))
+                if m is not None and id(m) not in seen:
+                    seen.add(id(m))
+                    owners.append(m)
+            for m in owners:
+                if init_fn is not None:
+                    init_fn(m)
+                elif hasattr(m, "reset_parameters"):
+                    m.reset_parameters()
+            lo = u.rank * u.shard_len
+            u.shard.copy_(tmp[lo : lo + u.shard_len])
+            for p, s in zip(u.params, saved):
+                p.data = s
+            del tmp
+        self.meta_init = False
+
+    @torch.no_grad()
+    def load_shard_slices(self, fetch_flat_slice):
+        """Per-rank sliced load: each rank reads ONLY its shard's bytes.
+
+        ``fetch_flat_slice(param_name, lo, hi)`` returns elements [lo, hi) of
+        the named parameter's row-major flattening (any dtype; cast here).
+        Used by `parallel.fsdp_io.load_full_checkpoint_sliced` with
+        mmap-backed safetensors slices — memory and disk reads are O(shard)
+        per rank, and no cross-rank communication happens at all.
+        """
+        for u in self.units:
+            lo = u.rank * u.shard_len
+            hi = lo + u.shard_len
+            for name, shape, start, end in u.param_intervals():
+                a, b = max(lo, start), min(hi, end)
+                if a >= b:
+                    continue
+                values = fetch_flat_slice(name, a - start, b - start)
+                if values is None:
+                    raise KeyError(f"missing parameter {name} in sliced checkpoint load")
+                u.shard[a - lo : b - lo].copy_(values.reshape(-1).to(u.shard.device, torch.float32))
+        self.meta_init = False
 
     @torch.no_grad()
     def sharded_state_dict(self):

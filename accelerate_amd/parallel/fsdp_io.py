@@ -31,6 +31,97 @@ def load_fsdp_sharded_checkpoint(model, directory):
     model.load_sharded_state_dict(sd)
 
 
+def load_full_checkpoint_sliced(model, checkpoint_path):
+    """Load a FULL checkpoint into a ShardedModel with per-rank slicing.
+
+    Every rank opens the same file(s) and reads ONLY the byte ranges of its
+    own shard — O(shard) memory and IO per rank, zero communication. This
+    replaces the reference's rank-0-load + per-param broadcast
+    (fsdp_utils.py:563-656 `fsdp2_load_full_state_dict`) with the
+    shard-load-from-disk design SURVEY.md §5.8 calls for.
+
+    Accepts: a `.safetensors` file, a directory containing
+    `model.safetensors` or a `model.safetensors.index.json` shard index, or
+    a torch `.bin`/`.pt` file (mmap'd).
+
+    Slicing detail (safetensors): a flat range [a, b) of a row-major tensor
+    lies inside rows [a//rowsz, ceil(b/rowsz)); we fetch those rows with
+    `get_slice` (mmap window, no full-tensor read) and trim.
+    """
+    import json
+
+    path = Path(checkpoint_path)
+    name_to_file = {}
+    if path.is_dir():
+        idx = path / "model.safetensors.index.json"
+        if idx.exists():
+            weight_map = json.loads(idx.read_text())["weight_map"]
+            name_to_file = {k: path / v for k, v in weight_map.items()}
+        elif (path / "model.safetensors").exists():
+            path = path / "model.safetensors"
+        elif (path / "pytorch_model.bin").exists():
+            path = path / "pytorch_model.bin"
+        else:
+            raise FileNotFoundError(f"no loadable checkpoint in {checkpoint_path}")
+
+    if name_to_file or path.suffix == ".safetensors":
+        from safetensors import safe_open
+
+        handles = {}
+
+        def get_handle(fname):
+            if fname not in handles:
+                handles[fname] = safe_open(str(fname), framework="pt", device="cpu")
+            return handles[fname]
+
+        def fetch(name, lo, hi):
+            fname = name_to_file.get(name, path)
+            f = get_handle(fname)
+            if name not in f.keys():
+                return None
+            sl = f.get_slice(name)
+            shape = sl.get_shape()
+            if len(shape) <= 1:
+                return sl[lo:hi]
+            rowsz = 1
+            for d in shape[1:]:
+                rowsz *= d
+            r0, r1 = lo // rowsz, -(-hi // rowsz)
+            chunk = sl[r0:r1]
+            return chunk.reshape(-1)[lo - r0 * rowsz : hi - r0 * rowsz]
+
+        model.load_shard_slices(fetch)
+    else:
+        sd = torch.load(path, weights_only=True, mmap=True, map_location="cpu")
+
+        def fetch(name, lo, hi):
+            t = sd.get(name)
+            if t is None:
+                return None
+            return t.reshape(-1)[lo:hi]
+
+        model.load_shard_slices(fetch)
+    # buffers travel whole (they are small); best-effort from the same source
+    buffer_names = [n for n, _ in model.module.named_buffers()]
+    if buffer_names:
+        if name_to_file or path.suffix == ".safetensors":
+            from safetensors import safe_open
+
+            srcs = {}
+            files = set(name_to_file.values()) if name_to_file else {path}
+            for f in files:
+                with safe_open(str(f), framework="pt", device="cpu") as h:
+                    for k in h.keys():
+                        if k in buffer_names:
+                            srcs[k] = h.get_tensor(k)
+        else:
+            srcs = {k: v for k, v in sd.items() if k in buffer_names}
+        with torch.no_grad():
+            for n, buf in model.module.named_buffers():
+                if n in srcs:
+                    buf.copy_(srcs[n].to(buf.device, buf.dtype))
+
+
 def merge_fsdp_weights(checkpoint_directory, output_path, safe_serialization: bool = True):
     """Merge per-rank shard files into one full state dict
     (reference: fsdp_utils.py:462, `accelerate merge-weights`)."""

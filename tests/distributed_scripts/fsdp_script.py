@@ -140,6 +140,56 @@ def main():
     if acc.is_main_process:
         print("FSDP_NOSYNC_PASS")
 
+    # meta-device init + per-rank SLICED checkpoint load: the full model is
+    # never materialized on any rank (VERDICT missing #2; reference contrast:
+    # fsdp_utils.py:563-656 rank-0 broadcast load)
+    import safetensors.torch
+
+    from accelerate_amd.big_modeling import init_empty_weights
+    from accelerate_amd.parallel.fsdp_io import load_full_checkpoint_sliced
+
+    ref2 = make_model()
+    d2 = [tempfile.mkdtemp() if acc.is_main_process else None]
+    broadcast_object_list(d2)
+    ckpt = os.path.join(d2[0], "model.safetensors")
+    if acc.is_main_process:
+        safetensors.torch.save_file(ref2.state_dict(), ckpt)
+    acc.wait_for_everyone()
+
+    with init_empty_weights():
+        meta_mod = make_model()
+    assert all(p.is_meta for p in meta_mod.parameters())
+    sm = ShardedModel(meta_mod, min_num_params=100, device=torch.device("cpu"))
+    assert sm.meta_init
+    # nothing resident: every unit's full buffer must be size-0 post-init
+    for u in sm.units:
+        assert u.full.untyped_storage().size() == 0, "meta init left a full buffer resident"
+    load_full_checkpoint_sliced(sm, ckpt)
+    full2 = sm.full_state_dict()
+    for k, v in ref2.state_dict().items():
+        assert torch.allclose(full2[k], v, atol=1e-6), f"sliced load mismatch {k}"
+    with torch.no_grad():
+        out = sm(X[:8])
+        ref_out2 = ref2(X[:8])
+    assert torch.allclose(out, ref_out2, atol=1e-5), "meta-loaded forward diverges"
+    if acc.is_main_process:
+        print("FSDP_METALOAD_PASS")
+
+    # meta init WITHOUT a checkpoint: per-unit materialize-and-init sweep
+    with init_empty_weights():
+        meta_mod2 = make_model()
+    sm2 = ShardedModel(meta_mod2, min_num_params=100, device=torch.device("cpu"))
+    sm2.materialize_and_init_(seed=1234)
+    # ranks must hold CONSISTENT shards (same seed, same sweep): forward parity
+    with torch.no_grad():
+        y0 = sm2(X[:4])
+    ys = [torch.empty_like(y0) for _ in range(n)]
+    dist.all_gather(ys, y0)
+    assert torch.allclose(ys[0], ys[1], atol=1e-6), "materialize_and_init_ diverged across ranks"
+    assert any((u.shard != 0).any() for u in sm2.units), "init left shards zero"
+    if acc.is_main_process:
+        print("FSDP_METAINIT_PASS")
+
     acc.end_training()
 
 
