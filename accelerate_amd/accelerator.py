@@ -422,6 +422,41 @@ class Accelerator:
                     " Please rerun your script specifying `--num_processes=1` or by launching with `python {{myscript.py}}`."
                 )
 
+        # multi-dimensional worlds (TP/CP x DP) route through the
+        # ParallelismConfig groups (reference: accelerator.py:1531-1560
+        # _prepare_tp/_prepare_cp): shard each model by its tp_plan, register
+        # the sequence-parallel context, then fall through so DP wrapping and
+        # dataloader sharding happen on the dp dimension only.
+        pc = self.parallelism_config
+        if pc is not None and (pc.tp_size > 1 or pc.cp_size > 1):
+            if not pc._groups:
+                pc.build_groups()
+            if pc.tp_size > 1:
+                from .parallel.tp import apply_tp_plan
+
+                args = tuple(
+                    apply_tp_plan(obj, group=pc._groups["tp"]) if isinstance(obj, torch.nn.Module) else obj
+                    for obj in args
+                )
+                # re-point already-constructed optimizers at the sharded
+                # params (reference: accelerator.py:1647-1654 optimizer remap)
+                swap = {}
+                for obj in args:
+                    if isinstance(obj, torch.nn.Module):
+                        swap.update(getattr(obj, "_tp_param_swap", {}))
+                for obj in args:
+                    if isinstance(obj, torch.optim.Optimizer):
+                        if len(obj.state) > 0:
+                            raise RuntimeError(
+                                "prepare() with tensor parallelism must run before the "
+                                "optimizer takes its first step (param swap would orphan state)"
+                            )
+                        for group in obj.param_groups:
+                            group["params"] = [swap.get(id(p), p) for p in group["params"]]
+            # cp activation is scoped to `maybe_context_parallel` (the
+            # per-step ctx manager, reference accelerator.py:4111) so
+            # unprepared/eval models are never caught by the ambient context
+
         if self.distributed_type == DistributedType.FSDP:
             from .parallel.fsdp import fsdp_prepare
 
@@ -466,13 +501,21 @@ class Accelerator:
         if device_placement and not self.verify_device_map(model):
             model = model.to(self.device)
 
-        if not evaluation_mode and self.use_distributed and self.distributed_type == DistributedType.MULTI_GPU:
+        pc = self.parallelism_config
+        multi_dim = pc is not None and (pc.tp_size > 1 or pc.cp_size > 1) and pc._groups
+        if not evaluation_mode and self.use_distributed and self.distributed_type in (
+            DistributedType.MULTI_GPU,
+            DistributedType.MULTI_CPU,
+        ):
             ddp_kwargs = self.ddp_handler.to_dict() if self.ddp_handler is not None else {}
-            model = DistributedDataParallelEngine(model, **ddp_kwargs)
-        elif not evaluation_mode and self.use_distributed and self.distributed_type == DistributedType.MULTI_CPU:
-            # CPU gloo world (tests): same engine, gloo collectives
-            ddp_kwargs = self.ddp_handler.to_dict() if self.ddp_handler is not None else {}
-            model = DistributedDataParallelEngine(model, **ddp_kwargs)
+            if multi_dim:
+                # gradient averaging happens on the dp x cp domain; tp ranks
+                # hold DIFFERENT shards that must not be averaged (cp ranks
+                # replicate params over different sequence shards)
+                if pc.dp_size * pc.cp_size > 1:
+                    model = DistributedDataParallelEngine(model, process_group=pc._groups["grad"], **ddp_kwargs)
+            else:
+                model = DistributedDataParallelEngine(model, **ddp_kwargs)
         model._is_accelerate_prepared = True
         self._models.append(model)
         return model
@@ -485,11 +528,19 @@ class Accelerator:
             return data_loader
         if device_placement is None:
             device_placement = self.device_placement
+        # Under TP/CP the batch is sharded over the DP dimension only: every
+        # rank in the same tp/cp group must see the SAME batch (reference:
+        # data_loader.py:1129-1165 process_index // (tp*cp) remap).
+        dl_num, dl_idx = self.num_processes, self.process_index
+        pc = self.parallelism_config
+        if pc is not None and (pc.tp_size > 1 or pc.cp_size > 1):
+            dl_num = pc.dp_size
+            dl_idx = self.data_parallel_rank
         prepared = prepare_data_loader(
             data_loader,
             self.device,
-            num_processes=self.num_processes,
-            process_index=self.process_index,
+            num_processes=dl_num,
+            process_index=dl_idx,
             split_batches=self.split_batches,
             put_on_device=device_placement,
             rng_types=self.rng_types.copy() if self.rng_types else None,
@@ -526,11 +577,14 @@ class Accelerator:
             if getattr(scheduler, "optimizer", None) == opt.optimizer:
                 optimizer = opt
                 break
+        pc = self.parallelism_config
+        shards = pc.dp_size if pc is not None and (pc.tp_size > 1 or pc.cp_size > 1) else None
         scheduler = AcceleratedScheduler(
             scheduler,
             optimizer,
             step_with_optimizer=self.step_scheduler_with_optimizer,
             split_batches=self.split_batches,
+            num_batch_shards=shards,
         )
         scheduler._is_accelerate_prepared = True
         self._schedulers.append(scheduler)
@@ -767,13 +821,24 @@ class Accelerator:
         group = None
         if pc is not None and pc.cp_size > 1 and pc._groups:
             group = pc._groups.get("cp")
-        if group is not None and buffers:
+        if group is None:
+            yield
+            return
+        if buffers:
             from .parallel.cp import shard_sequence
 
             dims = buffer_seq_dims or [1] * len(buffers)
             for i, (buf, dim) in enumerate(zip(buffers, dims)):
                 buffers[i] = shard_sequence(buf, group=group, dim=dim)
-        yield
+        # activate the sequence-parallel collective pattern for the scope of
+        # this step: models route attention through ops.attention.dispatch
+        from .ops.attention import set_sequence_parallel
+
+        set_sequence_parallel(pc.cp_impl, group)
+        try:
+            yield
+        finally:
+            set_sequence_parallel(None)
 
     @contextmanager
     def autocast(self, autocast_handler: AutocastKwargs = None):

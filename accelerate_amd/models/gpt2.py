@@ -51,10 +51,21 @@ class GPT2Attention(nn.Module):
 
     def forward(self, x, kv_cache=None):
         B, S, H = x.shape
-        q, k, v = self.c_attn(x).split(H, dim=2)
+        y = self.c_attn(x)
+        # split by the PROJECTION's width, not the input width: under tensor
+        # parallelism c_attn is column-sharded and yields [q_loc|k_loc|v_loc]
+        q, k, v = y.split(y.shape[-1] // 3, dim=2)
         q = q.view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
         k = k.view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
         v = v.view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
+        from ..ops.attention import dispatch_attention, sequence_parallel_info
+
+        sp_mode, _, _, _ = sequence_parallel_info()
+        if sp_mode is not None and kv_cache is None and S > 1:
+            # sequence shard path: positions were offset in the model forward
+            ctx = dispatch_attention(q, k, v, causal=True)
+            ctx = ctx.transpose(1, 2).reshape(B, S, -1)
+            return self.c_proj(ctx)
         past = 0
         if kv_cache is not None:
             past = kv_cache["len"]
@@ -76,7 +87,7 @@ class GPT2Attention(nn.Module):
                 scores = scores.masked_fill(~mask, torch.finfo(scores.dtype).min)
             probs = F.softmax(scores.float(), dim=-1).to(v.dtype)
             ctx = torch.matmul(probs, v)
-        ctx = ctx.transpose(1, 2).reshape(B, S, H)
+        ctx = ctx.transpose(1, 2).reshape(B, S, -1)
         return self.c_proj(ctx)
 
 
@@ -105,6 +116,16 @@ class GPT2Block(nn.Module):
 
 
 class GPT2LMHeadModel(nn.Module):
+    # model-generic TP contract consumed by parallel.tp.apply_tp_plan via
+    # Accelerator.prepare; c_attn is a fused [q|k|v] projection
+    tp_plan = {
+        "h.*.attn.c_attn": "colwise_fused3",
+        "h.*.attn.c_proj": "rowwise",
+        "h.*.mlp.c_fc": "colwise",
+        "h.*.mlp.c_proj": "rowwise",
+    }
+    tp_shard_attrs = {"h.*.attn": ("n_heads",)}
+
     def __init__(self, config: GPT2Config = None):
         super().__init__()
         c = self.config = config or GPT2Config()
@@ -125,7 +146,11 @@ class GPT2LMHeadModel(nn.Module):
     def forward(self, input_ids, labels=None, kv_caches=None):
         B, S = input_ids.shape
         past = kv_caches[0]["len"] if kv_caches is not None else 0
-        pos = torch.arange(past, past + S, device=input_ids.device)
+        from ..ops.attention import sequence_parallel_info
+
+        sp_mode, _, sp_rank, _ = sequence_parallel_info()
+        pos0 = past + (sp_rank * S if sp_mode is not None and kv_caches is None else 0)
+        pos = torch.arange(pos0, pos0 + S, device=input_ids.device)
         x = self.wte(input_ids) + self.wpe(pos)[None]
         for i, block in enumerate(self.h):
             x = block(x, kv_caches[i] if kv_caches is not None else None)

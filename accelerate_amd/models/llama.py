@@ -105,6 +105,18 @@ class LlamaAttention(nn.Module):
         q = self.q_proj(x).view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
         k = self.k_proj(x).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
         v = self.v_proj(x).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
+        # sequence parallelism (CP/Ulysses): x is this rank's sequence shard;
+        # RoPE uses the shard's ABSOLUTE positions and the attention runs
+        # through the registered collective pattern (ops.attention dispatch)
+        from ..ops.attention import dispatch_attention, sequence_parallel_info
+
+        sp_mode, _, sp_rank, _ = sequence_parallel_info()
+        if sp_mode is not None and kv_cache is None and S > 1:
+            q = apply_rope(q, cos[sp_rank * S :], sin[sp_rank * S :])
+            k = apply_rope(k, cos[sp_rank * S :], sin[sp_rank * S :])
+            ctx = dispatch_attention(q, k, v, causal=True)
+            ctx = ctx.transpose(1, 2).reshape(B, S, -1)
+            return self.o_proj(ctx)
         q = apply_rope(q, cos, sin)
         k = apply_rope(k, cos, sin)
         past_len = 0
@@ -166,6 +178,19 @@ class LlamaDecoderLayer(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
+    # model-generic TP contract consumed by parallel.tp.apply_tp_plan via
+    # Accelerator.prepare (the transformers `tp_plan` idiom)
+    tp_plan = {
+        "layers.*.self_attn.q_proj": "colwise",
+        "layers.*.self_attn.k_proj": "colwise",
+        "layers.*.self_attn.v_proj": "colwise",
+        "layers.*.self_attn.o_proj": "rowwise",
+        "layers.*.mlp.gate_proj": "colwise",
+        "layers.*.mlp.up_proj": "colwise",
+        "layers.*.mlp.down_proj": "rowwise",
+    }
+    tp_shard_attrs = {"layers.*.self_attn": ("n_heads", "n_kv")}
+
     def __init__(self, config: LlamaConfig = None, attn_impl: str = "chunked"):
         super().__init__()
         self.config = config or LlamaConfig()

@@ -188,3 +188,62 @@ def flash_attention(
 
 # alias used by models (the fused HIP kernel will take over this symbol)
 flash_attention_forward = flash_attention
+
+
+# ---------------------------------------------------------------------------
+# Sequence-parallel dispatch (the model-generic CP/SP route)
+#
+# `Accelerator.prepare` registers the active context here when
+# ParallelismConfig declares cp_size > 1; any model whose attention goes
+# through `dispatch_attention` (all in-repo families) is then context- or
+# Ulysses-parallel with NO model-specific patching (reference contrast:
+# accelerator.py:1658 _prepare_cp requires torch experimental
+# context_parallel + SDPA hooks).
+# ---------------------------------------------------------------------------
+
+_SEQ_PARALLEL = {"mode": None, "group": None}
+
+
+def set_sequence_parallel(mode: Optional[str], group=None):
+    """mode: None (off) | 'allgather' (CP: KV all-gather) | 'ulysses'
+    (SP: dual all-to-all head resharding)."""
+    if mode not in (None, "allgather", "ulysses"):
+        raise ValueError(f"unknown sequence-parallel mode {mode!r}")
+    _SEQ_PARALLEL["mode"] = mode
+    _SEQ_PARALLEL["group"] = group
+
+
+def sequence_parallel_info():
+    """(mode, group, rank, world) of the active sequence-parallel context;
+    (None, None, 0, 1) when off."""
+    import torch.distributed as dist
+
+    mode, group = _SEQ_PARALLEL["mode"], _SEQ_PARALLEL["group"]
+    if mode is None or not dist.is_initialized():
+        return None, None, 0, 1
+    world = dist.get_world_size(group)
+    if world == 1:
+        return None, None, 0, 1
+    return mode, group, dist.get_rank(group), world
+
+
+def dispatch_attention(q, k, v, causal: bool = True):
+    """Attention entry point for models: local flash attention normally;
+    under an active sequence-parallel context, q/k/v are this rank's
+    sequence shard and the collective pattern of the registered mode runs
+    around the kernel. GQA inputs are head-expanded first on the parallel
+    paths (the collectives need matching head counts)."""
+    mode, group, _, _ = sequence_parallel_info()
+    if mode is None:
+        return flash_attention(q, k, v, causal=causal)
+    if k.shape[1] != q.shape[1]:
+        rep = q.shape[1] // k.shape[1]
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    if mode == "allgather":
+        from ..parallel.cp import context_parallel_attention
+
+        return context_parallel_attention(q, k, v, group=group, causal=causal)
+    from ..parallel.sp import ulysses_attention
+
+    return ulysses_attention(q, k, v, group=group, causal=causal)

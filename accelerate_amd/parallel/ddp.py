@@ -331,7 +331,15 @@ class DistributedDataParallelEngine(nn.Module):
             # collective launch per step instead of one per buffer (per-step
             # latency tax at N=8 otherwise).
             if self._float_buffers is None:
-                self._float_buffers = [b for b in self.module.buffers() if b.is_floating_point() and b.numel() > 0]
+                # non-persistent buffers (absent from state_dict) are derived
+                # constants (e.g. RoPE tables) — identical by construction,
+                # never broadcast
+                persistent = set(self.module.state_dict(keep_vars=True))
+                self._float_buffers = [
+                    b
+                    for n, b in self.module.named_buffers()
+                    if n in persistent and b.is_floating_point() and b.numel() > 0
+                ]
                 if self._float_buffers:
                     total = sum(b.numel() for b in self._float_buffers)
                     # fp32 wire is lossless for bf16/fp16/fp32 buffers; widen

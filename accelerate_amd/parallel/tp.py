@@ -149,24 +149,105 @@ class RowParallelLinear(nn.Module):
         return mod
 
 
-def tp_parallelize_llama(model, group=None):
-    """Shard a LlamaForCausalLM in place (megatron pattern): column q/k/v +
-    gate/up, row o/down; per-rank head counts. One all-reduce per attention
-    and one per MLP per layer."""
+def _fused_colwise_from_linear(linear: nn.Linear, n_fused: int, group=None):
+    """Column-shard a FUSED projection (e.g. GPT-2's c_attn = [q|k|v]):
+    each of the ``n_fused`` output blocks is sharded by this rank's slice so
+    the local output stays ``[q_loc|k_loc|v_loc]`` and the model can split it
+    by ``out_local // n_fused``."""
+    world = dist.get_world_size(group) if dist.is_initialized() else 1
+    rank = dist.get_rank(group) if dist.is_initialized() else 0
+    block = linear.out_features // n_fused
+    assert block % world == 0, "fused block size must divide tp world"
+    per = block // world
+    rows = []
+    for j in range(n_fused):
+        rows.append(torch.arange(j * block + rank * per, j * block + (rank + 1) * per))
+    rows = torch.cat(rows)
+    mod = ColumnParallelLinear.__new__(ColumnParallelLinear)
+    nn.Module.__init__(mod)
+    mod.group = group
+    mod.gather_output = False
+    mod.out_per_rank = n_fused * per
+    mod.weight = nn.Parameter(linear.weight[rows].detach().clone())
+    mod.bias = nn.Parameter(linear.bias[rows].detach().clone()) if linear.bias is not None else None
+    return mod
+
+
+def apply_tp_plan(model, group=None, plan=None, shard_attrs=None):
+    """Model-GENERIC tensor parallelism: shard by module-name patterns.
+
+    ``plan`` maps fnmatch patterns of module names to a sharding kind —
+    'colwise' | 'colwise_gather' | 'rowwise' | 'colwise_fusedN' (N fused
+    output blocks, e.g. GPT-2 c_attn = colwise_fused3). ``shard_attrs`` maps
+    patterns to int attribute names divided by the tp degree (per-rank head
+    counts). Defaults come from the model's ``tp_plan`` / ``tp_shard_attrs``
+    class attributes (the transformers ``tp_plan`` idiom; reference routes
+    this in accelerator.py:1531-1560 via DTensors — ours is megatron-style
+    in-place resharding with ONE all-reduce per attention/MLP).
+    """
+    import fnmatch
+
     world = dist.get_world_size(group) if dist.is_initialized() else 1
     if world == 1:
         return model
-    for layer in model.layers:
-        attn = layer.self_attn
-        assert attn.n_heads % world == 0 and attn.n_kv % world == 0, "heads must divide tp world"
-        attn.q_proj = ColumnParallelLinear.from_linear(attn.q_proj, group)
-        attn.k_proj = ColumnParallelLinear.from_linear(attn.k_proj, group)
-        attn.v_proj = ColumnParallelLinear.from_linear(attn.v_proj, group)
-        attn.o_proj = RowParallelLinear.from_linear(attn.o_proj, group)
-        attn.n_heads //= world
-        attn.n_kv //= world
-        mlp = layer.mlp
-        mlp.gate_proj = ColumnParallelLinear.from_linear(mlp.gate_proj, group)
-        mlp.up_proj = ColumnParallelLinear.from_linear(mlp.up_proj, group)
-        mlp.down_proj = RowParallelLinear.from_linear(mlp.down_proj, group)
+    plan = plan if plan is not None else getattr(model, "tp_plan", None)
+    if plan is None:
+        raise ValueError(
+            "no tensor-parallel plan: pass `plan=` or define a `tp_plan` "
+            "dict of module-name patterns on the model class"
+        )
+    shard_attrs = shard_attrs if shard_attrs is not None else getattr(model, "tp_shard_attrs", {})
+
+    # old param -> sharded param, so Accelerator.prepare can re-point an
+    # already-constructed optimizer (reference: accelerator.py:1647-1654)
+    swap: dict = {}
+    replaced = 0
+    for name, module in list(model.named_modules()):
+        kind = None
+        for pattern, k in plan.items():
+            if fnmatch.fnmatchcase(name, pattern):
+                kind = k
+                break
+        if kind is None:
+            continue
+        if not isinstance(module, nn.Linear):
+            raise TypeError(f"tp_plan pattern matched non-Linear module {name} ({type(module).__name__})")
+        if kind == "colwise":
+            new = ColumnParallelLinear.from_linear(module, group)
+        elif kind == "colwise_gather":
+            new = ColumnParallelLinear.from_linear(module, group, gather_output=True)
+        elif kind == "rowwise":
+            new = RowParallelLinear.from_linear(module, group)
+        elif kind.startswith("colwise_fused"):
+            new = _fused_colwise_from_linear(module, int(kind[len("colwise_fused") :]), group)
+        else:
+            raise ValueError(f"unknown tp_plan kind {kind!r} for {name}")
+        parent_name, _, leaf = name.rpartition(".")
+        parent = model.get_submodule(parent_name) if parent_name else model
+        swap[id(module.weight)] = new.weight
+        if module.bias is not None and new.bias is not None:
+            swap[id(module.bias)] = new.bias
+        setattr(parent, leaf, new)
+        replaced += 1
+    if replaced == 0:
+        raise ValueError("tp_plan matched no modules — check the patterns against named_modules()")
+
+    for name, module in model.named_modules():
+        for pattern, attrs in shard_attrs.items():
+            if fnmatch.fnmatchcase(name, pattern):
+                for a in attrs:
+                    val = getattr(module, a)
+                    if val % world != 0:
+                        raise ValueError(f"{name}.{a} ({val}) must divide the tp degree ({world})")
+                    setattr(module, a, val // world)
+    model._tp_group = group
+    model._tp_param_swap = swap
+    model.tp_size = world
     return model
+
+
+def tp_parallelize_llama(model, group=None):
+    """Shard a LlamaForCausalLM in place (megatron pattern) — the historical
+    entry point, now a thin wrapper over the generic `apply_tp_plan` with
+    the Llama class plan."""
+    return apply_tp_plan(model, group=group)

@@ -29,6 +29,10 @@ class ParallelismConfig:
     tp_size: int = 1
     cp_size: int = 1
     ep_size: int = 1
+    # sequence-parallel collective pattern for the cp dimension:
+    # 'allgather' = KV all-gather CP (reference torch CP default rotate),
+    # 'ulysses'   = dual all-to-all head resharding (reference DeepSpeed SP)
+    cp_impl: str = "allgather"
 
     _groups: Dict[str, Optional[object]] = field(default_factory=dict, repr=False)
 
@@ -41,6 +45,8 @@ class ParallelismConfig:
         return self.dp_replicate_size * self.dp_shard_size
 
     def validate(self, world_size: int):
+        if self.cp_impl not in ("allgather", "ulysses"):
+            raise ValueError(f"cp_impl must be 'allgather' or 'ulysses', got {self.cp_impl!r}")
         if self.total_size != world_size:
             raise ValueError(
                 f"ParallelismConfig total_size ({self.total_size}) != world size ({world_size}); "
@@ -103,11 +109,9 @@ class ParallelismConfig:
                 if rank in ranks:
                     mine = pg
             result[dim] = mine
-        # combined DP group (replicate x shard): gradient averaging domain
-        if self.dp_size == world:
-            result["dp"] = None  # whole world — use the default group
-        elif self.dp_size == 1:
-            result["dp"] = None
+        # combined DP group (replicate x shard): the batch-sharding domain
+        if self.dp_size in (world, 1):
+            result["dp"] = None  # whole world / off — use the default group
         else:
             mine = None
             tpcp = self.tp_size * self.cp_size
@@ -119,5 +123,24 @@ class ParallelismConfig:
                 if rank in ranks:
                     mine = pg
             result["dp"] = mine
+        # gradient-averaging domain = dp x cp (cp ranks hold REPLICATED
+        # params over different sequence shards; a mean-reduced loss means
+        # the true gradient is the cp-average too — the reference's
+        # dp_shard_cp mesh flattening, parallelism_config.py:136-143)
+        grad_size = self.dp_size * self.cp_size
+        if grad_size in (world, 1):
+            result["grad"] = None
+        elif self.cp_size == 1:
+            result["grad"] = result["dp"]
+        else:
+            mine = None
+            lists = {}
+            for r in range(world):
+                lists.setdefault(r % self.tp_size, []).append(r)
+            for ranks in lists.values():
+                pg = dist.new_group(ranks)
+                if rank in ranks:
+                    mine = pg
+            result["grad"] = mine
         self._groups = result
         return result

@@ -16,12 +16,22 @@ from .state import AcceleratorState, GradientState
 
 
 class AcceleratedScheduler:
-    def __init__(self, scheduler, optimizers, step_with_optimizer: bool = True, split_batches: bool = False):
+    def __init__(
+        self,
+        scheduler,
+        optimizers,
+        step_with_optimizer: bool = True,
+        split_batches: bool = False,
+        num_batch_shards: int = None,
+    ):
         self.scheduler = scheduler
         self.optimizers = list(optimizers) if isinstance(optimizers, (list, tuple)) else [optimizers]
         self.split_batches = split_batches
         self.step_with_optimizer = step_with_optimizer
         self.gradient_state = GradientState()
+        # how many ways the global batch is sharded: the dp degree under
+        # TP/CP worlds (tp/cp ranks replicate the batch), else num_processes
+        self.num_batch_shards = num_batch_shards
 
     # -- stepping ----------------------------------------------------------
 
@@ -47,7 +57,12 @@ class AcceleratedScheduler:
             return
         if self._any_optimizer_skipped():
             return  # fp16 inf/nan step: LR stays put
-        ticks = 1 if self.split_batches else AcceleratorState().num_processes
+        if self.split_batches:
+            ticks = 1
+        elif self.num_batch_shards is not None:
+            ticks = self.num_batch_shards
+        else:
+            ticks = AcceleratorState().num_processes
         for _ in range(ticks):
             self._advance(*args, **kwargs)
 
