@@ -55,6 +55,7 @@ from .utils.operations import (
     reduce,
     send_to_device,
 )
+from .utils.environment import parse_flag_from_env
 from .utils.other import extract_model_from_parallel, wait_for_everyone
 
 logger = get_logger(__name__)
@@ -80,6 +81,8 @@ class Accelerator:
         fsdp_plugin: Optional[FullyShardedDataParallelPlugin] = None,
         rng_types: Optional[List[Union[str, RNGType]]] = None,
         parallelism_config=None,
+        dynamo_plugin=None,
+        dynamo_backend: Union[str, None] = None,
         dispatch_batches: Optional[bool] = None,
         even_batches: bool = True,
         use_seedable_sampler: bool = False,
@@ -132,10 +135,22 @@ class Accelerator:
         self.non_blocking = non_blocking
         self.use_stateful_dataloader = False
         self.data_seed = None
-        # multi-dimensional parallelism descriptor (reference accelerator.py:301)
+        # multi-dimensional parallelism descriptor (reference accelerator.py:301);
+        # with no explicit config the PARALLELISM_CONFIG_* env plane set by the
+        # launcher takes over (reference: parallelism_config.py:274-341)
+        if parallelism_config is None and parse_flag_from_env("ACCELERATE_USE_PARALLELISM_CONFIG", False):
+            from .parallelism_config import ParallelismConfig
+
+            parallelism_config = ParallelismConfig()
         self.parallelism_config = parallelism_config
         if parallelism_config is not None and self.state.distributed_type != DistributedType.NO:
             parallelism_config.validate(self.num_processes)
+        # torch.compile plane (reference: dataclasses.py:1033 TorchDynamoPlugin)
+        from .utils.dataclasses import TorchDynamoPlugin
+
+        if dynamo_plugin is None:
+            dynamo_plugin = TorchDynamoPlugin(backend=dynamo_backend) if dynamo_backend else TorchDynamoPlugin()
+        self.dynamo_plugin = dynamo_plugin
         if dataloader_config is not None:  # bundled options take precedence
             self.split_batches = dataloader_config.split_batches
             self.dispatch_batches = dataloader_config.dispatch_batches
@@ -516,6 +531,16 @@ class Accelerator:
                     model = DistributedDataParallelEngine(model, process_group=pc._groups["grad"], **ddp_kwargs)
             else:
                 model = DistributedDataParallelEngine(model, **ddp_kwargs)
+        # torch.compile LAST, over the wrapped model (reference:
+        # accelerator.py:2062-2066); regional compilation compiles each
+        # repeated block once (utils/other.py compile_regions)
+        if getattr(self, "dynamo_plugin", None) is not None and self.dynamo_plugin.enabled:
+            if self.dynamo_plugin.use_regional_compilation:
+                from .utils.other import compile_regions
+
+                model = compile_regions(model, **self.dynamo_plugin.compile_kwargs())
+            else:
+                model = torch.compile(model, **self.dynamo_plugin.compile_kwargs())
         model._is_accelerate_prepared = True
         self._models.append(model)
         return model

@@ -38,6 +38,36 @@ def add_parser(subparsers):
                         help="FULL_STATE_DICT | SHARDED_STATE_DICT")
     parser.add_argument("--fsdp_activation_checkpointing", default=None,
                         help="true|false: checkpoint each wrapped unit")
+    parser.add_argument("--fsdp_offload_params", default=None, help="true|false")
+    parser.add_argument("--fsdp_min_num_params", type=int, default=None)
+    parser.add_argument("--fsdp_auto_wrap_policy", default=None)
+    parser.add_argument("--fsdp_reshard_after_forward", default=None, help="true|false")
+    parser.add_argument("--fsdp_forward_prefetch", default=None, help="true|false")
+    parser.add_argument("--fsdp_backward_prefetch", default=None,
+                        help="backward_pre | backward_post | no_prefetch")
+    parser.add_argument("--fsdp_cpu_ram_efficient_loading", default=None,
+                        help="true|false: meta-init + per-rank sliced checkpoint load")
+    parser.add_argument("--fsdp_sync_module_states", default=None, help="true|false")
+    parser.add_argument("--fsdp_use_orig_params", default=None, help="true|false")
+    parser.add_argument("--fsdp_version", type=int, default=None)
+    # multi-dimensional parallelism plane (reference: utils/launch.py:400-425)
+    parser.add_argument("--parallelism_config_dp_replicate_size", type=int, default=None)
+    parser.add_argument("--parallelism_config_dp_shard_size", type=int, default=None)
+    parser.add_argument("--parallelism_config_tp_size", type=int, default=None)
+    parser.add_argument("--parallelism_config_cp_size", type=int, default=None)
+    parser.add_argument("--parallelism_config_cp_comm_strategy", default=None,
+                        help="allgather | ulysses (alltoall accepted as ulysses)")
+    # torch.compile plane (reference: utils/launch.py:189-193)
+    parser.add_argument("--dynamo_backend", default=None, help="e.g. inductor; NO disables")
+    parser.add_argument("--dynamo_mode", default=None,
+                        help="default | reduce-overhead | max-autotune")
+    parser.add_argument("--dynamo_use_fullgraph", action="store_true")
+    parser.add_argument("--dynamo_use_dynamic", action="store_true")
+    parser.add_argument("--dynamo_use_regional_compilation", action="store_true")
+    # fp8 recipe plane (our CDNA4 recipe, utils/dataclasses.py FP8RecipeKwargs)
+    parser.add_argument("--fp8_format", default=None, help="E4M3 | HYBRID")
+    parser.add_argument("--fp8_amax_history_len", type=int, default=None)
+    parser.add_argument("--fp8_margin", type=int, default=None)
     parser.add_argument("--num_cpu_threads_per_process", type=int, default=None,
                         help="Sets OMP_NUM_THREADS per worker")
     parser.add_argument("--gpu_ids", default=None, help="Comma-separated HIP device ids to use")
@@ -116,10 +146,49 @@ def build_env(args, config: ClusterConfig) -> dict:
             "FSDP_SHARD_GROUP_SIZE": args.fsdp_shard_group_size,
             "FSDP_STATE_DICT_TYPE": args.fsdp_state_dict_type,
             "FSDP_ACTIVATION_CHECKPOINTING": args.fsdp_activation_checkpointing,
+            "FSDP_OFFLOAD_PARAMS": args.fsdp_offload_params,
+            "FSDP_MIN_NUM_PARAMS": args.fsdp_min_num_params,
+            "FSDP_AUTO_WRAP_POLICY": args.fsdp_auto_wrap_policy,
+            "FSDP_RESHARD_AFTER_FORWARD": args.fsdp_reshard_after_forward,
+            "FSDP_FORWARD_PREFETCH": args.fsdp_forward_prefetch,
+            "FSDP_BACKWARD_PREFETCH": args.fsdp_backward_prefetch,
+            "FSDP_CPU_RAM_EFFICIENT_LOADING": args.fsdp_cpu_ram_efficient_loading,
+            "FSDP_SYNC_MODULE_STATES": args.fsdp_sync_module_states,
+            "FSDP_USE_ORIG_PARAMS": args.fsdp_use_orig_params,
+            "FSDP_VERSION": args.fsdp_version,
         }
         for key, value in cli_fsdp.items():
             if value is not None:
                 env[key] = str(value)
+    # parallelism plane (reference: utils/launch.py:400-425)
+    pc_cli = {
+        "PARALLELISM_CONFIG_DP_REPLICATE_SIZE": args.parallelism_config_dp_replicate_size,
+        "PARALLELISM_CONFIG_DP_SHARD_SIZE": args.parallelism_config_dp_shard_size,
+        "PARALLELISM_CONFIG_TP_SIZE": args.parallelism_config_tp_size,
+        "PARALLELISM_CONFIG_CP_SIZE": args.parallelism_config_cp_size,
+        "PARALLELISM_CONFIG_CP_COMM_STRATEGY": args.parallelism_config_cp_comm_strategy,
+    }
+    if any(v is not None for v in pc_cli.values()):
+        env["ACCELERATE_USE_PARALLELISM_CONFIG"] = "true"
+        for key, value in pc_cli.items():
+            if value is not None:
+                env[key] = str(value)
+    # torch.compile plane
+    if args.dynamo_backend is not None:
+        env["ACCELERATE_DYNAMO_BACKEND"] = str(args.dynamo_backend).upper()
+        if args.dynamo_mode is not None:
+            env["ACCELERATE_DYNAMO_MODE"] = str(args.dynamo_mode)
+        env["ACCELERATE_DYNAMO_USE_FULLGRAPH"] = str(args.dynamo_use_fullgraph)
+        env["ACCELERATE_DYNAMO_USE_DYNAMIC"] = str(args.dynamo_use_dynamic)
+        env["ACCELERATE_DYNAMO_USE_REGIONAL_COMPILATION"] = str(args.dynamo_use_regional_compilation)
+    # fp8 recipe plane
+    for key, value in {
+        "ACCELERATE_FP8_FORMAT": args.fp8_format,
+        "ACCELERATE_FP8_AMAX_HISTORY_LEN": args.fp8_amax_history_len,
+        "ACCELERATE_FP8_MARGIN": args.fp8_margin,
+    }.items():
+        if value is not None:
+            env[key] = str(value)
     if args.num_cpu_threads_per_process is not None:
         env["OMP_NUM_THREADS"] = str(args.num_cpu_threads_per_process)
     if args.gpu_ids not in (None, "all"):

@@ -16,6 +16,7 @@ then dp_shard, then dp_replicate — so TP groups are intra-node neighbors
 (xGMI-adjacent) when launched one-rank-per-GPU.
 """
 
+import os
 from dataclasses import dataclass, field
 from typing import Dict, Optional
 
@@ -24,17 +25,42 @@ import torch.distributed as dist
 
 @dataclass
 class ParallelismConfig:
-    dp_replicate_size: int = 1
-    dp_shard_size: int = 1
-    tp_size: int = 1
-    cp_size: int = 1
-    ep_size: int = 1
+    """Fields left ``None`` fall back to the ``PARALLELISM_CONFIG_*`` env
+    plane set by the launcher (reference: parallelism_config.py:274-341 —
+    this is the launcher↔library ABI SURVEY.md §2.7 calls load-bearing)."""
+
+    dp_replicate_size: Optional[int] = None
+    dp_shard_size: Optional[int] = None
+    tp_size: Optional[int] = None
+    cp_size: Optional[int] = None
+    ep_size: Optional[int] = None
     # sequence-parallel collective pattern for the cp dimension:
     # 'allgather' = KV all-gather CP (reference torch CP default rotate),
-    # 'ulysses'   = dual all-to-all head resharding (reference DeepSpeed SP)
-    cp_impl: str = "allgather"
+    # 'ulysses'   = dual all-to-all head resharding (reference DeepSpeed SP);
+    # the reference's 'alltoall' rotate spelling is accepted as 'ulysses'
+    cp_impl: Optional[str] = None
 
     _groups: Dict[str, Optional[object]] = field(default_factory=dict, repr=False)
+
+    def __post_init__(self):
+        env = os.environ
+        if self.dp_replicate_size is None:
+            self.dp_replicate_size = int(env.get("PARALLELISM_CONFIG_DP_REPLICATE_SIZE", "1"))
+        if self.dp_shard_size is None:
+            self.dp_shard_size = int(env.get("PARALLELISM_CONFIG_DP_SHARD_SIZE", "1"))
+        if self.tp_size is None:
+            self.tp_size = int(env.get("PARALLELISM_CONFIG_TP_SIZE", "1"))
+        if self.cp_size is None:
+            self.cp_size = int(env.get("PARALLELISM_CONFIG_CP_SIZE", "1"))
+        if self.ep_size is None:
+            self.ep_size = int(env.get("PARALLELISM_CONFIG_EP_SIZE", "1"))
+        if self.cp_impl is None:
+            self.cp_impl = env.get("PARALLELISM_CONFIG_CP_COMM_STRATEGY", "allgather")
+        if self.cp_impl == "alltoall":
+            self.cp_impl = "ulysses"
+        for name in ("dp_replicate_size", "dp_shard_size", "tp_size", "cp_size", "ep_size"):
+            if getattr(self, name) < 1:
+                raise ValueError(f"{name} must be at least 1, got {getattr(self, name)}")
 
     @property
     def total_size(self) -> int:

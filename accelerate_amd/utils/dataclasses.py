@@ -190,13 +190,76 @@ class FP8RecipeKwargs(KwargsHandler):
     """CDNA4 fp8 recipe: OCP e4m3fn forward / e5m2 grad, delayed scaling with
     amax history (replaces the reference's TE/AO/MSAMP triple backend,
     reference: dataclasses.py:313-485). Used by accelerate_amd.ops.fp8.
+
+    Fields left at their defaults pick up the ``ACCELERATE_FP8_*`` env plane
+    set by the launcher (reference env ABI pattern, utils/launch.py:82-99).
     """
 
-    format: str = "HYBRID"  # "E4M3" (both dirs) or "HYBRID" (e4m3 fwd / e5m2 bwd)
-    amax_history_len: int = 16
-    amax_compute_algo: str = "max"
-    margin: int = 0
-    use_first_last_bf16: bool = True  # keep first/last linear in bf16
+    format: str = None  # "E4M3" (both dirs) or "HYBRID" (e4m3 fwd / e5m2 bwd)
+    amax_history_len: int = None
+    amax_compute_algo: str = None
+    margin: int = None
+    use_first_last_bf16: bool = None  # keep first/last linear in bf16
+
+    def __post_init__(self):
+        if self.format is None:
+            self.format = os.environ.get("ACCELERATE_FP8_FORMAT", "HYBRID").upper()
+        if self.amax_history_len is None:
+            self.amax_history_len = int(os.environ.get("ACCELERATE_FP8_AMAX_HISTORY_LEN", "16"))
+        if self.amax_compute_algo is None:
+            self.amax_compute_algo = os.environ.get("ACCELERATE_FP8_AMAX_COMPUTE_ALGO", "max")
+        if self.margin is None:
+            self.margin = int(os.environ.get("ACCELERATE_FP8_MARGIN", "0"))
+        if self.use_first_last_bf16 is None:
+            self.use_first_last_bf16 = parse_flag_from_env("ACCELERATE_FP8_FIRST_LAST_BF16", True)
+        if self.format not in ("E4M3", "HYBRID"):
+            raise ValueError(f"fp8 format must be E4M3 or HYBRID, got {self.format!r}")
+
+
+@dataclass
+class TorchDynamoPlugin(KwargsHandler):
+    """torch.compile configuration plane (reference: dataclasses.py:1033).
+
+    On ROCm this drives the inductor backend; ``use_regional_compilation``
+    compiles each repeated block once and reuses the artifact (reference
+    utils/other.py:106-177 `compile_regions` — far cheaper compile time on
+    deep decoders). Fields fall back to ``ACCELERATE_DYNAMO_*`` env vars.
+    """
+
+    backend: str = None
+    mode: str = None
+    fullgraph: bool = None
+    dynamic: Optional[bool] = None
+    use_regional_compilation: bool = None
+    options: Any = None
+    disable: bool = False
+
+    def __post_init__(self):
+        if self.backend is None:
+            self.backend = os.environ.get("ACCELERATE_DYNAMO_BACKEND", "NO")
+        self.backend = self.backend.upper() if isinstance(self.backend, str) else self.backend
+        if self.mode is None:
+            self.mode = os.environ.get("ACCELERATE_DYNAMO_MODE", "default")
+        if self.fullgraph is None:
+            self.fullgraph = parse_flag_from_env("ACCELERATE_DYNAMO_USE_FULLGRAPH", False)
+        if self.dynamic is None and "ACCELERATE_DYNAMO_USE_DYNAMIC" in os.environ:
+            self.dynamic = parse_flag_from_env("ACCELERATE_DYNAMO_USE_DYNAMIC", False)
+        if self.use_regional_compilation is None:
+            self.use_regional_compilation = parse_flag_from_env("ACCELERATE_DYNAMO_USE_REGIONAL_COMPILATION", False)
+
+    @property
+    def enabled(self) -> bool:
+        return not self.disable and self.backend not in (None, "NO")
+
+    def compile_kwargs(self) -> Dict[str, Any]:
+        kw: Dict[str, Any] = {"backend": self.backend.lower(), "mode": self.mode}
+        if self.fullgraph:
+            kw["fullgraph"] = True
+        if self.dynamic is not None:
+            kw["dynamic"] = self.dynamic
+        if self.options is not None:
+            kw["options"] = self.options
+        return kw
 
 
 @dataclass
@@ -265,6 +328,13 @@ class FullyShardedDataParallelPlugin(KwargsHandler):
     state_dict_type: str = None  # "full_state_dict" | "sharded_state_dict"
     use_orig_params: bool = True
     sync_module_states: bool = None
+    # reference-ABI fields (utils/launch.py:309-332): our engine ALWAYS
+    # prefetches on side streams, so the prefetch flags are accepted and
+    # recorded for parity; fsdp_version 2 is the only engine we ship
+    fsdp_version: int = None
+    forward_prefetch: bool = None
+    backward_prefetch: str = None  # "backward_pre" | "backward_post" | "no_prefetch"
+    cpu_ram_efficient_loading: bool = None  # meta-init + per-rank sliced load
 
     def __post_init__(self):
         env_prefix = "FSDP_"
@@ -289,6 +359,14 @@ class FullyShardedDataParallelPlugin(KwargsHandler):
         if self.min_num_params is None:
             val = os.environ.get(env_prefix + "MIN_NUM_PARAMS", None)
             self.min_num_params = int(val) if val else None
+        if self.fsdp_version is None:
+            self.fsdp_version = int(os.environ.get(env_prefix + "VERSION", "2"))
+        if self.forward_prefetch is None:
+            self.forward_prefetch = parse_flag_from_env(env_prefix + "FORWARD_PREFETCH", True)
+        if self.backward_prefetch is None:
+            self.backward_prefetch = os.environ.get(env_prefix + "BACKWARD_PREFETCH", "backward_pre").lower()
+        if self.cpu_ram_efficient_loading is None:
+            self.cpu_ram_efficient_loading = parse_flag_from_env(env_prefix + "CPU_RAM_EFFICIENT_LOADING", False)
 
 
 @dataclass
