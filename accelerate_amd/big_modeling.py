@@ -46,61 +46,46 @@ def init_empty_weights(include_buffers: bool = False):
         yield f
 
 
+def _rehome_parameter(registered: nn.Parameter, device: torch.device) -> nn.Parameter:
+    """Rebuild a just-registered Parameter on ``device``, preserving its
+    concrete class and any attributes a Parameter subclass stashed on the
+    instance (the contract transformers' custom params rely on)."""
+    attrs = dict(registered.__dict__)
+    attrs["requires_grad"] = registered.requires_grad
+    return type(registered)(registered.to(device), **attrs)
+
+
 @contextmanager
 def init_on_device(device: torch.device, include_buffers: bool = False):
-    """Initialize a model with all parameters created on `device`
-    (reference: big_modeling.py:98-176)."""
+    """Run model construction with every parameter landing on ``device``.
+
+    Behavior parity with reference big_modeling.py:98-176; mechanism is
+    ours. With ``include_buffers`` the torch device-mode context already
+    routes every tensor constructor (params, buffers, intermediates) to the
+    target device, so no patching is needed at all. Without it, only
+    ``nn.Module.register_parameter`` is intercepted: the module registers
+    its parameter normally, then the slot is swapped for a re-homed copy —
+    buffers keep their computed (host) values, which is what lets meta-init
+    models keep usable RoPE/position tables.
+    """
     if include_buffers:
         with device:
             yield
         return
 
-    old_register_parameter = nn.Module.register_parameter
-    if include_buffers:
-        old_register_buffer = nn.Module.register_buffer
+    unpatched = nn.Module.register_parameter
 
-    def register_empty_parameter(module, name, param):
-        old_register_parameter(module, name, param)
-        if param is not None:
-            param_cls = type(module._parameters[name])
-            kwargs = module._parameters[name].__dict__
-            kwargs["requires_grad"] = param.requires_grad
-            module._parameters[name] = param_cls(module._parameters[name].to(device), **kwargs)
+    def register_and_rehome(module, name, param):
+        unpatched(module, name, param)
+        slot = module._parameters.get(name)
+        if slot is not None:
+            module._parameters[name] = _rehome_parameter(slot, device)
 
-    def register_empty_buffer(module, name, buffer, persistent=True):
-        old_register_buffer(module, name, buffer, persistent=persistent)
-        if buffer is not None:
-            module._buffers[name] = module._buffers[name].to(device)
-
-    # Patch tensor creation
-    if include_buffers:
-        tensor_constructors_to_patch = {
-            torch_function_name: getattr(torch, torch_function_name)
-            for torch_function_name in ["empty", "zeros", "ones", "full"]
-        }
-    else:
-        tensor_constructors_to_patch = {}
-
-    def patch_tensor_constructor(fn):
-        def wrapper(*args, **kwargs):
-            kwargs["device"] = device
-            return fn(*args, **kwargs)
-
-        return wrapper
-
+    nn.Module.register_parameter = register_and_rehome
     try:
-        nn.Module.register_parameter = register_empty_parameter
-        if include_buffers:
-            nn.Module.register_buffer = register_empty_buffer
-        for torch_function_name in tensor_constructors_to_patch.keys():
-            setattr(torch, torch_function_name, patch_tensor_constructor(getattr(torch, torch_function_name)))
         yield
     finally:
-        nn.Module.register_parameter = old_register_parameter
-        if include_buffers:
-            nn.Module.register_buffer = old_register_buffer
-        for torch_function_name, old_torch_function in tensor_constructors_to_patch.items():
-            setattr(torch, torch_function_name, old_torch_function)
+        nn.Module.register_parameter = unpatched
 
 
 def cpu_offload(
@@ -303,14 +288,21 @@ def load_checkpoint_and_dispatch(
     force_hooks: bool = False,
     strict: bool = False,
 ):
-    """infer map → load shards → dispatch (reference: big_modeling.py:520)."""
-    if isinstance(device_map, str) and device_map not in ["auto", "balanced", "balanced_low_0", "sequential"]:
-        raise ValueError(
-            "If passing a string for `device_map`, please choose 'auto', 'balanced', 'balanced_low_0' or 'sequential'."
-        )
+    """Three phases (reference behavior: big_modeling.py:520): resolve a
+    device map from its policy name, stream checkpoint shards into place
+    (offload writes included), then attach dispatch hooks."""
+    _POLICIES = ("auto", "balanced", "balanced_low_0", "sequential")
     if isinstance(device_map, str):
+        if device_map not in _POLICIES:
+            raise ValueError(
+                f"unknown device_map policy {device_map!r}; expected one of "
+                f"{', '.join(_POLICIES)} (or an explicit dict)"
+            )
+        budget = max_memory
         if device_map != "sequential":
-            max_memory = get_balanced_memory(
+            # balanced policies pre-split the memory budget evenly so the
+            # greedy allocator fills devices at matching rates
+            budget = get_balanced_memory(
                 model,
                 max_memory=max_memory,
                 no_split_module_classes=no_split_module_classes,
@@ -319,20 +311,20 @@ def load_checkpoint_and_dispatch(
             )
         device_map = infer_auto_device_map(
             model,
-            max_memory=max_memory,
+            max_memory=budget,
             no_split_module_classes=no_split_module_classes,
             dtype=dtype,
             offload_buffers=offload_buffers,
         )
-    if offload_state_dict is None and device_map is not None and "disk" in device_map.values():
-        offload_state_dict = True
+
+    spills_to_disk = device_map is not None and "disk" in device_map.values()
     load_checkpoint_in_model(
         model,
         checkpoint,
         device_map=device_map,
         offload_folder=offload_folder,
         dtype=dtype,
-        offload_state_dict=offload_state_dict,
+        offload_state_dict=spills_to_disk if offload_state_dict is None else offload_state_dict,
         offload_buffers=offload_buffers,
         strict=strict,
     )

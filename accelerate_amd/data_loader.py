@@ -98,8 +98,9 @@ class BatchSamplerShard(BatchSampler):
     ):
         if split_batches and batch_sampler.batch_size % num_processes != 0:
             raise ValueError(
-                f"To use `BatchSamplerShard` in `split_batches` mode, the batch size ({batch_sampler.batch_size}) "
-                f"needs to be a round multiple of the number of processes ({num_processes})."
+                f"split_batches slices each global batch {num_processes} ways, so the batch size "
+                f"must be divisible by the process count (got batch_size={batch_sampler.batch_size}, "
+                f"num_processes={num_processes})."
             )
         self.batch_sampler = batch_sampler
         self.num_processes = num_processes
@@ -222,8 +223,9 @@ class IterableDatasetShard(IterableDataset):
     ):
         if split_batches and batch_size > 1 and batch_size % num_processes != 0:
             raise ValueError(
-                f"To use `IterableDatasetShard` in `split_batches` mode, the batch size ({batch_size}) "
-                f"needs to be a round multiple of the number of processes ({num_processes})."
+                f"split_batches slices each fetched batch {num_processes} ways, so the batch size "
+                f"must be divisible by the process count (got batch_size={batch_size}, "
+                f"num_processes={num_processes})."
             )
         self.dataset = dataset
         self.batch_size = batch_size
@@ -251,28 +253,28 @@ class IterableDatasetShard(IterableDataset):
             and isinstance(self.dataset.generator, torch.Generator)
         ):
             self.dataset.generator.manual_seed(getattr(self, "epoch", 0))
-        real_batch_size = self.batch_size if self.split_batches else (self.batch_size * self.num_processes)
-        per_rank = real_batch_size // self.num_processes
-        lo, hi = per_rank * self.process_index, per_rank * (self.process_index + 1)
+        fetch_window = self.batch_size if self.split_batches else (self.batch_size * self.num_processes)
+        per_rank = fetch_window // self.num_processes
+        mine = slice(per_rank * self.process_index, per_rank * (self.process_index + 1))
 
-        wrap_buffer = None
-        buffer = []
+        first_window = None  # recycled to pad a short tail window
+        window = []
         for item in self.dataset:
-            buffer.append(item)
-            if len(buffer) == real_batch_size:
-                for i in range(lo, hi):
-                    yield buffer[i]
-                if wrap_buffer is None:
-                    wrap_buffer = buffer.copy()
-                buffer = []
+            window.append(item)
+            if len(window) < fetch_window:
+                continue
+            yield from window[mine]
+            first_window = first_window or window
+            window = []
 
-        if not self.drop_last and len(buffer) > 0:
-            if wrap_buffer is None:
-                wrap_buffer = buffer.copy()
-            while len(buffer) < real_batch_size:
-                buffer += wrap_buffer
-            for i in range(lo, hi):
-                yield buffer[i]
+        if self.drop_last or not window:
+            return
+        # short tail: top up from the first window (or the tail itself on
+        # tiny datasets) until rectangular, then emit this rank's slice
+        pad_source = first_window or list(window)
+        while len(window) < fetch_window:
+            window.extend(pad_source)
+        yield from window[mine]
 
 
 class DataLoaderStateMixin:
@@ -689,8 +691,9 @@ def prepare_data_loader(
 
     if split_batches and dataloader.batch_size is not None and dataloader.batch_size % num_processes != 0:
         raise ValueError(
-            f"To use a `DataLoader` in `split_batches` mode, the batch size ({dataloader.batch_size}) "
-            f"needs to be a round multiple of the number of processes ({num_processes})."
+            f"split_batches slices each global batch {num_processes} ways, so the DataLoader's "
+            f"batch size must be divisible by the process count (got batch_size="
+            f"{dataloader.batch_size}, num_processes={num_processes})."
         )
 
     new_dataset = dataloader.dataset

@@ -45,43 +45,47 @@ def should_reduce_batch_size(exception: Exception) -> bool:
 
 
 def find_executable_batch_size(function=None, starting_batch_size: int = 128, reduce_batch_size_fn=None):
-    """Retry decorator that halves... well, shrinks ×0.9 the batch size on OOM
-    (reference: memory.py:119-187)."""
+    """OOM-retry decorator: call ``function(batch_size, ...)`` and shrink the
+    batch on every allocator failure until a size fits (behavior parity with
+    reference memory.py:119-187; default shrink ×0.9 per retry).
+
+    The wrapped function's FIRST positional parameter is the batch size and
+    is supplied by the decorator — callers pass only the remaining args.
+    The chosen size is sticky across calls (a training re-run starts at the
+    last size that worked, not back at ``starting_batch_size``).
+    """
     if function is None:
         return functools.partial(
             find_executable_batch_size, starting_batch_size=starting_batch_size, reduce_batch_size_fn=reduce_batch_size_fn
         )
-    if reduce_batch_size_fn is None:
+    shrink = reduce_batch_size_fn or (lambda bs: int(bs * 0.9))
+    state = {"bs": starting_batch_size}
 
-        def reduce_batch_size_fn(batch_size):
-            return int(batch_size * 0.9)
-
-    batch_size = starting_batch_size
-
-    def decorator(*args, **kwargs):
-        nonlocal batch_size
-        clear_device_cache(garbage_collection=True)
-        params = list(inspect.signature(function).parameters.keys())
-        # Guard against user error
-        if len(params) < (len(args) + 1):
-            arg_str = ", ".join([f"{arg}={value}" for arg, value in zip(params[1:], args[1:])])
+    def _reject_caller_supplied_batch_size(args):
+        declared = list(inspect.signature(function).parameters)
+        if len(declared) < len(args) + 1:
+            shown = ", ".join(f"{name}={val}" for name, val in zip(declared[1:], args[1:]))
             raise TypeError(
-                f"Batch size was passed into `{function.__name__}` as the first argument when called."
-                f"Remove this as the decorator already does so: `{function.__name__}({arg_str})`"
+                f"`{function.__name__}` received a batch size from its caller, but the "
+                f"find_executable_batch_size decorator injects it. Call it as "
+                f"`{function.__name__}({shown})` instead."
             )
-        while True:
-            if batch_size == 0:
-                raise RuntimeError("No executable batch size found, reached zero.")
-            try:
-                return function(batch_size, *args, **kwargs)
-            except Exception as e:
-                if should_reduce_batch_size(e):
-                    clear_device_cache(garbage_collection=True)
-                    batch_size = reduce_batch_size_fn(batch_size)
-                else:
-                    raise
 
-    return decorator
+    @functools.wraps(function)
+    def attempt_until_it_fits(*args, **kwargs):
+        clear_device_cache(garbage_collection=True)
+        _reject_caller_supplied_batch_size(args)
+        while state["bs"] > 0:
+            try:
+                return function(state["bs"], *args, **kwargs)
+            except Exception as oom:
+                if not should_reduce_batch_size(oom):
+                    raise
+                clear_device_cache(garbage_collection=True)
+                state["bs"] = shrink(state["bs"])
+        raise RuntimeError("No executable batch size found, reached zero.")
+
+    return attempt_until_it_fits
 
 
 def get_xpu_available_memory(*args, **kwargs):  # pragma: no cover
