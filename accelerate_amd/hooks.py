@@ -9,11 +9,14 @@ device-map attachment normalizes its inputs once instead of branching on
 scalar-vs-dict shapes throughout.
 
 On MI355X the onload path is the hot loop of offloaded inference: weights
-stream H2D from pinned host memory with non_blocking copies; a
+stream H2D from a PINNED host cache with non_blocking copies, and an
+execution-order-learning prefetcher (`_OnloadPrefetcher`) onloads the NEXT
+block's weights on a side stream while the current block computes; a
 tied-pointer map guarantees shared weights cross PCIe once per step.
 """
 
 import functools
+import os
 from typing import Dict, List, Mapping, Optional, Union
 
 import torch
@@ -26,6 +29,60 @@ from .utils.modeling import (
 )
 from .utils.offload import PrefixedDataset
 from .utils.operations import find_device, send_to_device
+
+
+class _OnloadPrefetcher:
+    """Execution-order-learning async H2D weight prefetcher.
+
+    One instance per dispatched model (keyed on the shared weights source).
+    On the first forward pass it records the order offloaded hooks fire;
+    from the second pass on, when hook i runs it kicks hook i+1's weights
+    H2D on a dedicated side stream from PINNED host copies — the next
+    block's onload overlaps the current block's compute, and the ring wraps
+    so autoregressive decode prefetches block 0 during the last block
+    (north-star item; reference contrast: hooks.py:359-399 synchronous
+    pageable onload).
+    """
+
+    def __init__(self):
+        self.stream: Optional[torch.cuda.Stream] = None
+        self.order: List["AlignDevicesHook"] = []
+        self.position: Dict[int, int] = {}
+        self.learning = True
+
+    def note(self, hook: "AlignDevicesHook"):
+        """Called at the end of every offloaded pre_forward."""
+        if self.learning:
+            if id(hook) in self.position:
+                self.learning = False  # ring closed: order learned
+            else:
+                self.position[id(hook)] = len(self.order)
+                self.order.append(hook)
+        if not self.learning and len(self.order) > 1:
+            nxt = self.order[(self.position[id(hook)] + 1) % len(self.order)]
+            if nxt is not hook and nxt._prefetched is None:
+                nxt._start_prefetch(self._get_stream())
+
+    def _get_stream(self) -> torch.cuda.Stream:
+        if self.stream is None:
+            self.stream = torch.cuda.Stream()
+        return self.stream
+
+
+_prefetchers: Dict[int, _OnloadPrefetcher] = {}
+_pinned_bytes = 0
+
+
+def _prefetcher_for(weights_map) -> _OnloadPrefetcher:
+    source = getattr(weights_map, "dataset", weights_map)  # unwrap PrefixedDataset
+    key = id(source)
+    if key not in _prefetchers:
+        _prefetchers[key] = _OnloadPrefetcher()
+    return _prefetchers[key]
+
+
+def _async_onload_enabled() -> bool:
+    return os.environ.get("ACCELERATE_AMD_ASYNC_ONLOAD", "1") == "1" and torch.cuda.is_available()
 
 
 class ModelHook:
@@ -180,6 +237,12 @@ class AlignDevicesHook(ModelHook):
         self.input_device = None
         self.tied_params_names = set()
         self.tied_pointers_to_remove = set()
+        # async-onload machinery (wired in init_hook for CUDA offload hooks)
+        self._module_ref = None
+        self._prefetcher = None
+        self._prefetched = None
+        self._prefetch_event = None
+        self._pinned_cache: Dict[str, torch.Tensor] = {}
 
     def __repr__(self):
         return (
@@ -230,6 +293,14 @@ class AlignDevicesHook(ModelHook):
                 self.tied_params_names.add(name)
             set_module_tensor_to_device(module, name, "meta")
 
+        if (
+            _async_onload_enabled()
+            and self.execution_device is not None
+            and torch.device(self.execution_device).type == "cuda"
+        ):
+            self._module_ref = module
+            self._prefetcher = _prefetcher_for(self.weights_map)
+
         if self.execution_device is not None:
             if not self.offload_buffers:
                 # buffers stay resident on the execution device
@@ -242,12 +313,63 @@ class AlignDevicesHook(ModelHook):
                     set_module_tensor_to_device(module, name, self.execution_device)
         return module
 
+    def _pinned_value(self, name: str, value: torch.Tensor) -> torch.Tensor:
+        """Serve H2D copies from a per-hook pinned-host cache: the copy then
+        runs on the HIP copy engine without a staging bounce, and can be
+        issued asynchronously on the prefetch stream."""
+        if value.device.type != "cpu" or value.is_pinned():
+            return value
+        cached = self._pinned_cache.get(name)
+        if cached is None:
+            global _pinned_bytes
+            budget = int(os.environ.get("ACCELERATE_AMD_PINNED_CACHE_MB", "65536")) << 20
+            nbytes = value.numel() * value.element_size()
+            if _pinned_bytes + nbytes > budget:
+                return value  # cache full: fall back to a pageable copy
+            try:
+                cached = value.pin_memory()
+            except RuntimeError:
+                cached = value  # host pinning exhausted
+            else:
+                _pinned_bytes += nbytes
+            self._pinned_cache[name] = cached
+        return cached
+
+    def _start_prefetch(self, stream: torch.cuda.Stream):
+        """Issue this hook's weight onload on the side stream (called by the
+        prefetcher when the PREVIOUS block's pre_forward runs)."""
+        module = self._module_ref
+        if module is None:
+            return
+        staged = {}
+        with torch.cuda.stream(stream):
+            for name, _ in self._managed(module):
+                if name in self.tied_params_names:
+                    continue  # tied weights go through the dedup slow path
+                value = self.weights_map[name]
+                if value is None:
+                    continue
+                staged[name] = self._pinned_value(name, value).to(self.execution_device, non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record(stream)
+        self._prefetched, self._prefetch_event = staged, ev
+
     def pre_forward(self, module, *args, **kwargs):
         if self.io_same_device:
             self.input_device = find_device([args, kwargs])
         if self.offload:
             self.tied_pointers_to_remove = set()
+            staged = self._prefetched
+            self._prefetched = None
+            if staged is not None:
+                # compute stream consumes the side-stream copies
+                torch.cuda.current_stream().wait_event(self._prefetch_event)
             for name, _ in self._managed(module):
+                got = staged.pop(name, None) if staged else None
+                if got is not None:
+                    got.record_stream(torch.cuda.current_stream())
+                    set_module_tensor_to_device(module, name, self.execution_device, value=got)
+                    continue
                 value = self.weights_map[name]
                 if name in self.tied_params_names and value.data_ptr() not in self.tied_params_map:
                     self.tied_params_map[value.data_ptr()] = {}
@@ -257,9 +379,13 @@ class AlignDevicesHook(ModelHook):
                     and value.data_ptr() in self.tied_params_map
                 ):
                     self.tied_pointers_to_remove.add((value.data_ptr(), self.execution_device))
+                elif value is not None and self._prefetcher is not None:
+                    value = self._pinned_value(name, value)
                 set_module_tensor_to_device(
                     module, name, self.execution_device, value=value, tied_params_map=self.tied_params_map
                 )
+            if self._prefetcher is not None:
+                self._prefetcher.note(self)
         return (
             send_to_device(args, self.execution_device),
             send_to_device(kwargs, self.execution_device, skip_keys=self.skip_keys),

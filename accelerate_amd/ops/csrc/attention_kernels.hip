@@ -31,12 +31,22 @@ typedef __bf16 abf16;
 typedef abf16 bf16x8a __attribute__((ext_vector_type(8)));
 typedef float f32x4a __attribute__((ext_vector_type(4)));
 
+// f32 -> bf16 through the HARDWARE packed convert: a plain static_cast
+// compiles to v_cvt_pk_bf16_f32 (RNE, NaN-correct) and the compiler packs
+// adjacent pairs — the previous branchy bit-twiddled round-to-nearest cost
+// ~7 VALU ops per element and was the top VALU source in the disassembly
+// (2983 VALU vs 128 MFMA in fa_fwd<128>, docs/NEXT_STEPS round 1).
 __device__ __forceinline__ ushort af2bf(float f) {
-  union { float f; unsigned int i; } cv;
-  cv.f = f;
-  if ((cv.i & 0x7F800000u) == 0x7F800000u) return (ushort)(cv.i >> 16) | (ushort)((cv.i & 0xFFFFu) ? 0x40 : 0);
-  cv.i += 0x7FFFu + ((cv.i >> 16) & 1u);
-  return (ushort)(cv.i >> 16);
+  abf16 h = (abf16)f;
+  return *reinterpret_cast<ushort*>(&h);
+}
+
+
+// bit view of one lane of a bf16 vector (ext_vector elements have no address)
+__device__ __forceinline__ unsigned bfbits(abf16 h) {
+  union { abf16 h; ushort u; } cv;
+  cv.h = h;
+  return (unsigned)cv.u;
 }
 
 constexpr float kLog2e = 1.4426950408889634f;
@@ -49,7 +59,7 @@ constexpr float kLn2 = 0.6931471805599453f;
 // storage); lse:[B,Hq,Sq] fp32 natural-log contiguous.
 // past: causal offset — query i attends keys <= past + i.
 template <int D>
-__global__ __launch_bounds__(256) void fa_fwd_kernel(
+__global__ __launch_bounds__(256, 2) void fa_fwd_kernel(
     const abf16* __restrict__ q, const abf16* __restrict__ k,
     const abf16* __restrict__ v, ushort* __restrict__ out,
     float* __restrict__ lse, int Sq, int Sk, int past, int causal,
@@ -63,7 +73,12 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
   extern __shared__ char smem[];
   abf16* Ks = reinterpret_cast<abf16*>(smem);                 // [BN][KP]
   abf16* VTs = Ks + BN * KP;                                  // [D][NP]
-  abf16* Ps = VTs + D * NP;                                   // [BM][NP]
+  // P staging ALIASES the K tile at D=128 (exactly BM*NP == BN*KP bf16):
+  // K is fully consumed by the QK^T MFMAs, so after a barrier the same
+  // LDS holds P. This cuts the block from 102 KB to 68 KB — 2 blocks/CU
+  // instead of 1, which is the occupancy this kernel was starved of
+  // (round-1 PMC: ~80% stall at occupancy 1, profiles/llama8b_fa_pmc.md).
+  abf16* Ps = (BM * NP <= BN * KP) ? Ks : VTs + D * NP;       // [BM][NP]
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -104,21 +119,30 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
 
   for (int kb0 = 0; kb0 < k_hi; kb0 += BN) {
     // --- stage K [BN][D] row-major and V^T [D][BN] into LDS ---
-    // 256 threads x 16 B chunks; global reads coalesce along rows.
+    // Each thread owns a PAIR of adjacent key rows: the V transpose then
+    // writes packed 2-key b32 columns (8 ds_write_b32 per pair) instead of
+    // 16 scalar ds_write_b16 — half the LDS-write ops of the scatter.
     {
-      constexpr int chunks = BN * D / 8;         // 16 B chunks in the tile
+      constexpr int pair_chunks = (BN / 2) * (D / 8);
 #pragma unroll
-      for (int it = 0; it < chunks / 256; ++it) {
+      for (int it = 0; it < pair_chunks / 256; ++it) {
         const int cid = tid + it * 256;
-        const int row = cid / (D / 8);           // key within tile
+        const int r0 = (cid / (D / 8)) * 2;      // even key within tile
         const int col8 = (cid % (D / 8)) * 8;    // d offset
-        int64_t key = kb0 + row;
-        if (key >= Sk) key = Sk - 1;             // clamp (masked below)
-        bf16x8a kv8 = *reinterpret_cast<const bf16x8a*>(kb + key * sK.s + col8);
-        *reinterpret_cast<bf16x8a*>(Ks + row * KP + col8) = kv8;
-        bf16x8a vv8 = *reinterpret_cast<const bf16x8a*>(vb + key * sV.s + col8);
+        int64_t key0 = kb0 + r0, key1 = kb0 + r0 + 1;
+        if (key0 >= Sk) key0 = Sk - 1;           // clamp (masked below)
+        if (key1 >= Sk) key1 = Sk - 1;
+        *reinterpret_cast<bf16x8a*>(Ks + r0 * KP + col8) =
+            *reinterpret_cast<const bf16x8a*>(kb + key0 * sK.s + col8);
+        *reinterpret_cast<bf16x8a*>(Ks + (r0 + 1) * KP + col8) =
+            *reinterpret_cast<const bf16x8a*>(kb + key1 * sK.s + col8);
+        bf16x8a v0 = *reinterpret_cast<const bf16x8a*>(vb + key0 * sV.s + col8);
+        bf16x8a v1 = *reinterpret_cast<const bf16x8a*>(vb + key1 * sV.s + col8);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) VTs[(col8 + j) * NP + row] = vv8[j];
+        for (int j = 0; j < 8; ++j) {
+          const unsigned pk = bfbits(v0[j]) | (bfbits(v1[j]) << 16);
+          *reinterpret_cast<unsigned*>(VTs + (col8 + j) * NP + r0) = pk;
+        }
       }
     }
     __syncthreads();
@@ -129,6 +153,7 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
     for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
       for (int ni = 0; ni < BN / 16; ++ni) sacc[mi][ni] = f32x4a{};
+    __builtin_amdgcn_s_setprio(1);  // favor this wave while the MFMA cluster runs
 #pragma unroll
     for (int ni = 0; ni < BN / 16; ++ni) {
 #pragma unroll
@@ -140,19 +165,24 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
           sacc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[mi][kc], bf, sacc[mi][ni], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // --- mask (causal upper edge and Sk tail) in the RAW score domain ---
+    // row/col indices are affine in the loop counters — hoisted bases, one
+    // add per element instead of a 4-term recomputation
     const bool edge = (kb0 + BN > Sk) || (causal && kb0 + BN > past + q0);
     if (edge) {
+      const int kc0 = kb0 + (lane & 15);
+      const int qr0 = q0 + wave * 32 + (lane >> 4) * 4;
 #pragma unroll
       for (int ni = 0; ni < BN / 16; ++ni) {
-        const int kc = kb0 + ni * 16 + (lane & 15);
+        const int kc = kc0 + ni * 16;
 #pragma unroll
         for (int mi = 0; mi < 2; ++mi) {
+          const int qb16 = qr0 + mi * 16;
 #pragma unroll
           for (int j = 0; j < 4; ++j) {
-            const int qr = q0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
-            if (kc >= Sk || (causal && kc > past + qr)) sacc[mi][ni][j] = -INFINITY;
+            if (kc >= Sk || (causal && kc > past + qb16 + j)) sacc[mi][ni][j] = -INFINITY;
           }
         }
       }
@@ -189,20 +219,22 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
     }
 
     // --- P -> LDS (wave-private rows; C-layout scatter, bf16) ---
+    // Ps aliases Ks at D=128: EVERY wave must be done with its QK^T reads
+    // of the K tile before any wave's P lands on top of it.
+    __syncthreads();
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
-      for (int ni = 0; ni < BN / 16; ++ni)
+      for (int j = 0; j < 4; ++j) {
+        abf16* prow = Ps + (wave * 32 + mi * 16 + (lane >> 4) * 4 + j) * NP + (lane & 15);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const int row = wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
-          ushort pb = af2bf(sacc[mi][ni][j]);
-          Ps[row * NP + ni * 16 + (lane & 15)] = *reinterpret_cast<abf16*>(&pb);
-        }
+        for (int ni = 0; ni < BN / 16; ++ni) prow[ni * 16] = (abf16)sacc[mi][ni][j];
+      }
     // same-wave write->read: the compiler orders the ds ops (no barrier —
     // no wave reads another wave's P rows)
 
     // --- O += P V : A = P (rows, k=key), B = V^T rows are d ---
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kc = 0; kc < BN / 32; ++kc) {
       bf16x8a pf[2];
@@ -219,6 +251,7 @@ __global__ __launch_bounds__(256) void fa_fwd_kernel(
           oacc[mi][nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf[mi], vf, oacc[mi][nd], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();  // next iteration restages Ks/VTs
   }
 
@@ -336,24 +369,33 @@ __global__ __launch_bounds__(256) void fa_bwd_dkdv_kernel(
   const int q_lo = causal ? max(0, ((k0 - past) / BQ) * BQ) : 0;
 
   for (int q0 = q_lo; q0 < Sq; q0 += BQ) {
-    // stage Q/dO tiles (both layouts) + L*log2e + Drow
+    // stage Q/dO tiles (both layouts) + L*log2e + Drow; transposed copies
+    // write packed 2-row b32 columns (see the fwd staging comment)
     {
-      constexpr int chunks = BQ * D / 8;
+      constexpr int pair_chunks = (BQ / 2) * (D / 8);
 #pragma unroll
-      for (int it = 0; it < chunks / 256; ++it) {
+      for (int it = 0; it < pair_chunks / 256 + (pair_chunks % 256 != 0); ++it) {
         const int cid = tid + it * 256;
-        const int row = cid / (D / 8);
+        if (pair_chunks % 256 != 0 && cid >= pair_chunks) break;
+        const int r0 = (cid / (D / 8)) * 2;
         const int col8 = (cid % (D / 8)) * 8;
-        int64_t qr = q0 + row;
-        if (qr >= Sq) qr = Sq - 1;  // clamped; masked via p=0 below
-        bf16x8a q8 = *reinterpret_cast<const bf16x8a*>(qb + qr * sQ.s + col8);
-        *reinterpret_cast<bf16x8a*>(Qs + row * KP + col8) = q8;
-        bf16x8a d8 = *reinterpret_cast<const bf16x8a*>(dob + qr * sDo.s + col8);
-        *reinterpret_cast<bf16x8a*>(dOs + row * KP + col8) = d8;
+        int64_t qr0 = q0 + r0, qr1 = q0 + r0 + 1;
+        if (qr0 >= Sq) qr0 = Sq - 1;  // clamped; masked via p=0 below
+        if (qr1 >= Sq) qr1 = Sq - 1;
+        bf16x8a qa = *reinterpret_cast<const bf16x8a*>(qb + qr0 * sQ.s + col8);
+        bf16x8a qc = *reinterpret_cast<const bf16x8a*>(qb + qr1 * sQ.s + col8);
+        *reinterpret_cast<bf16x8a*>(Qs + r0 * KP + col8) = qa;
+        *reinterpret_cast<bf16x8a*>(Qs + (r0 + 1) * KP + col8) = qc;
+        bf16x8a da = *reinterpret_cast<const bf16x8a*>(dob + qr0 * sDo.s + col8);
+        bf16x8a dc = *reinterpret_cast<const bf16x8a*>(dob + qr1 * sDo.s + col8);
+        *reinterpret_cast<bf16x8a*>(dOs + r0 * KP + col8) = da;
+        *reinterpret_cast<bf16x8a*>(dOs + (r0 + 1) * KP + col8) = dc;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          QTs[(col8 + j) * QP + row] = q8[j];
-          dOTs[(col8 + j) * QP + row] = d8[j];
+          unsigned pq = bfbits(qa[j]) | (bfbits(qc[j]) << 16);
+          *reinterpret_cast<unsigned*>(QTs + (col8 + j) * QP + r0) = pq;
+          unsigned pd = bfbits(da[j]) | (bfbits(dc[j]) << 16);
+          *reinterpret_cast<unsigned*>(dOTs + (col8 + j) * QP + r0) = pd;
         }
       }
       if (tid < BQ) {
@@ -482,7 +524,7 @@ __global__ __launch_bounds__(256) void fa_bwd_dkdv_kernel(
 // Q and dO rows live in registers as A-fragments; K,V tiles of 64 keys are
 // staged per inner step (K in both layouts).
 template <int D>
-__global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
+__global__ __launch_bounds__(256, 2) void fa_bwd_dq_kernel(
     const abf16* __restrict__ q, const abf16* __restrict__ k,
     const abf16* __restrict__ v, const abf16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
@@ -535,20 +577,29 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
 
   for (int k0 = 0; k0 < k_hi; k0 += BN) {
     {
-      constexpr int chunks = BN * D / 8;
+      constexpr int pair_chunks = (BN / 2) * (D / 8);
 #pragma unroll
-      for (int it = 0; it < chunks / 256; ++it) {
+      for (int it = 0; it < pair_chunks / 256 + (pair_chunks % 256 != 0); ++it) {
         const int cid = tid + it * 256;
-        const int row = cid / (D / 8);
+        if (pair_chunks % 256 != 0 && cid >= pair_chunks) break;
+        const int r0 = (cid / (D / 8)) * 2;
         const int col8 = (cid % (D / 8)) * 8;
-        int64_t kr = k0 + row;
-        if (kr >= Sk) kr = Sk - 1;
-        bf16x8a k8 = *reinterpret_cast<const bf16x8a*>(kb + kr * sK.s + col8);
-        *reinterpret_cast<bf16x8a*>(Ks + row * KP + col8) = k8;
-        *reinterpret_cast<bf16x8a*>(Vs + row * KP + col8) =
-            *reinterpret_cast<const bf16x8a*>(vb + kr * sV.s + col8);
+        int64_t kr0 = k0 + r0, kr1 = k0 + r0 + 1;
+        if (kr0 >= Sk) kr0 = Sk - 1;
+        if (kr1 >= Sk) kr1 = Sk - 1;
+        bf16x8a ka = *reinterpret_cast<const bf16x8a*>(kb + kr0 * sK.s + col8);
+        bf16x8a kc8 = *reinterpret_cast<const bf16x8a*>(kb + kr1 * sK.s + col8);
+        *reinterpret_cast<bf16x8a*>(Ks + r0 * KP + col8) = ka;
+        *reinterpret_cast<bf16x8a*>(Ks + (r0 + 1) * KP + col8) = kc8;
+        *reinterpret_cast<bf16x8a*>(Vs + r0 * KP + col8) =
+            *reinterpret_cast<const bf16x8a*>(vb + kr0 * sV.s + col8);
+        *reinterpret_cast<bf16x8a*>(Vs + (r0 + 1) * KP + col8) =
+            *reinterpret_cast<const bf16x8a*>(vb + kr1 * sV.s + col8);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) KTs[(col8 + j) * NP + row] = k8[j];
+        for (int j = 0; j < 8; ++j) {
+          unsigned pk = bfbits(ka[j]) | (bfbits(kc8[j]) << 16);
+          *reinterpret_cast<unsigned*>(KTs + (col8 + j) * NP + r0) = pk;
+        }
       }
     }
     __syncthreads();
@@ -647,8 +698,10 @@ namespace {
 
 template <int D>
 int fa_lds_bytes() {
-  constexpr int BN = 128, NP = BN + 8;
-  return (BN * (D + 8) + D * NP + 128 * NP) * 2;
+  constexpr int BM = 128, BN = 128, NP = BN + 8, KP = D + 8;
+  // P aliases the K tile when it fits (D=128); else it gets its own region
+  const int elems = BN * KP + D * NP + (BM * NP <= BN * KP ? 0 : BM * NP);
+  return elems * 2;
 }
 
 template <int D>

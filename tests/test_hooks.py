@@ -72,3 +72,49 @@ def test_io_same_device():
     add_hook_to_module(model, hook)
     out = model(torch.randn(2, 4))
     assert out.device == torch.device("cpu")
+
+
+def test_onload_prefetcher_learns_ring_order():
+    """The async-onload prefetcher learns execution order on pass 1 and
+    prefetches hook i+1 (wrapping) from pass 2 on."""
+    from accelerate_amd.hooks import _OnloadPrefetcher
+
+    class StubHook:
+        def __init__(self):
+            self.kicked = 0
+            self._prefetched = None
+
+        def _start_prefetch(self, stream):
+            self.kicked += 1
+
+    p = _OnloadPrefetcher()
+    p._get_stream = lambda: None  # no CUDA in CPU CI
+    hooks = [StubHook() for _ in range(3)]
+    for h in hooks:  # pass 1: learning, no prefetches yet
+        p.note(h)
+    assert p.learning and all(h.kicked == 0 for h in hooks)
+    p.note(hooks[0])  # ring closes
+    assert not p.learning
+    assert hooks[1].kicked == 1  # successor of 0 prefetched
+    p.note(hooks[2])
+    assert hooks[0].kicked == 1  # wrap-around: last prefetches first
+    # an unconsumed prefetch is not re-issued
+    hooks[1]._prefetched = {}
+    p.note(hooks[0])
+    assert hooks[1].kicked == 1
+
+
+def test_pinned_cache_budget(monkeypatch):
+    """_pinned_value pins once per name and respects the byte budget."""
+    import accelerate_amd.hooks as H
+
+    hook = H.AlignDevicesHook(offload=True)
+    t = torch.randn(16)
+    monkeypatch.setenv("ACCELERATE_AMD_PINNED_CACHE_MB", "1")
+    a = hook._pinned_value("w", t)
+    b = hook._pinned_value("w", t)
+    assert a is b  # cached
+    monkeypatch.setenv("ACCELERATE_AMD_PINNED_CACHE_MB", "0")
+    hook2 = H.AlignDevicesHook(offload=True)
+    c = hook2._pinned_value("w", t)
+    assert c is t  # budget exhausted: pageable fallback
