@@ -494,7 +494,9 @@ class ShardedModel(nn.Module):
         stream's backward-prefetch all_gathers proceed concurrently.
         ``finalize_backward`` joins the streams before the optimizer runs.
         """
-        if self._reduce_stream is None:
+        if self._reduce_stream is None or (self.world <= 1 and self.replica_group is None):
+            # single-rank worlds have no collective to overlap — the stream
+            # ping-pong (wait_event per unit) would be pure launch overhead
             unit.reduce_grads()
             self._hsdp_allreduce(unit)
             return
