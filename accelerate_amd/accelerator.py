@@ -573,6 +573,7 @@ class Accelerator:
             even_batches=self.even_batches,
             slice_fn_for_dispatch=slice_fn_for_dispatch,
             use_seedable_sampler=self.use_seedable_sampler,
+            data_seed=self.data_seed,
             non_blocking=self.non_blocking,
         )
         prepared._is_accelerate_prepared = True

@@ -70,7 +70,10 @@ class SeedableRandomSampler(RandomSampler):
         if self.generator is None:
             self.generator = torch.Generator()
             if self.initial_seed is None:
-                self.initial_seed = torch.random.default_generator.seed() % 2**31
+                # DETERMINISTIC: the seed the process was last seeded with
+                # (set_seed), identical across ranks — never a fresh entropy
+                # draw (reference: data_loader.py:88 torch.random.initial_seed)
+                self.initial_seed = torch.random.initial_seed() % 2**31
         elif self.initial_seed is None:
             self.initial_seed = self.generator.initial_seed()
 
@@ -96,12 +99,6 @@ class BatchSamplerShard(BatchSampler):
         split_batches: bool = False,
         even_batches: bool = True,
     ):
-        if split_batches and batch_sampler.batch_size % num_processes != 0:
-            raise ValueError(
-                f"split_batches slices each global batch {num_processes} ways, so the batch size "
-                f"must be divisible by the process count (got batch_size={batch_sampler.batch_size}, "
-                f"num_processes={num_processes})."
-            )
         self.batch_sampler = batch_sampler
         self.num_processes = num_processes
         self.process_index = process_index
@@ -109,8 +106,17 @@ class BatchSamplerShard(BatchSampler):
         self.even_batches = even_batches
         self.batch_size = getattr(batch_sampler, "batch_size", None)
         self.drop_last = getattr(batch_sampler, "drop_last", False)
-        if self.batch_size is None and self.even_batches:
-            raise ValueError("You need to use `even_batches=False` when the batch sampler has no batch size.")
+        if split_batches and self.batch_size is None:
+            raise ValueError(
+                "split_batches needs a fixed batch size to slice; this batch sampler has none "
+                "(dynamic batch sizes work in the default deal mode)."
+            )
+        if split_batches and self.batch_size % num_processes != 0:
+            raise ValueError(
+                f"split_batches slices each global batch {num_processes} ways, so the batch size "
+                f"must be divisible by the process count (got batch_size={self.batch_size}, "
+                f"num_processes={num_processes})."
+            )
 
     @property
     def total_length(self):
@@ -156,7 +162,42 @@ class BatchSamplerShard(BatchSampler):
                         padded += wrap_pool[: self.batch_size - len(padded)]
                     yield padded[lo:hi]
 
+    def _iter_deal_dynamic(self):
+        """Deal mode for VARYING batch sizes (no batch_size attribute): batch
+        k goes to rank k % n verbatim; with even_batches the final incomplete
+        round is padded with WHOLE batches cycled from the first round
+        (reference: dynamic-batch support in BatchSamplerShard, pinned by
+        tests/test_data_loader.py varying-batch-size cases); with drop_last
+        only complete rounds are emitted."""
+        initial: List[list] = []  # first n batches — the batch-granular pad pool
+        held: List[list] = []  # current round, buffered when drop_last
+        idx = -1
+        for idx, batch in enumerate(self.batch_sampler):
+            if len(initial) < self.num_processes:
+                initial.append(list(batch))
+            pos = idx % self.num_processes
+            if self.drop_last:
+                held.append(batch)
+                if pos == self.num_processes - 1:
+                    yield held[self.process_index]
+                    held = []
+            elif pos == self.process_index:
+                yield list(batch)
+        total = idx + 1
+        if self.drop_last or total == 0 or not self.even_batches or total % self.num_processes == 0:
+            return
+        while len(initial) < self.num_processes:
+            initial = initial + initial
+        cycle = 0
+        for pos in range(total % self.num_processes, self.num_processes):
+            if pos == self.process_index:
+                yield list(initial[cycle])
+            cycle += 1
+
     def _iter_deal(self):
+        if self.batch_size is None:
+            yield from self._iter_deal_dynamic()
+            return
         wrap_pool = []  # indices from the first full cycle, used for tail padding
         pending = []  # this rank's batch, held until the cycle completes
         idx = -1
@@ -673,6 +714,7 @@ def prepare_data_loader(
     even_batches: bool = True,
     slice_fn_for_dispatch: Optional[Callable] = None,
     use_seedable_sampler: bool = False,
+    data_seed: Optional[int] = None,
     non_blocking: bool = False,
     use_stateful_dataloader: bool = False,
 ) -> DataLoader:
@@ -710,7 +752,8 @@ def prepare_data_loader(
             data_source=sampler.data_source,
             replacement=sampler.replacement,
             num_samples=sampler._num_samples,
-            generator=getattr(sampler, "generator", torch.Generator()),
+            generator=getattr(sampler, "generator", None),
+            seed=data_seed,
         )
     if num_processes != 1 and not dispatch_batches:
         if isinstance(new_dataset, IterableDataset):
@@ -875,6 +918,7 @@ def skip_first_batches(dataloader, num_batches=0):
         if new_batch_sampler is None:
             # Need to manually skip batches in the dataloader
             kwargs["skip_batches"] = num_batches
+        prior_iteration = dataloader.iteration
         dataloader = DataLoaderDispatcher(
             dataset,
             split_batches=dataloader.split_batches,
@@ -882,6 +926,9 @@ def skip_first_batches(dataloader, num_batches=0):
             _drop_last=dataloader._drop_last,
             **kwargs,
         )
+        # mid-epoch resume must NOT rewind the epoch: iteration carries over
+        # so __iter__'s set_epoch call replays the same sampler epoch
+        dataloader.iteration = prior_iteration
     elif isinstance(dataloader, DataLoaderShard):
         if new_batch_sampler is None:
             # Need to manually skip batches in the dataloader
@@ -891,6 +938,7 @@ def skip_first_batches(dataloader, num_batches=0):
             kwargs["batch_size"] = dataloader.batch_size
         else:
             kwargs["batch_sampler"] = new_batch_sampler
+        prior_iteration = dataloader.iteration
         dataloader = DataLoaderShard(
             dataset,
             device=dataloader.device,
@@ -898,6 +946,7 @@ def skip_first_batches(dataloader, num_batches=0):
             synchronized_generator=dataloader.synchronized_generator,
             **kwargs,
         )
+        dataloader.iteration = prior_iteration
     else:
         if new_batch_sampler is None:
             # Need to manually skip batches in the dataloader

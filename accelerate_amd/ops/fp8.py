@@ -90,11 +90,19 @@ class _FP8LinearFn(torch.autograd.Function):
         # refresh scales BEFORE casting: the device scale scalar must stay
         # untouched between a cast and its GEMM consumers (incl. backward)
         sx.roll_and_update(ext)
-        sw.roll_and_update(ext)
         # each operand cast ONCE producing both layouts: x8 (fwd) + x8t (wgrad),
         # w8 (fwd) + w8t (dgrad)
         x8, x8t = _cast_transpose_fp8(x2d, sx, e5m2=False)
-        w8, w8t = _cast_transpose_fp8(weight, sw, e5m2=False)
+        # the WEIGHT changes once per optimizer step, not once per forward:
+        # cache its fp8 casts keyed on torch's in-place version counter, so
+        # microbatched/accumulated steps pay the weight cast exactly once
+        cache = getattr(sw, "_w8_cache", None)
+        if cache is not None and cache[0] == weight._version:
+            w8, w8t = cache[1], cache[2]
+        else:
+            sw.roll_and_update(ext)
+            w8, w8t = _cast_transpose_fp8(weight, sw, e5m2=False)
+            sw._w8_cache = (weight._version, w8, w8t)
         # y = (x8 @ w8^T) * (1/sx) * (1/sw)  — hipBLASLt fp8 MFMA GEMM
         y = torch._scaled_mm(
             x8, w8.t(), scale_a=sx.scale_inv, scale_b=sw.scale_inv, bias=bias, out_dtype=torch.bfloat16

@@ -407,7 +407,6 @@ class GradientState:
         self.__dict__ = self._shared_state
         if not self.initialized:
             self.sync_gradients = True
-            self.active_dataloader = None
             self.dataloader_references = [None]
             self.plugin_kwargs = (
                 gradient_accumulation_plugin.to_kwargs() if gradient_accumulation_plugin is not None else {}
@@ -455,16 +454,21 @@ class GradientState:
     def _set_sync_gradients(self, sync_gradients: bool):
         self.sync_gradients = sync_gradients
 
+    @property
+    def active_dataloader(self):
+        # weakref-only registry: the GradientState singleton must never keep
+        # a dataloader alive (reference: state.py:1335-1366; a strong ref
+        # here leaks loaders abandoned mid-iteration)
+        last = self.dataloader_references[-1]
+        return last() if last is not None else None
+
     def _add_dataloader(self, dataloader):
-        self.active_dataloader = dataloader
-        self.dataloader_references.append(weakref.ref(self.active_dataloader))
+        self.dataloader_references.append(weakref.ref(dataloader))
 
     def _remove_dataloader(self, dataloader):
         self.dataloader_references = [
-            ref for ref in self.dataloader_references if ref is not None and ref() is not dataloader
+            ref for ref in self.dataloader_references if ref is None or ref() is not dataloader
         ] or [None]
-        last = self.dataloader_references[-1]
-        self.active_dataloader = last() if last is not None else None
 
     @property
     def in_dataloader(self) -> bool:
