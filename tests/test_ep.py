@@ -151,3 +151,17 @@ def test_pp_engine_validation_errors():
     # degenerate split still returns n stages
     stages = split_into_stages(nn.Sequential(nn.Linear(2, 2)), 3)
     assert len(stages) == 3
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(
+    not torch.cuda.is_available() or torch.cuda.device_count() < 2,
+    reason="PP-over-RCCL parity needs >= 2 GPUs (P2P recv on device tensors)",
+)
+def test_pp_train_gpu_parity_2dev():
+    """GPU parity for the PP training engine over RCCL (ADVICE round-1:
+    device-placed stages and P2P buffers)."""
+    out = launch_distributed(
+        "tests/distributed_scripts/pp_train_script.py", nproc=2, extra_env={"PP_GPU": "1"}
+    )
+    assert "PP_TRAIN_PASS" in out
