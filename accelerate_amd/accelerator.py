@@ -1002,7 +1002,23 @@ class Accelerator:
         os.makedirs(output_dir, exist_ok=True)
         logger.info(f"Saving current state to {output_dir}")
 
-        weights = [self.get_state_dict(m, unwrap=False) for m in self._models]
+        # SHARDED_STATE_DICT: every rank writes its own shard file with zero
+        # cross-rank communication (reference rationale: fsdp_utils.py:107-118
+        # — the DCP-timeout fix at 2800+ GPUs); FULL dicts all-gather per unit
+        # and rank 0 writes one file.
+        from .parallel.fsdp import ShardedModel
+
+        plugin = getattr(self.state, "fsdp_plugin", None)
+        sharded_dicts = plugin is not None and getattr(plugin, "state_dict_type", "") == "sharded_state_dict"
+        weights = []
+        for i, m in enumerate(self._models):
+            if sharded_dicts and isinstance(m, ShardedModel):
+                suffix = f"_{i}" if i > 0 else ""
+                shard_file = os.path.join(output_dir, f"model_fsdp{suffix}_rank{self.process_index}.bin")
+                torch.save(m.sharded_state_dict(), shard_file)
+                weights.append({"__fsdp_sharded__": True})  # placeholder: skip full save
+            else:
+                weights.append(self.get_state_dict(m, unwrap=False))
         # Save the samplers of the dataloaders
         dataloaders = self._dataloaders
 
@@ -1049,6 +1065,20 @@ class Accelerator:
         models = self._models
         for hook in self._load_model_state_pre_hook.values():
             hook(models, input_dir)
+
+        # sharded checkpoints load per-rank with no communication; those
+        # models are excluded from the full-dict loading below
+        from .parallel.fsdp import ShardedModel
+
+        skip_models = set()
+        for i, m in enumerate(models):
+            suffix = f"_{i}" if i > 0 else ""
+            shard_file = os.path.join(input_dir, f"model_fsdp{suffix}_rank{self.process_index}.bin")
+            if isinstance(m, ShardedModel) and os.path.exists(shard_file):
+                m.load_sharded_state_dict(torch.load(shard_file, weights_only=False))
+                skip_models.add(i)
+        # placeholders keep indices aligned with the on-disk _{i} suffixes
+        models = [None if i in skip_models else m for i, m in enumerate(models)]
 
         map_location = load_model_func_kwargs.pop("map_location", None)
         if map_location is None:

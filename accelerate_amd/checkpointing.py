@@ -49,6 +49,8 @@ def save_accelerator_state(
     output_dir = Path(output_dir)
     # Model states
     for i, state in enumerate(model_states):
+        if isinstance(state, dict) and state.get("__fsdp_sharded__"):
+            continue  # written per-rank by Accelerator.save_state (sharded)
         if safe_serialization and is_safetensors_available():
             import safetensors.torch
 
@@ -150,6 +152,8 @@ def load_accelerator_state(
 
     # Models
     for i, model in enumerate(models):
+        if model is None:
+            continue  # sharded: loaded per-rank by Accelerator.load_state
         ending = f"_{i}" if i > 0 else ""
         safe_path = input_dir / f"{SAFE_MODEL_NAME}{ending}.safetensors"
         bin_path = input_dir / f"{MODEL_NAME}{ending}.bin"
