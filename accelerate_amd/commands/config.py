@@ -47,13 +47,29 @@ class ClusterConfig:
             yaml.safe_dump(self.to_dict(), f)
         return path
 
+    _VALID_DISTRIBUTED = ("NO", "MULTI_GPU", "MULTI_CPU", "FSDP")
+    _VALID_PRECISION = ("no", "fp16", "bf16", "fp8")
+
     @classmethod
     def load(cls, path=None):
         path = Path(path or DEFAULT_CONFIG_FILE)
         with open(path) as f:
-            data = yaml.safe_load(f)
+            data = yaml.safe_load(f) or {}
         known = {f_.name for f_ in cls.__dataclass_fields__.values()}
-        return cls(**{k: v for k, v in data.items() if k in known})
+        unknown = sorted(set(data) - known)
+        if unknown:
+            raise ValueError(
+                f"Config file {path} contains unknown key(s): {', '.join(unknown)}. "
+                f"Valid keys: {', '.join(sorted(known))}."
+            )
+        cfg = cls(**data)
+        if cfg.distributed_type not in cls._VALID_DISTRIBUTED:
+            raise ValueError(
+                f"distributed_type {cfg.distributed_type!r} is not one of {cls._VALID_DISTRIBUTED}"
+            )
+        if cfg.mixed_precision not in cls._VALID_PRECISION:
+            raise ValueError(f"mixed_precision {cfg.mixed_precision!r} is not one of {cls._VALID_PRECISION}")
+        return cfg
 
 
 def _ask(prompt, default=None, cast=str, choices=None):

@@ -16,8 +16,28 @@ from pathlib import Path
 from .config import DEFAULT_CONFIG_FILE, ClusterConfig
 
 
+class _DashAliasParserProxy:
+    """Registers every ``--foo_bar`` flag with a ``--foo-bar`` alias so both
+    spellings work (reference: launch flags accept hyphens AND underscores,
+    tests/test_cli.py test_hyphen/test_underscore)."""
+
+    def __init__(self, parser):
+        self._parser = parser
+
+    def add_argument(self, name, *args, **kwargs):
+        names = [name]
+        if name.startswith("--") and "_" in name:
+            names.append(name.replace("_", "-"))
+            kwargs.setdefault("dest", name[2:])
+        return self._parser.add_argument(*names, *args, **kwargs)
+
+    def __getattr__(self, item):
+        return getattr(self._parser, item)
+
+
 def add_parser(subparsers):
-    parser = subparsers.add_parser("launch", help="Launch a training script on MI355X GPUs")
+    real_parser = subparsers.add_parser("launch", help="Launch a training script on MI355X GPUs")
+    parser = _DashAliasParserProxy(real_parser)
     parser.add_argument("--config_file", default=None)
     parser.add_argument("--num_processes", type=int, default=None, help="Total number of processes (one per GPU)")
     parser.add_argument("--num_machines", type=int, default=None)

@@ -140,6 +140,28 @@ def main():
     if acc.is_main_process:
         print("FSDP_NOSYNC_PASS")
 
+    # save_state/load_state round trip under SHARDED_STATE_DICT: every rank
+    # writes/reads its own model shard + optimizer file, no communication
+    acc.state.fsdp_plugin.state_dict_type = "sharded_state_dict"
+    d3 = [tempfile.mkdtemp() if acc.is_main_process else None]
+    broadcast_object_list(d3)
+    before = model.full_state_dict()
+    opt_before = {k: v for k, v in opt.state_dict().items()}
+    acc.save_state(d3[0])
+    acc.wait_for_everyone()
+    shard_files = [f for f in os.listdir(d3[0]) if f.startswith("model_fsdp")]
+    assert len(shard_files) == n, f"expected one shard file per rank, got {shard_files}"
+    with torch.no_grad():
+        for u in model.units:
+            u.shard.mul_(0.0)
+    acc.load_state(d3[0])
+    after = model.full_state_dict()
+    for k2 in before:
+        assert torch.allclose(before[k2], after[k2], atol=1e-7), f"sharded ckpt round-trip lost {k2}"
+    assert len(opt.state_dict()["state"]) == len(opt_before["state"])
+    if acc.is_main_process:
+        print("FSDP_SHARDED_CKPT_PASS")
+
     # meta-device init + per-rank SLICED checkpoint load: the full model is
     # never materialized on any rank (VERDICT missing #2; reference contrast:
     # fsdp_utils.py:563-656 rank-0 broadcast load)

@@ -107,3 +107,46 @@ def test_estimate_bundled_families():
     assert "131.4" in r.stdout
     r2 = run_cli("estimate", "mixtral-8x7b", "--dtypes", "bfloat16")
     assert r2.returncode == 0, r2.stderr
+
+
+def test_launch_flags_accept_hyphens_and_underscores():
+    """Both --foo-bar and --foo_bar spellings parse to the same dest
+    (reference tests/test_cli.py test_hyphen/test_underscore)."""
+    import argparse
+
+    from accelerate_amd.commands.launch import add_parser
+
+    parser = argparse.ArgumentParser()
+    add_parser(parser.add_subparsers())
+    a = parser.parse_args(["launch", "--fsdp_sharding_strategy", "hybrid_shard", "s.py"])
+    b = parser.parse_args(["launch", "--fsdp-sharding-strategy", "hybrid_shard", "s.py"])
+    assert a.fsdp_sharding_strategy == b.fsdp_sharding_strategy == "hybrid_shard"
+    c = parser.parse_args(["launch", "--parallelism-config-tp-size", "4", "s.py"])
+    assert c.parallelism_config_tp_size == 4
+
+
+def test_config_rejects_unknown_keys(tmp_path):
+    import pytest
+    import yaml
+
+    from accelerate_amd.commands.config import ClusterConfig
+
+    p = tmp_path / "cfg.yaml"
+    p.write_text(yaml.safe_dump({"num_processes": 2, "not_a_real_key": 1}))
+    with pytest.raises(ValueError, match="not_a_real_key"):
+        ClusterConfig.load(p)
+
+
+def test_config_rejects_invalid_values(tmp_path):
+    import pytest
+    import yaml
+
+    from accelerate_amd.commands.config import ClusterConfig
+
+    p = tmp_path / "cfg.yaml"
+    p.write_text(yaml.safe_dump({"distributed_type": "XPU_TPU"}))
+    with pytest.raises(ValueError, match="distributed_type"):
+        ClusterConfig.load(p)
+    p.write_text(yaml.safe_dump({"mixed_precision": "int3"}))
+    with pytest.raises(ValueError, match="mixed_precision"):
+        ClusterConfig.load(p)
