@@ -370,17 +370,30 @@ class Accelerator:
 
     @contextmanager
     def join_uneven_inputs(self, joinables, even_batches=None):
-        """Training on uneven per-rank inputs (reference: accelerator.py:1300).
+        """Training on UNEVEN per-rank inputs (reference: accelerator.py:1300
+        wrapping torch.distributed.algorithms.Join).
 
-        Our reducer communicates only from ``Accelerator.backward`` so ranks
-        that exhaust their data simply stop calling backward — but collective
-        counts must still match. We mirror the reference's contract by
-        overriding ``even_batches`` on prepared dataloaders inside the
-        context; with ``even_batches=True`` (default) inputs are already
-        rectangular and join is a no-op.
+        Protocol (our reducer's equivalent of the Join algorithm): while the
+        context is active every engine forward opens a liveness collective
+        round ``[n_live, n_live_syncing]``; a rank that exhausts its data
+        reaches the context exit and enters a drain loop that keeps
+        answering those rounds with zeros and SHADOWS each live step's
+        collectives (buffer broadcast + zero-contribution bucket
+        all-reduces, in the recorded launch order) until every rank has
+        joined — so uneven iteration counts cannot desynchronize or
+        deadlock the reducer. ``even_batches`` may additionally be
+        overridden on prepared dataloaders for the scope of the context.
         """
         if even_batches is None:
             even_batches = self.even_batches
+        engines = []
+        for j in joinables:
+            if isinstance(j, DistributedDataParallelEngine):
+                engines.append(j)
+            else:
+                for m in self._models:
+                    if isinstance(m, DistributedDataParallelEngine) and m.module is j:
+                        engines.append(m)
         iterable_dl_seen = False
         dl_even_batches_values = []
         for dl_idx, dl in enumerate(self._dataloaders):
@@ -392,9 +405,17 @@ class Accelerator:
                 dl.batch_sampler.even_batches = even_batches
         if iterable_dl_seen:
             warnings.warn("Overriding even_batches is only supported for map-style datasets; ignored for dispatchers.")
+        if self.use_distributed:
+            for e in engines:
+                e._join_active = True
         try:
             yield
         finally:
+            if self.use_distributed:
+                for e in engines:
+                    # this rank is done: shadow the stragglers' steps
+                    e.join_drain()
+                    e._join_active = False
             for dl_idx, value in dl_even_batches_values:
                 self._dataloaders[dl_idx].batch_sampler.even_batches = value
 

@@ -52,6 +52,7 @@ class _Bucket:
         "plan",
         "plan_key",
         "pinned",
+        "index",
     )
 
     def __init__(self, params: List[torch.nn.Parameter], dtype, comm_dtype, device):
@@ -164,6 +165,11 @@ class DistributedDataParallelEngine(nn.Module):
         # order (reference: torch DDP's reducer bucket rebuild)
         self._ready_order: List[torch.nn.Parameter] = []
         self._buckets_rebuilt = False
+        # join_uneven_inputs protocol state (liveness rounds + shadow steps)
+        self._join_active = False
+        self._join_steps = 0
+        self._launch_order: List[int] = []
+        self._cur_launch_order: List[int] = []
         self._build_buckets()
         self._register_hooks()
         if self._world_size > 1:
@@ -200,7 +206,8 @@ class DistributedDataParallelEngine(nn.Module):
             current_key = key
         if current:
             self._buckets.append(_Bucket(current, current_key[0], self.comm_dtype, current_key[1]))
-        for b in self._buckets:
+        for bi, b in enumerate(self._buckets):
+            b.index = bi
             for i, p in enumerate(b.params):
                 self._param_to_bucket[p] = (b, i)
 
@@ -252,6 +259,7 @@ class DistributedDataParallelEngine(nn.Module):
         if bucket.launched:
             return
         bucket.launched = True
+        self._cur_launch_order.append(bucket.index)
         if not self._fused_copy(bucket, to_flat=True):
             for i, p in enumerate(bucket.params):
                 lo = bucket.offsets[i]
@@ -307,6 +315,10 @@ class DistributedDataParallelEngine(nn.Module):
                     else:
                         p.grad.copy_(reduced, non_blocking=True)
             bucket.reset()
+        # remember the launch order for join shadow steps (deterministic:
+        # every rank's backward fires hooks in the same autograd order)
+        self._launch_order = self._cur_launch_order
+        self._cur_launch_order = []
         self._maybe_rebuild_buckets()
 
     def _maybe_rebuild_buckets(self):
@@ -323,46 +335,104 @@ class DistributedDataParallelEngine(nn.Module):
             self._build_buckets(ordered_params=self._ready_order)
         self._ready_order = []
 
-    def forward(self, *args, **kwargs):
-        if self.broadcast_buffers and self._world_size > 1 and self.module.training:
-            # per-iteration sync only for mutable float buffers (BN running
-            # stats); constant/int buffers were broadcast once at wrap time.
-            # All float buffers travel in ONE coalesced broadcast — a single
-            # collective launch per step instead of one per buffer (per-step
-            # latency tax at N=8 otherwise).
-            if self._float_buffers is None:
-                # non-persistent buffers (absent from state_dict) are derived
-                # constants (e.g. RoPE tables) — identical by construction,
-                # never broadcast
-                persistent = set(self.module.state_dict(keep_vars=True))
-                self._float_buffers = [
-                    b
-                    for n, b in self.module.named_buffers()
-                    if n in persistent and b.is_floating_point() and b.numel() > 0
-                ]
-                if self._float_buffers:
-                    total = sum(b.numel() for b in self._float_buffers)
-                    # fp32 wire is lossless for bf16/fp16/fp32 buffers; widen
-                    # to fp64 only if any buffer needs it
-                    wire = (
-                        torch.float64
-                        if any(b.dtype == torch.float64 for b in self._float_buffers)
-                        else torch.float32
-                    )
-                    self._buffer_flat = torch.empty(total, dtype=wire, device=self._float_buffers[0].device)
+    def _join_flag_round(self, live: int):
+        """One liveness collective of the join protocol:
+        [n_live, n_live_syncing] summed across ranks."""
+        dev = next(self.module.parameters()).device
+        flag = torch.tensor(
+            [float(live), float(live and self.require_backward_grad_sync)], device=dev
+        )
+        dist.all_reduce(flag, group=self.process_group)
+        return int(flag[0].item()), int(flag[1].item())
+
+    def _shadow_step(self, syncing: bool):
+        """Mirror one live step's collectives with zero gradient
+        contributions (torch Join semantics: the average still divides by
+        the full world size)."""
+        self._sync_float_buffers()
+        if not syncing:
+            return
+        order = self._launch_order or list(range(len(self._buckets)))
+        op = dist.ReduceOp.AVG if self._use_avg else dist.ReduceOp.SUM
+        for idx in order:
+            bucket = self._buckets[idx]
+            bucket.comm_flat.zero_()
+            dist.all_reduce(bucket.comm_flat, op=op, group=self.process_group)
+
+    def join_drain(self):
+        """Called when this rank exhausts its data inside
+        `Accelerator.join_uneven_inputs`: keep shadowing other ranks'
+        steps until every rank has joined."""
+        if self._world_size <= 1 or not dist.is_initialized():
+            return
+        while True:
+            n_live, n_sync = self._join_flag_round(0)
+            if n_live == 0:
+                break
+            self._shadow_step(syncing=n_sync > 0)
+        # final sync (torch Join's DDP post-hook): the AUTHORITATIVE rank —
+        # the one that ran the most steps (lowest rank breaks ties) —
+        # broadcasts its parameters/buffers so joined ranks catch up on the
+        # optimizer updates they shadowed but never applied
+        dev = next(self.module.parameters()).device
+        steps = torch.tensor([float(self._join_steps)], device=dev)
+        dist.all_reduce(steps, op=dist.ReduceOp.MAX, group=self.process_group)
+        me = dist.get_rank(self.process_group)
+        cand = torch.tensor(
+            [float(me if self._join_steps == int(steps.item()) else self._world_size)], device=dev
+        )
+        dist.all_reduce(cand, op=dist.ReduceOp.MIN, group=self.process_group)
+        src_rank = dist.get_global_rank(self.process_group, int(cand.item())) if self.process_group else int(cand.item())
+        for t in list(self.module.parameters()) + list(self.module.buffers()):
+            if t.numel() > 0 and not getattr(t, "_no_ddp_sync", False):
+                dist.broadcast(t.data, src=src_rank, group=self.process_group)
+        self._join_steps = 0
+
+    def _sync_float_buffers(self):
+        if not (self.broadcast_buffers and self._world_size > 1 and self.module.training):
+            return
+        if self._float_buffers is None:
+            # non-persistent buffers (absent from state_dict) are derived
+            # constants (e.g. RoPE tables) — identical by construction,
+            # never broadcast
+            persistent = set(self.module.state_dict(keep_vars=True))
+            self._float_buffers = [
+                b
+                for n, b in self.module.named_buffers()
+                if n in persistent and b.is_floating_point() and b.numel() > 0
+            ]
             if self._float_buffers:
-                flat = self._buffer_flat
-                if dist.get_rank(self.process_group) == 0:
-                    off = 0
-                    for b in self._float_buffers:
-                        flat[off : off + b.numel()].copy_(b.reshape(-1), non_blocking=True)
-                        off += b.numel()
-                dist.broadcast(flat, src=self._src_rank, group=self.process_group)
-                if dist.get_rank(self.process_group) != 0:
-                    off = 0
-                    for b in self._float_buffers:
-                        b.copy_(flat[off : off + b.numel()].view_as(b), non_blocking=True)
-                        off += b.numel()
+                total = sum(b.numel() for b in self._float_buffers)
+                # fp32 wire is lossless for bf16/fp16/fp32 buffers; widen
+                # to fp64 only if any buffer needs it
+                wire = (
+                    torch.float64
+                    if any(b.dtype == torch.float64 for b in self._float_buffers)
+                    else torch.float32
+                )
+                self._buffer_flat = torch.empty(total, dtype=wire, device=self._float_buffers[0].device)
+        if self._float_buffers:
+            flat = self._buffer_flat
+            if dist.get_rank(self.process_group) == 0:
+                off = 0
+                for b in self._float_buffers:
+                    flat[off : off + b.numel()].copy_(b.reshape(-1), non_blocking=True)
+                    off += b.numel()
+            dist.broadcast(flat, src=self._src_rank, group=self.process_group)
+            if dist.get_rank(self.process_group) != 0:
+                off = 0
+                for b in self._float_buffers:
+                    b.copy_(flat[off : off + b.numel()].view_as(b), non_blocking=True)
+                    off += b.numel()
+
+    def forward(self, *args, **kwargs):
+        if self._join_active and self._world_size > 1:
+            self._join_steps += 1
+            self._join_flag_round(1)
+        # per-iteration sync only for mutable float buffers (BN running
+        # stats) in ONE coalesced broadcast; constant/int and non-persistent
+        # buffers were handled at wrap time (see _sync_float_buffers)
+        self._sync_float_buffers()
         return self.module(*args, **kwargs)
 
     @contextmanager

