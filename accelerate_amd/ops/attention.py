@@ -48,6 +48,108 @@ def _fused_eligible(q, k, v) -> bool:
     )
 
 
+def _fwd_with_lse(q, k, v, causal, scale, past, q_block=1024, k_block=1024):
+    """Per-chunk forward returning (normalized out, NATURAL-log lse).
+    `past` = absolute query offset minus absolute key offset (query i
+    attends keys <= past + i). Shared by the autograd Function and the
+    ring-CP engine (parallel/ring.py)."""
+    B, H, Sq, Dh = q.shape
+    Sk = k.shape[2]
+
+    if _fused_eligible(q, k, v):
+        from . import _load_extension
+
+        ext = _load_extension(required=True)  # GPU boxes must run native
+        out, lse = ext.flash_attn_fwd(q, k, v, causal, scale, past)
+        return out, lse
+
+    # torch path computes with kv expanded to Hq heads
+    if k.shape[1] != H:
+        rep = H // k.shape[1]
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    out = torch.empty_like(q)
+    lse = torch.empty(B, H, Sq, dtype=torch.float32, device=q.device)
+    for q0 in range(0, Sq, q_block):
+        q1 = min(q0 + q_block, Sq)
+        qb = q[:, :, q0:q1]
+        acc = torch.zeros(B, H, q1 - q0, Dh, dtype=torch.float32, device=q.device)
+        m = torch.full((B, H, q1 - q0), -float("inf"), dtype=torch.float32, device=q.device)
+        denom = torch.zeros(B, H, q1 - q0, dtype=torch.float32, device=q.device)
+        k_hi = Sk if not causal else min(Sk, past + q1)
+        for k0 in range(0, k_hi, k_block):
+            k1 = min(k0 + k_block, k_hi)
+            s = torch.matmul(qb, k[:, :, k0:k1].transpose(-1, -2)).float() * scale
+            if causal and k1 > past + q0:
+                qi = torch.arange(q0, q1, device=q.device)[:, None]
+                ki = torch.arange(k0, k1, device=q.device)[None, :]
+                s = s.masked_fill(ki > past + qi, -float("inf"))
+            blk_max = s.amax(-1)
+            new_m = torch.maximum(m, blk_max)
+            corr = torch.exp(m - new_m)
+            p = torch.exp(s - new_m[..., None])
+            acc = acc * corr[..., None] + torch.matmul(p.to(v.dtype), v[:, :, k0:k1]).float()
+            denom = denom * corr + p.sum(-1)
+            m = new_m
+        out[:, :, q0:q1] = (acc / denom[..., None].clamp_min(1e-30)).to(q.dtype)
+        lse[:, :, q0:q1] = m + denom.clamp_min(1e-30).log()
+    return out, lse
+
+
+def _bwd_chunk(dout, q, k, v, out, lse, causal, scale, past, q_block=1024, k_block=1024):
+    """Per-chunk backward given the GLOBAL (merged) out/lse — returns
+    (dq_partial fp32, dk_chunk fp32, dv_chunk fp32). With a global lse the
+    chunk probabilities are exp(S - lse): partial rows of the full softmax,
+    which is exactly what ring attention accumulates per rotation step."""
+    B, H, Sq, Dh = q.shape
+    Hkv = k.shape[1]
+    rep = H // Hkv
+
+    if _fused_eligible(q, k, v) and os.environ.get("ACCELERATE_AMD_FA_BWD", "1") == "1":
+        from . import _load_extension
+
+        ext = _load_extension(required=True)
+        dq, dk, dv = ext.flash_attn_bwd(dout, q, k, v, out, lse, causal, scale, past)
+        if rep > 1:
+            dk = dk.view(B, Hkv, rep, *dk.shape[2:]).float().sum(2)
+            dv = dv.view(B, Hkv, rep, *dv.shape[2:]).float().sum(2)
+        return dq.float(), dk.float(), dv.float()
+    if rep > 1:
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    Sk = k.shape[2]
+    dq = torch.zeros_like(q, dtype=torch.float32)
+    dk = torch.zeros_like(k, dtype=torch.float32)
+    dv = torch.zeros_like(v, dtype=torch.float32)
+    Drow = (dout.float() * out.float()).sum(-1)  # [B,H,Sq]
+    for q0 in range(0, Sq, q_block):
+        q1 = min(q0 + q_block, Sq)
+        qb = q[:, :, q0:q1]
+        dob = dout[:, :, q0:q1]
+        Lb = lse[:, :, q0:q1]
+        Db = Drow[:, :, q0:q1]
+        k_hi = Sk if not causal else min(Sk, past + q1)
+        for k0 in range(0, k_hi, k_block):
+            k1 = min(k0 + k_block, k_hi)
+            kb, vb = k[:, :, k0:k1], v[:, :, k0:k1]
+            s = torch.matmul(qb, kb.transpose(-1, -2)).float() * scale
+            if causal and k1 > past + q0:
+                qi = torch.arange(q0, q1, device=q.device)[:, None]
+                ki = torch.arange(k0, k1, device=q.device)[None, :]
+                s = s.masked_fill(ki > past + qi, -float("inf"))
+            p = torch.exp(s - Lb[..., None])
+            pb = p.to(q.dtype)
+            dv[:, :, k0:k1] += torch.matmul(pb.transpose(-1, -2), dob).float()
+            dp = torch.matmul(dob, vb.transpose(-1, -2)).float()
+            ds = (p * (dp - Db[..., None]) * scale).to(q.dtype)
+            dq[:, :, q0:q1] += torch.matmul(ds, kb).float()
+            dk[:, :, k0:k1] += torch.matmul(ds.transpose(-1, -2), qb).float()
+    if rep > 1:
+        dk = dk.view(B, Hkv, rep, Sk, Dh).sum(2)
+        dv = dv.view(B, Hkv, rep, Sk, Dh).sum(2)
+    return dq, dk, dv
+
+
 class _FlashAttentionFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale, q_block, k_block, q_start=None):
@@ -205,9 +307,10 @@ _SEQ_PARALLEL = {"mode": None, "group": None}
 
 
 def set_sequence_parallel(mode: Optional[str], group=None):
-    """mode: None (off) | 'allgather' (CP: KV all-gather) | 'ulysses'
-    (SP: dual all-to-all head resharding)."""
-    if mode not in (None, "allgather", "ulysses"):
+    """mode: None (off) | 'allgather' (CP: KV all-gather) | 'ring' (CP:
+    P2P KV rotation, sequence-local memory) | 'ulysses' (SP: dual
+    all-to-all head resharding)."""
+    if mode not in (None, "allgather", "ring", "ulysses"):
         raise ValueError(f"unknown sequence-parallel mode {mode!r}")
     _SEQ_PARALLEL["mode"] = mode
     _SEQ_PARALLEL["group"] = group
@@ -244,6 +347,10 @@ def dispatch_attention(q, k, v, causal: bool = True):
         from ..parallel.cp import context_parallel_attention
 
         return context_parallel_attention(q, k, v, group=group, causal=causal)
+    if mode == "ring":
+        from ..parallel.cp import ring_attention
+
+        return ring_attention(q, k, v, group=group, causal=causal)
     from ..parallel.sp import ulysses_attention
 
     return ulysses_attention(q, k, v, group=group, causal=causal)

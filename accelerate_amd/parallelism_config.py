@@ -36,6 +36,7 @@ class ParallelismConfig:
     ep_size: Optional[int] = None
     # sequence-parallel collective pattern for the cp dimension:
     # 'allgather' = KV all-gather CP (reference torch CP default rotate),
+    # 'ring'      = P2P KV rotation (sequence-local memory, xGMI send/recv),
     # 'ulysses'   = dual all-to-all head resharding (reference DeepSpeed SP);
     # the reference's 'alltoall' rotate spelling is accepted as 'ulysses'
     cp_impl: Optional[str] = None
@@ -71,8 +72,10 @@ class ParallelismConfig:
         return self.dp_replicate_size * self.dp_shard_size
 
     def validate(self, world_size: int):
-        if self.cp_impl not in ("allgather", "ulysses"):
-            raise ValueError(f"cp_impl must be 'allgather' or 'ulysses', got {self.cp_impl!r}")
+        if self.cp_impl not in ("allgather", "ring", "ulysses"):
+            raise ValueError(
+                f"cp_impl must be 'allgather', 'ring' or 'ulysses', got {self.cp_impl!r}"
+            )
         if self.total_size != world_size:
             raise ValueError(
                 f"ParallelismConfig total_size ({self.total_size}) != world size ({world_size}); "

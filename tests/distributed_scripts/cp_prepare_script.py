@@ -3,7 +3,8 @@
 NO model-specific patching — models hit the registered collective pattern via
 `ops.attention.dispatch_attention` (VERDICT round-1 item 4).
 
-Matrix: Llama (GQA) x {allgather CP, ulysses SP} and GPT-2 x allgather.
+Matrix: Llama (GQA) x {allgather CP, ring CP, ulysses SP} and GPT-2 x
+allgather.
 Checks per case:
 - forward parity: local-shard logits == the reference full-sequence logits
   slice for this rank
@@ -73,6 +74,7 @@ def run_case(name, make_model, vocab, cp_impl):
 
 def main():
     run_case("LLAMA", lambda: LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2)), 1024, "allgather")
+    run_case("LLAMA", lambda: LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2)), 1024, "ring")
     run_case("LLAMA", lambda: LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2)), 1024, "ulysses")
     run_case("GPT2", lambda: GPT2LMHeadModel(GPT2Config.tiny()), 1024, "allgather")
     dist.barrier()

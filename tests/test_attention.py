@@ -282,3 +282,75 @@ def test_fused_kernel_gqa_and_strided_views(Hq, Hkv, D):
     for a, b, name in ((q.grad.float(), q2.grad.float(), "dq"), (k.grad.float(), ref_dk, "dk"), (v.grad.float(), ref_dv, "dv")):
         rel = (a - b).abs().max() / (b.abs().max() + 1e-6)
         assert rel < 0.04, f"{name}: {rel}"
+
+
+def _simulated_ring(q, k, v, causal, scale, world):
+    """Run the ring-CP chunk math single-process: per-rank partial attention
+    via `_fwd_with_lse` + lse merge, backward via `_bwd_chunk` with the
+    GLOBAL merged lse — exactly what `parallel.cp._RingAttention` executes,
+    minus the P2P shifts. Returns (out, dq, dk, dv) for rank r slices glued
+    back into full tensors."""
+    from accelerate_amd.ops.attention import _bwd_chunk, _fwd_with_lse
+    from accelerate_amd.parallel.cp import _merge_partial
+
+    B, H, S, D = q.shape
+    s = S // world
+    outs, dqs = [], []
+    dk = torch.zeros_like(k, dtype=torch.float32)
+    dv = torch.zeros_like(v, dtype=torch.float32)
+    dout = torch.ones_like(q)
+    for r in range(world):
+        qr = q[:, :, r * s : (r + 1) * s].contiguous()
+        out, lse = None, None
+        for c in range(world):
+            offset = (r - c) * s
+            if causal and offset < 0:
+                continue
+            o_i, l_i = _fwd_with_lse(
+                qr, k[:, :, c * s : (c + 1) * s].contiguous(),
+                v[:, :, c * s : (c + 1) * s].contiguous(), causal, scale, offset,
+            )
+            out, lse = _merge_partial(out, lse, o_i, l_i)
+        out = out.to(q.dtype)
+        outs.append(out)
+        dq = torch.zeros_like(qr, dtype=torch.float32)
+        for c in range(world):
+            offset = (r - c) * s
+            if causal and offset < 0:
+                continue
+            dq_i, dk_i, dv_i = _bwd_chunk(
+                dout[:, :, r * s : (r + 1) * s].contiguous(), qr,
+                k[:, :, c * s : (c + 1) * s].contiguous(),
+                v[:, :, c * s : (c + 1) * s].contiguous(), out, lse, causal, scale, offset,
+            )
+            dq += dq_i
+            dk[:, :, c * s : (c + 1) * s] += dk_i
+            dv[:, :, c * s : (c + 1) * s] += dv_i
+        dqs.append(dq)
+    return torch.cat(outs, dim=2), torch.cat(dqs, dim=2), dk, dv
+
+
+@gpu
+@pytest.mark.parametrize("causal", [False, True])
+def test_ring_chunk_math_matches_native_full(causal):
+    """The ring-CP partial-attention math (native fwd kernel + lse merge,
+    native bwd kernel with global lse) reproduces full-sequence flash
+    attention on GPU — validates parallel/cp.py's ring against the HIP
+    kernels without needing multi-rank."""
+    from accelerate_amd.ops.attention import flash_attention
+
+    torch.manual_seed(0)
+    B, H, S, D = 2, 4, 512, 128
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    scale = 1.0 / math.sqrt(D)
+
+    ref = flash_attention(q, k, v, causal=causal)
+    ref.sum().backward()
+
+    out, dq, dk, dv = _simulated_ring(q.detach(), k.detach(), v.detach(), causal, scale, world=4)
+    assert torch.allclose(out.float(), ref.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(dq, q.grad.float(), atol=8e-2, rtol=5e-2)
+    assert torch.allclose(dk, k.grad.float(), atol=8e-2, rtol=5e-2)
+    assert torch.allclose(dv, v.grad.float(), atol=8e-2, rtol=5e-2)

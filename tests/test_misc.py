@@ -188,5 +188,15 @@ def test_cp_through_prepare_2proc():
 
     out = launch_distributed("tests/distributed_scripts/cp_prepare_script.py", nproc=2, timeout=300)
     assert "CP_PREPARE_LLAMA_ALLGATHER_PASS" in out
+    assert "CP_PREPARE_LLAMA_RING_PASS" in out
     assert "CP_PREPARE_LLAMA_ULYSSES_PASS" in out
     assert "CP_PREPARE_GPT2_ALLGATHER_PASS" in out
+
+
+def test_ring_cp_2proc():
+    """Ring (P2P KV rotation) attention: fwd+bwd parity vs full-sequence
+    flash attention, incl. causal, non-causal, GQA, and dk/dv homing."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/ring_script.py", nproc=2, timeout=300)
+    assert "RING_CP_PASS" in out
