@@ -220,12 +220,13 @@ void unscale_and_check(std::vector<at::Tensor> grads, at::Tensor inv_scale, at::
 // norm kernels (norm_kernels.hip)
 __global__ void layernorm_fwd_bf16(const ushort*, const ushort*, const ushort*, ushort*,
                                    float*, float*, int64_t, int, float);
-__global__ void layernorm_bwd_bf16(const ushort*, const ushort*, const ushort*,
-                                   const float*, const float*, ushort*, float*, float*, int64_t, int);
+extern "C" hipError_t launch_layernorm_bwd(const void*, const void*, const void*, const void*,
+                                           const void*, void*, void*, void*, long long, int, int,
+                                           hipStream_t);
 __global__ void norm_fold_partials(const float*, const float*, ushort*, ushort*, int, int);
 __global__ void rmsnorm_fwd_bf16(const ushort*, const ushort*, ushort*, float*, int64_t, int, float);
-__global__ void rmsnorm_bwd_bf16(const ushort*, const ushort*, const ushort*, const float*,
-                                 ushort*, float*, int64_t, int);
+extern "C" hipError_t launch_rmsnorm_bwd(const void*, const void*, const void*, const void*,
+                                         void*, void*, long long, int, int, hipStream_t);
 
 static const ushort* bfp(const at::Tensor& t) { return reinterpret_cast<const ushort*>(t.data_ptr()); }
 static ushort* bfp_mut(at::Tensor& t) { return reinterpret_cast<ushort*>(t.data_ptr()); }
@@ -250,19 +251,22 @@ std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   const int d = (int)x.size(-1);
   TORCH_CHECK(d <= 2048, "layernorm_bwd: fused path supports inner dim <= 2048");
   const int64_t rows = x.numel() / d;
-  const int n_blocks = (int)std::min<int64_t>(rows, 512);
+  // wave-per-row: each block runs 4 rows concurrently; cap the partial count
+  const int n_blocks = (int)std::max<int64_t>(1, std::min<int64_t>((rows + 3) / 4, 1024));
+  const int n_partials = n_blocks * 4;
   auto dx = at::empty_like(x);
-  auto dw_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
-  auto db_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
+  auto dw_partial = at::empty({n_partials, d}, x.options().dtype(at::kFloat));
+  auto db_partial = at::empty({n_partials, d}, x.options().dtype(at::kFloat));
   auto dw = at::empty({d}, x.options());
   auto db = at::empty({d}, x.options());
   auto stream = at::hip::getCurrentHIPStream();
   auto dyc = dy.contiguous();
-  hipLaunchKernelGGL(layernorm_bwd_bf16, dim3(n_blocks), dim3(64), 0, stream.stream(),
-                     bfp(dyc), bfp(x), bfp(w), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     bfp_mut(dx), dw_partial.data_ptr<float>(), db_partial.data_ptr<float>(), rows, d);
+  auto err = launch_layernorm_bwd(bfp(dyc), bfp(x), bfp(w), mean.data_ptr<float>(),
+                                  rstd.data_ptr<float>(), bfp_mut(dx), dw_partial.data_ptr<float>(),
+                                  db_partial.data_ptr<float>(), rows, d, n_blocks, stream.stream());
+  TORCH_CHECK(err == hipSuccess, "layernorm_bwd: ", hipGetErrorString(err));
   hipLaunchKernelGGL(norm_fold_partials, dim3((d + 255) / 256), dim3(256), 0, stream.stream(),
-                     dw_partial.data_ptr<float>(), db_partial.data_ptr<float>(), bfp_mut(dw), bfp_mut(db), d, n_blocks);
+                     dw_partial.data_ptr<float>(), db_partial.data_ptr<float>(), bfp_mut(dw), bfp_mut(db), d, n_partials);
   return {dx, dw, db};
 }
 
@@ -283,17 +287,18 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, a
   const int d = (int)x.size(-1);
   TORCH_CHECK(d <= 8192, "rmsnorm_bwd: fused path supports inner dim <= 8192");
   const int64_t rows = x.numel() / d;
-  const int n_blocks = (int)std::min<int64_t>(rows, 512);
+  const int n_blocks = (int)std::max<int64_t>(1, std::min<int64_t>((rows + 3) / 4, 1024));
+  const int n_partials = n_blocks * 4;
   auto dx = at::empty_like(x);
-  auto dw_partial = at::empty({n_blocks, d}, x.options().dtype(at::kFloat));
+  auto dw_partial = at::empty({n_partials, d}, x.options().dtype(at::kFloat));
   auto dw = at::empty({d}, x.options());
   auto stream = at::hip::getCurrentHIPStream();
   auto dyc = dy.contiguous();
-  hipLaunchKernelGGL(rmsnorm_bwd_bf16, dim3(n_blocks), dim3(64), 0, stream.stream(),
-                     bfp(dyc), bfp(x), bfp(w), rstd.data_ptr<float>(), bfp_mut(dx),
-                     dw_partial.data_ptr<float>(), rows, d);
+  auto err = launch_rmsnorm_bwd(bfp(dyc), bfp(x), bfp(w), rstd.data_ptr<float>(), bfp_mut(dx),
+                                dw_partial.data_ptr<float>(), rows, d, n_blocks, stream.stream());
+  TORCH_CHECK(err == hipSuccess, "rmsnorm_bwd: ", hipGetErrorString(err));
   hipLaunchKernelGGL(norm_fold_partials, dim3((d + 255) / 256), dim3(256), 0, stream.stream(),
-                     dw_partial.data_ptr<float>(), nullptr, bfp_mut(dw), nullptr, d, n_blocks);
+                     dw_partial.data_ptr<float>(), nullptr, bfp_mut(dw), nullptr, d, n_partials);
   return {dx, dw};
 }
 

@@ -1,10 +1,12 @@
 // Fused LayerNorm / RMSNorm for CDNA4, bf16 in/out with fp32 statistics.
 //
-// One wave (64 lanes) per row; 16 B/lane vectorized loads (Guideline 13).
-// Row data is re-read in the normalize pass — a 1.5 KB row is L1-resident,
-// so the second pass costs L1 bandwidth, not HBM. Backward reduces dweight/
-// dbias through 64 fp32 partial rows (atomic contention 1/64 of direct
-// atomics — Guideline 12) + a small fold kernel.
+// Forward: one wave (64 lanes) per row; 16 B/lane vectorized loads
+// (Guideline 13); the 1.5 KB row is L1-resident for the normalize pass.
+// Backward: 256-thread blocks run 4 rows wave-per-row (the one-wave-per-
+// block shape left half the 1024 SIMDs idle); small d keeps dy/x in
+// registers for a single HBM pass; dweight/dbias accumulate in per-lane
+// register slots -> one non-atomic fp32 partial row per wave -> fold
+// kernel. d > 2048 (RMS) splits dx from a column-tiled dw kernel.
 
 #include "multi_tensor.h"
 
@@ -95,76 +97,88 @@ __global__ void layernorm_fwd_bf16(const ushort* __restrict__ x, const ushort* _
 //   dw_partial[row%64] += dy * xhat ; db_partial[row%64] += dy
 // ---------------------------------------------------------------------------
 
-// grid-stride over rows: each block accumulates its rows' dw/db in REGISTERS
-// (per-lane column slots), then writes ONE non-atomic partial row. No atomics
-// anywhere; a fold kernel sums the per-block partials.
-// MAX_COLS_PER_LANE bounds d at 64*8*4 = 2048 for LN (BERT 768 fits).
-#define LN_MAX_ITERS 4
-
-__global__ void layernorm_bwd_bf16(const ushort* __restrict__ dy, const ushort* __restrict__ x,
-                                   const ushort* __restrict__ w,
-                                   const float* __restrict__ mean, const float* __restrict__ rstd,
-                                   ushort* __restrict__ dx,
-                                   float* __restrict__ dw_partial, float* __restrict__ db_partial,
-                                   int64_t n_rows, int d) {
-  const int lane = threadIdx.x;
-  float accw[LN_MAX_ITERS][8];
-  float accb[LN_MAX_ITERS][8];
+// 256-thread blocks, one WAVE per row (4 rows in flight per block): the
+// old one-wave-per-block shape put only 512 waves on a 1024-SIMD chip.
+// Row data (dy, x) loaded once into registers when d <= ITERS*512 fits
+// (KEEP=true), so stats + dx are a single HBM pass; dweight/dbias
+// accumulate in per-lane register slots and each wave writes ONE
+// non-atomic partial row folded by norm_fold_partials.
+template <int ITERS, bool KEEP>
+__global__ __launch_bounds__(256) void layernorm_bwd_t(
+    const ushort* __restrict__ dy, const ushort* __restrict__ x,
+    const ushort* __restrict__ w,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    ushort* __restrict__ dx,
+    float* __restrict__ dw_partial, float* __restrict__ db_partial,
+    int64_t n_rows, int d) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  float accw[ITERS][8];
+  float accb[ITERS][8];
 #pragma unroll
-  for (int i = 0; i < LN_MAX_ITERS; ++i)
+  for (int i = 0; i < ITERS; ++i)
 #pragma unroll
     for (int k = 0; k < 8; ++k) { accw[i][k] = 0.f; accb[i][k] = 0.f; }
 
-  for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
+  float xv[KEEP ? ITERS : 1][8];
+  float gv[KEEP ? ITERS : 1][8];
+
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < n_rows; row += (int64_t)gridDim.x * 4) {
     const ushort* dyr = dy + row * d;
     const ushort* xr = x + row * d;
     ushort* dxr = dx + row * d;
     const float mu = mean[row], rs = rstd[row];
 
+    // pass 1: row stats. KEEP stashes RAW dy and x in registers; w is
+    // re-read in pass 2 (shared across all rows — L1-resident).
     float s1 = 0.f, s2 = 0.f;
-    for (int base = lane * 8; base < d; base += 64 * 8) {
-      ushort8 gv = *reinterpret_cast<const ushort8*>(dyr + base);
-      ushort8 xv8 = *reinterpret_cast<const ushort8*>(xr + base);
-      ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
+    int it = 0;
+    for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
+      ushort8 g8 = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 x8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = bf2f(gv[k]);
-        float xv = bf2f(xv8[k]);
-        float wk = bf2f(wv[k]);
-        float xhat = (xv - mu) * rs;
-        float gw = g * wk;
+        float g = bf2f(g8[k]);
+        float v = bf2f(x8[k]);
+        if (KEEP) { gv[KEEP ? it : 0][k] = g; xv[KEEP ? it : 0][k] = v; }
+        float gw = g * bf2f(w8[k]);
         s1 += gw;
-        s2 += gw * xhat;
+        s2 += gw * (v - mu) * rs;
       }
     }
     s1 = wave_sum(s1) / d;
     s2 = wave_sum(s2) / d;
 
-    int it = 0;
+    it = 0;
     for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
-      ushort8 gv = *reinterpret_cast<const ushort8*>(dyr + base);
-      ushort8 xv8 = *reinterpret_cast<const ushort8*>(xr + base);
-      ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
+      float gk[8], vk[8];
+      if (!KEEP) {
+        ushort8 g8 = *reinterpret_cast<const ushort8*>(dyr + base);
+        ushort8 x8 = *reinterpret_cast<const ushort8*>(xr + base);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) { gk[k] = bf2f(g8[k]); vk[k] = bf2f(x8[k]); }
+      } else {
+#pragma unroll
+        for (int k = 0; k < 8; ++k) { gk[k] = gv[KEEP ? it : 0][k]; vk[k] = xv[KEEP ? it : 0][k]; }
+      }
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
       ushort8 ov;
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = bf2f(gv[k]);
-        float xv = bf2f(xv8[k]);
-        float wk = bf2f(wv[k]);
-        float xhat = (xv - mu) * rs;
-        float dxv = rs * (g * wk - s1 - xhat * s2);
-        ushort r = f2bf(dxv);
-        ov[k] = r;
-        accw[it][k] += g * xhat;
-        accb[it][k] += g;
+        float xhat = (vk[k] - mu) * rs;
+        ov[k] = f2bf(rs * (gk[k] * bf2f(w8[k]) - s1 - xhat * s2));
+        accw[it][k] += gk[k] * xhat;
+        accb[it][k] += gk[k];
       }
       *reinterpret_cast<ushort8*>(dxr + base) = ov;
     }
   }
 
-  // one partial row per block, non-atomic
-  float* dwp = dw_partial + blockIdx.x * (int64_t)d;
-  float* dbp = db_partial + blockIdx.x * (int64_t)d;
+  // one partial row per WAVE, non-atomic
+  const int64_t pidx = (int64_t)blockIdx.x * 4 + wave;
+  float* dwp = dw_partial + pidx * d;
+  float* dbp = db_partial + pidx * d;
   int it = 0;
   for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
 #pragma unroll
@@ -173,6 +187,26 @@ __global__ void layernorm_bwd_bf16(const ushort* __restrict__ dy, const ushort* 
       dbp[base + k] = accb[it][k];
     }
   }
+}
+
+// host-side dispatch over the templated iteration counts (d <= ITERS*512)
+extern "C" hipError_t launch_layernorm_bwd(const void* dy, const void* x, const void* w,
+                                           const void* mean, const void* rstd, void* dx,
+                                           void* dw_partial, void* db_partial,
+                                           long long n_rows, int d, int n_blocks,
+                                           hipStream_t stream) {
+  dim3 g(n_blocks), b(256);
+#define LN_CASE(N, KEEP) \
+  hipLaunchKernelGGL((layernorm_bwd_t<N, KEEP>), g, b, 0, stream, \
+                     (const ushort*)dy, (const ushort*)x, (const ushort*)w, \
+                     (const float*)mean, (const float*)rstd, (ushort*)dx, \
+                     (float*)dw_partial, (float*)db_partial, (int64_t)n_rows, d)
+  if (d <= 512) LN_CASE(1, true);
+  else if (d <= 1024) LN_CASE(2, true);
+  else if (d <= 2048) LN_CASE(4, false);
+  else return hipErrorInvalidValue;
+#undef LN_CASE
+  return hipGetLastError();
 }
 
 // fold `n_partials` partial rows into bf16 dw/db
@@ -230,67 +264,178 @@ __global__ void rmsnorm_fwd_bf16(const ushort* __restrict__ x, const ushort* __r
   }
 }
 
-// grid-stride rows + register dw accumulation (see layernorm_bwd).
-// RMS_MAX_ITERS=16 bounds d at 64*8*16 = 8192 (Llama-70B hidden).
-#define RMS_MAX_ITERS 16
-
-__global__ void rmsnorm_bwd_bf16(const ushort* __restrict__ dy, const ushort* __restrict__ x,
-                                 const ushort* __restrict__ w, const float* __restrict__ rstd,
-                                 ushort* __restrict__ dx, float* __restrict__ dw_partial,
-                                 int64_t n_rows, int d) {
-  const int lane = threadIdx.x;
-  const int n_iters = (d + 64 * 8 - 1) / (64 * 8);
-  float accw[RMS_MAX_ITERS][8];
-  for (int i = 0; i < n_iters; ++i)
+// 256-thread blocks, one WAVE per row (see layernorm_bwd_t). KEEP=true
+// holds dy/x in registers across the stat + dx passes (d <= ITERS*512);
+// larger d (Llama 4096/8192) re-reads the L1/L2-resident row.
+template <int ITERS, bool KEEP>
+__global__ __launch_bounds__(256) void rmsnorm_bwd_t(
+    const ushort* __restrict__ dy, const ushort* __restrict__ x,
+    const ushort* __restrict__ w, const float* __restrict__ rstd,
+    ushort* __restrict__ dx, float* __restrict__ dw_partial,
+    int64_t n_rows, int d) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  float accw[ITERS][8];
+#pragma unroll
+  for (int i = 0; i < ITERS; ++i)
 #pragma unroll
     for (int k = 0; k < 8; ++k) accw[i][k] = 0.f;
 
-  for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
+  float xv[KEEP ? ITERS : 1][8];
+  float gv[KEEP ? ITERS : 1][8];
+
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < n_rows; row += (int64_t)gridDim.x * 4) {
     const ushort* dyr = dy + row * d;
     const ushort* xr = x + row * d;
     ushort* dxr = dx + row * d;
     const float rs = rstd[row];
 
-    float s = 0.f;
-    for (int base = lane * 8; base < d; base += 64 * 8) {
-      ushort8 gv = *reinterpret_cast<const ushort8*>(dyr + base);
-      ushort8 xv8 = *reinterpret_cast<const ushort8*>(xr + base);
-      ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
-#pragma unroll
-      for (int k = 0; k < 8; ++k) {
-        float g = bf2f(gv[k]);
-        float xv = bf2f(xv8[k]);
-        float wk = bf2f(wv[k]);
-        s += g * wk * xv;
-      }
-    }
-    s = wave_sum(s);
-    const float c = s * rs * rs * rs / d;
-
+    float sum = 0.f;
     int it = 0;
     for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
-      ushort8 gv = *reinterpret_cast<const ushort8*>(dyr + base);
-      ushort8 xv8 = *reinterpret_cast<const ushort8*>(xr + base);
-      ushort8 wv = *reinterpret_cast<const ushort8*>(w + base);
+      ushort8 g8 = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 x8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = bf2f(g8[k]);
+        float v = bf2f(x8[k]);
+        if (KEEP) { gv[KEEP ? it : 0][k] = g; xv[KEEP ? it : 0][k] = v; }
+        sum += g * bf2f(w8[k]) * v;
+      }
+    }
+    sum = wave_sum(sum);
+    const float c = sum * rs * rs * rs / d;
+
+    it = 0;
+    for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
+      float gk[8], vk[8];
+      if (!KEEP) {
+        ushort8 g8 = *reinterpret_cast<const ushort8*>(dyr + base);
+        ushort8 x8 = *reinterpret_cast<const ushort8*>(xr + base);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) { gk[k] = bf2f(g8[k]); vk[k] = bf2f(x8[k]); }
+      } else {
+#pragma unroll
+        for (int k = 0; k < 8; ++k) { gk[k] = gv[KEEP ? it : 0][k]; vk[k] = xv[KEEP ? it : 0][k]; }
+      }
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
       ushort8 ov;
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        float g = bf2f(gv[k]);
-        float xv = bf2f(xv8[k]);
-        float wk = bf2f(wv[k]);
-        float dxv = rs * g * wk - xv * c;
-        ushort r = f2bf(dxv);
-        ov[k] = r;
-        accw[it][k] += g * xv * rs;
+        ov[k] = f2bf(rs * gk[k] * bf2f(w8[k]) - vk[k] * c);
+        accw[it][k] += gk[k] * vk[k] * rs;
       }
       *reinterpret_cast<ushort8*>(dxr + base) = ov;
     }
   }
 
-  float* dwp = dw_partial + blockIdx.x * (int64_t)d;
+  float* dwp = dw_partial + ((int64_t)blockIdx.x * 4 + wave) * d;
   int it = 0;
   for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
 #pragma unroll
     for (int k = 0; k < 8; ++k) dwp[base + k] = accw[it][k];
   }
+}
+
+// Large-d (> 2048) RMS backward: the fused kernel's per-lane dw slots no
+// longer fit in registers (ITERS 8/16 spilled to scratch), so split into a
+// lean dx-only kernel (stats + dx, no accumulators — occupancy 8) and a
+// column-tiled dw kernel. dweight needs NO row statistics beyond the saved
+// mean/rstd, so the dw grid is 2D: blockIdx.y picks a 512-column tile
+// (one ushort8 per lane), blockIdx.x grid-strides rows wave-per-row; each
+// (row-block, tile) writes its 512-column slice of a full-width partial.
+__global__ __launch_bounds__(256) void rmsnorm_bwd_dx(
+    const ushort* __restrict__ dy, const ushort* __restrict__ x,
+    const ushort* __restrict__ w, const float* __restrict__ rstd,
+    ushort* __restrict__ dx, int64_t n_rows, int d) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < n_rows; row += (int64_t)gridDim.x * 4) {
+    const ushort* dyr = dy + row * d;
+    const ushort* xr = x + row * d;
+    ushort* dxr = dx + row * d;
+    const float rs = rstd[row];
+    float sum = 0.f;
+    for (int base = lane * 8; base < d; base += 64 * 8) {
+      ushort8 g8 = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 x8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) sum += bf2f(g8[k]) * bf2f(w8[k]) * bf2f(x8[k]);
+    }
+    sum = wave_sum(sum);
+    const float c = sum * rs * rs * rs / d;
+    for (int base = lane * 8; base < d; base += 64 * 8) {
+      ushort8 g8 = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 x8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
+      ushort8 ov;
+#pragma unroll
+      for (int k = 0; k < 8; ++k)
+        ov[k] = f2bf(rs * bf2f(g8[k]) * bf2f(w8[k]) - bf2f(x8[k]) * c);
+      *reinterpret_cast<ushort8*>(dxr + base) = ov;
+    }
+  }
+}
+
+// dw (and optionally db) over a 512-column tile; mean == nullptr => RMS
+// semantics (xhat = x * rstd), else LayerNorm (xhat = (x - mean) * rstd).
+__global__ __launch_bounds__(256) void norm_dw_tile(
+    const ushort* __restrict__ dy, const ushort* __restrict__ x,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    float* __restrict__ dw_partial, float* __restrict__ db_partial,
+    int64_t n_rows, int d) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int base = blockIdx.y * 512 + lane * 8;
+  float accw[8], accb[8];
+#pragma unroll
+  for (int k = 0; k < 8; ++k) { accw[k] = 0.f; accb[k] = 0.f; }
+  if (base < d) {
+    for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < n_rows; row += (int64_t)gridDim.x * 4) {
+      ushort8 g8 = *reinterpret_cast<const ushort8*>(dy + row * d + base);
+      ushort8 x8 = *reinterpret_cast<const ushort8*>(x + row * d + base);
+      const float mu = mean ? mean[row] : 0.f;
+      const float rs = rstd[row];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = bf2f(g8[k]);
+        accw[k] += g * (bf2f(x8[k]) - mu) * rs;
+        accb[k] += g;
+      }
+    }
+    float* dwp = dw_partial + ((int64_t)blockIdx.x * 4 + wave) * d + base;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) dwp[k] = accw[k];
+    if (db_partial != nullptr) {
+      float* dbp = db_partial + ((int64_t)blockIdx.x * 4 + wave) * d + base;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) dbp[k] = accb[k];
+    }
+  }
+}
+
+extern "C" hipError_t launch_rmsnorm_bwd(const void* dy, const void* x, const void* w,
+                                         const void* rstd, void* dx, void* dw_partial,
+                                         long long n_rows, int d, int n_blocks,
+                                         hipStream_t stream) {
+  dim3 g(n_blocks), b(256);
+#define RMS_CASE(N, KEEP) \
+  hipLaunchKernelGGL((rmsnorm_bwd_t<N, KEEP>), g, b, 0, stream, \
+                     (const ushort*)dy, (const ushort*)x, (const ushort*)w, \
+                     (const float*)rstd, (ushort*)dx, (float*)dw_partial, (int64_t)n_rows, d)
+  if (d <= 512) RMS_CASE(1, true);
+  else if (d <= 1024) RMS_CASE(2, true);
+  else if (d <= 2048) RMS_CASE(4, false);
+  else if (d <= 8192) {
+    hipLaunchKernelGGL(rmsnorm_bwd_dx, g, b, 0, stream, (const ushort*)dy, (const ushort*)x,
+                       (const ushort*)w, (const float*)rstd, (ushort*)dx, (int64_t)n_rows, d);
+    dim3 g2(n_blocks, (d + 511) / 512);
+    hipLaunchKernelGGL(norm_dw_tile, g2, b, 0, stream, (const ushort*)dy, (const ushort*)x,
+                       (const float*)nullptr, (const float*)rstd, (float*)dw_partial,
+                       (float*)nullptr, (int64_t)n_rows, d);
+  } else return hipErrorInvalidValue;
+#undef RMS_CASE
+  return hipGetLastError();
 }
