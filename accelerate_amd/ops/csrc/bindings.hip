@@ -223,7 +223,6 @@ __global__ void layernorm_fwd_bf16(const ushort*, const ushort*, const ushort*, 
 extern "C" hipError_t launch_layernorm_bwd(const void*, const void*, const void*, const void*,
                                            const void*, void*, void*, void*, long long, int, int,
                                            hipStream_t);
-__global__ void norm_fold_partials(const float*, const float*, ushort*, ushort*, int, int);
 __global__ void rmsnorm_fwd_bf16(const ushort*, const ushort*, ushort*, float*, int64_t, int, float);
 extern "C" hipError_t launch_rmsnorm_bwd(const void*, const void*, const void*, const void*,
                                          void*, void*, long long, int, int, hipStream_t);
@@ -246,28 +245,38 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w, at::Tensor b, 
   return {y, mean, rstd};
 }
 
+
+// wave-per-row backward: each block runs 4 rows concurrently. More blocks =
+// more waves in flight but more partial-row fold traffic; 512 (2048 waves,
+// 2/SIMD) balanced best on MI355X. ACCELERATE_AMD_NORM_BLOCKS overrides.
+static int norm_bwd_blocks(int64_t rows) {
+  static int cap = []() {
+    const char* e = getenv("ACCELERATE_AMD_NORM_BLOCKS");
+    return e ? atoi(e) : 512;
+  }();
+  return (int)std::max<int64_t>(1, std::min<int64_t>((rows + 3) / 4, cap));
+}
+
 std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
                                       at::Tensor mean, at::Tensor rstd) {
   const int d = (int)x.size(-1);
   TORCH_CHECK(d <= 2048, "layernorm_bwd: fused path supports inner dim <= 2048");
   const int64_t rows = x.numel() / d;
-  // wave-per-row: each block runs 4 rows concurrently; cap the partial count
-  const int n_blocks = (int)std::max<int64_t>(1, std::min<int64_t>((rows + 3) / 4, 1024));
-  const int n_partials = n_blocks * 4;
+  const int n_blocks = norm_bwd_blocks(rows);
+  const int n_partials = n_blocks * 4;  // one partial row per wave
   auto dx = at::empty_like(x);
-  auto dw_partial = at::empty({n_partials, d}, x.options().dtype(at::kFloat));
-  auto db_partial = at::empty({n_partials, d}, x.options().dtype(at::kFloat));
-  auto dw = at::empty({d}, x.options());
-  auto db = at::empty({d}, x.options());
+  // dw and db interleave in ONE [P][2d] buffer -> one fold launch
+  auto dwdb_partial = at::empty({n_partials, 2 * (int64_t)d}, x.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   auto dyc = dy.contiguous();
   auto err = launch_layernorm_bwd(bfp(dyc), bfp(x), bfp(w), mean.data_ptr<float>(),
-                                  rstd.data_ptr<float>(), bfp_mut(dx), dw_partial.data_ptr<float>(),
-                                  db_partial.data_ptr<float>(), rows, d, n_blocks, stream.stream());
+                                  rstd.data_ptr<float>(), bfp_mut(dx), dwdb_partial.data_ptr<float>(),
+                                  nullptr, rows, d, n_blocks, stream.stream());
   TORCH_CHECK(err == hipSuccess, "layernorm_bwd: ", hipGetErrorString(err));
-  hipLaunchKernelGGL(norm_fold_partials, dim3((d + 255) / 256), dim3(256), 0, stream.stream(),
-                     dw_partial.data_ptr<float>(), db_partial.data_ptr<float>(), bfp_mut(dw), bfp_mut(db), d, n_partials);
-  return {dx, dw, db};
+  // fold with torch's tree reduction (a serial-loop fold kernel was the old
+  // bottleneck: d/256 blocks looping n_partials rows, latency-bound)
+  auto folded = dwdb_partial.sum(0).to(x.scalar_type());
+  return {dx, folded.narrow(0, 0, d), folded.narrow(0, d, d)};
 }
 
 std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
@@ -287,7 +296,7 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, a
   const int d = (int)x.size(-1);
   TORCH_CHECK(d <= 8192, "rmsnorm_bwd: fused path supports inner dim <= 8192");
   const int64_t rows = x.numel() / d;
-  const int n_blocks = (int)std::max<int64_t>(1, std::min<int64_t>((rows + 3) / 4, 1024));
+  const int n_blocks = norm_bwd_blocks(rows);
   const int n_partials = n_blocks * 4;
   auto dx = at::empty_like(x);
   auto dw_partial = at::empty({n_partials, d}, x.options().dtype(at::kFloat));
@@ -297,8 +306,7 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, a
   auto err = launch_rmsnorm_bwd(bfp(dyc), bfp(x), bfp(w), rstd.data_ptr<float>(), bfp_mut(dx),
                                 dw_partial.data_ptr<float>(), rows, d, n_blocks, stream.stream());
   TORCH_CHECK(err == hipSuccess, "rmsnorm_bwd: ", hipGetErrorString(err));
-  hipLaunchKernelGGL(norm_fold_partials, dim3((d + 255) / 256), dim3(256), 0, stream.stream(),
-                     dw_partial.data_ptr<float>(), nullptr, bfp_mut(dw), nullptr, d, n_partials);
+  dw.copy_(dw_partial.sum(0));
   return {dx, dw};
 }
 

@@ -102,7 +102,7 @@ __global__ void layernorm_fwd_bf16(const ushort* __restrict__ x, const ushort* _
 // Row data (dy, x) loaded once into registers when d <= ITERS*512 fits
 // (KEEP=true), so stats + dx are a single HBM pass; dweight/dbias
 // accumulate in per-lane register slots and each wave writes ONE
-// non-atomic partial row folded by norm_fold_partials.
+// non-atomic partial row; the host folds partials with torch sum(0).
 template <int ITERS, bool KEEP>
 __global__ __launch_bounds__(256) void layernorm_bwd_t(
     const ushort* __restrict__ dy, const ushort* __restrict__ x,
@@ -175,16 +175,16 @@ __global__ __launch_bounds__(256) void layernorm_bwd_t(
     }
   }
 
-  // one partial row per WAVE, non-atomic
+  // one partial row per WAVE, non-atomic; dw and db share one [P][2d]
+  // buffer so the host folds both with a single sum(0) launch
   const int64_t pidx = (int64_t)blockIdx.x * 4 + wave;
-  float* dwp = dw_partial + pidx * d;
-  float* dbp = db_partial + pidx * d;
+  float* dwp = dw_partial + pidx * (2 * (int64_t)d);
   int it = 0;
   for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
       dwp[base + k] = accw[it][k];
-      dbp[base + k] = accb[it][k];
+      dwp[d + base + k] = accb[it][k];
     }
   }
 }
@@ -207,20 +207,6 @@ extern "C" hipError_t launch_layernorm_bwd(const void* dy, const void* x, const 
   else return hipErrorInvalidValue;
 #undef LN_CASE
   return hipGetLastError();
-}
-
-// fold `n_partials` partial rows into bf16 dw/db
-__global__ void norm_fold_partials(const float* __restrict__ dw_partial, const float* __restrict__ db_partial,
-                                   ushort* __restrict__ dw, ushort* __restrict__ db, int d, int n_partials) {
-  const int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= d) return;
-  float sw = 0.f, sb = 0.f;
-  for (int r = 0; r < n_partials; ++r) {
-    sw += dw_partial[(int64_t)r * d + col];
-    if (db_partial != nullptr) sb += db_partial[(int64_t)r * d + col];
-  }
-  dw[col] = f2bf(sw);
-  if (db != nullptr) db[col] = f2bf(sb);
 }
 
 // ---------------------------------------------------------------------------
