@@ -21,6 +21,8 @@
 // l (denominator) live per (row) in 4 registers per lane.
 
 #include <hip/hip_runtime.h>
+#include <cstdlib>
+#include <cstring>
 #include <hip/hip_bf16.h>
 
 #include "attention.h"
@@ -273,6 +275,256 @@ __global__ __launch_bounds__(256, 2) void fa_fwd_kernel(
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Swapped-QK^T forward (CDNA4 8-warp 32x32 ladder, D=128).
+//
+// Each of 8 waves owns 32 query rows (BM=256), key tiles of BN=64.
+// S^T is computed as mfma(K, Q) on 32x32x16 MFMA so each lane holds a
+// full half-row of P for ONE query row (q = lane&31): row statistics are
+// 31 in-register fmax/adds + ONE permlane32_swap — no cross-lane shuffles
+// ladders, and P never round-trips through LDS: the bf16 pair-pack +
+// permlane32_swap relayout (verified by benchmarks/fa_swapped_probe.hip)
+// turns the C-layout scores directly into PV's A-fragments.
+// Softmax runs in the base-2 domain; defer-max (skip the O rescale while
+// the running max grows by < 8) removes most O-wide rescale passes.
+// ---------------------------------------------------------------------------
+
+namespace {
+typedef abf16 bf16x2a __attribute__((ext_vector_type(2)));
+typedef float f32x16a __attribute__((ext_vector_type(16)));
+typedef unsigned uint2a __attribute__((ext_vector_type(2)));
+
+__device__ __forceinline__ unsigned pack_bf16_pair(float lo, float hi) {
+  union { bf16x2a h; unsigned u; } cv;
+  cv.h = bf16x2a{(abf16)lo, (abf16)hi};
+  return cv.u;
+}
+}  // namespace
+
+template <int D>
+__global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
+    const abf16* __restrict__ q, const abf16* __restrict__ k,
+    const abf16* __restrict__ v, ushort* __restrict__ out,
+    float* __restrict__ lse, int Sq, int Sk, int past, int causal,
+    float scale, int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sO) {
+  static_assert(D == 128, "swapped ladder is tuned for head_dim 128");
+  constexpr int BM = 256, BN = 64;
+  // +8 padding measured BETTER than the XOR-chunk swizzle here (630 vs
+  // 591 TF/s non-causal): the b128 fragment reads are 4-way-conflict
+  // bounded either way, and padding keeps the addressing immediate-only.
+  constexpr int KP = D + 8;   // K tile row stride
+  constexpr int NP = BN + 8;  // V^T row stride
+  extern __shared__ char smem[];
+  abf16* Ks = reinterpret_cast<abf16*>(smem);  // [BN][KP]
+  abf16* VTs = Ks + BN * KP;                   // [D][NP]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int hi = lane >> 5;
+  const int col = lane & 31;     // this lane's query row within the wave tile
+  // causal blocks are launched heavy-first (largest q0 has the most key
+  // tiles): the light diagonal-top blocks then pack the scheduling tail
+  const int q0 = (causal ? (int)(gridDim.x - 1 - blockIdx.x) : (int)blockIdx.x) * BM;
+  const int64_t bh = blockIdx.y;
+  const int64_t b = bh / Hq, h = bh % Hq, hk = h / (Hq / Hkv);
+  const abf16* qb = q + b * sQ.b + h * sQ.h;
+  const abf16* kb = k + b * sK.b + hk * sK.h;
+  const abf16* vb = v + b * sV.b + hk * sV.h;
+
+  // Q as B-operand fragments, loaded once: lane holds Q row
+  // q0 + wave*32 + col, chunk t covers d = t*16 + hi*8 + 0..7
+  int qr_mine = q0 + wave * 32 + col;
+  if (qr_mine >= Sq) qr_mine = Sq - 1;  // clamp; padded rows never stored
+  bf16x8a qf[D / 16];
+#pragma unroll
+  for (int t = 0; t < D / 16; ++t)
+    qf[t] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr_mine * sQ.s + t * 16 + hi * 8);
+
+  f32x16a oacc[D / 32] = {};
+  float m_run = -INFINITY;  // base-2 scaled domain, row = col
+  float l_run = 0.f;        // THIS lane's half-row partial denominator
+
+  const float c = scale * kLog2e;
+  const int k_hi = causal ? min(Sk, past + q0 + BM) : Sk;
+  // this wave's own causal horizon: rows q0+wave*32 .. +32
+  const int k_hi_wave = causal ? min(Sk, past + q0 + wave * 32 + 32) : Sk;
+
+  // async-STAGE (T14): tile kb0+BN's global loads are ISSUED right after
+  // this tile's QK^T MFMAs and land in registers while softmax/PV run;
+  // the LDS write happens after the end-of-iteration barrier.
+  const int kr_row = tid / (D / 8);
+  const int kr_col8 = (tid % (D / 8)) * 8;
+  const int vr_r0 = (tid / (D / 8)) * 2;
+  bf16x8a krg[2], vrg0, vrg1;
+  auto issue_loads = [&](int kt0) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int64_t key = kt0 + kr_row + it * (512 / (D / 8));
+      if (key >= Sk) key = Sk - 1;  // clamp (masked below)
+      krg[it] = *reinterpret_cast<const bf16x8a*>(kb + key * sK.s + kr_col8);
+    }
+    int64_t key0 = kt0 + vr_r0, key1 = kt0 + vr_r0 + 1;
+    if (key0 >= Sk) key0 = Sk - 1;
+    if (key1 >= Sk) key1 = Sk - 1;
+    vrg0 = *reinterpret_cast<const bf16x8a*>(vb + key0 * sV.s + kr_col8);
+    vrg1 = *reinterpret_cast<const bf16x8a*>(vb + key1 * sV.s + kr_col8);
+  };
+  auto write_tile = [&]() {
+#pragma unroll
+    for (int it = 0; it < 2; ++it)
+      *reinterpret_cast<bf16x8a*>(Ks + (kr_row + it * (512 / (D / 8))) * KP + kr_col8) = krg[it];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const unsigned pk = bfbits(vrg0[j]) | (bfbits(vrg1[j]) << 16);
+      *reinterpret_cast<unsigned*>(VTs + (kr_col8 + j) * NP + vr_r0) = pk;
+    }
+  };
+
+  issue_loads(0);
+  for (int kb0 = 0; kb0 < k_hi; kb0 += BN) {
+    write_tile();
+    __syncthreads();
+
+    f32x16a sacc[2] = {f32x16a{}, f32x16a{}};
+    const bool live = kb0 < k_hi_wave;  // fully-masked tiles: barriers only
+    if (live) {
+      // --- S^T = mfma(K, Q): sacc[kt][r] = S[col][kt*32 + crow(r,hi)] ---
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+        for (int t = 0; t < D / 16; ++t) {
+          bf16x8a kf = *reinterpret_cast<const bf16x8a*>(
+              Ks + (kt * 32 + col) * KP + t * 16 + hi * 8);
+          sacc[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[t], sacc[kt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    if (kb0 + BN < k_hi) issue_loads(kb0 + BN);  // in flight under softmax+PV
+    if (live) {
+      // --- mask (causal diagonal / Sk tail) in the raw domain ---
+      const int qabs = q0 + wave * 32 + col;
+      const bool edge = (kb0 + BN > Sk) || (causal && kb0 + BN > past + q0 + wave * 32);
+      if (edge) {
+#pragma unroll
+        for (int kt = 0; kt < 2; ++kt) {
+          const int kbase = kb0 + kt * 32 + 4 * hi;
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kabs = kbase + (r & 3) + 8 * (r >> 2);
+            if (kabs >= Sk || (causal && kabs > past + qabs)) sacc[kt][r] = -INFINITY;
+          }
+        }
+      }
+
+      // --- in-register softmax: 31 fmax + one permlane32_swap ---
+      float pmax = sacc[0][0];
+#pragma unroll
+      for (int r = 1; r < 16; ++r) pmax = fmaxf(pmax, sacc[0][r]);
+#pragma unroll
+      for (int r = 0; r < 16; ++r) pmax = fmaxf(pmax, sacc[1][r]);
+      pmax *= c;  // c > 0: max commutes with the base-2 scale fold
+      {
+        uint2a sw = __builtin_amdgcn_permlane32_swap(__float_as_uint(pmax),
+                                                     __float_as_uint(pmax), false, false);
+        pmax = fmaxf(__uint_as_float(sw.x), __uint_as_float(sw.y));
+      }
+      // defer-max: keep m_run while the max grows slowly (P bounded by 2^8)
+      const bool rescale = !__all(pmax - m_run <= 8.f);
+      if (rescale) {
+        const float new_m = fmaxf(m_run, pmax);
+        const float corr = (m_run == -INFINITY) ? 0.f : exp2f(m_run - new_m);
+        m_run = new_m;
+        l_run *= corr;
+        // redistribute corr to O rows (row of value r is crow(r,hi))
+        float cr[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          cr[r] = __shfl(corr, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
+#pragma unroll
+        for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) oacc[dt][r] *= cr[r];
+      }
+
+      // --- P = exp2(S*c - m), accumulate this lane's half-row sum ---
+      float rs = 0.f;
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const float pv = exp2f(sacc[kt][r] * c - m_run);
+          sacc[kt][r] = pv;
+          rs += pv;
+        }
+      l_run += rs;
+
+      // --- relayout P into PV A-fragments (pack + permlane32_swap) ---
+      unsigned pa[4][4] __attribute__((aligned(16)));
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        const f32x16a& pp = sacc[ks >> 1];
+        const int base = 8 * (ks & 1);
+        uint2a r01 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(pp[base + 0], pp[base + 1]),
+            pack_bf16_pair(pp[base + 4], pp[base + 5]), false, false);
+        uint2a r23 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(pp[base + 2], pp[base + 3]),
+            pack_bf16_pair(pp[base + 6], pp[base + 7]), false, false);
+        pa[ks][0] = r01.x;
+        pa[ks][1] = r23.x;
+        pa[ks][2] = r01.y;
+        pa[ks][3] = r23.y;
+      }
+
+      // --- O += P V : B-frags are V^T rows (plain b128 reads) ---
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          bf16x8a vf = *reinterpret_cast<const bf16x8a*>(
+              VTs + (dt * 32 + col) * NP + ks * 16 + hi * 8);
+          oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<bf16x8a*>(pa[ks]), vf, oacc[dt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();  // next iteration overwrites Ks/VTs
+  }
+
+  // --- epilogue: combine half-row denominators, normalize, store ---
+  {
+    uint2a sw = __builtin_amdgcn_permlane32_swap(__float_as_uint(l_run),
+                                                 __float_as_uint(l_run), false, false);
+    l_run = __uint_as_float(sw.x) + __uint_as_float(sw.y);
+  }
+  const float inv_l_mine = 1.f / fmaxf(l_run, 1e-30f);
+  float il[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    il[r] = __shfl(inv_l_mine, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
+
+  ushort* ob = out + b * sO.b + h * sO.h;
+  float* lb = lse + bh * Sq;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int qr = q0 + wave * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+    if (qr >= Sq) continue;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+      ob[(int64_t)qr * sO.s + dt * 32 + col] = af2bf(oacc[dt][r] * il[r]);
+  }
+  const int qr_lse = q0 + wave * 32 + col;
+  if (hi == 0 && qr_lse < Sq)
+    lb[qr_lse] = (m_run + __log2f(fmaxf(l_run, 1e-30f))) * kLn2;
+}
+
+template __global__ void fa_fwd_swapped_kernel<128>(const abf16*, const abf16*, const abf16*,
+                                                    ushort*, float*, int, int, int, int, float,
+                                                    int, int, Str3, Str3, Str3, Str3);
 
 template __global__ void fa_fwd_kernel<64>(const abf16*, const abf16*, const abf16*,
                                            ushort*, float*, int, int, int, int, float,
@@ -705,10 +957,39 @@ int fa_lds_bytes() {
   return elems * 2;
 }
 
+// 0 = legacy 4-wave 16x16 kernel, 1 = swapped-QK^T 8-wave 32x32 ladder
+static int fa_fwd_impl() {
+  static int impl = []() {
+    const char* e = getenv("ACCELERATE_AMD_FA_FWD");
+    if (e && strcmp(e, "legacy") == 0) return 0;
+    if (e && strcmp(e, "swapped") == 0) return 1;
+    return 0;  // default until the swapped ladder is fully measured
+  }();
+  return impl;
+}
+
 template <int D>
 hipError_t launch_impl(const void* q, const void* k, const void* v, void* out, float* lse,
                        int64_t bh, int Sq, int Sk, int past, int causal, float scale,
                        int Hq, int Hkv, const Str3* strides, hipStream_t stream) {
+  if (D == 128 && fa_fwd_impl() == 1) {
+    constexpr int lds_sw = (64 * (D + 8) + D * (64 + 8)) * 2;
+    static bool attr_sw = false;
+    if (!attr_sw) {
+      hipError_t e =
+          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_fwd_swapped_kernel<128>),
+                              hipFuncAttributeMaxDynamicSharedMemorySize, lds_sw);
+      if (e != hipSuccess) return e;
+      attr_sw = true;
+    }
+    dim3 grid((Sq + 255) / 256, (unsigned)bh);
+    hipLaunchKernelGGL(fa_fwd_swapped_kernel<128>, grid, dim3(512), lds_sw, stream,
+                       reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                       reinterpret_cast<const abf16*>(v), reinterpret_cast<ushort*>(out),
+                       lse, Sq, Sk, past, causal, scale, Hq, Hkv,
+                       strides[0], strides[1], strides[2], strides[3]);
+    return hipGetLastError();
+  }
   static bool attr_set = false;
   const int lds = fa_lds_bytes<D>();
   if (!attr_set) {
