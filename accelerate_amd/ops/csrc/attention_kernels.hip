@@ -938,6 +938,174 @@ template __global__ void fa_bwd_dkdv_kernel<128, 32>(const abf16*, const abf16*,
                                                      ushort*, ushort*, int, int, int, int, float,
                                                      int, int, Str3, Str3, Str3, Str3);
 
+// Swapped-orientation dq (D=128): same 8-wave 32x32 ladder as the forward.
+// Per wave, 32 query rows are lane-resident (Q and dOut B-fragments, lse
+// and Drow are per-lane scalars); K tiles of 64 stream through LDS as
+// row-major A-sources (S^T = mfma(K,Q), dP^T = mfma(V,dOut)) plus a
+// transposed K^T copy for the dq += dS^T-relayout x K^T MFMAs. No online
+// softmax (lse recompute), so the per-tile VALU cost is one exp2 + the
+// dS arithmetic per score.
+template <int D>
+__global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
+    const abf16* __restrict__ q, const abf16* __restrict__ k,
+    const abf16* __restrict__ v, const abf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    ushort* __restrict__ dq, int Sq, int Sk, int past, int causal, float scale,
+    int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sDo) {
+  static_assert(D == 128, "swapped dq is tuned for head_dim 128");
+  constexpr int BM = 256, BN = 64;
+  constexpr int KP = D + 8;
+  constexpr int NP = BN + 8;
+  extern __shared__ char smem[];
+  abf16* Ks = reinterpret_cast<abf16*>(smem);  // [BN][KP] row-major
+  abf16* Vs = Ks + BN * KP;                    // [BN][KP] row-major
+  abf16* KTs = Vs + BN * KP;                   // [D][NP]  K^T
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int hi = lane >> 5;
+  const int col = lane & 31;
+  const int q0 = (causal ? (int)(gridDim.x - 1 - blockIdx.x) : (int)blockIdx.x) * BM;
+  const int64_t bh = blockIdx.y;
+  const int64_t b = bh / Hq, h = bh % Hq, hk = h / (Hq / Hkv);
+  const abf16* qb = q + b * sQ.b + h * sQ.h;
+  const abf16* kb = k + b * sK.b + hk * sK.h;
+  const abf16* vb = v + b * sV.b + hk * sV.h;
+  const abf16* dob = dout + b * sDo.b + h * sDo.h;
+
+  int qr_mine = q0 + wave * 32 + col;
+  if (qr_mine >= Sq) qr_mine = Sq - 1;
+  bf16x8a qf[D / 16], dof[D / 16];
+#pragma unroll
+  for (int t = 0; t < D / 16; ++t) {
+    qf[t] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr_mine * sQ.s + t * 16 + hi * 8);
+    dof[t] = *reinterpret_cast<const bf16x8a*>(dob + (int64_t)qr_mine * sDo.s + t * 16 + hi * 8);
+  }
+  const float lse2 = lse[bh * Sq + qr_mine] * kLog2e;  // base-2 domain
+  const float dr_m = drow[bh * Sq + qr_mine];
+
+  f32x16a dqacc[D / 32] = {};
+  const float c = scale * kLog2e;
+  const int qabs = q0 + wave * 32 + col;
+  const int k_hi = causal ? min(Sk, past + q0 + BM) : Sk;
+  const int k_hi_wave = causal ? min(Sk, past + q0 + wave * 32 + 32) : Sk;
+
+  for (int kb0 = 0; kb0 < k_hi; kb0 += BN) {
+    // --- stage K, V row-major and K^T (pair-packed) ---
+    {
+#pragma unroll
+      for (int it = 0; it < 2 * BN * (D / 8) / 512; ++it) {
+        const int cid = tid + it * 512;
+        const int r = (cid / (D / 8)) % BN;
+        const bool is_v = cid >= BN * (D / 8);
+        const int col8 = (cid % (D / 8)) * 8;
+        int64_t key = kb0 + r;
+        if (key >= Sk) key = Sk - 1;  // clamp (P zeroed below)
+        if (is_v)
+          *reinterpret_cast<bf16x8a*>(Vs + r * KP + col8) =
+              *reinterpret_cast<const bf16x8a*>(vb + key * sV.s + col8);
+        else
+          *reinterpret_cast<bf16x8a*>(Ks + r * KP + col8) =
+              *reinterpret_cast<const bf16x8a*>(kb + key * sK.s + col8);
+      }
+      const int r0 = (tid / (D / 8)) * 2;
+      const int col8 = (tid % (D / 8)) * 8;
+      int64_t key0 = kb0 + r0, key1 = kb0 + r0 + 1;
+      if (key0 >= Sk) key0 = Sk - 1;
+      if (key1 >= Sk) key1 = Sk - 1;
+      bf16x8a k0 = *reinterpret_cast<const bf16x8a*>(kb + key0 * sK.s + col8);
+      bf16x8a k1 = *reinterpret_cast<const bf16x8a*>(kb + key1 * sK.s + col8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const unsigned pk = bfbits(k0[j]) | (bfbits(k1[j]) << 16);
+        *reinterpret_cast<unsigned*>(KTs + (col8 + j) * NP + r0) = pk;
+      }
+    }
+    __syncthreads();
+
+    if (kb0 < k_hi_wave) {
+      // --- S^T = mfma(K,Q), dP^T = mfma(V,dOut): col = q, rows = k ---
+      f32x16a sacc[2] = {f32x16a{}, f32x16a{}};
+      f32x16a dpacc[2] = {f32x16a{}, f32x16a{}};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+        for (int t = 0; t < D / 16; ++t) {
+          bf16x8a kf = *reinterpret_cast<const bf16x8a*>(
+              Ks + (kt * 32 + col) * KP + t * 16 + hi * 8);
+          sacc[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[t], sacc[kt], 0, 0, 0);
+          bf16x8a vf = *reinterpret_cast<const bf16x8a*>(
+              Vs + (kt * 32 + col) * KP + t * 16 + hi * 8);
+          dpacc[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[t], dpacc[kt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+
+      // --- dS = P (dP - Drow) scale, with P zeroed on masked positions ---
+      const bool edge = (kb0 + BN > Sk) || (causal && kb0 + BN > past + q0 + wave * 32);
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt) {
+        const int kbase = kb0 + kt * 32 + 4 * hi;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float pv = exp2f(sacc[kt][r] * c - lse2);
+          if (edge) {
+            const int kabs = kbase + (r & 3) + 8 * (r >> 2);
+            if (kabs >= Sk || (causal && kabs > past + qabs)) pv = 0.f;
+          }
+          sacc[kt][r] = pv * (dpacc[kt][r] - dr_m) * scale;
+        }
+      }
+
+      // --- relayout dS into A-fragments, dq += dS^T-frags x K^T ---
+      unsigned pa[4][4] __attribute__((aligned(16)));
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        const f32x16a& pp = sacc[ks >> 1];
+        const int base = 8 * (ks & 1);
+        uint2a r01 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(pp[base + 0], pp[base + 1]),
+            pack_bf16_pair(pp[base + 4], pp[base + 5]), false, false);
+        uint2a r23 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(pp[base + 2], pp[base + 3]),
+            pack_bf16_pair(pp[base + 6], pp[base + 7]), false, false);
+        pa[ks][0] = r01.x;
+        pa[ks][1] = r23.x;
+        pa[ks][2] = r01.y;
+        pa[ks][3] = r23.y;
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          bf16x8a ktf = *reinterpret_cast<const bf16x8a*>(
+              KTs + (dt * 32 + col) * NP + ks * 16 + hi * 8);
+          dqacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<bf16x8a*>(pa[ks]), ktf, dqacc[dt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+  }
+
+  ushort* dqb = dq + bh * Sq * (int64_t)D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int qr = q0 + wave * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+    if (qr >= Sq) continue;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+      dqb[(int64_t)qr * D + dt * 32 + col] = af2bf(dqacc[dt][r]);
+  }
+}
+
+template __global__ void fa_bwd_dq_swapped_kernel<128>(const abf16*, const abf16*, const abf16*,
+                                                       const abf16*, const float*, const float*,
+                                                       ushort*, int, int, int, int, float,
+                                                       int, int, Str3, Str3, Str3, Str3);
+
 template __global__ void fa_bwd_dq_kernel<64>(const abf16*, const abf16*, const abf16*,
                                               const abf16*, const float*, const float*,
                                               ushort*, int, int, int, int, float,
@@ -1066,12 +1234,32 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
                      lse, drow, reinterpret_cast<ushort*>(dk), reinterpret_cast<ushort*>(dv),
                      Sq, Sk, past, causal, scale, Hq, Hkv,
                      strides[0], strides[1], strides[2], strides[3]);
-  hipLaunchKernelGGL(fa_bwd_dq_kernel<D>, dim3((Sq + 127) / 128, (unsigned)bh), dim3(256),
-                     lds_dq, stream,
-                     reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
-                     reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
-                     lse, drow, reinterpret_cast<ushort*>(dq), Sq, Sk, past, causal, scale,
-                     Hq, Hkv, strides[0], strides[1], strides[2], strides[3]);
+  const char* dq_env = getenv("ACCELERATE_AMD_FA_BWD_DQ");
+  const bool dq_swapped = (D == 128) && !(dq_env && strcmp(dq_env, "legacy") == 0);
+  if (dq_swapped) {
+    constexpr int lds_dq_sw = (2 * 64 * (D + 8) + D * (64 + 8)) * 2;
+    static bool attr_sw = false;
+    if (!attr_sw) {
+      hipError_t e =
+          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dq_swapped_kernel<128>),
+                              hipFuncAttributeMaxDynamicSharedMemorySize, lds_dq_sw);
+      if (e != hipSuccess) return e;
+      attr_sw = true;
+    }
+    hipLaunchKernelGGL(fa_bwd_dq_swapped_kernel<128>, dim3((Sq + 255) / 256, (unsigned)bh),
+                       dim3(512), lds_dq_sw, stream,
+                       reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                       reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
+                       lse, drow, reinterpret_cast<ushort*>(dq), Sq, Sk, past, causal, scale,
+                       Hq, Hkv, strides[0], strides[1], strides[2], strides[3]);
+  } else {
+    hipLaunchKernelGGL(fa_bwd_dq_kernel<D>, dim3((Sq + 127) / 128, (unsigned)bh), dim3(256),
+                       lds_dq, stream,
+                       reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                       reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
+                       lse, drow, reinterpret_cast<ushort*>(dq), Sq, Sk, past, causal, scale,
+                       Hq, Hkv, strides[0], strides[1], strides[2], strides[3]);
+  }
   return hipGetLastError();
 }
 
