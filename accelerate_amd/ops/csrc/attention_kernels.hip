@@ -938,6 +938,527 @@ template __global__ void fa_bwd_dkdv_kernel<128, 32>(const abf16*, const abf16*,
                                                      ushort*, ushort*, int, int, int, int, float,
                                                      int, int, Str3, Str3, Str3, Str3);
 
+// Swapped-orientation dv / dk (D=128), split into two kernels so each
+// stays inside the 256-VGPR budget at 8 waves (the fused variant needs
+// K+V residency + two 64-reg accumulators + two 32-reg score fragments
+// at once). Each wave owns 32 KEY rows lane-resident (K rows for dv,
+// K+V rows for dk); 64-row query tiles stream through LDS. The score
+// orientation is mfma(Q_lds, K_reg) -> C[q][k] with col = k (the lane's
+// resident key), so the P^T/dS^T relayout feeds the dv/dk MFMAs with the
+// same pack+permlane32_swap recipe as the forward.
+
+template <int D>
+__global__ __launch_bounds__(512, 1) void fa_bwd_dv_swapped_kernel(
+    const abf16* __restrict__ q, const abf16* __restrict__ k,
+    const abf16* __restrict__ dout, const float* __restrict__ lse,
+    ushort* __restrict__ dv, int Sq, int Sk, int past, int causal, float scale,
+    int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sDo) {
+  static_assert(D == 128);
+  constexpr int BK = 256, BQ = 64;
+  constexpr int KP = D + 8;
+  constexpr int NP = BQ + 8;
+  extern __shared__ char smem[];
+  abf16* Qs = reinterpret_cast<abf16*>(smem);  // [BQ][KP] row-major
+  abf16* DOTs = Qs + BQ * KP;                  // [D][NP] dOut^T
+  float* lse_s = reinterpret_cast<float*>(DOTs + D * NP);  // [BQ]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int hi = lane >> 5;
+  const int col = lane & 31;
+  // causal: blocks over LOW keys see the most q-tiles -> launch them first
+  const int k0 = (causal ? (int)blockIdx.x : (int)blockIdx.x) * BK;
+  const int64_t bh = blockIdx.y;
+  const int64_t b = bh / Hq, h = bh % Hq, hk = h / (Hq / Hkv);
+  const abf16* qb = q + b * sQ.b + h * sQ.h;
+  const abf16* kb = k + b * sK.b + hk * sK.h;
+  const abf16* dob = dout + b * sDo.b + h * sDo.h;
+
+  int kr_mine = k0 + wave * 32 + col;
+  if (kr_mine >= Sk) kr_mine = Sk - 1;  // clamp (stores guarded)
+  bf16x8a kfr[D / 16];
+#pragma unroll
+  for (int t = 0; t < D / 16; ++t)
+    kfr[t] = *reinterpret_cast<const bf16x8a*>(kb + (int64_t)kr_mine * sK.s + t * 16 + hi * 8);
+
+  f32x16a dvacc[D / 32] = {};
+  const float c = scale * kLog2e;
+  const int kabs = k0 + wave * 32 + col;  // this lane's key (column)
+  // causal: key kb attends only q >= kb - past
+  const int q_lo_blk = causal ? max(0, ((k0 - past) / BQ) * BQ) : 0;
+  const int q_lo_wave = causal ? (k0 + wave * 32 - past) : 0;
+
+  for (int qt0 = q_lo_blk; qt0 < Sq; qt0 += BQ) {
+    {  // stage Q rows, dOut^T, lse tile
+#pragma unroll
+      for (int it = 0; it < BQ * (D / 8) / 512; ++it) {
+        const int cid = tid + it * 512;
+        const int r = cid / (D / 8);
+        const int col8 = (cid % (D / 8)) * 8;
+        int64_t qrow = qt0 + r;
+        if (qrow >= Sq) qrow = Sq - 1;
+        *reinterpret_cast<bf16x8a*>(Qs + r * KP + col8) =
+            *reinterpret_cast<const bf16x8a*>(qb + qrow * sQ.s + col8);
+      }
+      const int r0 = (tid / (D / 8)) * 2;
+      const int col8 = (tid % (D / 8)) * 8;
+      int64_t qr0 = qt0 + r0, qr1 = qt0 + r0 + 1;
+      if (qr0 >= Sq) qr0 = Sq - 1;
+      if (qr1 >= Sq) qr1 = Sq - 1;
+      bf16x8a d0 = *reinterpret_cast<const bf16x8a*>(dob + qr0 * sDo.s + col8);
+      bf16x8a d1 = *reinterpret_cast<const bf16x8a*>(dob + qr1 * sDo.s + col8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const unsigned pk = bfbits(d0[j]) | (bfbits(d1[j]) << 16);
+        *reinterpret_cast<unsigned*>(DOTs + (col8 + j) * NP + r0) = pk;
+      }
+      if (tid < BQ) lse_s[tid] = lse[bh * Sq + min(qt0 + tid, Sq - 1)];
+    }
+    __syncthreads();
+
+    if (qt0 + BQ > q_lo_wave) {  // tiles fully below the horizon: barriers only
+      f32x16a sacc[2] = {f32x16a{}, f32x16a{}};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int qt = 0; qt < 2; ++qt)
+#pragma unroll
+        for (int t = 0; t < D / 16; ++t) {
+          bf16x8a qff = *reinterpret_cast<const bf16x8a*>(
+              Qs + (qt * 32 + col) * KP + t * 16 + hi * 8);
+          sacc[qt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qff, kfr[t], sacc[qt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+
+      // P^T = exp2(S*c - lse[q]); zero masked (causal below-horizon, q tail)
+      const bool edge = (qt0 + BQ > Sq) || (causal && qt0 < q_lo_wave + 32);
+#pragma unroll
+      for (int qt = 0; qt < 2; ++qt) {
+        const int qbase = qt0 + qt * 32 + 4 * hi;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qrow = qt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float pv = exp2f(sacc[qt][r] * c - lse_s[qrow] * kLog2e);
+          if (edge) {
+            const int qa = qbase + (r & 3) + 8 * (r >> 2);
+            if (qa >= Sq || (causal && kabs > past + qa)) pv = 0.f;
+          }
+          sacc[qt][r] = pv;
+        }
+      }
+
+      unsigned pa[4][4] __attribute__((aligned(16)));
+#pragma unroll
+      for (int qs = 0; qs < 4; ++qs) {
+        const f32x16a& pp = sacc[qs >> 1];
+        const int base = 8 * (qs & 1);
+        uint2a r01 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(pp[base + 0], pp[base + 1]),
+            pack_bf16_pair(pp[base + 4], pp[base + 5]), false, false);
+        uint2a r23 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(pp[base + 2], pp[base + 3]),
+            pack_bf16_pair(pp[base + 6], pp[base + 7]), false, false);
+        pa[qs][0] = r01.x;
+        pa[qs][1] = r23.x;
+        pa[qs][2] = r01.y;
+        pa[qs][3] = r23.y;
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int qs = 0; qs < 4; ++qs) {
+          bf16x8a dof = *reinterpret_cast<const bf16x8a*>(
+              DOTs + (dt * 32 + col) * NP + qs * 16 + hi * 8);
+          dvacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<bf16x8a*>(pa[qs]), dof, dvacc[dt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+  }
+
+  ushort* dvb = dv + bh * Sk * (int64_t)D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kr = k0 + wave * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+    if (kr >= Sk) continue;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+      dvb[(int64_t)kr * D + dt * 32 + col] = af2bf(dvacc[dt][r]);
+  }
+}
+
+template <int D>
+__global__ __launch_bounds__(512, 1) void fa_bwd_dk_swapped_kernel(
+    const abf16* __restrict__ q, const abf16* __restrict__ k,
+    const abf16* __restrict__ v, const abf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    ushort* __restrict__ dk, int Sq, int Sk, int past, int causal, float scale,
+    int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sDo) {
+  static_assert(D == 128);
+  constexpr int BK = 256, BQ = 64;
+  constexpr int KP = D + 8;
+  constexpr int NP = BQ + 8;
+  extern __shared__ char smem[];
+  abf16* Qs = reinterpret_cast<abf16*>(smem);  // [BQ][KP]
+  abf16* DOs = Qs + BQ * KP;                   // [BQ][KP] dOut rows
+  abf16* QTs = DOs + BQ * KP;                  // [D][NP]  Q^T
+  float* lse_s = reinterpret_cast<float*>(QTs + D * NP);   // [BQ]
+  float* drow_s = lse_s + BQ;                              // [BQ]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int hi = lane >> 5;
+  const int col = lane & 31;
+  const int k0 = (int)blockIdx.x * BK;
+  const int64_t bh = blockIdx.y;
+  const int64_t b = bh / Hq, h = bh % Hq, hk = h / (Hq / Hkv);
+  const abf16* qb = q + b * sQ.b + h * sQ.h;
+  const abf16* kb = k + b * sK.b + hk * sK.h;
+  const abf16* vb = v + b * sV.b + hk * sV.h;
+  const abf16* dob = dout + b * sDo.b + h * sDo.h;
+
+  int kr_mine = k0 + wave * 32 + col;
+  if (kr_mine >= Sk) kr_mine = Sk - 1;
+  bf16x8a kfr[D / 16], vfr[D / 16];
+#pragma unroll
+  for (int t = 0; t < D / 16; ++t) {
+    kfr[t] = *reinterpret_cast<const bf16x8a*>(kb + (int64_t)kr_mine * sK.s + t * 16 + hi * 8);
+    vfr[t] = *reinterpret_cast<const bf16x8a*>(vb + (int64_t)kr_mine * sV.s + t * 16 + hi * 8);
+  }
+
+  f32x16a dkacc[D / 32] = {};
+  const float c = scale * kLog2e;
+  const int kabs = k0 + wave * 32 + col;
+  const int q_lo_blk = causal ? max(0, ((k0 - past) / BQ) * BQ) : 0;
+  const int q_lo_wave = causal ? (k0 + wave * 32 - past) : 0;
+
+  for (int qt0 = q_lo_blk; qt0 < Sq; qt0 += BQ) {
+    {  // stage Q rows, dOut rows, Q^T, lse+drow tiles
+#pragma unroll
+      for (int it = 0; it < 2 * BQ * (D / 8) / 512; ++it) {
+        const int cid = tid + it * 512;
+        const int r = (cid / (D / 8)) % BQ;
+        const bool is_do = cid >= BQ * (D / 8);
+        const int col8 = (cid % (D / 8)) * 8;
+        int64_t qrow = qt0 + r;
+        if (qrow >= Sq) qrow = Sq - 1;
+        if (is_do)
+          *reinterpret_cast<bf16x8a*>(DOs + r * KP + col8) =
+              *reinterpret_cast<const bf16x8a*>(dob + qrow * sDo.s + col8);
+        else
+          *reinterpret_cast<bf16x8a*>(Qs + r * KP + col8) =
+              *reinterpret_cast<const bf16x8a*>(qb + qrow * sQ.s + col8);
+      }
+      const int r0 = (tid / (D / 8)) * 2;
+      const int col8 = (tid % (D / 8)) * 8;
+      int64_t qr0 = qt0 + r0, qr1 = qt0 + r0 + 1;
+      if (qr0 >= Sq) qr0 = Sq - 1;
+      if (qr1 >= Sq) qr1 = Sq - 1;
+      bf16x8a a0 = *reinterpret_cast<const bf16x8a*>(qb + qr0 * sQ.s + col8);
+      bf16x8a a1 = *reinterpret_cast<const bf16x8a*>(qb + qr1 * sQ.s + col8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const unsigned pk = bfbits(a0[j]) | (bfbits(a1[j]) << 16);
+        *reinterpret_cast<unsigned*>(QTs + (col8 + j) * NP + r0) = pk;
+      }
+      if (tid < BQ) {
+        lse_s[tid] = lse[bh * Sq + min(qt0 + tid, Sq - 1)];
+        drow_s[tid] = drow[bh * Sq + min(qt0 + tid, Sq - 1)];
+      }
+    }
+    __syncthreads();
+
+    if (qt0 + BQ > q_lo_wave) {
+      f32x16a sacc[2] = {f32x16a{}, f32x16a{}};
+      f32x16a dpacc[2] = {f32x16a{}, f32x16a{}};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int qt = 0; qt < 2; ++qt)
+#pragma unroll
+        for (int t = 0; t < D / 16; ++t) {
+          bf16x8a qff = *reinterpret_cast<const bf16x8a*>(
+              Qs + (qt * 32 + col) * KP + t * 16 + hi * 8);
+          sacc[qt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qff, kfr[t], sacc[qt], 0, 0, 0);
+          bf16x8a dff = *reinterpret_cast<const bf16x8a*>(
+              DOs + (qt * 32 + col) * KP + t * 16 + hi * 8);
+          dpacc[qt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dff, vfr[t], dpacc[qt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+
+      // dS^T = P^T (dP^T - Drow[q]) scale (masked -> 0)
+      const bool edge = (qt0 + BQ > Sq) || (causal && qt0 < q_lo_wave + 32);
+#pragma unroll
+      for (int qt = 0; qt < 2; ++qt) {
+        const int qbase = qt0 + qt * 32 + 4 * hi;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qrow = qt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float pv = exp2f(sacc[qt][r] * c - lse_s[qrow] * kLog2e);
+          if (edge) {
+            const int qa = qbase + (r & 3) + 8 * (r >> 2);
+            if (qa >= Sq || (causal && kabs > past + qa)) pv = 0.f;
+          }
+          sacc[qt][r] = pv * (dpacc[qt][r] - drow_s[qrow]) * scale;
+        }
+      }
+
+      unsigned pa[4][4] __attribute__((aligned(16)));
+#pragma unroll
+      for (int qs = 0; qs < 4; ++qs) {
+        const f32x16a& pp = sacc[qs >> 1];
+        const int base = 8 * (qs & 1);
+        uint2a r01 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(pp[base + 0], pp[base + 1]),
+            pack_bf16_pair(pp[base + 4], pp[base + 5]), false, false);
+        uint2a r23 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(pp[base + 2], pp[base + 3]),
+            pack_bf16_pair(pp[base + 6], pp[base + 7]), false, false);
+        pa[qs][0] = r01.x;
+        pa[qs][1] = r23.x;
+        pa[qs][2] = r01.y;
+        pa[qs][3] = r23.y;
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int qs = 0; qs < 4; ++qs) {
+          bf16x8a qtf = *reinterpret_cast<const bf16x8a*>(
+              QTs + (dt * 32 + col) * NP + qs * 16 + hi * 8);
+          dkacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<bf16x8a*>(pa[qs]), qtf, dkacc[dt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+  }
+
+  ushort* dkb = dk + bh * Sk * (int64_t)D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kr = k0 + wave * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+    if (kr >= Sk) continue;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt)
+      dkb[(int64_t)kr * D + dt * 32 + col] = af2bf(dkacc[dt][r]);
+  }
+}
+
+template __global__ void fa_bwd_dv_swapped_kernel<128>(const abf16*, const abf16*, const abf16*,
+                                                       const float*, ushort*, int, int, int, int,
+                                                       float, int, int, Str3, Str3, Str3);
+template __global__ void fa_bwd_dk_swapped_kernel<128>(const abf16*, const abf16*, const abf16*,
+                                                       const abf16*, const float*, const float*,
+                                                       ushort*, int, int, int, int, float,
+                                                       int, int, Str3, Str3, Str3, Str3);
+
+// Fused swapped dk+dv (D=128, BQ=32): one pass recomputes S once for both
+// grads. Register liveness is sequenced so the P fragments are consumed by
+// the dv MFMAs BEFORE dP materializes (peak stays under the 256-VGPR cap):
+//   S -> P(in place) -> pa_p -> dv += pa_p x dOut^T -> dP -> dS(in place)
+//   -> pa_ds -> dk += pa_ds x Q^T
+template <int D>
+__global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
+    const abf16* __restrict__ q, const abf16* __restrict__ k,
+    const abf16* __restrict__ v, const abf16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    ushort* __restrict__ dk, ushort* __restrict__ dv,
+    int Sq, int Sk, int past, int causal, float scale,
+    int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sDo) {
+  static_assert(D == 128);
+  constexpr int BK = 256, BQ = 32;
+  constexpr int KP = D + 8;
+  constexpr int NP = BQ + 8;
+  extern __shared__ char smem[];
+  abf16* Qs = reinterpret_cast<abf16*>(smem);  // [BQ][KP]
+  abf16* DOs = Qs + BQ * KP;                   // [BQ][KP]
+  abf16* QTs = DOs + BQ * KP;                  // [D][NP]
+  abf16* DOTs = QTs + D * NP;                  // [D][NP]
+  float* lse_s = reinterpret_cast<float*>(DOTs + D * NP);  // [BQ] (base-2)
+  float* drow_s = lse_s + BQ;                              // [BQ]
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int hi = lane >> 5;
+  const int col = lane & 31;
+  const int k0 = (int)blockIdx.x * BK;  // low keys = most q-tiles: heavy-first
+  const int64_t bh = blockIdx.y;
+  const int64_t b = bh / Hq, h = bh % Hq, hk = h / (Hq / Hkv);
+  const abf16* qb = q + b * sQ.b + h * sQ.h;
+  const abf16* kb = k + b * sK.b + hk * sK.h;
+  const abf16* vb = v + b * sV.b + hk * sV.h;
+  const abf16* dob = dout + b * sDo.b + h * sDo.h;
+
+  int kr_mine = k0 + wave * 32 + col;
+  if (kr_mine >= Sk) kr_mine = Sk - 1;
+  bf16x8a kfr[D / 16], vfr[D / 16];
+#pragma unroll
+  for (int t = 0; t < D / 16; ++t) {
+    kfr[t] = *reinterpret_cast<const bf16x8a*>(kb + (int64_t)kr_mine * sK.s + t * 16 + hi * 8);
+    vfr[t] = *reinterpret_cast<const bf16x8a*>(vb + (int64_t)kr_mine * sV.s + t * 16 + hi * 8);
+  }
+
+  f32x16a dkacc[D / 32] = {}, dvacc[D / 32] = {};
+  const float c = scale * kLog2e;
+  const int kabs = k0 + wave * 32 + col;
+  const int q_lo_blk = causal ? max(0, ((k0 - past) / BQ) * BQ) : 0;
+  const int q_lo_wave = causal ? (k0 + wave * 32 - past) : 0;
+
+  for (int qt0 = q_lo_blk; qt0 < Sq; qt0 += BQ) {
+    {  // stage: Q rows + dOut rows (BQ*(D/8) = 512 chunks each)
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        const bool second = it == 1;
+        const int r = tid / (D / 8);
+        const int col8 = (tid % (D / 8)) * 8;
+        int64_t qrow = qt0 + r;
+        if (qrow >= Sq) qrow = Sq - 1;
+        if (second)
+          *reinterpret_cast<bf16x8a*>(DOs + r * KP + col8) =
+              *reinterpret_cast<const bf16x8a*>(dob + qrow * sDo.s + col8);
+        else
+          *reinterpret_cast<bf16x8a*>(Qs + r * KP + col8) =
+              *reinterpret_cast<const bf16x8a*>(qb + qrow * sQ.s + col8);
+      }
+      // transposes: threads 0-255 pair-pack Q^T, 256-511 dOut^T
+      const bool tsecond = tid >= 256;
+      const int tp = tid & 255;
+      const int r0 = (tp / (D / 8)) * 2;
+      const int col8t = (tp % (D / 8)) * 8;
+      int64_t qr0 = qt0 + r0, qr1 = qt0 + r0 + 1;
+      if (qr0 >= Sq) qr0 = Sq - 1;
+      if (qr1 >= Sq) qr1 = Sq - 1;
+      const abf16* src = tsecond ? dob : qb;
+      const int64_t ss = tsecond ? sDo.s : sQ.s;
+      abf16* dst = tsecond ? DOTs : QTs;
+      bf16x8a a0 = *reinterpret_cast<const bf16x8a*>(src + qr0 * ss + col8t);
+      bf16x8a a1 = *reinterpret_cast<const bf16x8a*>(src + qr1 * ss + col8t);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const unsigned pk = bfbits(a0[j]) | (bfbits(a1[j]) << 16);
+        *reinterpret_cast<unsigned*>(dst + (col8t + j) * NP + r0) = pk;
+      }
+      if (tid < BQ) {
+        lse_s[tid] = lse[bh * Sq + min(qt0 + tid, Sq - 1)] * kLog2e;
+        drow_s[tid] = drow[bh * Sq + min(qt0 + tid, Sq - 1)];
+      }
+    }
+    __syncthreads();
+
+    if (qt0 + BQ > q_lo_wave) {
+      // --- S^T = mfma(Q, K): col = k, rows = q ---
+      f32x16a sacc = {};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int t = 0; t < D / 16; ++t) {
+        bf16x8a qff = *reinterpret_cast<const bf16x8a*>(Qs + col * KP + t * 16 + hi * 8);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qff, kfr[t], sacc, 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // P^T (masked -> 0)
+      const bool edge = (qt0 + BQ > Sq) || (causal && qt0 < q_lo_wave + 32);
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float pv = exp2f(sacc[r] * c - lse_s[qrow]);
+        if (edge) {
+          const int qa = qt0 + qrow;
+          if (qa >= Sq || (causal && kabs > past + qa)) pv = 0.f;
+        }
+        sacc[r] = pv;
+      }
+      unsigned pa[2][4] __attribute__((aligned(16)));
+#pragma unroll
+      for (int qs = 0; qs < 2; ++qs) {
+        const int base = 8 * qs;
+        uint2a r01 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(sacc[base + 0], sacc[base + 1]),
+            pack_bf16_pair(sacc[base + 4], sacc[base + 5]), false, false);
+        uint2a r23 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(sacc[base + 2], sacc[base + 3]),
+            pack_bf16_pair(sacc[base + 6], sacc[base + 7]), false, false);
+        pa[qs][0] = r01.x;
+        pa[qs][1] = r23.x;
+        pa[qs][2] = r01.y;
+        pa[qs][3] = r23.y;
+      }
+      // dv += P^T x dOut^T (consumes pa before dP lives)
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int qs = 0; qs < 2; ++qs) {
+          bf16x8a dof = *reinterpret_cast<const bf16x8a*>(
+              DOTs + (dt * 32 + col) * NP + qs * 16 + hi * 8);
+          dvacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<bf16x8a*>(pa[qs]), dof, dvacc[dt], 0, 0, 0);
+        }
+      // dP^T = mfma(dOut, V)
+      f32x16a dpacc = {};
+#pragma unroll
+      for (int t = 0; t < D / 16; ++t) {
+        bf16x8a dff = *reinterpret_cast<const bf16x8a*>(DOs + col * KP + t * 16 + hi * 8);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dff, vfr[t], dpacc, 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+      // dS^T = P (dP - Drow) scale
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        sacc[r] = sacc[r] * (dpacc[r] - drow_s[qrow]) * scale;
+      }
+#pragma unroll
+      for (int qs = 0; qs < 2; ++qs) {
+        const int base = 8 * qs;
+        uint2a r01 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(sacc[base + 0], sacc[base + 1]),
+            pack_bf16_pair(sacc[base + 4], sacc[base + 5]), false, false);
+        uint2a r23 = __builtin_amdgcn_permlane32_swap(
+            pack_bf16_pair(sacc[base + 2], sacc[base + 3]),
+            pack_bf16_pair(sacc[base + 6], sacc[base + 7]), false, false);
+        pa[qs][0] = r01.x;
+        pa[qs][1] = r23.x;
+        pa[qs][2] = r01.y;
+        pa[qs][3] = r23.y;
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dt = 0; dt < D / 32; ++dt)
+#pragma unroll
+        for (int qs = 0; qs < 2; ++qs) {
+          bf16x8a qtf = *reinterpret_cast<const bf16x8a*>(
+              QTs + (dt * 32 + col) * NP + qs * 16 + hi * 8);
+          dkacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<bf16x8a*>(pa[qs]), qtf, dkacc[dt], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+  }
+
+  ushort* dkb = dk + bh * Sk * (int64_t)D;
+  ushort* dvb = dv + bh * Sk * (int64_t)D;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kr = k0 + wave * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+    if (kr >= Sk) continue;
+#pragma unroll
+    for (int dt = 0; dt < D / 32; ++dt) {
+      dkb[(int64_t)kr * D + dt * 32 + col] = af2bf(dkacc[dt][r]);
+      dvb[(int64_t)kr * D + dt * 32 + col] = af2bf(dvacc[dt][r]);
+    }
+  }
+}
+
+template __global__ void fa_bwd_dkdv_swapped_kernel<128>(
+    const abf16*, const abf16*, const abf16*, const abf16*, const float*, const float*,
+    ushort*, ushort*, int, int, int, int, float, int, int, Str3, Str3, Str3, Str3);
+
 // Swapped-orientation dq (D=128): same 8-wave 32x32 ladder as the forward.
 // Per wave, 32 query rows are lane-resident (Q and dOut B-fragments, lse
 // and Drow are per-lane scalars); K tiles of 64 stream through LDS as
@@ -1226,14 +1747,64 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
                      drow, n_rows, D, Hq, Sq, strides[3], strides[4]);
   // hipLaunchKernelGGL is a macro: template-ids with commas must go
   // through a function pointer
-  auto dkdv_fn = &fa_bwd_dkdv_kernel<D, BQSEL>;
-  hipLaunchKernelGGL(dkdv_fn, dim3((Sk + 127) / 128, (unsigned)bh), dim3(256),
-                     lds_dkdv, stream,
-                     reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
-                     reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
-                     lse, drow, reinterpret_cast<ushort*>(dk), reinterpret_cast<ushort*>(dv),
-                     Sq, Sk, past, causal, scale, Hq, Hkv,
-                     strides[0], strides[1], strides[2], strides[3]);
+  // dkdv impl: fused swapped (default) | split | legacy (env)
+  const char* dkdv_env = getenv("ACCELERATE_AMD_FA_BWD_DKDV");
+  const int dkdv_mode = (D != 128) ? 0
+                        : (dkdv_env && strcmp(dkdv_env, "legacy") == 0) ? 0
+                        : (dkdv_env && strcmp(dkdv_env, "split") == 0)  ? 1
+                                                                        : 2;
+  if (dkdv_mode == 1) {
+    constexpr int lds_dv = (64 * (D + 8) + D * (64 + 8)) * 2 + 64 * 4;
+    constexpr int lds_dk = (2 * 64 * (D + 8) + D * (64 + 8)) * 2 + 2 * 64 * 4;
+    static bool attr_sw2 = false;
+    if (!attr_sw2) {
+      hipError_t e =
+          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dv_swapped_kernel<128>),
+                              hipFuncAttributeMaxDynamicSharedMemorySize, lds_dv);
+      if (e != hipSuccess) return e;
+      e = hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dk_swapped_kernel<128>),
+                              hipFuncAttributeMaxDynamicSharedMemorySize, lds_dk);
+      if (e != hipSuccess) return e;
+      attr_sw2 = true;
+    }
+    dim3 gsw((Sk + 255) / 256, (unsigned)bh);
+    hipLaunchKernelGGL(fa_bwd_dv_swapped_kernel<128>, gsw, dim3(512), lds_dv, stream,
+                       reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                       reinterpret_cast<const abf16*>(dout), lse,
+                       reinterpret_cast<ushort*>(dv), Sq, Sk, past, causal, scale, Hq, Hkv,
+                       strides[0], strides[1], strides[3]);
+    hipLaunchKernelGGL(fa_bwd_dk_swapped_kernel<128>, gsw, dim3(512), lds_dk, stream,
+                       reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                       reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
+                       lse, drow, reinterpret_cast<ushort*>(dk), Sq, Sk, past, causal, scale,
+                       Hq, Hkv, strides[0], strides[1], strides[2], strides[3]);
+  } else if (dkdv_mode == 2) {
+    constexpr int lds_f = (2 * 32 * (D + 8) + 2 * D * (32 + 8)) * 2 + 2 * 32 * 4;
+    static bool attr_f = false;
+    if (!attr_f) {
+      hipError_t e =
+          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dkdv_swapped_kernel<128>),
+                              hipFuncAttributeMaxDynamicSharedMemorySize, lds_f);
+      if (e != hipSuccess) return e;
+      attr_f = true;
+    }
+    hipLaunchKernelGGL(fa_bwd_dkdv_swapped_kernel<128>, dim3((Sk + 255) / 256, (unsigned)bh),
+                       dim3(512), lds_f, stream,
+                       reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                       reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
+                       lse, drow, reinterpret_cast<ushort*>(dk), reinterpret_cast<ushort*>(dv),
+                       Sq, Sk, past, causal, scale, Hq, Hkv,
+                       strides[0], strides[1], strides[2], strides[3]);
+  } else {
+    auto dkdv_fn = &fa_bwd_dkdv_kernel<D, BQSEL>;
+    hipLaunchKernelGGL(dkdv_fn, dim3((Sk + 127) / 128, (unsigned)bh), dim3(256),
+                       lds_dkdv, stream,
+                       reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
+                       reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
+                       lse, drow, reinterpret_cast<ushort*>(dk), reinterpret_cast<ushort*>(dv),
+                       Sq, Sk, past, causal, scale, Hq, Hkv,
+                       strides[0], strides[1], strides[2], strides[3]);
+  }
   const char* dq_env = getenv("ACCELERATE_AMD_FA_BWD_DQ");
   const bool dq_swapped = (D == 128) && !(dq_env && strcmp(dq_env, "legacy") == 0);
   if (dq_swapped) {
