@@ -243,8 +243,53 @@ class LlamaForCausalLM(nn.Module):
             for _ in range(c.num_hidden_layers)
         ]
 
+    def _decode_step_static(self, x_tok, pos, caches, arange_cache, rope_dev):
+        """One greedy decode step with STATIC shapes and device-tensor
+        position: every op (rope index_select, cache index_copy_, full-buffer
+        masked attention, argmax, pos.add_) is hipGraph-capturable, so the
+        whole 80-layer step replays as ONE graph launch (the eager decode
+        loop is launch-bound: BENCHMARKS.md 70B demo).
+        Mutates x_tok (next token) and pos in place; returns nothing."""
+        x = self.embed_tokens(x_tok)
+        cos = rope_dev[0].index_select(0, pos)
+        sin = rope_dev[1].index_select(0, pos)
+        scale = 1.0 / math.sqrt(self.layers[0].self_attn.head_dim)
+        # keys strictly after the write position are masked
+        key_mask = (arange_cache > pos)[None, None, None, :]
+        for layer, cache in zip(self.layers, caches):
+            attn = layer.self_attn
+            h = layer.input_layernorm(x)
+            B = h.shape[0]
+            q = attn.q_proj(h).view(B, 1, attn.n_heads, attn.head_dim).transpose(1, 2)
+            k = attn.k_proj(h).view(B, 1, attn.n_kv, attn.head_dim).transpose(1, 2)
+            v = attn.v_proj(h).view(B, 1, attn.n_kv, attn.head_dim).transpose(1, 2)
+            q = apply_rope(q, cos, sin)
+            k = apply_rope(k, cos, sin)
+            cache["k"].index_copy_(2, pos, k)
+            cache["v"].index_copy_(2, pos, v)
+            kf, vf = cache["k"], cache["v"]
+            if attn.n_kv != attn.n_heads:
+                rep = attn.n_heads // attn.n_kv
+                kf = kf.repeat_interleave(rep, dim=1)
+                vf = vf.repeat_interleave(rep, dim=1)
+            scores = torch.matmul(q, kf.transpose(-1, -2)) * scale
+            scores = scores.masked_fill(key_mask, torch.finfo(scores.dtype).min)
+            probs = F.softmax(scores.float(), dim=-1).to(q.dtype)
+            ctx = torch.matmul(probs, vf).transpose(1, 2).reshape(B, 1, -1)
+            x = x + attn.o_proj(ctx)
+            x = x + layer.mlp(layer.post_attention_layernorm(x))
+        logits = self.lm_head(self.norm(x))
+        x_tok.copy_(logits[:, -1].argmax(-1, keepdim=True))
+        pos.add_(1)
+
     @torch.no_grad()
-    def generate(self, input_ids, max_new_tokens: int = 32):
+    def generate(self, input_ids, max_new_tokens: int = 32, graph_decode=None):
+        """Greedy generation. ``graph_decode=True`` captures the decode
+        step into a hipGraph (one replay per token, zero host work) — a win
+        for small launch-bound models; measured at PARITY-to-slower on 70B
+        (decode there is GEMV-bandwidth-bound and capture-mode BLAS picks
+        conservative algorithms: benchmarks/decode_ab.py, 38.9 ms eager vs
+        46 ms graphed), so the default stays eager."""
         device = input_ids.device
         dtype = self.lm_head.weight.dtype
         caches = self.make_kv_caches(input_ids.shape[0], input_ids.shape[1] + max_new_tokens, device, dtype)
@@ -252,10 +297,44 @@ class LlamaForCausalLM(nn.Module):
         tokens = [input_ids]
         next_tok = out["logits"][:, -1].argmax(-1, keepdim=True)
         tokens.append(next_tok)
+        if max_new_tokens <= 1:
+            return torch.cat(tokens, dim=1)
+        if graph_decode is None:
+            graph_decode = False  # see docstring: eager wins on big models
+        graph_decode = graph_decode and input_ids.is_cuda and torch.cuda.is_available()
+        if not graph_decode:
+            for _ in range(max_new_tokens - 1):
+                out = self(next_tok, kv_caches=caches)
+                next_tok = out["logits"][:, -1].argmax(-1, keepdim=True)
+                tokens.append(next_tok)
+            return torch.cat(tokens, dim=1)
+
+        # --- hipGraph decode: capture one static step, replay per token ---
+        max_len = caches[0]["k"].shape[2]
+        start = caches[0]["len"]
+        x_tok = next_tok.clone()
+        pos = torch.tensor([start], device=device, dtype=torch.long)
+        arange_cache = torch.arange(max_len, device=device)
+        rope_dev = (self.rope_cos.to(device), self.rope_sin.to(device))
+        # warm up allocations/algorithms on a side stream, then rewind state
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):
+                self._decode_step_static(x_tok, pos, caches, arange_cache, rope_dev)
+        torch.cuda.current_stream().wait_stream(side)
+        pos.fill_(start)
+        x_tok.copy_(next_tok)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            self._decode_step_static(x_tok, pos, caches, arange_cache, rope_dev)
+        # capture RECORDS without executing: state (x_tok=next_tok,
+        # pos=start) is untouched — every generated token is one replay
         for _ in range(max_new_tokens - 1):
-            out = self(next_tok, kv_caches=caches)
-            next_tok = out["logits"][:, -1].argmax(-1, keepdim=True)
-            tokens.append(next_tok)
+            graph.replay()
+            tokens.append(x_tok.clone())
+        for c in caches:
+            c["len"] = start + max_new_tokens - 1
         return torch.cat(tokens, dim=1)
 
 

@@ -354,3 +354,35 @@ def test_ring_chunk_math_matches_native_full(causal):
     assert torch.allclose(dq, q.grad.float(), atol=8e-2, rtol=5e-2)
     assert torch.allclose(dk, k.grad.float(), atol=8e-2, rtol=5e-2)
     assert torch.allclose(dv, v.grad.float(), atol=8e-2, rtol=5e-2)
+
+
+def test_llama_generate_eager_cpu():
+    """Greedy generate on CPU (graph path auto-disabled) still works after
+    the hipGraph-decode refactor and is deterministic."""
+    from accelerate_amd.models.llama import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2)).eval()
+    ids = torch.randint(0, 1024, (2, 8))
+    with torch.no_grad():
+        out1 = model.generate(ids, max_new_tokens=6)
+        out2 = model.generate(ids, max_new_tokens=6)
+    assert out1.shape == (2, 14)
+    assert torch.equal(out1, out2)
+
+
+@gpu
+def test_llama_graph_decode_matches_eager():
+    """hipGraph-captured decode produces the same greedy tokens as the
+    eager per-token loop."""
+    from accelerate_amd.models.llama import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig.tiny(num_hidden_layers=4)
+    model = LlamaForCausalLM(cfg).eval().cuda().to(torch.bfloat16)
+    ids = torch.randint(0, 1024, (2, 12), device="cuda")
+    with torch.no_grad():
+        eager = model.generate(ids, max_new_tokens=10, graph_decode=False)
+        graphed = model.generate(ids, max_new_tokens=10, graph_decode=True)
+    assert eager.shape == graphed.shape == (2, 22)
+    assert torch.equal(eager, graphed), (eager[:, 12:], graphed[:, 12:])
