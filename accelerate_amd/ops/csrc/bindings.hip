@@ -556,12 +556,37 @@ at::Tensor w8a16_gemv(at::Tensor q, at::Tensor scale, at::Tensor x, c10::optiona
   return y;
 }
 
+extern "C" hipError_t launch_gemv_bf16(const void*, const void*, const void*, void*,
+                                       long long, long long, int, hipStream_t);
+
+at::Tensor gemv_bf16(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias) {
+  // y = x @ w^T for skinny x (decode): x [M, K] (M <= 8), w [N, K]
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kBFloat16,
+              "gemv_bf16: bf16 cuda");
+  auto xc = x.contiguous();
+  TORCH_CHECK(w.is_contiguous(), "gemv_bf16: weight must be contiguous");
+  const int64_t K = w.size(1), N = w.size(0);
+  const int64_t M = xc.numel() / K;
+  TORCH_CHECK(M >= 1 && M <= 8 && xc.size(-1) == K && K % 8 == 0, "gemv_bf16: shape");
+  auto y = at::empty({M, N}, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  auto err = launch_gemv_bf16(w.data_ptr(), xc.data_ptr(),
+                              bias.has_value() ? bias->data_ptr() : nullptr, y.data_ptr(),
+                              N, K, (int)M, stream.stream());
+  TORCH_CHECK(err == hipSuccess, "gemv_bf16: ", hipGetErrorString(err));
+  std::vector<int64_t> shape(x.sizes().begin(), x.sizes().end());
+  shape.back() = N;
+  return y.view(shape);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "fused bf16 LayerNorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused bf16 RMSNorm forward");
   m.def("rope_bf16", &rope_bf16, "fused rotary embedding (bf16, half-split layout)");
   m.def("mfma_gemm_bt", &mfma_gemm_bt, "hand-written MFMA bf16 GEMM (A @ B^T + bias)");
+  m.def("gemv_bf16", &gemv_bf16, "fused bf16 decode GEMV (x @ W^T, M <= 8)",
+        py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused bf16 RMSNorm backward");
   m.def("fp8_cast_amax", &fp8_cast_amax, "bf16 -> fp8 cast with fused amax (gfx950)");
   m.def("fp8_cast_transpose", &fp8_cast_transpose,

@@ -145,3 +145,90 @@ __global__ __launch_bounds__(256) void gemm_bt_bf16_kernel(
     }
   }
 }
+
+// ---------------------------------------------------------------------------
+// Decode GEMV: y[m, n] = x[m, :] . W[n, :]  (nn.Linear weight layout),
+// bf16 in/out, fp32 accumulate via V_DOT2_F32_BF16 (one VALU per bf16
+// pair keeps the loop HBM-bound, not conversion-bound — the reason the
+// rocBLAS skinny-GEMM path measured only ~3.4 TB/s on 70B decode).
+// One wave per output row, grid-stride; x rows (M <= 8) stay L1-resident.
+// ---------------------------------------------------------------------------
+
+namespace {
+typedef unsigned short gus16x2 __attribute__((ext_vector_type(2)));
+typedef unsigned short gus16x8 __attribute__((ext_vector_type(8)));
+
+__device__ __forceinline__ ushort gvf2bf(float f) {
+  union { float f; unsigned i; } cv;
+  cv.f = f;
+  if ((cv.i & 0x7F800000u) == 0x7F800000u)
+    return (ushort)(cv.i >> 16) | (ushort)((cv.i & 0xFFFFu) ? 0x40 : 0);
+  cv.i += 0x7FFFu + ((cv.i >> 16) & 1u);
+  return (ushort)(cv.i >> 16);
+}
+}  // namespace
+
+template <int M>
+__global__ __launch_bounds__(256) void gemv_bf16_kernel(
+    const ushort* __restrict__ w, const ushort* __restrict__ x,
+    const ushort* __restrict__ bias, ushort* __restrict__ y,
+    int64_t N, int64_t K) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < N; row += (int64_t)gridDim.x * 4) {
+    const ushort* wr = w + row * K;
+    float acc[M];
+#pragma unroll
+    for (int b = 0; b < M; ++b) acc[b] = 0.f;
+    for (int64_t k0 = (int64_t)lane * 8; k0 < K; k0 += 64 * 8) {
+      gus16x8 w8 = *reinterpret_cast<const gus16x8*>(wr + k0);
+#pragma unroll
+      for (int b = 0; b < M; ++b) {
+        gus16x8 x8 = *reinterpret_cast<const gus16x8*>(x + (int64_t)b * K + k0);
+#pragma unroll
+        for (int p = 0; p < 4; ++p) {
+          gus16x2 wp = {w8[2 * p], w8[2 * p + 1]};
+          gus16x2 xp = {x8[2 * p], x8[2 * p + 1]};
+          acc[b] = __builtin_amdgcn_fdot2_f32_bf16(wp, xp, acc[b], false);
+        }
+      }
+    }
+#pragma unroll
+    for (int b = 0; b < M; ++b) {
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) acc[b] += __shfl_down(acc[b], off, 64);
+    }
+    if (lane == 0) {
+      float bv = 0.f;
+      if (bias != nullptr) {
+        union { ushort u; } ub{bias[row]};
+        unsigned wide = ((unsigned)ub.u) << 16;
+        bv = *reinterpret_cast<float*>(&wide);
+      }
+#pragma unroll
+      for (int b = 0; b < M; ++b) y[(int64_t)b * N + row] = gvf2bf(acc[b] + bv);
+    }
+  }
+}
+
+extern "C" hipError_t launch_gemv_bf16(const void* w, const void* x, const void* bias, void* y,
+                                       long long N, long long K, int M, hipStream_t stream) {
+  const int n_blocks = (int)std::min<long long>((N + 3) / 4, 4096);
+  dim3 g(n_blocks), b(256);
+#define GEMV_CASE(MM) \
+  hipLaunchKernelGGL((gemv_bf16_kernel<MM>), g, b, 0, stream, (const ushort*)w, \
+                     (const ushort*)x, (const ushort*)bias, (ushort*)y, (int64_t)N, (int64_t)K)
+  switch (M) {
+    case 1: GEMV_CASE(1); break;
+    case 2: GEMV_CASE(2); break;
+    case 3: GEMV_CASE(3); break;
+    case 4: GEMV_CASE(4); break;
+    case 5: GEMV_CASE(5); break;
+    case 6: GEMV_CASE(6); break;
+    case 7: GEMV_CASE(7); break;
+    case 8: GEMV_CASE(8); break;
+    default: return hipErrorInvalidValue;
+  }
+#undef GEMV_CASE
+  return hipGetLastError();
+}

@@ -33,6 +33,10 @@ def main():
         torch.set_default_dtype(torch.bfloat16)
         model = LlamaForCausalLM(config).eval()
         torch.set_default_dtype(torch.float32)
+    if not int8:
+        from accelerate_amd.ops.linear import convert_linears_for_inference
+
+        convert_linears_for_inference(model)  # fused decode GEMV
     if int8:
         # weight-only int8: halves HBM residency AND decode time (decode is
         # weight-bandwidth-bound; the fused w8a16 GEMV reads int8 directly)

@@ -201,3 +201,40 @@ def test_multi_tensor_copy_roundtrip():
     torch.cuda.synchronize()
     for t, o in zip(tensors, offsets):
         assert torch.equal(t.reshape(-1), flat[o : o + t.numel()])
+
+
+@gpu
+@pytest.mark.parametrize("M,N,K", [(1, 1024, 4096), (4, 512, 2048), (2, 28672, 4096)])
+def test_gemv_bf16_matches_linear(M, N, K):
+    """Fused decode GEMV vs F.linear (fp32 reference tolerance)."""
+    import torch.nn.functional as F
+
+    from accelerate_amd.ops import _load_extension
+
+    ext = _load_extension(required=True)
+    torch.manual_seed(0)
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.05
+    b = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+    y = ext.gemv_bf16(x, w, b)
+    ref = F.linear(x.float(), w.float(), b.float())
+    assert y.shape == (M, N)
+    assert (y.float() - ref).abs().max() < 0.05 * ref.abs().max()
+
+
+@gpu
+def test_fast_linear_decode_route():
+    """FastLinear matches nn.Linear on decode shapes and 3-D inputs."""
+    from accelerate_amd.ops.linear import convert_linears_for_inference
+
+    torch.manual_seed(0)
+    m = torch.nn.Sequential(torch.nn.Linear(256, 512), torch.nn.Linear(512, 128)).cuda().to(torch.bfloat16)
+    ref = [p.detach().clone() for p in m.parameters()]
+    convert_linears_for_inference(m)
+    x = torch.randn(2, 1, 256, device="cuda", dtype=torch.bfloat16)
+    with torch.no_grad():
+        y = m(x)
+        y_ref = torch.nn.functional.linear(
+            torch.nn.functional.linear(x, ref[0], ref[1]), ref[2], ref[3]
+        )
+    assert (y.float() - y_ref.float()).abs().max() < 0.1
