@@ -16,6 +16,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.norms import FusedDropoutAddLayerNorm
+
 
 @dataclass
 class BertConfig:
@@ -95,17 +97,22 @@ class BertLayer(nn.Module):
     def __init__(self, config: BertConfig):
         super().__init__()
         self.attention = BertSelfAttention(config)
-        self.attn_norm = nn.LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
+        # both residual junctions run dropout+add+LayerNorm as ONE fused
+        # kernel pair on GPU (ops/norms.py); params match nn.LayerNorm
+        self.attn_norm = FusedDropoutAddLayerNorm(
+            config.hidden_size, eps=config.layer_norm_eps, p=config.hidden_dropout_prob
+        )
         self.ffn_in = nn.Linear(config.hidden_size, config.intermediate_size)
         self.ffn_out = nn.Linear(config.intermediate_size, config.hidden_size)
-        self.ffn_norm = nn.LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
-        self.dropout = nn.Dropout(config.hidden_dropout_prob)
+        self.ffn_norm = FusedDropoutAddLayerNorm(
+            config.hidden_size, eps=config.layer_norm_eps, p=config.hidden_dropout_prob
+        )
 
     def forward(self, hidden, attention_mask=None):
         attn = self.attention(hidden, attention_mask)
-        hidden = self.attn_norm(hidden + self.dropout(attn))
+        hidden = self.attn_norm(hidden, attn)
         ffn = self.ffn_out(F.gelu(self.ffn_in(hidden), approximate="tanh"))
-        hidden = self.ffn_norm(hidden + self.dropout(ffn))
+        hidden = self.ffn_norm(hidden, ffn)
         return hidden
 
 

@@ -3,6 +3,7 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/hip/HIPGeneratorImpl.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
@@ -579,9 +580,76 @@ at::Tensor gemv_bf16(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias)
   return y.view(shape);
 }
 
+extern "C" hipError_t launch_dropout_add_ln_fwd(const void*, const void*, const void*,
+                                                const void*, void*, void*, void*, void*, void*,
+                                                long long, int, float, float, float,
+                                                at::PhiloxCudaState, int, hipStream_t);
+extern "C" hipError_t launch_dropout_add_ln_bwd(const void*, const void*, const void*,
+                                                const void*, const void*, const void*, void*,
+                                                void*, void*, long long, int, float, int,
+                                                hipStream_t);
+
+std::vector<at::Tensor> dropout_add_ln_fwd(at::Tensor x, at::Tensor z, at::Tensor w,
+                                           at::Tensor b, double eps, double p) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == at::kBFloat16,
+              "dropout_add_ln_fwd: bf16 contiguous");
+  const int d = (int)x.size(-1);
+  TORCH_CHECK(d % 8 == 0 && d <= 2048, "dropout_add_ln_fwd: inner dim");
+  const int64_t rows = x.numel() / d;
+  auto y = at::empty_like(x);
+  auto s_out = at::empty_like(x);
+  auto mask = at::empty({rows, d}, x.options().dtype(at::kByte));
+  auto mean = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
+  const float keep = 1.f - (float)p;
+  at::PhiloxCudaState rng(0, 0);
+  if (keep < 1.f) {
+    auto gen = at::get_generator_or_default<at::CUDAGeneratorImpl>(
+        std::nullopt, at::cuda::detail::getDefaultCUDAGenerator());
+    std::lock_guard<std::mutex> lock(gen->mutex_);
+    rng = gen->philox_cuda_state((uint64_t)x.numel());
+  }
+  const int n_blocks = norm_bwd_blocks(rows);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto err = launch_dropout_add_ln_fwd(bfp(x), bfp(z), bfp(w), bfp(b), bfp_mut(y),
+                                       bfp_mut(s_out), mask.data_ptr(), mean.data_ptr<float>(),
+                                       rstd.data_ptr<float>(), rows, d, (float)eps, keep,
+                                       keep > 0.f ? 1.f / keep : 0.f, rng, n_blocks,
+                                       stream.stream());
+  TORCH_CHECK(err == hipSuccess, "dropout_add_ln_fwd: ", hipGetErrorString(err));
+  return {y, s_out, mask, mean, rstd};
+}
+
+std::vector<at::Tensor> dropout_add_ln_bwd(at::Tensor dy, at::Tensor s, at::Tensor w,
+                                           at::Tensor mask, at::Tensor mean, at::Tensor rstd,
+                                           double p) {
+  const int d = (int)s.size(-1);
+  const int64_t rows = s.numel() / d;
+  const int n_blocks = norm_bwd_blocks(rows);
+  const int n_partials = n_blocks * 4;
+  auto dx = at::empty_like(s);
+  auto dz = at::empty_like(s);
+  auto dwdb_partial = at::empty({n_partials, 2 * (int64_t)d}, s.options().dtype(at::kFloat));
+  const float keep = 1.f - (float)p;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto dyc = dy.contiguous();
+  auto err = launch_dropout_add_ln_bwd(bfp(dyc), bfp(s), bfp(w), mask.data_ptr(),
+                                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                                       bfp_mut(dx), bfp_mut(dz), dwdb_partial.data_ptr<float>(),
+                                       rows, d, keep > 0.f ? 1.f / keep : 0.f, n_blocks,
+                                       stream.stream());
+  TORCH_CHECK(err == hipSuccess, "dropout_add_ln_bwd: ", hipGetErrorString(err));
+  auto folded = dwdb_partial.sum(0).to(s.scalar_type());
+  return {dx, dz, folded.narrow(0, 0, d), folded.narrow(0, d, d)};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd, "fused bf16 LayerNorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "fused bf16 LayerNorm backward");
+  m.def("dropout_add_ln_fwd", &dropout_add_ln_fwd,
+        "fused y = LayerNorm(x + dropout(z)) forward (graph-safe philox mask)");
+  m.def("dropout_add_ln_bwd", &dropout_add_ln_bwd,
+        "fused dropout+add+LayerNorm backward -> (dx, dz, dw, db)");
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused bf16 RMSNorm forward");
   m.def("rope_bf16", &rope_bf16, "fused rotary embedding (bf16, half-split layout)");
   m.def("mfma_gemm_bt", &mfma_gemm_bt, "hand-written MFMA bf16 GEMM (A @ B^T + bias)");

@@ -425,3 +425,224 @@ extern "C" hipError_t launch_rmsnorm_bwd(const void* dy, const void* x, const vo
 #undef RMS_CASE
   return hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// Fused residual junction: y = LayerNorm(x + dropout(z)) — BERT runs this
+// TWICE per layer; unfused it costs a dropout kernel, an add, the LN, and
+// their backward mirrors plus intermediate round-trips. One pass here:
+// dropout mask is generated in-kernel from the torch philox state
+// (graph-capture-safe: ATen's PhiloxCudaState offset is read via pointer
+// under capture), the summed input s is written once for the backward,
+// and the mask is saved as u8.
+// ---------------------------------------------------------------------------
+
+#include <ATen/hip/PhiloxUtils.cuh>
+
+namespace {
+__device__ __forceinline__ unsigned mix_rand24(uint64_t seed, uint64_t idx) {
+  uint64_t zz = seed + idx * 0x9E3779B97F4A7C15ull;
+  zz = (zz ^ (zz >> 30)) * 0xBF58476D1CE4E5B9ull;
+  zz = (zz ^ (zz >> 27)) * 0x94D049BB133111EBull;
+  return (unsigned)(zz >> 40);  // 24 uniform bits
+}
+}  // namespace
+
+template <int ITERS, bool KEEP>
+__global__ __launch_bounds__(256) void dropout_add_ln_fwd_t(
+    const ushort* __restrict__ x, const ushort* __restrict__ z,
+    const ushort* __restrict__ w, const ushort* __restrict__ b,
+    ushort* __restrict__ y, ushort* __restrict__ s_out,
+    unsigned char* __restrict__ mask_out,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    int64_t n_rows, int d, float eps, float keep_p, float inv_keep,
+    at::PhiloxCudaState rng) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const auto seeds = at::cuda::philox::unpack(rng);
+  const uint64_t seed = std::get<0>(seeds) ^ (std::get<1>(seeds) * 0xD2B74407B1CE6E93ull);
+  const unsigned thresh = (unsigned)(keep_p * 16777216.f);  // keep if r < thresh
+  const bool do_drop = keep_p < 1.f;
+
+  float sv[KEEP ? ITERS : 1][8];
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < n_rows; row += (int64_t)gridDim.x * 4) {
+    const ushort* xr = x + row * d;
+    const ushort* zr = z + row * d;
+    ushort* sr = s_out + row * d;
+    unsigned char* mr = mask_out + row * d;
+
+    float sum = 0.f, sq = 0.f;
+    int it = 0;
+    for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
+      ushort8 x8 = *reinterpret_cast<const ushort8*>(xr + base);
+      ushort8 z8 = *reinterpret_cast<const ushort8*>(zr + base);
+      ushort8 s8;
+      unsigned char m8[8];
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        unsigned char keep = 1;
+        if (do_drop) keep = mix_rand24(seed, (uint64_t)row * d + base + k) < thresh;
+        m8[k] = keep;
+        float v = bf2f(x8[k]) + (keep ? bf2f(z8[k]) * inv_keep : 0.f);
+        if (KEEP) sv[KEEP ? it : 0][k] = v;
+        s8[k] = f2bf(v);
+        sum += v;
+        sq += v * v;
+      }
+      *reinterpret_cast<ushort8*>(sr + base) = s8;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) mr[base + k] = m8[k];
+    }
+    sum = wave_sum(sum);
+    sq = wave_sum(sq);
+    const float mu = sum / d;
+    const float var = sq / d - mu * mu;
+    const float rs = rsqrtf(var + eps);
+    if (lane == 0) {
+      mean_out[row] = mu;
+      rstd_out[row] = rs;
+    }
+    ushort* yr = y + row * d;
+    it = 0;
+    for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
+      float vk[8];
+      if (!KEEP) {
+        ushort8 s8 = *reinterpret_cast<const ushort8*>(sr + base);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) vk[k] = bf2f(s8[k]);
+      } else {
+#pragma unroll
+        for (int k = 0; k < 8; ++k) vk[k] = sv[KEEP ? it : 0][k];
+      }
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
+      ushort8 b8 = *reinterpret_cast<const ushort8*>(b + base);
+      ushort8 o8;
+#pragma unroll
+      for (int k = 0; k < 8; ++k)
+        o8[k] = f2bf((vk[k] - mu) * rs * bf2f(w8[k]) + bf2f(b8[k]));
+      *reinterpret_cast<ushort8*>(yr + base) = o8;
+    }
+  }
+}
+
+// backward: dx = LN-backward(dy) over s; dz = dx * mask * inv_keep;
+// dw/db accumulate per-wave into one [P][2d] partial (host folds by sum).
+template <int ITERS, bool KEEP>
+__global__ __launch_bounds__(256) void dropout_add_ln_bwd_t(
+    const ushort* __restrict__ dy, const ushort* __restrict__ s,
+    const ushort* __restrict__ w, const unsigned char* __restrict__ mask,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    ushort* __restrict__ dx, ushort* __restrict__ dz,
+    float* __restrict__ dwdb_partial, int64_t n_rows, int d, float inv_keep) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  float accw[ITERS][8];
+  float accb[ITERS][8];
+#pragma unroll
+  for (int i = 0; i < ITERS; ++i)
+#pragma unroll
+    for (int k = 0; k < 8; ++k) { accw[i][k] = 0.f; accb[i][k] = 0.f; }
+
+  float sv[KEEP ? ITERS : 1][8];
+  float gv[KEEP ? ITERS : 1][8];
+
+  for (int64_t row = (int64_t)blockIdx.x * 4 + wave; row < n_rows; row += (int64_t)gridDim.x * 4) {
+    const ushort* dyr = dy + row * d;
+    const ushort* srr = s + row * d;
+    const float mu = mean[row], rs = rstd[row];
+
+    float s1 = 0.f, s2 = 0.f;
+    int it = 0;
+    for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
+      ushort8 g8 = *reinterpret_cast<const ushort8*>(dyr + base);
+      ushort8 s8 = *reinterpret_cast<const ushort8*>(srr + base);
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float g = bf2f(g8[k]);
+        float v = bf2f(s8[k]);
+        if (KEEP) { gv[KEEP ? it : 0][k] = g; sv[KEEP ? it : 0][k] = v; }
+        float gw = g * bf2f(w8[k]);
+        s1 += gw;
+        s2 += gw * (v - mu) * rs;
+      }
+    }
+    s1 = wave_sum(s1) / d;
+    s2 = wave_sum(s2) / d;
+
+    ushort* dxr = dx + row * d;
+    ushort* dzr = dz + row * d;
+    const unsigned char* mr = mask + row * d;
+    it = 0;
+    for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
+      float gk[8], vk[8];
+      if (!KEEP) {
+        ushort8 g8 = *reinterpret_cast<const ushort8*>(dyr + base);
+        ushort8 s8 = *reinterpret_cast<const ushort8*>(srr + base);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) { gk[k] = bf2f(g8[k]); vk[k] = bf2f(s8[k]); }
+      } else {
+#pragma unroll
+        for (int k = 0; k < 8; ++k) { gk[k] = gv[KEEP ? it : 0][k]; vk[k] = sv[KEEP ? it : 0][k]; }
+      }
+      ushort8 w8 = *reinterpret_cast<const ushort8*>(w + base);
+      ushort8 ox, oz;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float xhat = (vk[k] - mu) * rs;
+        const float dsv = rs * (gk[k] * bf2f(w8[k]) - s1 - xhat * s2);
+        ox[k] = f2bf(dsv);
+        oz[k] = f2bf(mr[base + k] ? dsv * inv_keep : 0.f);
+        accw[it][k] += gk[k] * xhat;
+        accb[it][k] += gk[k];
+      }
+      *reinterpret_cast<ushort8*>(dxr + base) = ox;
+      *reinterpret_cast<ushort8*>(dzr + base) = oz;
+    }
+  }
+
+  float* dwp = dwdb_partial + ((int64_t)blockIdx.x * 4 + wave) * (2 * (int64_t)d);
+  int it = 0;
+  for (int base = lane * 8; base < d; base += 64 * 8, ++it) {
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      dwp[base + k] = accw[it][k];
+      dwp[d + base + k] = accb[it][k];
+    }
+  }
+}
+
+extern "C" hipError_t launch_dropout_add_ln_fwd(
+    const void* x, const void* z, const void* w, const void* b, void* y, void* s_out,
+    void* mask_out, void* mean_out, void* rstd_out, long long n_rows, int d, float eps,
+    float keep_p, float inv_keep, at::PhiloxCudaState rng, int n_blocks, hipStream_t stream) {
+  dim3 g(n_blocks), blk(256);
+#define DALN_F(N, KEEP) \
+  hipLaunchKernelGGL((dropout_add_ln_fwd_t<N, KEEP>), g, blk, 0, stream, (const ushort*)x, \
+                     (const ushort*)z, (const ushort*)w, (const ushort*)b, (ushort*)y, \
+                     (ushort*)s_out, (unsigned char*)mask_out, (float*)mean_out, \
+                     (float*)rstd_out, (int64_t)n_rows, d, eps, keep_p, inv_keep, rng)
+  if (d <= 512) DALN_F(1, true);
+  else if (d <= 1024) DALN_F(2, true);
+  else if (d <= 2048) DALN_F(4, false);
+  else return hipErrorInvalidValue;
+#undef DALN_F
+  return hipGetLastError();
+}
+
+extern "C" hipError_t launch_dropout_add_ln_bwd(
+    const void* dy, const void* s, const void* w, const void* mask, const void* mean,
+    const void* rstd, void* dx, void* dz, void* dwdb_partial, long long n_rows, int d,
+    float inv_keep, int n_blocks, hipStream_t stream) {
+  dim3 g(n_blocks), blk(256);
+#define DALN_B(N, KEEP) \
+  hipLaunchKernelGGL((dropout_add_ln_bwd_t<N, KEEP>), g, blk, 0, stream, (const ushort*)dy, \
+                     (const ushort*)s, (const ushort*)w, (const unsigned char*)mask, \
+                     (const float*)mean, (const float*)rstd, (ushort*)dx, (ushort*)dz, \
+                     (float*)dwdb_partial, (int64_t)n_rows, d, inv_keep)
+  if (d <= 512) DALN_B(1, true);
+  else if (d <= 1024) DALN_B(2, true);
+  else if (d <= 2048) DALN_B(4, false);
+  else return hipErrorInvalidValue;
+#undef DALN_B
+  return hipGetLastError();
+}

@@ -92,3 +92,58 @@ def convert_to_fused_norms(model: nn.Module) -> nn.Module:
                 fused.weight = child.weight
                 setattr(module, child_name, fused)
     return model
+
+
+class _FusedDropoutAddLNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, z, weight, bias, eps, p):
+        ext = _load_extension(required=True)
+        y, s, mask, mean, rstd = ext.dropout_add_ln_fwd(x, z, weight, bias, eps, p)
+        ctx.save_for_backward(s, weight, mask, mean, rstd)
+        ctx.p = p
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _load_extension(required=True)
+        s, weight, mask, mean, rstd = ctx.saved_tensors
+        dx, dz, dw, db = ext.dropout_add_ln_bwd(dy, s, weight, mask, mean, rstd, ctx.p)
+        return dx, dz, dw, db, None, None
+
+
+class FusedDropoutAddLayerNorm(nn.LayerNorm):
+    """``LayerNorm(x + dropout(z))`` in ONE kernel pair — the residual
+    junction BERT executes twice per layer. The dropout mask comes from the
+    torch philox state in-kernel (hipGraph-capture-safe) and is saved as u8
+    for the backward, which produces dx and the mask-scaled dz together
+    with dweight/dbias in a single pass. Parameter names/shapes match
+    nn.LayerNorm, so checkpoints are interchangeable. Falls back to the
+    composite ops off-GPU / non-bf16."""
+
+    def __init__(self, normalized_shape, eps=1e-12, p=0.0):
+        super().__init__(normalized_shape, eps=eps)
+        self.p = p
+
+    def forward(self, x, z=None):
+        if z is None:  # plain LayerNorm use (no junction)
+            return super().forward(x)
+        p = self.p if self.training else 0.0
+        # the single-kernel path measured 0.60x the composite at BERT
+        # shapes (the u8 mask + saved-sum writes outweigh the saved
+        # launches once torch's dropout/add/LN are graph-captured), so it
+        # is opt-in until the mask is bit-packed:
+        import os
+
+        if (
+            os.environ.get("ACCELERATE_AMD_FUSED_JUNCTION") == "1"
+            and _use_fused(x, self.weight, self.bias)
+            and x.shape == z.shape
+            and z.dtype == x.dtype
+            and len(self.normalized_shape) == 1
+            and x.shape[-1] <= 2048
+        ):
+            return _FusedDropoutAddLNFn.apply(
+                x.contiguous(), z.contiguous(), self.weight, self.bias, self.eps, p
+            )
+        h = x + torch.nn.functional.dropout(z, p=p, training=p > 0)
+        return super().forward(h)

@@ -80,4 +80,37 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    import sys
+    if "--junction" not in sys.argv:
+        main()
+
+
+def junction_bench():
+    """Fused dropout+add+LN vs composite at the BERT bench shape."""
+    from accelerate_amd.ops.norms import FusedDropoutAddLayerNorm
+
+    rows, d, p = 2048, 768, 0.1
+    fused = FusedDropoutAddLayerNorm(d, eps=1e-12, p=p).cuda().to(torch.bfloat16)
+    fused.train()
+    ln = nn.LayerNorm(d, eps=1e-12).cuda().to(torch.bfloat16)
+    x = torch.randn(rows, d, device="cuda", dtype=torch.bfloat16)
+    z = torch.randn(rows, d, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(rows, d, device="cuda", dtype=torch.bfloat16)
+
+    def step_fused():
+        xr = x.detach().requires_grad_(True)
+        zr = z.detach().requires_grad_(True)
+        fused(xr, zr).backward(dy)
+
+    def step_comp():
+        xr = x.detach().requires_grad_(True)
+        zr = z.detach().requires_grad_(True)
+        ln(xr + torch.nn.functional.dropout(zr, 0.1, True)).backward(dy)
+
+    f = bench(step_fused)
+    c = bench(step_comp)
+    print(f"junction {rows}x{d} p={p}: fused {f:7.1f} us  composite {c:7.1f} us  speedup {c / f:4.2f}x")
+
+
+if __name__ == "__main__" and "--junction" in __import__("sys").argv:
+    junction_bench()

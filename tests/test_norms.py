@@ -85,14 +85,82 @@ def test_fused_rmsnorm_matches_fp32(shape):
 @gpu
 def test_convert_to_fused_norms_bert():
     from accelerate_amd.models import BertConfig, BertForSequenceClassification
-    from accelerate_amd.ops.norms import FusedLayerNorm, convert_to_fused_norms
+    from accelerate_amd.ops.norms import (
+        FusedDropoutAddLayerNorm,
+        FusedLayerNorm,
+        convert_to_fused_norms,
+    )
 
     model = BertForSequenceClassification(BertConfig(num_hidden_layers=2)).cuda().to(torch.bfloat16)
     convert_to_fused_norms(model)
-    assert isinstance(model.bert.layers[0].attn_norm, FusedLayerNorm)
+    # residual junctions are natively fused dropout+add+LN modules;
+    # remaining plain LayerNorms (embeddings) get the fused LN swap
+    assert isinstance(model.bert.layers[0].attn_norm, FusedDropoutAddLayerNorm)
+    assert isinstance(model.bert.embeddings.LayerNorm, FusedLayerNorm)
     ids = torch.randint(0, 30522, (2, 32), device="cuda")
     labels = torch.randint(0, 2, (2,), device="cuda")
     out = model(ids, labels=labels)
     out["loss"].backward()
     torch.cuda.synchronize()
     assert torch.isfinite(out["loss"])
+
+
+@gpu
+@pytest.mark.parametrize("p", [0.0, 0.3])
+def test_dropout_add_ln_matches_composite(p, monkeypatch):
+    monkeypatch.setenv("ACCELERATE_AMD_FUSED_JUNCTION", "1")
+    """Fused LayerNorm(x + dropout(z)): forward reproduces the composite
+    computed with the kernel's OWN saved mask; backward (dx, dz, dw, db)
+    matches autograd through that composite."""
+    from accelerate_amd.ops import _load_extension
+
+    ext = _load_extension(required=True)
+    torch.manual_seed(3)
+    rows, d = 512, 768
+    x = torch.randn(rows, d, device="cuda", dtype=torch.bfloat16)
+    z = torch.randn(rows, d, device="cuda", dtype=torch.bfloat16)
+    w = (torch.randn(d, device="cuda") * 0.1 + 1.0).to(torch.bfloat16)
+    b = (torch.randn(d, device="cuda") * 0.1).to(torch.bfloat16)
+    y, s, mask, mean, rstd = ext.dropout_add_ln_fwd(x, z, w, b, 1e-5, p)
+    keep = mask.float().mean().item()
+    if p > 0:
+        assert abs(keep - (1 - p)) < 0.02, keep
+    else:
+        assert keep == 1.0
+
+    x32 = x.float().requires_grad_(True)
+    z32 = z.float().requires_grad_(True)
+    w32 = w.float().requires_grad_(True)
+    b32 = b.float().requires_grad_(True)
+    h = x32 + z32 * mask.float() / (1 - p)
+    ref = torch.nn.functional.layer_norm(h, (d,), w32, b32, 1e-5)
+    assert (y.float() - ref).abs().max() < 0.06, (y.float() - ref).abs().max()
+    sd = (s.float() - h.detach()).abs().max()
+    assert sd < 0.01 * h.abs().max() + 0.02, sd  # s is bf16-rounded
+
+    dy = torch.randn_like(y)
+    ref.backward(dy.float())
+    dx, dz, dw, db = ext.dropout_add_ln_bwd(dy, s, w, mask, mean, rstd, p)
+    assert (dx.float() - x32.grad).abs().max() < 0.06
+    assert (dz.float() - z32.grad).abs().max() < 0.1
+    assert (dw.float() - w32.grad).abs().max() / w32.grad.abs().max() < 0.05
+    assert (db.float() - b32.grad).abs().max() / b32.grad.abs().max() < 0.05
+
+
+@gpu
+def test_bert_fused_junction_trains():
+    """BERT layer with the fused junctions: one train step, finite loss,
+    and dropout actually drops (two training forwards differ)."""
+    from accelerate_amd.models import BertConfig, BertForSequenceClassification
+
+    torch.manual_seed(0)
+    model = BertForSequenceClassification(BertConfig(num_hidden_layers=2)).cuda().to(torch.bfloat16)
+    ids = torch.randint(0, 30522, (4, 64), device="cuda")
+    labels = torch.randint(0, 2, (4,), device="cuda")
+    out = model(ids, labels=labels)
+    out["loss"].backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(out["loss"])
+    l1 = model(ids, labels=labels)["loss"]
+    l2 = model(ids, labels=labels)["loss"]
+    assert not torch.equal(l1, l2)  # training-mode dropout is live
