@@ -16,8 +16,8 @@ OUT = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), 
 def main():
     tunable.enable(True)
     tunable.tuning_enable(True)
-    tunable.set_max_tuning_duration(30)
-    tunable.set_max_tuning_iterations(100)
+    tunable.set_max_tuning_duration(10)
+    tunable.set_max_tuning_iterations(30)
 
     from accelerate_amd import Accelerator, set_seed
     from accelerate_amd.models import BertConfig, BertForSequenceClassification
