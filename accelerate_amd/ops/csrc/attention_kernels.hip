@@ -302,7 +302,13 @@ __device__ __forceinline__ unsigned pack_bf16_pair(float lo, float hi) {
 }
 }  // namespace
 
-template <int D>
+typedef abf16 bf16x4t __attribute__((ext_vector_type(4)));
+
+// TR=true stores the V tile as [k/4][d/16][4][16] subtiles and reads PV's
+// B-fragments with ds_read_b64_tr_b16 (hardware transpose; each 16-lane
+// group's addresses tile one 4x16 subtile window — fa_swapped_probe
+// stage E), replacing the pair-packed V^T staging scatter.
+template <int D, bool TR>
 __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
     const abf16* __restrict__ q, const abf16* __restrict__ k,
     const abf16* __restrict__ v, ushort* __restrict__ out,
@@ -375,10 +381,19 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
 #pragma unroll
     for (int it = 0; it < 2; ++it)
       *reinterpret_cast<bf16x8a*>(Ks + (kr_row + it * (512 / (D / 8))) * KP + kr_col8) = krg[it];
+    if (TR) {
+      // subtiled [k/4][d/16][4][16]: natural vectorized stores, no packing
+      const int db = (kr_col8 >> 4) * 64 + (kr_col8 & 15);
+      *reinterpret_cast<bf16x8a*>(VTs + ((vr_r0 >> 2) * (D / 16)) * 64 + db +
+                                  (vr_r0 & 3) * 16) = vrg0;
+      *reinterpret_cast<bf16x8a*>(VTs + (((vr_r0 + 1) >> 2) * (D / 16)) * 64 + db +
+                                  ((vr_r0 + 1) & 3) * 16) = vrg1;
+    } else {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const unsigned pk = bfbits(vrg0[j]) | (bfbits(vrg1[j]) << 16);
-      *reinterpret_cast<unsigned*>(VTs + (kr_col8 + j) * NP + vr_r0) = pk;
+      for (int j = 0; j < 8; ++j) {
+        const unsigned pk = bfbits(vrg0[j]) | (bfbits(vrg1[j]) << 16);
+        *reinterpret_cast<unsigned*>(VTs + (kr_col8 + j) * NP + vr_r0) = pk;
+      }
     }
   };
 
@@ -485,8 +500,19 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
       for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
-          bf16x8a vf = *reinterpret_cast<const bf16x8a*>(
-              VTs + (dt * 32 + col) * NP + ks * 16 + hi * 8);
+          bf16x8a vf;
+          if (TR) {
+            const unsigned a0 =
+                64u * (32 * ks + 16 * hi + 2 * dt + ((lane >> 4) & 1)) + (lane & 15) * 4;
+            bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (__attribute__((address_space(3))) bf16x4t*)(VTs + a0));
+            bf16x4t hh = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+                (__attribute__((address_space(3))) bf16x4t*)(VTs + a0 + 512));
+            vf = bf16x8a{lo[0], lo[1], lo[2], lo[3], hh[0], hh[1], hh[2], hh[3]};
+          } else {
+            vf = *reinterpret_cast<const bf16x8a*>(
+                VTs + (dt * 32 + col) * NP + ks * 16 + hi * 8);
+          }
           oacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<bf16x8a*>(pa[ks]), vf, oacc[dt], 0, 0, 0);
         }
@@ -522,9 +548,12 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
     lb[qr_lse] = (m_run + __log2f(fmaxf(l_run, 1e-30f))) * kLn2;
 }
 
-template __global__ void fa_fwd_swapped_kernel<128>(const abf16*, const abf16*, const abf16*,
-                                                    ushort*, float*, int, int, int, int, float,
-                                                    int, int, Str3, Str3, Str3, Str3);
+template __global__ void fa_fwd_swapped_kernel<128, false>(
+    const abf16*, const abf16*, const abf16*, ushort*, float*, int, int, int, int, float,
+    int, int, Str3, Str3, Str3, Str3);
+template __global__ void fa_fwd_swapped_kernel<128, true>(
+    const abf16*, const abf16*, const abf16*, ushort*, float*, int, int, int, int, float,
+    int, int, Str3, Str3, Str3, Str3);
 
 template __global__ void fa_fwd_kernel<64>(const abf16*, const abf16*, const abf16*,
                                            ushort*, float*, int, int, int, int, float,
@@ -1652,7 +1681,8 @@ static int fa_fwd_impl() {
     const char* e = getenv("ACCELERATE_AMD_FA_FWD");
     if (e && strcmp(e, "legacy") == 0) return 0;
     if (e && strcmp(e, "swapped") == 0) return 1;
-    return 1;  // swapped ladder: 652 vs 437 TF/s (gpurun_out/fa_cmp3.log)
+    if (e && strcmp(e, "swapped_tr") == 0) return 2;
+    return 2;  // tr_read V path: 782 vs 652 TF/s (gpurun_out/fa_tr2.log)
   }();
   return impl;
 }
@@ -1661,18 +1691,19 @@ template <int D>
 hipError_t launch_impl(const void* q, const void* k, const void* v, void* out, float* lse,
                        int64_t bh, int Sq, int Sk, int past, int causal, float scale,
                        int Hq, int Hkv, const Str3* strides, hipStream_t stream) {
-  if (D == 128 && fa_fwd_impl() == 1) {
+  if (D == 128 && fa_fwd_impl() >= 1) {
     constexpr int lds_sw = (64 * (D + 8) + D * (64 + 8)) * 2;
     static bool attr_sw = false;
+    auto kfn = fa_fwd_impl() == 2 ? &fa_fwd_swapped_kernel<128, true>
+                                  : &fa_fwd_swapped_kernel<128, false>;
     if (!attr_sw) {
-      hipError_t e =
-          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_fwd_swapped_kernel<128>),
-                              hipFuncAttributeMaxDynamicSharedMemorySize, lds_sw);
+      hipError_t e = hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),
+                                         hipFuncAttributeMaxDynamicSharedMemorySize, lds_sw);
       if (e != hipSuccess) return e;
       attr_sw = true;
     }
     dim3 grid((Sq + 255) / 256, (unsigned)bh);
-    hipLaunchKernelGGL(fa_fwd_swapped_kernel<128>, grid, dim3(512), lds_sw, stream,
+    hipLaunchKernelGGL(kfn, grid, dim3(512), lds_sw, stream,
                        reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
                        reinterpret_cast<const abf16*>(v), reinterpret_cast<ushort*>(out),
                        lse, Sq, Sk, past, causal, scale, Hq, Hkv,

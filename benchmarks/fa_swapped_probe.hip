@@ -118,6 +118,54 @@ __global__ void probe_kernel(const abf16* Q, const abf16* K, const abf16* V,
     }
 }
 
+// ---------------------------------------------------------------------------
+// Stage E (separate kernel): ds_read_b64_tr_b16 B-fragment path for PV.
+// V stored in LDS as [k/4][d/16][4][16] subtiles; each 16-lane group's
+// addresses tile one 128 B subtile window (lane slot = (l&15)*8 B) and the
+// transpose delivers column (l&15): lane l gets
+// V[k0 + 8*(l>>5) + j][d0 + (l&31)] for j=0..3 (second read +512 elements
+// covers j=4..7). Verified against the direct loads.
+// ---------------------------------------------------------------------------
+
+typedef __bf16 bf16x4p __attribute__((ext_vector_type(4)));
+
+__global__ void probe_tr_kernel(const abf16* V, float* err_count) {
+  constexpr int D = 128;
+  __shared__ abf16 Vt[64 * 128];
+  const int lane = threadIdx.x;
+  const int hi = lane >> 5;
+  const int col = lane & 31;
+  // stage V[64][128] into subtiled layout
+  for (int i = lane; i < 64 * (D / 8); i += 64) {
+    const int k = i / (D / 8);
+    const int d0 = (i % (D / 8)) * 8;
+    const int base = ((k >> 2) * 8 + (d0 >> 4)) * 64 + (k & 3) * 16 + (d0 & 15);
+    *reinterpret_cast<bf16x8*>(Vt + base) = *reinterpret_cast<const bf16x8*>(V + k * D + d0);
+  }
+  __syncthreads();
+  int bad = 0;
+  for (int dt = 0; dt < D / 32; ++dt)
+    for (int ks = 0; ks < 4; ++ks) {
+      const int k0 = ks * 16;
+      const int d0 = dt * 32;
+      // per-lane slot: 16-lane groups tile a 128 B window; lane l reads
+      // column (l&15) of subtile (kb = k0/4 + 2*hi, db = d0/16 + ((l>>4)&1))
+      const unsigned addr_elems =
+          64u * ((k0 / 4 + 2 * hi) * 8 + d0 / 16 + ((lane >> 4) & 1)) + (lane & 15) * 4;
+      auto p0 = (__attribute__((address_space(3))) bf16x4p*)(Vt + addr_elems);
+      auto p1 = (__attribute__((address_space(3))) bf16x4p*)(Vt + addr_elems + 512);
+      bf16x4p lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
+      bf16x4p hi4 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+      for (int j = 0; j < 4; ++j) {
+        float want_lo = (float)V[(k0 + hi * 8 + j) * D + d0 + col];
+        float want_hi = (float)V[(k0 + 4 + hi * 8 + j) * D + d0 + col];
+        if ((float)lo[j] != want_lo) ++bad;
+        if ((float)hi4[j] != want_hi) ++bad;
+      }
+    }
+  atomicAdd(err_count, (float)bad);
+}
+
 int main() {
   constexpr int D = 128;
   srand(7);
@@ -135,8 +183,13 @@ int main() {
   for (int i = 0; i < 64 * D; ++i) V[i] = (abf16)frand();
 
   hipLaunchKernelGGL(probe_kernel, dim3(1), dim3(64), 0, 0, Q, K, V, S, P, O);
+  float* errc;
+  hipMallocManaged(&errc, sizeof(float));
+  *errc = 0.f;
+  hipLaunchKernelGGL(probe_tr_kernel, dim3(1), dim3(64), 0, 0, V, errc);
   hipError_t err = hipDeviceSynchronize();
   if (err != hipSuccess) { printf("HIP ERROR: %s\n", hipGetErrorString(err)); return 2; }
+  printf("stage E (tr_read V subtile): %s (%g bad)\n", *errc == 0.f ? "PASS" : "FAIL", *errc);
 
   // host references
   int bad = 0;
@@ -174,3 +227,4 @@ int main() {
   printf("stage D (PV via PA frags): %s\n", bado ? "FAIL" : "PASS");
   return (bad || badp || bado) ? 1 : 0;
 }
+
