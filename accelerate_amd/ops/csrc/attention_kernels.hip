@@ -1304,9 +1304,9 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
   extern __shared__ char smem[];
   abf16* Qs = reinterpret_cast<abf16*>(smem);  // [BQ][KP]
   abf16* DOs = Qs + BQ * KP;                   // [BQ][KP]
-  abf16* QTs = DOs + BQ * KP;                  // [D][NP]
-  abf16* DOTs = QTs + D * NP;                  // [D][NP]
-  float* lse_s = reinterpret_cast<float*>(DOTs + D * NP);  // [BQ] (base-2)
+  abf16* QTs = DOs + BQ * KP;                  // Q subtiled [q/4][d/16][4][16]
+  abf16* DOTs = QTs + BQ * D;                  // dOut subtiled likewise
+  float* lse_s = reinterpret_cast<float*>(DOTs + BQ * D);  // [BQ] (base-2)
   float* drow_s = lse_s + BQ;                              // [BQ]
 
   const int tid = threadIdx.x;
@@ -1366,11 +1366,10 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
       abf16* dst = tsecond ? DOTs : QTs;
       bf16x8a a0 = *reinterpret_cast<const bf16x8a*>(src + qr0 * ss + col8t);
       bf16x8a a1 = *reinterpret_cast<const bf16x8a*>(src + qr1 * ss + col8t);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const unsigned pk = bfbits(a0[j]) | (bfbits(a1[j]) << 16);
-        *reinterpret_cast<unsigned*>(dst + (col8t + j) * NP + r0) = pk;
-      }
+      const int dbq = (col8t >> 4) * 64 + (col8t & 15);
+      *reinterpret_cast<bf16x8a*>(dst + ((r0 >> 2) * (D / 16)) * 64 + dbq + (r0 & 3) * 16) = a0;
+      *reinterpret_cast<bf16x8a*>(dst + (((r0 + 1) >> 2) * (D / 16)) * 64 + dbq +
+                                  ((r0 + 1) & 3) * 16) = a1;
       if (tid < BQ) {
         lse_s[tid] = lse[bh * Sq + min(qt0 + tid, Sq - 1)] * kLog2e;
         drow_s[tid] = drow[bh * Sq + min(qt0 + tid, Sq - 1)];
@@ -1422,8 +1421,13 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
       for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
         for (int qs = 0; qs < 2; ++qs) {
-          bf16x8a dof = *reinterpret_cast<const bf16x8a*>(
-              DOTs + (dt * 32 + col) * NP + qs * 16 + hi * 8);
+          const unsigned a0 =
+              64u * (32 * qs + 16 * hi + 2 * dt + ((lane >> 4) & 1)) + (lane & 15) * 4;
+          bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (__attribute__((address_space(3))) bf16x4t*)(DOTs + a0));
+          bf16x4t hh = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (__attribute__((address_space(3))) bf16x4t*)(DOTs + a0 + 512));
+          bf16x8a dof = bf16x8a{lo[0], lo[1], lo[2], lo[3], hh[0], hh[1], hh[2], hh[3]};
           dvacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<bf16x8a*>(pa[qs]), dof, dvacc[dt], 0, 0, 0);
         }
@@ -1460,8 +1464,13 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
       for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
         for (int qs = 0; qs < 2; ++qs) {
-          bf16x8a qtf = *reinterpret_cast<const bf16x8a*>(
-              QTs + (dt * 32 + col) * NP + qs * 16 + hi * 8);
+          const unsigned a0 =
+              64u * (32 * qs + 16 * hi + 2 * dt + ((lane >> 4) & 1)) + (lane & 15) * 4;
+          bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (__attribute__((address_space(3))) bf16x4t*)(QTs + a0));
+          bf16x4t hh = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (__attribute__((address_space(3))) bf16x4t*)(QTs + a0 + 512));
+          bf16x8a qtf = bf16x8a{lo[0], lo[1], lo[2], lo[3], hh[0], hh[1], hh[2], hh[3]};
           dkacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<bf16x8a*>(pa[qs]), qtf, dkacc[dt], 0, 0, 0);
         }
@@ -1509,7 +1518,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
   extern __shared__ char smem[];
   abf16* Ks = reinterpret_cast<abf16*>(smem);  // [BN][KP] row-major
   abf16* Vs = Ks + BN * KP;                    // [BN][KP] row-major
-  abf16* KTs = Vs + BN * KP;                   // [D][NP]  K^T
+  abf16* KTs = Vs + BN * KP;                   // K subtiled [k/4][d/16][4][16] (tr_read)
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -1566,11 +1575,10 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
       if (key1 >= Sk) key1 = Sk - 1;
       bf16x8a k0 = *reinterpret_cast<const bf16x8a*>(kb + key0 * sK.s + col8);
       bf16x8a k1 = *reinterpret_cast<const bf16x8a*>(kb + key1 * sK.s + col8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const unsigned pk = bfbits(k0[j]) | (bfbits(k1[j]) << 16);
-        *reinterpret_cast<unsigned*>(KTs + (col8 + j) * NP + r0) = pk;
-      }
+      const int dbq = (col8 >> 4) * 64 + (col8 & 15);
+      *reinterpret_cast<bf16x8a*>(KTs + ((r0 >> 2) * (D / 16)) * 64 + dbq + (r0 & 3) * 16) = k0;
+      *reinterpret_cast<bf16x8a*>(KTs + (((r0 + 1) >> 2) * (D / 16)) * 64 + dbq +
+                                  ((r0 + 1) & 3) * 16) = k1;
     }
     __syncthreads();
 
@@ -1630,8 +1638,13 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
       for (int dt = 0; dt < D / 32; ++dt)
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
-          bf16x8a ktf = *reinterpret_cast<const bf16x8a*>(
-              KTs + (dt * 32 + col) * NP + ks * 16 + hi * 8);
+          const unsigned a0 =
+              64u * (32 * ks + 16 * hi + 2 * dt + ((lane >> 4) & 1)) + (lane & 15) * 4;
+          bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (__attribute__((address_space(3))) bf16x4t*)(KTs + a0));
+          bf16x4t hh = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+              (__attribute__((address_space(3))) bf16x4t*)(KTs + a0 + 512));
+          bf16x8a ktf = bf16x8a{lo[0], lo[1], lo[2], lo[3], hh[0], hh[1], hh[2], hh[3]};
           dqacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<bf16x8a*>(pa[ks]), ktf, dqacc[dt], 0, 0, 0);
         }
@@ -1810,7 +1823,7 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
                        lse, drow, reinterpret_cast<ushort*>(dk), Sq, Sk, past, causal, scale,
                        Hq, Hkv, strides[0], strides[1], strides[2], strides[3]);
   } else if (dkdv_mode == 2) {
-    constexpr int lds_f = (2 * 32 * (D + 8) + 2 * D * (32 + 8)) * 2 + 2 * 32 * 4;
+    constexpr int lds_f = (2 * 32 * (D + 8) + 2 * 32 * D) * 2 + 2 * 32 * 4;
     static bool attr_f = false;
     if (!attr_f) {
       hipError_t e =
@@ -1839,7 +1852,7 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
   const char* dq_env = getenv("ACCELERATE_AMD_FA_BWD_DQ");
   const bool dq_swapped = (D == 128) && !(dq_env && strcmp(dq_env, "legacy") == 0);
   if (dq_swapped) {
-    constexpr int lds_dq_sw = (2 * 64 * (D + 8) + D * (64 + 8)) * 2;
+    constexpr int lds_dq_sw = (2 * 64 * (D + 8) + 64 * D) * 2;
     static bool attr_sw = false;
     if (!attr_sw) {
       hipError_t e =
