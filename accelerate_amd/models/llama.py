@@ -135,19 +135,27 @@ class LlamaAttention(nn.Module):
 
             ctx = flash_attention(q, k, v, causal=True)
         else:
-            # math path (decode / CPU): expand kv heads
-            if self.n_kv != self.n_heads:
-                rep = self.n_heads // self.n_kv
-                k = k.repeat_interleave(rep, dim=1)
-                v = v.repeat_interleave(rep, dim=1)
+            # math path (decode / CPU). GQA via batched-GEMM broadcasting:
+            # q grouped [B, Hkv, rep*S, D] against the UNexpanded cache —
+            # no repeat_interleave copy of K/V per layer per token
             scale = 1.0 / math.sqrt(self.head_dim)
-            scores = torch.matmul(q, k.transpose(-1, -2)) * scale
+            rep = self.n_heads // self.n_kv
             total = k.shape[2]
+            if rep > 1:
+                qg = q.reshape(B, self.n_kv, rep * S, self.head_dim)
+                scores = torch.matmul(qg, k.transpose(-1, -2)) * scale
+                scores = scores.view(B, self.n_heads, S, total)
+            else:
+                scores = torch.matmul(q, k.transpose(-1, -2)) * scale
             if S > 1:
                 causal = torch.ones(S, total, dtype=torch.bool, device=x.device).tril(diagonal=past_len)
                 scores = scores.masked_fill(~causal, torch.finfo(scores.dtype).min)
             probs = F.softmax(scores.float(), dim=-1).to(q.dtype)
-            ctx = torch.matmul(probs, v)
+            if rep > 1:
+                ctx = torch.matmul(probs.view(B, self.n_kv, rep * S, total), v)
+                ctx = ctx.view(B, self.n_heads, S, self.head_dim)
+            else:
+                ctx = torch.matmul(probs, v)
         ctx = ctx.transpose(1, 2).reshape(B, S, -1)
         return self.o_proj(ctx)
 

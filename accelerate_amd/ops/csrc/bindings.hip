@@ -225,6 +225,7 @@ extern "C" hipError_t launch_layernorm_bwd(const void*, const void*, const void*
                                            const void*, void*, void*, void*, long long, int, int,
                                            hipStream_t);
 __global__ void rmsnorm_fwd_bf16(const ushort*, const ushort*, ushort*, float*, int64_t, int, float);
+__global__ void rmsnorm_fwd_wide(const ushort*, const ushort*, ushort*, float*, int64_t, int, float);
 extern "C" hipError_t launch_rmsnorm_bwd(const void*, const void*, const void*, const void*,
                                          void*, void*, long long, int, int, hipStream_t);
 
@@ -288,8 +289,12 @@ std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
   auto y = at::empty_like(x);
   auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(rmsnorm_fwd_bf16, dim3(rows), dim3(64), 0, stream.stream(),
-                     bfp(x), bfp(w), bfp_mut(y), rstd.data_ptr<float>(), rows, d, (float)eps);
+  if (rows < 4)  // decode shapes: 256 threads cooperate per row
+    hipLaunchKernelGGL(rmsnorm_fwd_wide, dim3(rows), dim3(256), 0, stream.stream(),
+                       bfp(x), bfp(w), bfp_mut(y), rstd.data_ptr<float>(), rows, d, (float)eps);
+  else
+    hipLaunchKernelGGL(rmsnorm_fwd_bf16, dim3(rows), dim3(64), 0, stream.stream(),
+                       bfp(x), bfp(w), bfp_mut(y), rstd.data_ptr<float>(), rows, d, (float)eps);
   return {y, rstd};
 }
 

@@ -164,3 +164,20 @@ def test_bert_fused_junction_trains():
     l1 = model(ids, labels=labels)["loss"]
     l2 = model(ids, labels=labels)["loss"]
     assert not torch.equal(l1, l2)  # training-mode dropout is live
+
+
+@gpu
+def test_rmsnorm_decode_shape_matches():
+    """rows<4 cooperative path (decode) matches the reference math."""
+    from accelerate_amd.ops.norms import FusedRMSNorm
+
+    torch.manual_seed(0)
+    for rows, d in [(1, 8192), (2, 4096), (3, 768)]:
+        x32 = torch.randn(rows, d, device="cuda")
+        fused = FusedRMSNorm(d).cuda().to(torch.bfloat16)
+        with torch.no_grad():
+            w = torch.randn(d, device="cuda") * 0.1 + 1.0
+            fused.weight.copy_(w.to(torch.bfloat16))
+        y = fused(x32.to(torch.bfloat16))
+        ref = w * (x32 * torch.rsqrt(x32.pow(2).mean(-1, keepdim=True) + 1e-5))
+        assert (y.float() - ref).abs().max() < 0.05, (rows, d)
