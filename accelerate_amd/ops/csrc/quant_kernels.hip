@@ -97,7 +97,10 @@ __global__ void w8a16_gemv_kernel(const char* __restrict__ q,
   const char* wrow = q + row * cols;
   for (int b = 0; b < batch; ++b) {
     const ushort* xb = x + (int64_t)b * cols;
-    float acc = 0.f;
+    // two independent accumulators + 2x unroll: the single-acc version's
+    // 16-deep FMA chain per 16 B made the loop latency-bound (~1.9 TB/s)
+    float acc0 = 0.f, acc1 = 0.f;
+#pragma unroll 2
     for (int64_t k0 = (int64_t)lane * 16; k0 < cols; k0 += 64 * 16) {
       char16_t_v w16 = *reinterpret_cast<const char16_t_v*>(wrow + k0);
       ushort8 xa = *reinterpret_cast<const ushort8*>(xb + k0);
@@ -105,10 +108,11 @@ __global__ void w8a16_gemv_kernel(const char* __restrict__ q,
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
         ushort ua = xa[k], uc = xc[k];
-        acc += (float)w16[k] * __bfloat162float(*reinterpret_cast<__hip_bfloat16*>(&ua));
-        acc += (float)w16[k + 8] * __bfloat162float(*reinterpret_cast<__hip_bfloat16*>(&uc));
+        acc0 += (float)w16[k] * __bfloat162float(*reinterpret_cast<__hip_bfloat16*>(&ua));
+        acc1 += (float)w16[k + 8] * __bfloat162float(*reinterpret_cast<__hip_bfloat16*>(&uc));
       }
     }
+    float acc = acc0 + acc1;
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
     if (lane == 0) {
