@@ -343,16 +343,23 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
   // q0 + wave*32 + col, chunk t covers d = t*16 + hi*8 + 0..7
   int qr_mine = q0 + wave * 32 + col;
   if (qr_mine >= Sq) qr_mine = Sq - 1;  // clamp; padded rows never stored
+  // fold c = scale*log2e INTO the Q fragments once (64 converts here)
+  // instead of scaling every score every tile (32 VALU/lane/tile): S then
+  // lands directly in the base-2 domain. bf16 re-round only shifts the
+  // exponent (c is a pure magnitude), so precision is unchanged.
+  const float c = scale * kLog2e;
   bf16x8a qf[D / 16];
 #pragma unroll
-  for (int t = 0; t < D / 16; ++t)
-    qf[t] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr_mine * sQ.s + t * 16 + hi * 8);
+  for (int t = 0; t < D / 16; ++t) {
+    bf16x8a raw = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr_mine * sQ.s + t * 16 + hi * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) raw[j] = (abf16)((float)raw[j] * c);
+    qf[t] = raw;
+  }
 
   f32x16a oacc[D / 32] = {};
   float m_run = -INFINITY;  // base-2 scaled domain, row = col
   float l_run = 0.f;        // THIS lane's half-row partial denominator
-
-  const float c = scale * kLog2e;
   const int k_hi = causal ? min(Sk, past + q0 + BM) : Sk;
   // this wave's own causal horizon: rows q0+wave*32 .. +32
   const int k_hi_wave = causal ? min(Sk, past + q0 + wave * 32 + 32) : Sk;
@@ -440,7 +447,6 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
       for (int r = 1; r < 16; ++r) pmax = fmaxf(pmax, sacc[0][r]);
 #pragma unroll
       for (int r = 0; r < 16; ++r) pmax = fmaxf(pmax, sacc[1][r]);
-      pmax *= c;  // c > 0: max commutes with the base-2 scale fold
       {
         uint2a sw = __builtin_amdgcn_permlane32_swap(__float_as_uint(pmax),
                                                      __float_as_uint(pmax), false, false);
@@ -470,7 +476,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
       for (int kt = 0; kt < 2; ++kt)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float pv = exp2f(sacc[kt][r] * c - m_run);
+          const float pv = exp2f(sacc[kt][r] - m_run);
           sacc[kt][r] = pv;
           rs += pv;
         }
@@ -1541,7 +1547,9 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
     qf[t] = *reinterpret_cast<const bf16x8a*>(qb + (int64_t)qr_mine * sQ.s + t * 16 + hi * 8);
     dof[t] = *reinterpret_cast<const bf16x8a*>(dob + (int64_t)qr_mine * sDo.s + t * 16 + hi * 8);
   }
-  const float lse2 = lse[bh * Sq + qr_mine] * kLog2e;  // base-2 domain
+  // scale folds into the exp argument: dS = P*(dP-D)*scale
+  //   = exp2(s*c - lse*log2e + log2(scale)) * (dP - D)
+  const float lse2 = lse[bh * Sq + qr_mine] * kLog2e - __log2f(scale);
   const float dr_m = drow[bh * Sq + qr_mine];
 
   f32x16a dqacc[D / 32] = {};
@@ -1612,7 +1620,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
             const int kabs = kbase + (r & 3) + 8 * (r >> 2);
             if (kabs >= Sk || (causal && kabs > past + qabs)) pv = 0.f;
           }
-          sacc[kt][r] = pv * (dpacc[kt][r] - dr_m) * scale;
+          sacc[kt][r] = pv * (dpacc[kt][r] - dr_m);
         }
       }
 
