@@ -314,7 +314,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
     const abf16* __restrict__ v, ushort* __restrict__ out,
     float* __restrict__ lse, int Sq, int Sk, int past, int causal,
     float scale, int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sO) {
-  static_assert(D == 128, "swapped ladder is tuned for head_dim 128");
+  static_assert(D == 64 || D == 128, "swapped ladder supports head_dim 64/128");
   constexpr int BM = 256, BN = 64;
   // +8 padding measured BETTER than the XOR-chunk swizzle here (630 vs
   // 591 TF/s non-causal): the b128 fragment reads are 4-way-conflict
@@ -370,36 +370,45 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
   const int kr_row = tid / (D / 8);
   const int kr_col8 = (tid % (D / 8)) * 8;
   const int vr_r0 = (tid / (D / 8)) * 2;
-  bf16x8a krg[2], vrg0, vrg1;
+  // K tile = BN*(D/8) b128 chunks over 512 threads: 2 per thread at D=128,
+  // 1 at D=64 (a second pass would walk past the 64-key tile)
+  constexpr int KIT = BN * (D / 8) / 512;
+  bf16x8a krg[KIT], vrg0, vrg1;
   auto issue_loads = [&](int kt0) {
 #pragma unroll
-    for (int it = 0; it < 2; ++it) {
+    for (int it = 0; it < KIT; ++it) {
       int64_t key = kt0 + kr_row + it * (512 / (D / 8));
       if (key >= Sk) key = Sk - 1;  // clamp (masked below)
       krg[it] = *reinterpret_cast<const bf16x8a*>(kb + key * sK.s + kr_col8);
     }
-    int64_t key0 = kt0 + vr_r0, key1 = kt0 + vr_r0 + 1;
-    if (key0 >= Sk) key0 = Sk - 1;
-    if (key1 >= Sk) key1 = Sk - 1;
-    vrg0 = *reinterpret_cast<const bf16x8a*>(vb + key0 * sV.s + kr_col8);
-    vrg1 = *reinterpret_cast<const bf16x8a*>(vb + key1 * sV.s + kr_col8);
+    if (vr_r0 < 64) {  // D=64: only BN/2*(D/8)=256 pair-chunks
+      int64_t key0 = kt0 + vr_r0, key1 = kt0 + vr_r0 + 1;
+      if (key0 >= Sk) key0 = Sk - 1;
+      if (key1 >= Sk) key1 = Sk - 1;
+      vrg0 = *reinterpret_cast<const bf16x8a*>(vb + key0 * sV.s + kr_col8);
+      vrg1 = *reinterpret_cast<const bf16x8a*>(vb + key1 * sV.s + kr_col8);
+    }
   };
   auto write_tile = [&]() {
 #pragma unroll
-    for (int it = 0; it < 2; ++it)
+    for (int it = 0; it < KIT; ++it)
       *reinterpret_cast<bf16x8a*>(Ks + (kr_row + it * (512 / (D / 8))) * KP + kr_col8) = krg[it];
     if (TR) {
       // subtiled [k/4][d/16][4][16]: natural vectorized stores, no packing
-      const int db = (kr_col8 >> 4) * 64 + (kr_col8 & 15);
-      *reinterpret_cast<bf16x8a*>(VTs + ((vr_r0 >> 2) * (D / 16)) * 64 + db +
-                                  (vr_r0 & 3) * 16) = vrg0;
-      *reinterpret_cast<bf16x8a*>(VTs + (((vr_r0 + 1) >> 2) * (D / 16)) * 64 + db +
-                                  ((vr_r0 + 1) & 3) * 16) = vrg1;
+      if (vr_r0 < 64) {
+        const int db = (kr_col8 >> 4) * 64 + (kr_col8 & 15);
+        *reinterpret_cast<bf16x8a*>(VTs + ((vr_r0 >> 2) * (D / 16)) * 64 + db +
+                                    (vr_r0 & 3) * 16) = vrg0;
+        *reinterpret_cast<bf16x8a*>(VTs + (((vr_r0 + 1) >> 2) * (D / 16)) * 64 + db +
+                                    ((vr_r0 + 1) & 3) * 16) = vrg1;
+      }
     } else {
+      if (vr_r0 < 64) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const unsigned pk = bfbits(vrg0[j]) | (bfbits(vrg1[j]) << 16);
-        *reinterpret_cast<unsigned*>(VTs + (kr_col8 + j) * NP + vr_r0) = pk;
+        for (int j = 0; j < 8; ++j) {
+          const unsigned pk = bfbits(vrg0[j]) | (bfbits(vrg1[j]) << 16);
+          *reinterpret_cast<unsigned*>(VTs + (kr_col8 + j) * NP + vr_r0) = pk;
+        }
       }
     }
   };
@@ -509,11 +518,12 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
           bf16x8a vf;
           if (TR) {
             const unsigned a0 =
-                64u * (32 * ks + 16 * hi + 2 * dt + ((lane >> 4) & 1)) + (lane & 15) * 4;
+                64u * ((4 * ks + 2 * hi) * (D / 16) + 2 * dt + ((lane >> 4) & 1)) +
+                (lane & 15) * 4;
             bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
                 (__attribute__((address_space(3))) bf16x4t*)(VTs + a0));
             bf16x4t hh = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-                (__attribute__((address_space(3))) bf16x4t*)(VTs + a0 + 512));
+                (__attribute__((address_space(3))) bf16x4t*)(VTs + a0 + 64 * (D / 16)));
             vf = bf16x8a{lo[0], lo[1], lo[2], lo[3], hh[0], hh[1], hh[2], hh[3]};
           } else {
             vf = *reinterpret_cast<const bf16x8a*>(
@@ -558,6 +568,12 @@ template __global__ void fa_fwd_swapped_kernel<128, false>(
     const abf16*, const abf16*, const abf16*, ushort*, float*, int, int, int, int, float,
     int, int, Str3, Str3, Str3, Str3);
 template __global__ void fa_fwd_swapped_kernel<128, true>(
+    const abf16*, const abf16*, const abf16*, ushort*, float*, int, int, int, int, float,
+    int, int, Str3, Str3, Str3, Str3);
+template __global__ void fa_fwd_swapped_kernel<64, false>(
+    const abf16*, const abf16*, const abf16*, ushort*, float*, int, int, int, int, float,
+    int, int, Str3, Str3, Str3, Str3);
+template __global__ void fa_fwd_swapped_kernel<64, true>(
     const abf16*, const abf16*, const abf16*, ushort*, float*, int, int, int, int, float,
     int, int, Str3, Str3, Str3, Str3);
 
@@ -1303,7 +1319,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
     ushort* __restrict__ dk, ushort* __restrict__ dv,
     int Sq, int Sk, int past, int causal, float scale,
     int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sDo) {
-  static_assert(D == 128);
+  static_assert(D == 64 || D == 128);
   constexpr int BK = 256, BQ = 32;
   constexpr int KP = D + 8;
   constexpr int NP = BQ + 8;
@@ -1350,6 +1366,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
         const bool second = it == 1;
         const int r = tid / (D / 8);
         const int col8 = (tid % (D / 8)) * 8;
+        if (r >= BQ) break;  // D=64: BQ*(D/8)=256 chunks per tensor
         int64_t qrow = qt0 + r;
         if (qrow >= Sq) qrow = Sq - 1;
         if (second)
@@ -1364,18 +1381,20 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
       const int tp = tid & 255;
       const int r0 = (tp / (D / 8)) * 2;
       const int col8t = (tp % (D / 8)) * 8;
-      int64_t qr0 = qt0 + r0, qr1 = qt0 + r0 + 1;
-      if (qr0 >= Sq) qr0 = Sq - 1;
-      if (qr1 >= Sq) qr1 = Sq - 1;
-      const abf16* src = tsecond ? dob : qb;
-      const int64_t ss = tsecond ? sDo.s : sQ.s;
-      abf16* dst = tsecond ? DOTs : QTs;
-      bf16x8a a0 = *reinterpret_cast<const bf16x8a*>(src + qr0 * ss + col8t);
-      bf16x8a a1 = *reinterpret_cast<const bf16x8a*>(src + qr1 * ss + col8t);
-      const int dbq = (col8t >> 4) * 64 + (col8t & 15);
-      *reinterpret_cast<bf16x8a*>(dst + ((r0 >> 2) * (D / 16)) * 64 + dbq + (r0 & 3) * 16) = a0;
-      *reinterpret_cast<bf16x8a*>(dst + (((r0 + 1) >> 2) * (D / 16)) * 64 + dbq +
-                                  ((r0 + 1) & 3) * 16) = a1;
+      if (r0 < BQ) {  // D=64: (BQ/2)*(D/8)=128 pair-chunks per tensor
+        int64_t qr0 = qt0 + r0, qr1 = qt0 + r0 + 1;
+        if (qr0 >= Sq) qr0 = Sq - 1;
+        if (qr1 >= Sq) qr1 = Sq - 1;
+        const abf16* src = tsecond ? dob : qb;
+        const int64_t ss = tsecond ? sDo.s : sQ.s;
+        abf16* dst = tsecond ? DOTs : QTs;
+        bf16x8a a0 = *reinterpret_cast<const bf16x8a*>(src + qr0 * ss + col8t);
+        bf16x8a a1 = *reinterpret_cast<const bf16x8a*>(src + qr1 * ss + col8t);
+        const int dbq = (col8t >> 4) * 64 + (col8t & 15);
+        *reinterpret_cast<bf16x8a*>(dst + ((r0 >> 2) * (D / 16)) * 64 + dbq + (r0 & 3) * 16) = a0;
+        *reinterpret_cast<bf16x8a*>(dst + (((r0 + 1) >> 2) * (D / 16)) * 64 + dbq +
+                                    ((r0 + 1) & 3) * 16) = a1;
+      }
       if (tid < BQ) {
         lse_s[tid] = lse[bh * Sq + min(qt0 + tid, Sq - 1)] * kLog2e;
         drow_s[tid] = drow[bh * Sq + min(qt0 + tid, Sq - 1)];
@@ -1428,11 +1447,12 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
 #pragma unroll
         for (int qs = 0; qs < 2; ++qs) {
           const unsigned a0 =
-              64u * (32 * qs + 16 * hi + 2 * dt + ((lane >> 4) & 1)) + (lane & 15) * 4;
+              64u * ((4 * qs + 2 * hi) * (D / 16) + 2 * dt + ((lane >> 4) & 1)) +
+              (lane & 15) * 4;
           bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
               (__attribute__((address_space(3))) bf16x4t*)(DOTs + a0));
           bf16x4t hh = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-              (__attribute__((address_space(3))) bf16x4t*)(DOTs + a0 + 512));
+              (__attribute__((address_space(3))) bf16x4t*)(DOTs + a0 + 64 * (D / 16)));
           bf16x8a dof = bf16x8a{lo[0], lo[1], lo[2], lo[3], hh[0], hh[1], hh[2], hh[3]};
           dvacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<bf16x8a*>(pa[qs]), dof, dvacc[dt], 0, 0, 0);
@@ -1471,11 +1491,12 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
 #pragma unroll
         for (int qs = 0; qs < 2; ++qs) {
           const unsigned a0 =
-              64u * (32 * qs + 16 * hi + 2 * dt + ((lane >> 4) & 1)) + (lane & 15) * 4;
+              64u * ((4 * qs + 2 * hi) * (D / 16) + 2 * dt + ((lane >> 4) & 1)) +
+              (lane & 15) * 4;
           bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
               (__attribute__((address_space(3))) bf16x4t*)(QTs + a0));
           bf16x4t hh = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-              (__attribute__((address_space(3))) bf16x4t*)(QTs + a0 + 512));
+              (__attribute__((address_space(3))) bf16x4t*)(QTs + a0 + 64 * (D / 16)));
           bf16x8a qtf = bf16x8a{lo[0], lo[1], lo[2], lo[3], hh[0], hh[1], hh[2], hh[3]};
           dkacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<bf16x8a*>(pa[qs]), qtf, dkacc[dt], 0, 0, 0);
@@ -1502,6 +1523,9 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
 template __global__ void fa_bwd_dkdv_swapped_kernel<128>(
     const abf16*, const abf16*, const abf16*, const abf16*, const float*, const float*,
     ushort*, ushort*, int, int, int, int, float, int, int, Str3, Str3, Str3, Str3);
+template __global__ void fa_bwd_dkdv_swapped_kernel<64>(
+    const abf16*, const abf16*, const abf16*, const abf16*, const float*, const float*,
+    ushort*, ushort*, int, int, int, int, float, int, int, Str3, Str3, Str3, Str3);
 
 // Swapped-orientation dq (D=128): same 8-wave 32x32 ladder as the forward.
 // Per wave, 32 query rows are lane-resident (Q and dOut B-fragments, lse
@@ -1517,7 +1541,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
     const float* __restrict__ lse, const float* __restrict__ drow,
     ushort* __restrict__ dq, int Sq, int Sk, int past, int causal, float scale,
     int Hq, int Hkv, Str3 sQ, Str3 sK, Str3 sV, Str3 sDo) {
-  static_assert(D == 128, "swapped dq is tuned for head_dim 128");
+  static_assert(D == 64 || D == 128, "swapped dq supports head_dim 64/128");
   constexpr int BM = 256, BN = 64;
   constexpr int KP = D + 8;
   constexpr int NP = BN + 8;
@@ -1578,15 +1602,17 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
       }
       const int r0 = (tid / (D / 8)) * 2;
       const int col8 = (tid % (D / 8)) * 8;
-      int64_t key0 = kb0 + r0, key1 = kb0 + r0 + 1;
-      if (key0 >= Sk) key0 = Sk - 1;
-      if (key1 >= Sk) key1 = Sk - 1;
-      bf16x8a k0 = *reinterpret_cast<const bf16x8a*>(kb + key0 * sK.s + col8);
-      bf16x8a k1 = *reinterpret_cast<const bf16x8a*>(kb + key1 * sK.s + col8);
-      const int dbq = (col8 >> 4) * 64 + (col8 & 15);
-      *reinterpret_cast<bf16x8a*>(KTs + ((r0 >> 2) * (D / 16)) * 64 + dbq + (r0 & 3) * 16) = k0;
-      *reinterpret_cast<bf16x8a*>(KTs + (((r0 + 1) >> 2) * (D / 16)) * 64 + dbq +
-                                  ((r0 + 1) & 3) * 16) = k1;
+      if (r0 < BN) {  // D=64: only BN/2*(D/8)=256 pair-chunks
+        int64_t key0 = kb0 + r0, key1 = kb0 + r0 + 1;
+        if (key0 >= Sk) key0 = Sk - 1;
+        if (key1 >= Sk) key1 = Sk - 1;
+        bf16x8a k0 = *reinterpret_cast<const bf16x8a*>(kb + key0 * sK.s + col8);
+        bf16x8a k1 = *reinterpret_cast<const bf16x8a*>(kb + key1 * sK.s + col8);
+        const int dbq = (col8 >> 4) * 64 + (col8 & 15);
+        *reinterpret_cast<bf16x8a*>(KTs + ((r0 >> 2) * (D / 16)) * 64 + dbq + (r0 & 3) * 16) = k0;
+        *reinterpret_cast<bf16x8a*>(KTs + (((r0 + 1) >> 2) * (D / 16)) * 64 + dbq +
+                                    ((r0 + 1) & 3) * 16) = k1;
+      }
     }
     __syncthreads();
 
@@ -1647,11 +1673,12 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
 #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
           const unsigned a0 =
-              64u * (32 * ks + 16 * hi + 2 * dt + ((lane >> 4) & 1)) + (lane & 15) * 4;
+              64u * ((4 * ks + 2 * hi) * (D / 16) + 2 * dt + ((lane >> 4) & 1)) +
+              (lane & 15) * 4;
           bf16x4t lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
               (__attribute__((address_space(3))) bf16x4t*)(KTs + a0));
           bf16x4t hh = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
-              (__attribute__((address_space(3))) bf16x4t*)(KTs + a0 + 512));
+              (__attribute__((address_space(3))) bf16x4t*)(KTs + a0 + 64 * (D / 16)));
           bf16x8a ktf = bf16x8a{lo[0], lo[1], lo[2], lo[3], hh[0], hh[1], hh[2], hh[3]};
           dqacc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *reinterpret_cast<bf16x8a*>(pa[ks]), ktf, dqacc[dt], 0, 0, 0);
@@ -1676,6 +1703,10 @@ template __global__ void fa_bwd_dq_swapped_kernel<128>(const abf16*, const abf16
                                                        const abf16*, const float*, const float*,
                                                        ushort*, int, int, int, int, float,
                                                        int, int, Str3, Str3, Str3, Str3);
+template __global__ void fa_bwd_dq_swapped_kernel<64>(const abf16*, const abf16*, const abf16*,
+                                                      const abf16*, const float*, const float*,
+                                                      ushort*, int, int, int, int, float,
+                                                      int, int, Str3, Str3, Str3, Str3);
 
 template __global__ void fa_bwd_dq_kernel<64>(const abf16*, const abf16*, const abf16*,
                                               const abf16*, const float*, const float*,
@@ -1712,11 +1743,11 @@ template <int D>
 hipError_t launch_impl(const void* q, const void* k, const void* v, void* out, float* lse,
                        int64_t bh, int Sq, int Sk, int past, int causal, float scale,
                        int Hq, int Hkv, const Str3* strides, hipStream_t stream) {
-  if (D == 128 && fa_fwd_impl() >= 1) {
+  if (fa_fwd_impl() >= 1) {
     constexpr int lds_sw = (64 * (D + 8) + D * (64 + 8)) * 2;
     static bool attr_sw = false;
-    auto kfn = fa_fwd_impl() == 2 ? &fa_fwd_swapped_kernel<128, true>
-                                  : &fa_fwd_swapped_kernel<128, false>;
+    auto kfn = fa_fwd_impl() == 2 ? &fa_fwd_swapped_kernel<D, true>
+                                  : &fa_fwd_swapped_kernel<D, false>;
     if (!attr_sw) {
       hipError_t e = hipFuncSetAttribute(reinterpret_cast<const void*>(kfn),
                                          hipFuncAttributeMaxDynamicSharedMemorySize, lds_sw);
@@ -1801,10 +1832,9 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
   // through a function pointer
   // dkdv impl: fused swapped (default) | split | legacy (env)
   const char* dkdv_env = getenv("ACCELERATE_AMD_FA_BWD_DKDV");
-  const int dkdv_mode = (D != 128) ? 0
-                        : (dkdv_env && strcmp(dkdv_env, "legacy") == 0) ? 0
-                        : (dkdv_env && strcmp(dkdv_env, "split") == 0)  ? 1
-                                                                        : 2;
+  const int dkdv_mode = (dkdv_env && strcmp(dkdv_env, "legacy") == 0) ? 0
+                        : (D == 128 && dkdv_env && strcmp(dkdv_env, "split") == 0) ? 1
+                                                                                   : 2;
   if (dkdv_mode == 1) {
     constexpr int lds_dv = (64 * (D + 8) + D * (64 + 8)) * 2 + 64 * 4;
     constexpr int lds_dk = (2 * 64 * (D + 8) + D * (64 + 8)) * 2 + 2 * 64 * 4;
@@ -1835,12 +1865,12 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
     static bool attr_f = false;
     if (!attr_f) {
       hipError_t e =
-          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dkdv_swapped_kernel<128>),
+          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dkdv_swapped_kernel<D>),
                               hipFuncAttributeMaxDynamicSharedMemorySize, lds_f);
       if (e != hipSuccess) return e;
       attr_f = true;
     }
-    hipLaunchKernelGGL(fa_bwd_dkdv_swapped_kernel<128>, dim3((Sk + 255) / 256, (unsigned)bh),
+    hipLaunchKernelGGL(fa_bwd_dkdv_swapped_kernel<D>, dim3((Sk + 255) / 256, (unsigned)bh),
                        dim3(512), lds_f, stream,
                        reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
                        reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
@@ -1858,18 +1888,18 @@ hipError_t launch_bwd_impl(const void* q, const void* k, const void* v, const vo
                        strides[0], strides[1], strides[2], strides[3]);
   }
   const char* dq_env = getenv("ACCELERATE_AMD_FA_BWD_DQ");
-  const bool dq_swapped = (D == 128) && !(dq_env && strcmp(dq_env, "legacy") == 0);
+  const bool dq_swapped = !(dq_env && strcmp(dq_env, "legacy") == 0);
   if (dq_swapped) {
     constexpr int lds_dq_sw = (2 * 64 * (D + 8) + 64 * D) * 2;
     static bool attr_sw = false;
     if (!attr_sw) {
       hipError_t e =
-          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dq_swapped_kernel<128>),
+          hipFuncSetAttribute(reinterpret_cast<const void*>(&fa_bwd_dq_swapped_kernel<D>),
                               hipFuncAttributeMaxDynamicSharedMemorySize, lds_dq_sw);
       if (e != hipSuccess) return e;
       attr_sw = true;
     }
-    hipLaunchKernelGGL(fa_bwd_dq_swapped_kernel<128>, dim3((Sq + 255) / 256, (unsigned)bh),
+    hipLaunchKernelGGL(fa_bwd_dq_swapped_kernel<D>, dim3((Sq + 255) / 256, (unsigned)bh),
                        dim3(512), lds_dq_sw, stream,
                        reinterpret_cast<const abf16*>(q), reinterpret_cast<const abf16*>(k),
                        reinterpret_cast<const abf16*>(v), reinterpret_cast<const abf16*>(dout),
