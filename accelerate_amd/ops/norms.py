@@ -7,6 +7,8 @@ CPU or non-bf16 inputs fall back to the eager math so the modules stay
 correct everywhere.
 """
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -132,8 +134,6 @@ class FusedDropoutAddLayerNorm(nn.LayerNorm):
         # shapes (the u8 mask + saved-sum writes outweigh the saved
         # launches once torch's dropout/add/LN are graph-captured), so it
         # is opt-in until the mask is bit-packed:
-        import os
-
         if (
             os.environ.get("ACCELERATE_AMD_FUSED_JUNCTION") == "1"
             and _use_fused(x, self.weight, self.bias)
