@@ -238,3 +238,19 @@ def test_fast_linear_decode_route():
             torch.nn.functional.linear(x, ref[0], ref[1]), ref[2], ref[3]
         )
     assert (y.float() - y_ref.float()).abs().max() < 0.1
+
+
+def test_fast_linear_cpu_fallback():
+    """FastLinear off-GPU is exactly nn.Linear; the converter shares
+    parameters (no copies) and builds on meta (no 2x host RAM)."""
+    from accelerate_amd.ops.linear import FastLinear, convert_linears_for_inference
+
+    torch.manual_seed(0)
+    m = torch.nn.Sequential(torch.nn.Linear(32, 48), torch.nn.ReLU(), torch.nn.Linear(48, 8))
+    w0 = m[0].weight
+    x = torch.randn(3, 32)
+    ref = m(x)
+    convert_linears_for_inference(m)
+    assert isinstance(m[0], FastLinear)
+    assert m[0].weight is w0  # shared, not copied
+    assert torch.equal(m(x), ref)
