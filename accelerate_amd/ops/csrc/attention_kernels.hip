@@ -321,9 +321,12 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
   // bounded either way, and padding keeps the addressing immediate-only.
   constexpr int KP = D + 8;   // K tile row stride
   constexpr int NP = BN + 8;  // V^T row stride
+  // DOUBLE-BUFFERED tiles: one barrier per iteration instead of two, so
+  // the two co-resident waves per SIMD can drift a phase apart and
+  // interleave softmax VALU with the other wave's MFMA clusters.
+  constexpr int TILE = BN * KP + D * NP;  // one K+V buffer (elements)
   extern __shared__ char smem[];
-  abf16* Ks = reinterpret_cast<abf16*>(smem);  // [BN][KP]
-  abf16* VTs = Ks + BN * KP;                   // [D][NP]
+  abf16* tiles = reinterpret_cast<abf16*>(smem);  // [2][TILE]
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -389,7 +392,9 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
       vrg1 = *reinterpret_cast<const bf16x8a*>(vb + key1 * sV.s + kr_col8);
     }
   };
-  auto write_tile = [&]() {
+  auto write_tile = [&](int buf) {
+    abf16* Ks = tiles + buf * TILE;
+    abf16* VTs = Ks + BN * KP;
 #pragma unroll
     for (int it = 0; it < KIT; ++it)
       *reinterpret_cast<bf16x8a*>(Ks + (kr_row + it * (512 / (D / 8))) * KP + kr_col8) = krg[it];
@@ -414,12 +419,16 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
   };
 
   issue_loads(0);
+  write_tile(0);
+  int buf = 0;
   for (int kb0 = 0; kb0 < k_hi; kb0 += BN) {
-    write_tile();
-    __syncthreads();
+    __syncthreads();  // staging of `buf` done; prior reads of buf^1 done
+    const abf16* Ks = tiles + buf * TILE;
+    const abf16* VTs = Ks + BN * KP;
+    if (kb0 + BN < k_hi) issue_loads(kb0 + BN);
 
     f32x16a sacc[2] = {f32x16a{}, f32x16a{}};
-    const bool live = kb0 < k_hi_wave;  // fully-masked tiles: barriers only
+    const bool live = kb0 < k_hi_wave;  // fully-masked tiles: no compute
     if (live) {
       // --- S^T = mfma(K, Q): sacc[kt][r] = S[col][kt*32 + crow(r,hi)] ---
       __builtin_amdgcn_s_setprio(1);
@@ -433,7 +442,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
         }
       __builtin_amdgcn_s_setprio(0);
     }
-    if (kb0 + BN < k_hi) issue_loads(kb0 + BN);  // in flight under softmax+PV
+    if (kb0 + BN < k_hi) write_tile(buf ^ 1);  // regs -> other buffer
     if (live) {
       // --- mask (causal diagonal / Sk tail) in the raw domain ---
       const int qabs = q0 + wave * 32 + col;
@@ -534,7 +543,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
         }
       __builtin_amdgcn_s_setprio(0);
     }
-    __syncthreads();  // next iteration overwrites Ks/VTs
+    buf ^= 1;
   }
 
   // --- epilogue: combine half-row denominators, normalize, store ---
@@ -1744,7 +1753,7 @@ hipError_t launch_impl(const void* q, const void* k, const void* v, void* out, f
                        int64_t bh, int Sq, int Sk, int past, int causal, float scale,
                        int Hq, int Hkv, const Str3* strides, hipStream_t stream) {
   if (fa_fwd_impl() >= 1) {
-    constexpr int lds_sw = (64 * (D + 8) + D * (64 + 8)) * 2;
+    constexpr int lds_sw = 2 * (64 * (D + 8) + D * (64 + 8)) * 2;
     static bool attr_sw = false;
     auto kfn = fa_fwd_impl() == 2 ? &fa_fwd_swapped_kernel<D, true>
                                   : &fa_fwd_swapped_kernel<D, false>;
