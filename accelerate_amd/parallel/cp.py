@@ -65,10 +65,19 @@ def context_parallel_attention(q, k, v, group=None, causal=True):
 
 
 def shard_sequence(tensor: torch.Tensor, group=None, dim: int = 1) -> torch.Tensor:
-    """Slice this rank's contiguous sequence shard (default dim 1 = [B, S])."""
+    """Slice this rank's contiguous sequence shard (default dim 1 = [B, S]).
+
+    The sequence length must divide the cp world size — a silent
+    truncation of the tail would train on different data than the
+    unsharded run (the reference's torch CP has the same requirement)."""
     world = dist.get_world_size(group) if dist.is_initialized() else 1
     if world == 1:
         return tensor
+    if tensor.shape[dim] % world != 0:
+        raise ValueError(
+            f"context parallelism requires the sequence length ({tensor.shape[dim]}, dim {dim}) "
+            f"to divide cp_size ({world}); pad the batch to a multiple"
+        )
     rank = dist.get_rank(group)
     s = tensor.shape[dim] // world
     return tensor.narrow(dim, rank * s, s).contiguous()

@@ -70,6 +70,15 @@ def main():
     check_ring_parity(causal=False, hkv=4, tag="full")
     check_ring_parity(causal=True, hkv=2, tag="gqa")
 
+    # non-divisible sequence -> loud error, not silent truncation
+    from accelerate_amd.parallel.cp import shard_sequence
+
+    try:
+        shard_sequence(torch.randn(2, world * 5 + 1), dim=1)
+        raise AssertionError("expected ValueError for non-divisible sequence")
+    except ValueError:
+        pass
+
     if acc.is_main_process:
         print("RING_CP_PASS")
     acc.end_training()

@@ -200,3 +200,20 @@ def test_ring_cp_2proc():
 
     out = launch_distributed("tests/distributed_scripts/ring_script.py", nproc=2, timeout=300)
     assert "RING_CP_PASS" in out
+
+
+def test_shard_sequence_world1_passthrough():
+    """No-dist world-1 path returns the tensor untouched (the non-divisible
+    ValueError path is exercised by the gloo scripts at world > 1)."""
+    from accelerate_amd.parallel.cp import shard_sequence
+
+    t = torch.randn(2, 10)
+    assert shard_sequence(t) is t
+
+
+def test_ring_script_world3_shapes():
+    """Ring CP oracle also holds at world 3 (multi-hop rotation)."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/ring_script.py", nproc=3, timeout=300)
+    assert "RING_CP_PASS" in out
