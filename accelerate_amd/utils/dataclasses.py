@@ -150,6 +150,13 @@ class DistributedDataParallelKwargs(KwargsHandler):
             if hook != DDPCommunicationHookType.NO:
                 self.comm_dtype = hook.value
 
+    def to_dict(self):
+        # comm_hook is a reference-API alias folded into comm_dtype above;
+        # the engine itself only takes comm_dtype
+        d = super().to_dict()
+        d.pop("comm_hook", None)
+        return d
+
 
 @dataclass
 class GradScalerKwargs(KwargsHandler):

@@ -63,3 +63,20 @@ def test_synchronize_rng_states_generator():
     synchronize_rng_states(["generator"], generator=g)
     assert torch.equal(g.get_state(), before)
     _fresh()
+
+
+def test_ddp_kwargs_comm_hook_alias_feeds_engine():
+    """comm_hook (reference alias) folds into comm_dtype and must NOT leak
+    into the engine kwargs — regression: the SCALE launch crashed with
+    'unexpected keyword argument comm_hook' at world > 1."""
+    import torch.nn as nn
+
+    from accelerate_amd.parallel.ddp import DistributedDataParallelEngine
+    from accelerate_amd.utils import DDPCommunicationHookType, DistributedDataParallelKwargs
+
+    h = DistributedDataParallelKwargs(comm_hook=DDPCommunicationHookType.BF16)
+    d = h.to_dict()
+    assert "comm_hook" not in d
+    assert d["comm_dtype"] == "bf16"
+    engine = DistributedDataParallelEngine(nn.Linear(4, 4), **d)  # must not raise
+    assert engine.comm_dtype is not None
