@@ -56,6 +56,13 @@ constexpr float kLn2 = 0.6931471805599453f;
 
 }  // namespace
 
+namespace {
+// exp2f lowers to ldexp+v_exp range-fixup pairs without -ffast-math; the
+// softmax arguments are bounded (s - m <= defer-threshold, s - lse <= 0 on
+// real rows), so the raw hardware v_exp_f32 is exact where it matters.
+__device__ __forceinline__ float fast_exp2(float x) { return __builtin_amdgcn_exp2f(x); }
+}  // namespace
+
 // D = head_dim (64 or 128). q:[B,Hq,Sq,D] k,v:[B,Hkv,Sk,D] bf16 views
 // (strides in elements, dim 3 contiguous). out: written through sO (BSHD
 // storage); lse:[B,Hq,Sq] fp32 natural-log contiguous.
@@ -203,12 +210,12 @@ __global__ __launch_bounds__(256, 2) void fa_fwd_kernel(
         for (int x = 1; x < 16; x <<= 1) bm = fmaxf(bm, __shfl_xor(bm, x, 64));
         const float new_m = fmaxf(m_run[mi][j], bm);
         // every processed row has >= 1 unmasked key (causal rows see key 0)
-        const float corr = (m_run[mi][j] == -INFINITY) ? 0.f : exp2f(m_run[mi][j] - new_m);
+        const float corr = (m_run[mi][j] == -INFINITY) ? 0.f : fast_exp2(m_run[mi][j] - new_m);
         m_run[mi][j] = new_m;
         float rs = 0.f;
 #pragma unroll
         for (int ni = 0; ni < BN / 16; ++ni) {
-          const float p = exp2f(sacc[mi][ni][j] * c - new_m);
+          const float p = fast_exp2(sacc[mi][ni][j] * c - new_m);
           sacc[mi][ni][j] = p;
           rs += p;
         }
@@ -474,7 +481,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
       const bool rescale = !__all(pmax - m_run <= 8.f);
       if (rescale) {
         const float new_m = fmaxf(m_run, pmax);
-        const float corr = (m_run == -INFINITY) ? 0.f : exp2f(m_run - new_m);
+        const float corr = (m_run == -INFINITY) ? 0.f : fast_exp2(m_run - new_m);
         m_run = new_m;
         l_run *= corr;
         // redistribute corr to O rows (row of value r is crow(r,hi))
@@ -494,7 +501,7 @@ __global__ __launch_bounds__(512, 1) void fa_fwd_swapped_kernel(
       for (int kt = 0; kt < 2; ++kt)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float pv = exp2f(sacc[kt][r] - m_run);
+          const float pv = fast_exp2(sacc[kt][r] - m_run);
           sacc[kt][r] = pv;
           rs += pv;
         }
@@ -752,7 +759,7 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dkdv_kernel(
         for (int j = 0; j < 4; ++j) {
           const int kr = k0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
           const bool dead = qcol >= Sq || kr >= Sk || (causal && kr > past + qcol);
-          const float p = dead ? 0.f : exp2f(st[mi][ni][j] * c - l2);
+          const float p = dead ? 0.f : fast_exp2(st[mi][ni][j] * c - l2);
           st[mi][ni][j] = p;
           ushort pb = af2bf(p);
           Ws[(wave * 32 + mi * 16 + (lane >> 4) * 4 + j) * QP + ni * 16 + (lane & 15)] =
@@ -947,7 +954,7 @@ __global__ __launch_bounds__(256, 2) void fa_bwd_dq_kernel(
         for (int j = 0; j < 4; ++j) {
           const int qr = q0 + wave * 32 + mi * 16 + (lane >> 4) * 4 + j;
           const bool dead = kc >= Sk || (causal && kc > past + qr);
-          const float p = dead ? 0.f : exp2f(st[mi][ni][j] * c - l2r[mi][j]);
+          const float p = dead ? 0.f : fast_exp2(st[mi][ni][j] * c - l2r[mi][j]);
           const float ds = p * (dpt[mi][ni][j] - drr[mi][j]) * scale;
           ushort b = af2bf(ds);
           dSs[(wave * 32 + mi * 16 + (lane >> 4) * 4 + j) * NP + ni * 16 + (lane & 15)] =
@@ -1098,7 +1105,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dv_swapped_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int qrow = qt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          float pv = exp2f(sacc[qt][r] * c - lse_s[qrow] * kLog2e);
+          float pv = fast_exp2(sacc[qt][r] * c - lse_s[qrow] * kLog2e);
           if (edge) {
             const int qa = qbase + (r & 3) + 8 * (r >> 2);
             if (qa >= Sq || (causal && kabs > past + qa)) pv = 0.f;
@@ -1256,7 +1263,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dk_swapped_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int qrow = qt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          float pv = exp2f(sacc[qt][r] * c - lse_s[qrow] * kLog2e);
+          float pv = fast_exp2(sacc[qt][r] * c - lse_s[qrow] * kLog2e);
           if (edge) {
             const int qa = qbase + (r & 3) + 8 * (r >> 2);
             if (qa >= Sq || (causal && kabs > past + qa)) pv = 0.f;
@@ -1427,7 +1434,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dkdv_swapped_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float pv = exp2f(sacc[r] * c - lse_s[qrow]);
+        float pv = fast_exp2(sacc[r] * c - lse_s[qrow]);
         if (edge) {
           const int qa = qt0 + qrow;
           if (qa >= Sq || (causal && kabs > past + qa)) pv = 0.f;
@@ -1650,7 +1657,7 @@ __global__ __launch_bounds__(512, 1) void fa_bwd_dq_swapped_kernel(
         const int kbase = kb0 + kt * 32 + 4 * hi;
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          float pv = exp2f(sacc[kt][r] * c - lse2);
+          float pv = fast_exp2(sacc[kt][r] * c - lse2);
           if (edge) {
             const int kabs = kbase + (r & 3) + 8 * (r >> 2);
             if (kabs >= Sk || (causal && kabs > past + qabs)) pv = 0.f;
