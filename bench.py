@@ -71,7 +71,13 @@ def main():
     device = accelerator.device
     on_gpu = device.type == "cuda"
 
-    config = BertConfig.bert_base()
+    if os.environ.get("BENCH_TINY") == "1":
+        # CI plumbing smoke ONLY (2-proc gloo test): tiny model, NOT the
+        # measured configuration — the emitted config block says so
+        config = BertConfig(num_hidden_layers=2, hidden_size=128, num_attention_heads=2,
+                            intermediate_size=256)
+    else:
+        config = BertConfig.bert_base()
     model = BertForSequenceClassification(config)
     if bf16_weights:
         model = model.to(torch.bfloat16)
@@ -234,7 +240,7 @@ def main():
             "dtype": "bf16" if on_gpu else "fp32(cpu-ci)",
             "data": "synthetic",
             "config": {
-                "model": "bert-base",
+                "model": "bert-tiny(ci-smoke)" if os.environ.get("BENCH_TINY") == "1" else "bert-base",
                 "global_batch": global_batch,
                 "per_gpu_batch": PER_GPU_BATCH,
                 "seq_len": SEQ_LEN,

@@ -42,3 +42,26 @@ def test_notebook_launcher_single_process():
 
     notebook_launcher(fn, args=(21,), num_processes=1)
     assert out == [42]
+
+
+def test_bench_contract_2proc_plumbing():
+    """The driver launches bench.py via torchrun at N>1: the full
+    prepare/DDP/loader plumbing must survive world>1 (regression: a
+    kwargs-handler field leak crashed every SCALE run). BENCH_TINY=1
+    shrinks the model so this is a plumbing smoke, not a measurement."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    env = dict(os.environ, BENCH_TINY="1", PYTHONPATH=os.getcwd())
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "2",
+         "--master-addr", "127.0.0.1", "--master-port", "29555",
+         "bench.py", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=280, env=env, cwd=os.getcwd(),
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2 and rec["config"]["parallelism"] == "dp2"
