@@ -217,3 +217,12 @@ def test_ring_script_world3_shapes():
 
     out = launch_distributed("tests/distributed_scripts/ring_script.py", nproc=3, timeout=300)
     assert "RING_CP_PASS" in out
+
+
+def test_dp_cp_combined_4proc():
+    """dp_replicate=2 x cp=2: DDP over the flattened dp x cp grad group +
+    per-step sequence sharding, one train step == single-process reference."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/dpcp_script.py", nproc=4, timeout=300)
+    assert "DPCP_PREPARE_PASS" in out
