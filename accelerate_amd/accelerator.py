@@ -538,6 +538,19 @@ class Accelerator:
             model = model.to(self.device)
 
         pc = self.parallelism_config
+        if (
+            pc is not None
+            and pc.dp_shard_size > 1
+            and getattr(self.state, "fsdp_plugin", None) is None
+        ):
+            # mirror the reference's requirement (state.py:995-1007):
+            # dp_shard is the FSDP dimension — silently replicating here
+            # would "work" while delivering none of the promised sharding
+            raise ValueError(
+                "ParallelismConfig.dp_shard_size > 1 requires an FSDP plugin "
+                "(fsdp_plugin=FullyShardedDataParallelPlugin(...) or the FSDP_* env plane); "
+                "use dp_replicate_size for replicated data parallelism."
+            )
         multi_dim = pc is not None and (pc.tp_size > 1 or pc.cp_size > 1) and pc._groups
         if not evaluation_mode and self.use_distributed and self.distributed_type in (
             DistributedType.MULTI_GPU,

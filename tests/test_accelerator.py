@@ -333,3 +333,23 @@ def test_free_memory_clears_references():
     (model,) = acc.free_memory(model)
     assert acc._models == []
     assert model is None  # slots come back None so callers drop references
+
+
+def test_dp_shard_without_fsdp_plugin_raises():
+    """dp_shard is the FSDP dimension (reference state.py:995-1007):
+    without a plugin, prepare_model must refuse rather than silently
+    replicate where the user asked for sharding."""
+    import pytest as _pytest
+    import torch.nn as nn
+
+    from accelerate_amd import Accelerator, ParallelismConfig
+
+    acc = Accelerator(cpu=True)
+    # bypass the world-size check (single process here): the plugin
+    # requirement must trip FIRST in prepare_model regardless
+    pc = ParallelismConfig.__new__(ParallelismConfig)
+    pc.dp_replicate_size, pc.dp_shard_size, pc.tp_size, pc.cp_size = 1, 2, 1, 1
+    pc._groups = {}
+    acc.parallelism_config = pc
+    with _pytest.raises(ValueError, match="FSDP plugin"):
+        acc.prepare_model(nn.Linear(4, 4))
