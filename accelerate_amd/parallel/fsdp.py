@@ -369,6 +369,13 @@ class ShardedModel(nn.Module):
 
     @staticmethod
     def _select_units(module, policy, transformer_cls_names, min_num_params):
+        # containers without a forward of their own (ModuleList/-Dict) must
+        # never BE a unit: the unit's pre-forward unshard hook would never
+        # fire (nothing calls the container), so its parameters would be
+        # used while resharded — out-of-bounds shard reads. Recurse into
+        # them / expand them to their children instead.
+        container_types = (nn.ModuleList, nn.ModuleDict, nn.ParameterList, nn.ParameterDict)
+
         units = []
         if callable(policy):
             for name, m in module.named_modules():
@@ -384,7 +391,7 @@ class ShardedModel(nn.Module):
             def walk(prefix, m):
                 n_params = sum(p.numel() for p in m.parameters())
                 children = list(m.named_children())
-                if prefix and n_params >= min_num_params:
+                if prefix and n_params >= min_num_params and not isinstance(m, container_types):
                     big_children = [
                         (f"{prefix}.{cn}", c)
                         for cn, c in children
@@ -398,6 +405,17 @@ class ShardedModel(nn.Module):
                     walk(f"{prefix}.{cn}" if prefix else cn, c)
 
             walk("", module)
+        # expand any container unit (from a user policy) into its children
+        expanded = []
+        for name, m in units:
+            if isinstance(m, container_types):
+                expanded.extend(
+                    (f"{name}.{cn}", c) for cn, c in m.named_children()
+                    if any(True for _ in c.parameters())
+                )
+            else:
+                expanded.append((name, m))
+        units = expanded
         # drop nested units (keep outermost)
         kept = []
         for name, m in units:

@@ -144,3 +144,44 @@ def test_405b_fsdp_training_plan():
     assert peak64 < 288e9 * 0.7, f"405B at world 64 should fit: {peak64/1e9:.1f} GB/rank"
     # sanity: the master shards really are ~1/64 of fp32 model bytes
     assert abs(plan64["per_rank_master_bytes"] - plan64["total_numel"] * 4 / 64) < 1e9
+
+
+def test_auto_wrap_never_units_containers():
+    """A ModuleList must never become a unit: its pre-forward unshard hook
+    would never fire (containers have no forward), so its params would be
+    consumed while resharded — this segfaulted with out-of-bounds shard
+    reads before the fix (size-based policy on a model whose layer list
+    crossed min_num_params while no single layer did)."""
+    import torch.nn as nn
+
+    from accelerate_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    from accelerate_amd.parallel.fsdp import ShardedModel
+
+    model = LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2))
+    sm = ShardedModel(model, min_num_params=1_000_000)
+    for u in sm.units:
+        mod = dict(sm.module.named_modules())[u.name] if u.name else sm.module
+        assert not isinstance(mod, (nn.ModuleList, nn.ModuleDict)), u.name
+    ids = torch.randint(0, 1024, (2, 16))
+    out = sm(ids)["logits"]
+    out.float().pow(2).mean().backward()
+    assert torch.isfinite(out).all()
+
+    # a user policy that selects the container gets expanded to its children
+    sm2 = ShardedModel(
+        LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2)),
+        auto_wrap_policy=lambda m: isinstance(m, nn.ModuleList),
+    )
+    names = [u.name for u in sm2.units]
+    assert "layers" not in names
+    assert any(n.startswith("layers.") for n in names), names
+
+
+def test_fsdp_cp_combined_4proc():
+    """FSDP (dp_shard=2) x CP (cp=2): flat-shards over the flattened
+    dp_shard x cp domain + per-step sequence sharding == single-process
+    reference (the reference's FSDP2+context_parallel pairing)."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/fsdp_cp_script.py", nproc=4, timeout=300)
+    assert "FSDP_CP_PASS" in out
