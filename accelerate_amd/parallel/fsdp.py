@@ -823,6 +823,17 @@ def fsdp_prepare(accelerator, args, device_placement):
                 import os as _os
 
                 shard_group_size = int(_os.environ.get("FSDP_SHARD_GROUP_SIZE", "0")) or None
+            # under a multi-dim ParallelismConfig the flat-shard (and grad
+            # reduce) domain is the dp x cp group for THIS tp coordinate —
+            # sharding over the default world group would average DIFFERENT
+            # tp shards together (reference: the FSDP2 mesh's flattened
+            # ["dp_shard_cp"] dim, fsdp_utils.py:770)
+            shard_pg = None
+            pc = accelerator.parallelism_config
+            if pc is not None and pc.tp_size > 1:
+                if not pc._groups:
+                    pc.build_groups()
+                shard_pg = pc._groups.get("grad")
             wrapped_model = ShardedModel(
                 obj,
                 transformer_cls_names=plugin.transformer_cls_names_to_wrap,
@@ -830,6 +841,7 @@ def fsdp_prepare(accelerator, args, device_placement):
                 compute_dtype=compute_dtype,
                 reduce_dtype=mp.get("reduce_dtype"),
                 reshard_after_forward=plugin.reshard_after_forward,
+                process_group=shard_pg,
                 device=accelerator.device,
                 activation_checkpointing=plugin.activation_checkpointing,
                 shard_group_size=shard_group_size,

@@ -185,3 +185,13 @@ def test_fsdp_cp_combined_4proc():
 
     out = launch_distributed("tests/distributed_scripts/fsdp_cp_script.py", nproc=4, timeout=300)
     assert "FSDP_CP_PASS" in out
+
+
+def test_tp_fsdp_combined_4proc():
+    """TP (tp=2) x FSDP (dp_shard=2): the flat-shard engine must shard and
+    reduce over the dp group of THIS tp coordinate (sharding over the world
+    group would average different tp shards together)."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/tp_fsdp_script.py", nproc=4, timeout=300)
+    assert "TP_FSDP_PASS" in out
