@@ -226,3 +226,12 @@ def test_dp_cp_combined_4proc():
 
     out = launch_distributed("tests/distributed_scripts/dpcp_script.py", nproc=4, timeout=300)
     assert "DPCP_PREPARE_PASS" in out
+
+
+def test_tp_cp_combined_4proc():
+    """TP (tp=2) x CP (cp=2): tp-sharded attention heads + cp KV all-gather
+    + grad averaging over the cp group — one step == single-process ref."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/tp_cp_script.py", nproc=4, timeout=300)
+    assert "TP_CP_PASS" in out
