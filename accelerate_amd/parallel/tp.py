@@ -76,6 +76,13 @@ class _GatherFromTP(torch.autograd.Function):
 
 
 class ColumnParallelLinear(nn.Module):
+    # checkpoints stay tp-degree-agnostic: state_dict() all-gathers the FULL
+    # weight (a collective — call it on every tp rank, as save_state does)
+    # and _load_from_state_dict re-slices this rank's rows, so save/load
+    # round-trips across different tp layouts (reference parity: DTensor
+    # state dicts materialize full tensors).
+    n_fused = 1
+
     def __init__(self, in_features, out_features, bias=True, gather_output=False, group=None, dtype=None):
         super().__init__()
         world = dist.get_world_size(group) if dist.is_initialized() else 1
@@ -86,6 +93,47 @@ class ColumnParallelLinear(nn.Module):
         self.weight = nn.Parameter(torch.empty(self.out_per_rank, in_features, dtype=dtype))
         self.bias = nn.Parameter(torch.zeros(self.out_per_rank, dtype=dtype)) if bias else None
         nn.init.kaiming_uniform_(self.weight, a=5**0.5)
+
+    def _world(self):
+        return dist.get_world_size(self.group) if dist.is_initialized() else 1
+
+    def _row_index(self, world):
+        rank = dist.get_rank(self.group) if dist.is_initialized() else 0
+        per = self.out_per_rank // self.n_fused
+        block = per * world
+        rows = [torch.arange(j * block + rank * per, j * block + (rank + 1) * per) for j in range(self.n_fused)]
+        return torch.cat(rows)
+
+    def _unshard_rows(self, shards, world):
+        """Reassemble the full dim-0 tensor from per-rank shards (fused
+        blocks interleave: full block j = cat over ranks of each shard's
+        j-th sub-block)."""
+        per = self.out_per_rank // self.n_fused
+        blocks = []
+        for j in range(self.n_fused):
+            blocks.extend(shards[r][j * per : (j + 1) * per] for r in range(world))
+        return torch.cat(blocks, dim=0)
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        world = self._world()
+        if world == 1:
+            return super()._save_to_state_dict(destination, prefix, keep_vars)
+        for name, t in (("weight", self.weight), ("bias", self.bias)):
+            if t is None:
+                continue
+            shards = [torch.empty_like(t) for _ in range(world)]
+            dist.all_gather(shards, t.detach().contiguous(), group=self.group)
+            destination[prefix + name] = self._unshard_rows(shards, world)
+
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+        world = self._world()
+        if world > 1:
+            rows = self._row_index(world)
+            for name in ("weight", "bias"):
+                full = state_dict.get(prefix + name)
+                if full is not None and full.shape[0] == self.out_per_rank * world:
+                    state_dict[prefix + name] = full[rows]
+        return super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
 
     def forward(self, x):
         x = _CopyToTP.apply(x, self.group)
@@ -112,6 +160,7 @@ class ColumnParallelLinear(nn.Module):
 
 
 class RowParallelLinear(nn.Module):
+    # see ColumnParallelLinear: full-tensor state dicts, dim-1 sharding
     def __init__(self, in_features, out_features, bias=True, input_is_parallel=True, group=None, dtype=None):
         super().__init__()
         world = dist.get_world_size(group) if dist.is_initialized() else 1
@@ -133,6 +182,25 @@ class RowParallelLinear(nn.Module):
         if self.bias is not None:
             y = y + self.bias
         return y
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        world = dist.get_world_size(self.group) if dist.is_initialized() else 1
+        if world == 1:
+            return super()._save_to_state_dict(destination, prefix, keep_vars)
+        shards = [torch.empty_like(self.weight) for _ in range(world)]
+        dist.all_gather(shards, self.weight.detach().contiguous(), group=self.group)
+        destination[prefix + "weight"] = torch.cat(shards, dim=1)
+        if self.bias is not None:  # bias is replicated (applied post-reduce)
+            destination[prefix + "bias"] = self.bias.detach()
+
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+        world = dist.get_world_size(self.group) if dist.is_initialized() else 1
+        if world > 1:
+            full = state_dict.get(prefix + "weight")
+            if full is not None and full.shape[1] == self.in_per_rank * world:
+                rank = dist.get_rank(self.group) if dist.is_initialized() else 0
+                state_dict[prefix + "weight"] = full[:, rank * self.in_per_rank : (rank + 1) * self.in_per_rank]
+        return super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
 
     @classmethod
     def from_linear(cls, linear: nn.Linear, group=None, input_is_parallel=True):
@@ -167,6 +235,7 @@ def _fused_colwise_from_linear(linear: nn.Linear, n_fused: int, group=None):
     nn.Module.__init__(mod)
     mod.group = group
     mod.gather_output = False
+    mod.n_fused = n_fused
     mod.out_per_rank = n_fused * per
     mod.weight = nn.Parameter(linear.weight[rows].detach().clone())
     mod.bias = nn.Parameter(linear.bias[rows].detach().clone()) if linear.bias is not None else None

@@ -100,6 +100,21 @@ def run_family(acc, pc, name, make_model, vocab):
             assert torch.allclose(
                 block.mlp.c_proj.weight, C[:, tp * perc : (tp + 1) * perc], atol=1e-5
             ), f"{name} mlp.c_proj mismatch {i}"
+    # 4. tp-degree-agnostic checkpoints: state_dict() gathers FULL weights
+    # (collective — every rank calls it), and loading a full dict re-slices
+    # this rank's shard, so save_state/load_state round-trips under TP
+    sd = m.state_dict()
+    ref_sd = ref.state_dict()
+    for k, v in ref_sd.items():
+        assert k in sd and sd[k].shape == v.shape, f"{name}: state_dict shape {k}: {sd.get(k, None) is not None and sd[k].shape} vs {v.shape}"
+        assert torch.allclose(sd[k], v, atol=1e-5), f"{name}: state_dict value {k}: {(sd[k] - v).abs().max()}"
+    m.load_state_dict(sd)  # full dict loads back into the sharded module
+    with torch.no_grad():
+        out_rt = model(xb)
+        out_rt = out_rt["logits"] if isinstance(out_rt, dict) else out_rt
+        ref_rt = ref(xb)
+        ref_rt = ref_rt["logits"] if isinstance(ref_rt, dict) else ref_rt
+    assert torch.allclose(out_rt, ref_rt, atol=1e-4), f"{name}: post-roundtrip forward diverges"
     if acc.is_main_process:
         print(f"TP_PREPARE_{name}_PASS")
 
