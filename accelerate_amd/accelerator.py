@@ -489,6 +489,11 @@ class Accelerator:
                             )
                         for group in obj.param_groups:
                             group["params"] = [swap.get(id(p), p) for p in group["params"]]
+                        # moments now live on tp SHARDS: checkpointing must
+                        # write one optimizer file per rank (like FSDP), not
+                        # main-rank-only (which would load rank 0's shard
+                        # moments everywhere)
+                        obj._sharded = True
             # cp activation is scoped to `maybe_context_parallel` (the
             # per-step ctx manager, reference accelerator.py:4111) so
             # unprepared/eval models are never caught by the ambient context
@@ -1074,6 +1079,10 @@ class Accelerator:
         for i, obj in enumerate(self._custom_objects):
             save_custom_state(obj, output_dir, i, save_on_each_node=self.project_configuration.save_on_each_node)
         self.project_configuration.iteration += 1
+        # every file durable before ANY rank proceeds: a non-main rank that
+        # immediately calls load_state (or reads the dir) must not race the
+        # main rank's writes
+        self.wait_for_everyone()
         return save_location
 
     def load_state(self, input_dir: str = None, load_kwargs=None, **load_model_func_kwargs):

@@ -48,6 +48,10 @@ class AcceleratedOptimizer(torch.optim.Optimizer):
         self.gradient_state = GradientState()
         self.device_placement = device_placement
         self._last_step_skipped = False
+        # per-rank-state marker (tp/fsdp shard moments) set on the RAW
+        # optimizer before wrapping must survive onto the wrapper, which is
+        # what checkpointing inspects
+        self._sharded = getattr(optimizer, "_sharded", False)
         if device_placement:
             migrated = _state_to_device(optimizer.state_dict(), self.accelerator_state.device)
             optimizer.load_state_dict(migrated)

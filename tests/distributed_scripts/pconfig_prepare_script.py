@@ -119,6 +119,54 @@ def run_family(acc, pc, name, make_model, vocab):
         print(f"TP_PREPARE_{name}_PASS")
 
 
+def check_save_state_roundtrip(acc, pc):
+    """save_state under TP writes PER-RANK optimizer files (moments live on
+    tp shards); load_state restores THIS rank's moments exactly."""
+    import tempfile
+
+    import torch as _torch
+
+    # fresh registries: the earlier run_family models would otherwise be
+    # re-saved (and their per-rank tp gathers re-run) as models 0..1
+    acc._models.clear()
+    acc._optimizers.clear()
+    acc._schedulers.clear()
+    acc._dataloaders.clear()
+    set_seed(0)
+    model = LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2))
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    model, opt = acc.prepare(model, opt)
+    ids = torch.randint(0, 1024, (2, 16), generator=torch.Generator().manual_seed(5))
+    opt.zero_grad()
+    loss = model(ids)["logits"].float().pow(2).mean()
+    acc.backward(loss)
+    opt.step()  # materialize per-shard Adam moments
+
+    tmp = tempfile.mkdtemp(prefix=f"tp_ckpt_{acc.process_index}_")
+    # shared dir decided by rank 0
+    holder = [tmp]
+    dist.broadcast_object_list(holder, src=0)
+    tmp = holder[0]
+    acc.save_state(tmp)
+    snap = {
+        id(p): {k: v.clone() for k, v in st.items() if _torch.is_tensor(v)}
+        for p, st in opt.optimizer.state.items()
+    }
+    with _torch.no_grad():
+        for st in opt.optimizer.state.values():
+            for v in st.values():
+                if _torch.is_tensor(v) and v.is_floating_point():
+                    v.add_(1.0)
+    acc.load_state(tmp)
+    for p, st in opt.optimizer.state.items():
+        for k, v in st.items():
+            if _torch.is_tensor(v) and id(p) in snap and k in snap[id(p)]:
+                assert _torch.allclose(v, snap[id(p)][k]), f"optimizer state {k} not restored per-rank"
+    dist.barrier()
+    if acc.is_main_process:
+        print("TP_SAVE_STATE_PASS")
+
+
 def main():
     pc = ParallelismConfig(dp_replicate_size=2, tp_size=2)
     acc = Accelerator(cpu=True, parallelism_config=pc)
@@ -126,6 +174,7 @@ def main():
 
     run_family(acc, pc, "LLAMA", lambda: LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2)), 1024)
     run_family(acc, pc, "GPT2", lambda: GPT2LMHeadModel(GPT2Config.tiny()), 1024)
+    check_save_state_roundtrip(acc, pc)
 
     acc.end_training()
 

@@ -178,6 +178,7 @@ def test_tp_through_prepare_4proc():
     from tests.testing_utils import launch_distributed
 
     out = launch_distributed("tests/distributed_scripts/pconfig_prepare_script.py", nproc=4, timeout=300)
+    assert "TP_SAVE_STATE_PASS" in out
     assert "TP_PREPARE_LLAMA_PASS" in out
     assert "TP_PREPARE_GPT2_PASS" in out
 
