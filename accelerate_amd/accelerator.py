@@ -537,6 +537,13 @@ class Accelerator:
         if self.mixed_precision == "fp8":
             from .ops.fp8 import convert_linears_to_fp8
 
+            pc_fp8 = self.parallelism_config
+            if pc_fp8 is not None and pc_fp8.tp_size > 1:
+                logger.warning(
+                    "fp8 conversion targets plain nn.Linear modules: tensor-parallel-sharded "
+                    "projections stay in bf16 (fp8 column/row-parallel linears are not yet "
+                    "implemented). Expect fp8 gains only on unsharded layers."
+                )
             model = convert_linears_to_fp8(model, recipe=self.fp8_recipe_handler)
 
         if device_placement and not self.verify_device_map(model):
@@ -706,6 +713,44 @@ class Accelerator:
                 # model.clip_grad_norm_, accelerator.py:2977-3007)
                 if any(id(p) in model._unit_param_ids for p in parameters):
                     return model.clip_grad_norm_(max_norm, norm_type)
+        pc = self.parallelism_config
+        if pc is not None and pc.tp_size > 1 and any(
+            getattr(p, "_tp_sharded", False) for p in parameters
+        ):
+            # tensor parallelism: the global norm sums the SHARDED params'
+            # contributions over the tp group (each shard counted once)
+            # plus the replicated params' contribution counted ONCE; using
+            # the local norm would clip replicas by different coefficients
+            # and diverge them (reference: DTensor-aware clip)
+            grads = [(p, p.grad) for p in parameters if p.grad is not None]
+            if not grads:
+                return torch.tensor(0.0)
+            tp_group = pc._groups.get("tp")
+            dev = grads[0][1].device
+            if norm_type == float("inf"):
+                local = max(g.detach().abs().max() for _, g in grads)
+                total = local.clone().to(dev)
+                torch.distributed.all_reduce(total, op=torch.distributed.ReduceOp.MAX, group=tp_group)
+            else:
+                shard_sq = sum(
+                    g.detach().float().abs().pow(norm_type).sum()
+                    for p, g in grads
+                    if getattr(p, "_tp_sharded", False)
+                )
+                rep_sq = sum(
+                    g.detach().float().abs().pow(norm_type).sum()
+                    for p, g in grads
+                    if not getattr(p, "_tp_sharded", False)
+                )
+                t = torch.as_tensor(shard_sq, dtype=torch.float32, device=dev).clone()
+                torch.distributed.all_reduce(t, group=tp_group)
+                total = (t + torch.as_tensor(rep_sq, dtype=torch.float32, device=dev)).pow(
+                    1.0 / norm_type
+                )
+            coef = (max_norm / (total + 1e-6)).clamp(max=1.0)
+            for _, g in grads:
+                g.detach().mul_(coef.to(g.dtype))
+            return total
         from .ops.clip_grad import clip_grad_norm_ as _clip
 
         return _clip(parameters, max_norm, norm_type=norm_type)

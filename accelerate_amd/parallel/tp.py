@@ -92,6 +92,9 @@ class ColumnParallelLinear(nn.Module):
         self.out_per_rank = out_features // world
         self.weight = nn.Parameter(torch.empty(self.out_per_rank, in_features, dtype=dtype))
         self.bias = nn.Parameter(torch.zeros(self.out_per_rank, dtype=dtype)) if bias else None
+        self.weight._tp_sharded = True
+        if self.bias is not None:
+            self.bias._tp_sharded = True
         nn.init.kaiming_uniform_(self.weight, a=5**0.5)
 
     def _world(self):
@@ -156,6 +159,9 @@ class ColumnParallelLinear(nn.Module):
         mod.bias = (
             nn.Parameter(linear.bias[lo : lo + mod.out_per_rank].detach().clone()) if linear.bias is not None else None
         )
+        mod.weight._tp_sharded = True
+        if mod.bias is not None:
+            mod.bias._tp_sharded = True
         return mod
 
 
@@ -169,6 +175,7 @@ class RowParallelLinear(nn.Module):
         self.input_is_parallel = input_is_parallel
         self.in_per_rank = in_features // world
         self.weight = nn.Parameter(torch.empty(out_features, self.in_per_rank, dtype=dtype))
+        self.weight._tp_sharded = True  # bias stays replicated (post-reduce)
         self.bias = nn.Parameter(torch.zeros(out_features, dtype=dtype)) if bias else None
         nn.init.kaiming_uniform_(self.weight, a=5**0.5)
 
@@ -206,6 +213,7 @@ class RowParallelLinear(nn.Module):
     def from_linear(cls, linear: nn.Linear, group=None, input_is_parallel=True):
         world = dist.get_world_size(group) if dist.is_initialized() else 1
         rank = dist.get_rank(group) if dist.is_initialized() else 0
+        # (weight tagged _tp_sharded below; bias replicated)
         mod = cls.__new__(cls)
         nn.Module.__init__(mod)
         mod.group = group
@@ -213,6 +221,7 @@ class RowParallelLinear(nn.Module):
         mod.in_per_rank = linear.in_features // world
         lo = rank * mod.in_per_rank
         mod.weight = nn.Parameter(linear.weight[:, lo : lo + mod.in_per_rank].detach().clone())
+        mod.weight._tp_sharded = True
         mod.bias = nn.Parameter(linear.bias.detach().clone()) if linear.bias is not None else None
         return mod
 
@@ -239,6 +248,9 @@ def _fused_colwise_from_linear(linear: nn.Linear, n_fused: int, group=None):
     mod.out_per_rank = n_fused * per
     mod.weight = nn.Parameter(linear.weight[rows].detach().clone())
     mod.bias = nn.Parameter(linear.bias[rows].detach().clone()) if linear.bias is not None else None
+    mod.weight._tp_sharded = True
+    if mod.bias is not None:
+        mod.bias._tp_sharded = True
     return mod
 
 

@@ -119,6 +119,41 @@ def run_family(acc, pc, name, make_model, vocab):
         print(f"TP_PREPARE_{name}_PASS")
 
 
+def check_tp_clip(acc, pc):
+    """clip_grad_norm_ under TP: global norm sums sharded contributions over
+    the tp group + replicated once; returned norm and the clipped update
+    must match the single-process reference exactly."""
+    set_seed(0)
+    model = LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2))
+    set_seed(0)
+    ref = LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2))
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    ref_opt = torch.optim.SGD(ref.parameters(), lr=0.1)
+    model, opt = acc.prepare(model, opt)
+
+    ids = torch.randint(0, 1024, (2, 16), generator=torch.Generator().manual_seed(5))
+    opt.zero_grad()
+    loss = model(ids)["logits"].float().pow(2).mean()
+    acc.backward(loss)
+    norm = acc.clip_grad_norm_([p for p in acc.unwrap_model(model).parameters()], max_norm=0.05)
+    opt.step()
+
+    ref_opt.zero_grad()
+    ref(ids)["logits"].float().pow(2).mean().backward()
+    ref_norm = torch.nn.utils.clip_grad_norm_(ref.parameters(), max_norm=0.05)
+    ref_opt.step()
+    assert abs(float(norm) - float(ref_norm)) < 1e-4 * max(1.0, float(ref_norm)), (
+        f"tp clip norm {float(norm)} vs ref {float(ref_norm)}"
+    )
+    with torch.no_grad():
+        out = model(ids)["logits"]
+        want = ref(ids)["logits"]
+    assert torch.allclose(out, want, atol=1e-4), f"post-clip step diverges {(out - want).abs().max()}"
+    dist.barrier()
+    if acc.is_main_process:
+        print("TP_CLIP_PASS")
+
+
 def check_save_state_roundtrip(acc, pc):
     """save_state under TP writes PER-RANK optimizer files (moments live on
     tp shards); load_state restores THIS rank's moments exactly."""
@@ -175,6 +210,8 @@ def main():
     run_family(acc, pc, "LLAMA", lambda: LlamaForCausalLM(LlamaConfig.tiny(num_hidden_layers=2)), 1024)
     run_family(acc, pc, "GPT2", lambda: GPT2LMHeadModel(GPT2Config.tiny()), 1024)
     check_save_state_roundtrip(acc, pc)
+    acc._models.clear(); acc._optimizers.clear()
+    check_tp_clip(acc, pc)
 
     acc.end_training()
 

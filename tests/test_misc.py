@@ -179,6 +179,7 @@ def test_tp_through_prepare_4proc():
 
     out = launch_distributed("tests/distributed_scripts/pconfig_prepare_script.py", nproc=4, timeout=300)
     assert "TP_SAVE_STATE_PASS" in out
+    assert "TP_CLIP_PASS" in out
     assert "TP_PREPARE_LLAMA_PASS" in out
     assert "TP_PREPARE_GPT2_PASS" in out
 
