@@ -35,19 +35,36 @@ import torch.nn as nn
 
 
 def split_into_stages(model: nn.Sequential, n_stages: int) -> List[nn.Sequential]:
-    """Split an nn.Sequential into ``n_stages`` parameter-balanced chunks."""
+    """Split an nn.Sequential into ``n_stages`` parameter-balanced chunks.
+
+    Each cut lands at the layer boundary whose parameter prefix-sum is
+    CLOSEST to the ideal k/n split (a forward greedy cut can strand a
+    parameter-less activation layer as a whole stage — e.g. 3-way
+    splitting [Linear, Tanh, Linear, Tanh, Linear] used to produce a
+    [Tanh]-only stage, which breaks per-stage optimizers)."""
     layers = list(model.children())
     weights = [max(sum(p.numel() for p in layer.parameters()), 1) for layer in layers]
     total = sum(weights)
-    stages, current, acc = [], [], 0
-    target = total / n_stages
-    for layer, w in zip(layers, weights):
-        current.append(layer)
+    prefix = []
+    acc = 0
+    for w in weights:
         acc += w
-        if acc >= target * (len(stages) + 1) and len(stages) < n_stages - 1:
-            stages.append(nn.Sequential(*current))
-            current = []
-    stages.append(nn.Sequential(*current))
+        prefix.append(acc)
+    cuts = []
+    prev = 0
+    for k in range(1, n_stages):
+        ideal = total * k / n_stages
+        lo = prev + 1  # at least one layer per stage
+        hi = len(layers) - (n_stages - k)  # leave one layer per later stage
+        if lo > hi:
+            break
+        best = min(range(lo, hi + 1), key=lambda i: abs(prefix[i - 1] - ideal))
+        cuts.append(best)
+        prev = best
+    stages = []
+    bounds = [0] + cuts + [len(layers)]
+    for a, b in zip(bounds[:-1], bounds[1:]):
+        stages.append(nn.Sequential(*layers[a:b]))
     while len(stages) < n_stages:  # degenerate tiny models
         stages.append(nn.Sequential())
     return stages

@@ -165,3 +165,13 @@ def test_pp_train_gpu_parity_2dev():
         "tests/distributed_scripts/pp_train_script.py", nproc=2, extra_env={"PP_GPU": "1"}
     )
     assert "PP_TRAIN_PASS" in out
+
+
+def test_pp_train_3stage():
+    """PP training oracle at THREE stages (multi-hop sends; the stage
+    splitter must never strand a parameter-less stage when a param layer
+    is available — a greedy cut used to produce a [Tanh]-only stage)."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/pp_train_script.py", nproc=3, timeout=300)
+    assert "PP_TRAIN_PASS" in out
