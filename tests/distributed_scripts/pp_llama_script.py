@@ -50,7 +50,7 @@ def run(acc, schedule):
 
 def main():
     acc = Accelerator(cpu=True)
-    assert acc.num_processes == 2
+    assert acc.num_processes >= 2
     run(acc, "gpipe")
     run(acc, "1f1b")
     if acc.is_main_process:

@@ -175,3 +175,11 @@ def test_pp_train_3stage():
 
     out = launch_distributed("tests/distributed_scripts/pp_train_script.py", nproc=3, timeout=300)
     assert "PP_TRAIN_PASS" in out
+
+
+def test_pp_llama_3stage():
+    """Llama pipeline stages at 3 ranks (first/middle/last stage roles)."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/pp_llama_script.py", nproc=3, timeout=300)
+    assert "PP_LLAMA_PASS" in out
