@@ -42,12 +42,12 @@ def main():
     assert len(model.units) >= 2, f"expected multiple units, got {len(model.units)}"
 
     g = torch.Generator().manual_seed(3)
-    X = torch.randn(64, 8, generator=g)
-    Y = torch.randn(64, 1, generator=g)
+    X = torch.randn(48, 8, generator=g)  # 12/step divides world 2/3/4
+    Y = torch.randn(48, 1, generator=g)
 
     for step in range(4):
-        xb = X[step * 16 : (step + 1) * 16]
-        yb = Y[step * 16 : (step + 1) * 16]
+        xb = X[step * 12 : (step + 1) * 12]
+        yb = Y[step * 12 : (step + 1) * 12]
         opt.zero_grad()
         loss = ((model(xb[r::n]) - yb[r::n]) ** 2).mean()
         acc.backward(loss)

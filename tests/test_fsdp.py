@@ -195,3 +195,13 @@ def test_tp_fsdp_combined_4proc():
 
     out = launch_distributed("tests/distributed_scripts/tp_fsdp_script.py", nproc=4, timeout=300)
     assert "TP_FSDP_PASS" in out
+
+
+def test_fsdp_oracle_3proc():
+    """Full FSDP battery at an ODD world size (uneven flat-shard padding:
+    every unit pads to a multiple of 3 here)."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed(SCRIPT, nproc=3, timeout=300)
+    for marker in ("FSDP_PARITY_PASS", "FSDP_SHARDED_CKPT_PASS", "FSDP_METALOAD_PASS"):
+        assert marker in out
