@@ -115,7 +115,7 @@ def test_ddp_integration(acc):
 
 def main():
     acc = Accelerator(cpu=True)
-    assert acc.num_processes == 2
+    assert acc.num_processes >= 2
     test_forward_and_grads(acc)
     test_ddp_integration(acc)
     acc.end_training()

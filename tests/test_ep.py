@@ -183,3 +183,12 @@ def test_pp_llama_3stage():
 
     out = launch_distributed("tests/distributed_scripts/pp_llama_script.py", nproc=3, timeout=300)
     assert "PP_LLAMA_PASS" in out
+
+
+def test_ep_oracle_4proc():
+    """EP all-to-all dispatch at world 4 (8 experts, 2 per rank)."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/ep_script.py", nproc=4, timeout=300)
+    for m in ("EP_FWD_PASS", "EP_GRAD_PASS", "EP_DDP_PASS"):
+        assert m in out
