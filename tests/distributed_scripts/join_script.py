@@ -14,7 +14,7 @@ from accelerate_amd import Accelerator, set_seed
 def main():
     acc = Accelerator(cpu=True)
     n, r = acc.num_processes, acc.process_index
-    assert n == 2
+    assert n in (2, 3)
 
     set_seed(0)
     model = nn.Sequential(nn.Linear(8, 16), nn.Tanh(), nn.Linear(16, 2))
@@ -25,27 +25,26 @@ def main():
     ref_opt = torch.optim.SGD(ref.parameters(), lr=0.1)
 
     g = torch.Generator().manual_seed(5)
-    X = [torch.randn(4, 8, generator=g) for _ in range(10)]  # per-(rank,step) batches
-    n_steps = [5, 3]  # rank 0 runs 5 steps, rank 1 runs 3
+    X = [torch.randn(4, 8, generator=g) for _ in range(15)]  # per-(rank,step) batches
+    n_steps = [5, 3, 4][:n]  # uneven step counts per rank
 
     with acc.join_uneven_inputs([model]):
         for step in range(n_steps[r]):
             opt.zero_grad()
-            loss = model(X[step * 2 + r]).pow(2).mean()
+            loss = model(X[step * n + r]).pow(2).mean()
             acc.backward(loss)
             opt.step()
 
-    # reference semantics: steps 0-2 average both ranks' grads; steps 3-4
-    # average rank 0's grad with ZERO (divide by full world size = 2)
-    for step in range(5):
+    # reference semantics: each step averages the still-live ranks' grads
+    # with ZERO for joined ranks (divide by the FULL world size)
+    for step in range(max(n_steps)):
         ref_opt.zero_grad()
-        g0 = torch.autograd.grad(ref(X[step * 2 + 0]).pow(2).mean(), list(ref.parameters()))
-        if step < 3:
-            g1 = torch.autograd.grad(ref(X[step * 2 + 1]).pow(2).mean(), list(ref.parameters()))
-        else:
-            g1 = [torch.zeros_like(t) for t in g0]
-        for p, a, b in zip(ref.parameters(), g0, g1):
-            p.grad = (a + b) / 2
+        gs = []
+        for rr in range(n):
+            if step < n_steps[rr]:
+                gs.append(torch.autograd.grad(ref(X[step * n + rr]).pow(2).mean(), list(ref.parameters())))
+        for pi, p in enumerate(ref.parameters()):
+            p.grad = sum(g[pi] for g in gs) / n
         ref_opt.step()
 
     for (pn, p), (_, rp) in zip(acc.unwrap_model(model).named_parameters(), ref.named_parameters()):
@@ -54,7 +53,8 @@ def main():
     w = acc.unwrap_model(model)[0].weight.detach()
     ws = [torch.empty_like(w) for _ in range(n)]
     dist.all_gather(ws, w)
-    assert torch.equal(ws[0], ws[1])
+    for rr in range(1, n):
+        assert torch.equal(ws[0], ws[rr])
     if acc.is_main_process:
         print("JOIN_UNEVEN_PASS")
     acc.end_training()

@@ -33,3 +33,12 @@ def test_join_uneven_inputs_oracle():
     script = Path(__file__).parent / "distributed_scripts" / "join_script.py"
     out = launch_distributed(script, nproc=2, timeout=180)
     assert "JOIN_UNEVEN_PASS" in out
+
+
+def test_join_uneven_inputs_3proc():
+    """Join protocol at world 3 with three different exhaustion times
+    (5/3/4 steps): shadow collectives + authoritative final broadcast."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/join_script.py", nproc=3, timeout=300)
+    assert "JOIN_UNEVEN_PASS" in out
