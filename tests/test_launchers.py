@@ -54,10 +54,12 @@ def test_bench_contract_2proc_plumbing():
     import subprocess
     import sys
 
+    from tests.testing_utils import get_free_port
+
     env = dict(os.environ, BENCH_TINY="1", PYTHONPATH=os.getcwd())
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "2",
-         "--master-addr", "127.0.0.1", "--master-port", "29555",
+         "--master-addr", "127.0.0.1", "--master-port", str(get_free_port()),
          "bench.py", "--steps", "2", "--warmup", "1"],
         capture_output=True, text=True, timeout=280, env=env, cwd=os.getcwd(),
     )
