@@ -192,3 +192,35 @@ def test_ep_oracle_4proc():
     out = launch_distributed("tests/distributed_scripts/ep_script.py", nproc=4, timeout=300)
     for m in ("EP_FWD_PASS", "EP_GRAD_PASS", "EP_DDP_PASS"):
         assert m in out
+
+
+def test_split_into_stages_properties():
+    """Property sweep: every stage preserves layer order, covers all layers
+    exactly once, and no stage is parameter-less while a param-bearing
+    split exists (n_stages <= number of param layers)."""
+    import torch.nn as nn
+
+    from accelerate_amd.parallel.pp import split_into_stages
+
+    torch.manual_seed(0)
+    for trial in range(40):
+        n_layers = 2 + trial % 9
+        layers = []
+        for i in range(n_layers):
+            if (trial + i) % 3 == 0:
+                layers.append(nn.Tanh())
+            else:
+                layers.append(nn.Linear(8, 8))
+        model = nn.Sequential(*layers)
+        n_param_layers = sum(1 for l in layers if any(True for _ in l.parameters()))
+        for n_stages in range(2, min(n_layers, 5) + 1):
+            stages = split_into_stages(model, n_stages)
+            assert len(stages) == n_stages
+            flat = [m for s in stages for m in s]
+            assert len(flat) == n_layers
+            assert all(a is b for a, b in zip(flat, layers))  # order preserved
+            if n_stages <= n_param_layers:
+                for s in stages:
+                    assert any(True for _ in s.parameters()), (
+                        f"param-less stage at n_layers={n_layers} n_stages={n_stages}"
+                    )
