@@ -237,3 +237,23 @@ def test_tp_cp_combined_4proc():
 
     out = launch_distributed("tests/distributed_scripts/tp_cp_script.py", nproc=4, timeout=300)
     assert "TP_CP_PASS" in out
+
+
+def test_multiprocess_logger_adapter(caplog):
+    """Logging parity (reference tests/test_logging.py): get_logger wraps a
+    MultiProcessAdapter honoring main_process_only / in_order kwargs, and
+    warning_once emits exactly once."""
+    import logging
+
+    from accelerate_amd.logging import get_logger
+
+    logger = get_logger("accelerate_amd.test_logger")
+    with caplog.at_level(logging.INFO, logger="accelerate_amd.test_logger"):
+        logger.info("hello-main", main_process_only=True)
+        logger.info("hello-all", main_process_only=False)
+        logger.info("hello-order", in_order=True)
+        logger.warning_once("only-once")
+        logger.warning_once("only-once")
+    messages = [r.message for r in caplog.records]
+    assert "hello-main" in messages and "hello-all" in messages and "hello-order" in messages
+    assert messages.count("only-once") == 1
