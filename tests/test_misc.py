@@ -257,3 +257,19 @@ def test_multiprocess_logger_adapter(caplog):
     messages = [r.message for r in caplog.records]
     assert "hello-main" in messages and "hello-all" in messages and "hello-order" in messages
     assert messages.count("only-once") == 1
+
+
+def test_lazy_import_hygiene():
+    """Importing the package must not drag in heavy optional deps
+    (reference tests/test_imports.py): trackers and model libs load lazily."""
+    import subprocess
+    import sys
+
+    code = (
+        "import sys; import accelerate_amd; "
+        "heavy=[m for m in ('wandb','mlflow','comet_ml','aim','clearml','dvclive',"
+        "'swanlab','transformers','datasets','pandas') if m in sys.modules]; "
+        "assert not heavy, heavy; print('ok')"
+    )
+    out = subprocess.run([sys.executable, "-c", code], capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-1500:]
