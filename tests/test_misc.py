@@ -273,3 +273,23 @@ def test_lazy_import_hygiene():
     )
     out = subprocess.run([sys.executable, "-c", code], capture_output=True, text=True, timeout=120)
     assert out.returncode == 0, out.stderr[-1500:]
+
+
+def test_examples_parse():
+    """Every bundled example is at least syntactically valid and imports
+    only names the package exports (guards against API drift)."""
+    import ast
+    import pathlib
+
+    root = pathlib.Path(__file__).resolve().parent.parent / "examples"
+    assert root.is_dir()
+    files = sorted(root.glob("*.py"))
+    assert len(files) >= 6, files
+    import accelerate_amd
+
+    for f in files:
+        tree = ast.parse(f.read_text())
+        for node in ast.walk(tree):
+            if isinstance(node, ast.ImportFrom) and node.module == "accelerate_amd":
+                for alias in node.names:
+                    assert hasattr(accelerate_amd, alias.name), f"{f.name}: accelerate_amd.{alias.name} missing"
