@@ -93,6 +93,10 @@ def add_parser(subparsers):
     parser.add_argument("--gpu_ids", default=None, help="Comma-separated HIP device ids to use")
     parser.add_argument("--gradient_accumulation_steps", type=int, default=None)
     parser.add_argument("--debug", action="store_true")
+    parser.add_argument(
+        "--rccl_debug", default=None,
+        help="Set NCCL_DEBUG (=RCCL's debug plane on ROCm) for every worker, e.g. INFO or TRACE",
+    )
     parser.add_argument("--max_restarts", type=int, default=0)
     parser.add_argument("--monitor_interval", type=float, default=0.1)
     parser.add_argument("--rdzv_backend", default="static")
@@ -153,6 +157,9 @@ def build_env(args, config: ClusterConfig) -> dict:
     env["ACCELERATE_GRADIENT_ACCUMULATION_STEPS"] = str(args.gradient_accumulation_steps)
     if args.debug or config.debug:
         env["ACCELERATE_DEBUG_MODE"] = "1"
+    if args.rccl_debug:
+        # RCCL reads the NCCL_* env plane on ROCm (SURVEY §5.2 plumb-through)
+        env["NCCL_DEBUG"] = str(args.rccl_debug)
     if args.cpu:
         env["ACCELERATE_USE_CPU"] = "1"
     if args.use_fsdp:

@@ -150,3 +150,18 @@ def test_config_rejects_invalid_values(tmp_path):
     p.write_text(yaml.safe_dump({"mixed_precision": "int3"}))
     with pytest.raises(ValueError, match="mixed_precision"):
         ClusterConfig.load(p)
+
+
+def test_launch_rccl_debug_plumb():
+    """--rccl_debug sets the NCCL_* plane (RCCL on ROCm) for workers
+    (SURVEY §5.2 MI355X note)."""
+    import argparse
+
+    import accelerate_amd.commands.launch as launch_mod
+    from accelerate_amd.commands.config import ClusterConfig
+
+    parser = argparse.ArgumentParser()
+    launch_mod.add_parser(parser.add_subparsers())
+    args = parser.parse_args(["launch", "--rccl_debug", "INFO", "--cpu", "script.py"])
+    env = launch_mod.build_env(args, ClusterConfig())
+    assert env["NCCL_DEBUG"] == "INFO"
