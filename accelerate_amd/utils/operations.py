@@ -302,14 +302,18 @@ def gather(tensor):
 
 
 def gather_object(object: Any):
-    """All-gather picklable objects → list of length num_processes
-    (reference: operations.py:444-523)."""
+    """All-gather picklable objects, flattening ONE level: each rank
+    passes a LIST and the result concatenates every rank's items — the
+    reference semantics (operations.py:498-523 _gpu_gather_object) that
+    gather_for_metrics' remainder slicing depends on (slicing must drop
+    tail SAMPLES, not tail ranks). Non-distributed worlds return the
+    object unchanged, as the reference does."""
     state = PartialState()
     if not state.use_distributed:
-        return [object]
+        return object
     output_objects = [None for _ in range(state.num_processes)]
     torch.distributed.all_gather_object(output_objects, object)
-    return [x for y in output_objects for x in [y]]
+    return [x for y in output_objects for x in y]
 
 
 def _gpu_broadcast(data, src=0):

@@ -103,7 +103,7 @@ def test_collectives(acc):
     objs = acc.state
     from accelerate_amd.utils.operations import gather_object
 
-    objects = gather_object({"rank": r})
+    objects = gather_object([{"rank": r}])  # list input -> flat concat
     assert [o["rank"] for o in objects] == list(range(n))
     if acc.is_main_process:
         print("COLLECTIVES_PASS")

@@ -96,7 +96,7 @@ def test_listify():
 def test_single_process_collectives_passthrough():
     t = torch.randn(3)
     assert torch.equal(gather(t), t)
-    assert gather_object(["x"]) == [["x"]]
+    assert gather_object(["x"]) == ["x"]  # world 1: unchanged (reference parity)
     assert torch.equal(broadcast(t), t)
     r = reduce(t, "mean")
     assert torch.allclose(r, t)
@@ -165,3 +165,13 @@ class TestStructuredOpsProperties:
         out = concatenate(parts)
         assert out["x"].shape == (2 * n, 3)
         assert out["y"][0].shape == (4 * n,)
+
+
+def test_collective_ops_uneven_3proc():
+    """pad/gather/reduce/broadcast/gather_object at world 3 with
+    rank-uneven shapes (pad equalizes to the max before concatenation;
+    gather_object flattens each rank's list — reference semantics)."""
+    from tests.testing_utils import launch_distributed
+
+    out = launch_distributed("tests/distributed_scripts/operations_script.py", nproc=3, timeout=300)
+    assert "OPERATIONS_PASS" in out
