@@ -432,7 +432,9 @@ def reduce(tensor, reduction="mean", scale=1.0):
         cloned_tensor = tensor.clone()
         if not state.use_distributed:
             return cloned_tensor
-        op = torch.distributed.ReduceOp.SUM if reduction in ("sum", "mean") else torch.distributed.ReduceOp.MAX
+        # reference parity: only "max" maps to MAX; "sum"/"mean"/"none"
+        # all-reduce SUM (mean divides afterwards, none leaves the sum)
+        op = torch.distributed.ReduceOp.MAX if reduction == "max" else torch.distributed.ReduceOp.SUM
         torch.distributed.all_reduce(cloned_tensor, op)
         if reduction == "mean":
             cloned_tensor /= state.num_processes
