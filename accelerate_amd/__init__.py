@@ -28,6 +28,7 @@ from .state import AcceleratorState, GradientState, PartialState
 from .utils.random_utils import synchronize_rng_states  # noqa: F401
 from .utils.dataclasses import (
     AutocastKwargs,
+    DeepSpeedPlugin,
     DataLoaderConfiguration,
     DDPCommunicationHookType,
     DistributedDataParallelKwargs,

@@ -401,3 +401,28 @@ def add_model_config_to_megatron_parser(*args, **kwargs):  # pragma: no cover
         "Megatron-LM integration is an external-trainer delegation in the reference "
         "(utils/megatron_lm.py) and is out of scope for the MI355X-native framework."
     )
+
+
+@dataclass
+class DeepSpeedPlugin:
+    """API-surface shim for the reference's DeepSpeed delegation
+    (reference: dataclasses.py:1122). This framework does not delegate to
+    the DeepSpeed engine — the ZeRO-1/2/3 capabilities map onto the native
+    flat-shard engine (`FullyShardedDataParallelPlugin`: ZeRO-3 ≈ full
+    shard, ZeRO-2-style accumulation via sharded grads, fused HIP AdamW
+    instead of DeepSpeed's) — so constructing this raises with the
+    migration pointer instead of silently training differently."""
+
+    hf_ds_config: Any = None
+    gradient_accumulation_steps: int = None
+    zero_stage: int = None
+    offload_optimizer_device: str = None
+    offload_param_device: str = None
+
+    def __post_init__(self):
+        raise NotImplementedError(
+            "DeepSpeed delegation is out of scope on this MI355X-native stack. "
+            "Use FullyShardedDataParallelPlugin: zero_stage 2/3 capabilities map onto the "
+            "flat-shard engine (sharded grads + fp32 master shards + fused HIP AdamW); "
+            "see docs/COMPONENT_MAP.md 'Megatron / DeepSpeed'."
+        )
