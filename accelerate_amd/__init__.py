@@ -25,6 +25,10 @@ from .inference import prepare_pipeline, prepare_pippy
 from .local_sgd import LocalSGD
 from .parallelism_config import ParallelismConfig
 from .state import AcceleratorState, GradientState, PartialState
+from .utils.imports import is_rich_available  # noqa: F401
+
+if is_rich_available():  # reference: pretty tracebacks when rich is present
+    from .utils import rich  # noqa: F401
 from .utils.random_utils import synchronize_rng_states  # noqa: F401
 from .utils.dataclasses import (
     AutocastKwargs,
