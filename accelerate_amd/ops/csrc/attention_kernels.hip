@@ -1750,7 +1750,7 @@ static int fa_fwd_impl() {
     if (e && strcmp(e, "legacy") == 0) return 0;
     if (e && strcmp(e, "swapped") == 0) return 1;
     if (e && strcmp(e, "swapped_tr") == 0) return 2;
-    return 2;  // tr_read V path: 782 vs 652 TF/s (gpurun_out/fa_tr2.log)
+    return 2;  // tr_read + dbuf + v_exp ladder: 926 TF/s fwd (profiles/fa_swapped_r2.md)
   }();
   return impl;
 }
